@@ -1,0 +1,175 @@
+"""LoRA modules + HF-PEFT-compatible adapter checkpoint layout.
+
+Adapter layout contract (what the reference produces via PEFT
+`trainer.save_model` — cmd/tuning/train.py:300, SURVEY.md §5
+Checkpoint/resume): a directory with `adapter_config.json` and
+`adapter_model.safetensors` whose keys are
+`base_model.model.<module_path>.lora_A.weight` ([r, in]) and
+`...lora_B.weight` ([out, r]). Stock `peft.PeftModel.from_pretrained`
+can load our checkpoints.
+"""
+
+from __future__ import annotations
+
+import json
+import math
+import os
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops import lora_contract, lora_expand_add
+from ..ops.autograd import LoRALinear as _FusedLoRAFn
+
+
+class FrozenLinear(nn.Module):
+    """Bias-free frozen linear (Llama projections are bias-free)."""
+
+    def __init__(self, in_features: int, out_features: int,
+                 dtype=torch.bfloat16):
+        super().__init__()
+        self.in_features, self.out_features = in_features, out_features
+        self.weight = nn.Parameter(
+            torch.empty(out_features, in_features, dtype=dtype),
+            requires_grad=False)
+
+    def forward(self, x):
+        return x @ self.weight.t()
+
+
+class LoRAFunctionWithDropout(torch.autograd.Function):
+    """Fused LoRA linear with PEFT-style input dropout on the low-rank path.
+
+    mask is pre-scaled (0 or 1/keep) or None.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, a, b, scale, mask):
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1])
+        xd = x2 * mask if mask is not None else x2
+        y = x2 @ w.t()
+        t = lora_contract(xd, a)
+        lora_expand_add(y, t, b, scale)
+        ctx.save_for_backward(x2, w, a, b, t,
+                              mask if mask is not None else torch.empty(0))
+        ctx.scale, ctx.xshape = scale, xs
+        return y.reshape(*xs[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        from ..ops import lora_wgrad
+        x2, w, a, b, t, mask = ctx.saved_tensors
+        s = ctx.scale
+        has_mask = mask.numel() > 0
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = dy2 @ w
+        dt = lora_contract(dy2, b.t().contiguous())       # [M,r] = dy @ B
+        xd = x2 * mask if has_mask else x2
+        da = lora_wgrad(dt, xd, s)
+        db = lora_wgrad(t, dy2, s).t().contiguous()
+        if has_mask:
+            dxl = torch.zeros_like(dx)
+            lora_expand_add(dxl, dt, a.t().contiguous(), s)
+            dx = dx.addcmul_(dxl, mask)
+        else:
+            lora_expand_add(dx, dt, a.t().contiguous(), s)
+        return (dx.reshape(ctx.xshape), None, da.to(a.dtype),
+                db.to(b.dtype), None, None)
+
+
+class LoRALinearModule(nn.Module):
+    """Frozen base weight + trainable LoRA A/B (HF layouts: A [r,in],
+    B [out,r]); scale = alpha / r (reference defaults r=8 alpha=32 —
+    cmd/tuning/parser.py:138-149)."""
+
+    def __init__(self, in_features: int, out_features: int, r: int = 8,
+                 alpha: float = 32.0, dropout: float = 0.0,
+                 dtype=torch.bfloat16):
+        super().__init__()
+        self.in_features, self.out_features = in_features, out_features
+        self.r, self.alpha, self.dropout = r, alpha, dropout
+        self.scale = alpha / r
+        self.weight = nn.Parameter(
+            torch.empty(out_features, in_features, dtype=dtype),
+            requires_grad=False)
+        self.lora_A = nn.Parameter(torch.empty(r, in_features, dtype=dtype))
+        self.lora_B = nn.Parameter(torch.zeros(out_features, r, dtype=dtype))
+        nn.init.kaiming_uniform_(self.lora_A, a=math.sqrt(5))
+
+    def forward(self, x):
+        mask = None
+        if self.training and self.dropout > 0.0:
+            keep = 1.0 - self.dropout
+            x2 = x.reshape(-1, x.shape[-1])
+            mask = (torch.rand_like(x2, dtype=torch.float32) < keep)
+            mask = mask.to(x.dtype) / keep
+        if mask is None:
+            return _FusedLoRAFn.apply(x, self.weight, self.lora_A,
+                                      self.lora_B, self.scale)
+        return LoRAFunctionWithDropout.apply(x, self.weight, self.lora_A,
+                                             self.lora_B, self.scale, mask)
+
+    def merged_weight(self):
+        """W + s·B@A — used by the serving engine (no adapter overhead)."""
+        return (self.weight.float()
+                + self.scale * (self.lora_B.float() @ self.lora_A.float())
+                ).to(self.weight.dtype)
+
+
+def lora_state_dict(model: nn.Module, prefix: str = "base_model.model."):
+    out = {}
+    for name, mod in model.named_modules():
+        if isinstance(mod, LoRALinearModule):
+            out[f"{prefix}{name}.lora_A.weight"] = mod.lora_A.detach().cpu()
+            out[f"{prefix}{name}.lora_B.weight"] = mod.lora_B.detach().cpu()
+    return out
+
+
+def save_adapter(model: nn.Module, out_dir: str, *, r: int, alpha: float,
+                 dropout: float, target_modules: list[str],
+                 base_model_name_or_path: str = ""):
+    """Write HF-PEFT adapter layout (adapter_config.json +
+    adapter_model.safetensors)."""
+    os.makedirs(out_dir, exist_ok=True)
+    from safetensors.torch import save_file
+    sd = {k: v.contiguous() for k, v in lora_state_dict(model).items()}
+    save_file(sd, os.path.join(out_dir, "adapter_model.safetensors"))
+    cfg = {
+        "peft_type": "LORA",
+        "task_type": "CAUSAL_LM",
+        "r": r,
+        "lora_alpha": alpha,
+        "lora_dropout": dropout,
+        "target_modules": target_modules,
+        "base_model_name_or_path": base_model_name_or_path,
+        "bias": "none",
+        "fan_in_fan_out": False,
+        "inference_mode": False,
+        "modules_to_save": None,
+    }
+    with open(os.path.join(out_dir, "adapter_config.json"), "w") as f:
+        json.dump(cfg, f, indent=2)
+
+
+def load_adapter(model: nn.Module, adapter_dir: str,
+                 prefix: str = "base_model.model."):
+    from safetensors.torch import load_file
+    sd = load_file(os.path.join(adapter_dir, "adapter_model.safetensors"))
+    mods = {n: m for n, m in model.named_modules()
+            if isinstance(m, LoRALinearModule)}
+    n_loaded = 0
+    for key, tensor in sd.items():
+        if not key.startswith(prefix):
+            continue
+        rest = key[len(prefix):]
+        for suffix, attr in ((".lora_A.weight", "lora_A"),
+                             (".lora_B.weight", "lora_B")):
+            if rest.endswith(suffix):
+                mod_name = rest[: -len(suffix)]
+                if mod_name in mods:
+                    getattr(mods[mod_name], attr).data.copy_(
+                        tensor.to(getattr(mods[mod_name], attr).dtype))
+                    n_loaded += 1
+    return n_loaded

@@ -1,0 +1,152 @@
+"""torch.autograd.Function wrappers over the fused ops.
+
+These are what the model code calls; each routes to the HIP kernel on GPU
+and the torch reference on CPU via datatunerx_amd.ops dispatch.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import (attn_bwd, attn_fwd, lora_contract, lora_expand_add,
+               lora_wgrad, rmsnorm_bwd, rmsnorm_fwd, rope_bwd, rope_fwd,
+               softmax_xent_bwd, softmax_xent_fwd, swiglu_bwd, swiglu_fwd)
+
+
+class RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        y, inv = rmsnorm_fwd(x, w, eps)
+        ctx.save_for_backward(x, w, inv)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, inv = ctx.saved_tensors
+        dx, dw = rmsnorm_bwd(dy.contiguous(), x, w, inv)
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm(x, w, eps: float = 1e-5):
+    return RMSNorm.apply(x, w, eps)
+
+
+class Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin, pos0):
+        ctx.save_for_backward(cos, sin)
+        ctx.pos0 = pos0
+        return rope_fwd(x, cos, sin, pos0)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        return rope_bwd(dy.contiguous(), cos, sin, ctx.pos0), None, None, None
+
+
+def rope(x, cos, sin, pos0: int = 0):
+    return Rope.apply(x, cos, sin, pos0)
+
+
+class SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        ctx.save_for_backward(gate, up)
+        return swiglu_fwd(gate, up)
+
+    @staticmethod
+    def backward(ctx, dout):
+        gate, up = ctx.saved_tensors
+        dgate, dup = swiglu_bwd(dout.contiguous(), gate, up)
+        return dgate, dup
+
+
+def swiglu(gate, up):
+    return SwiGLU.apply(gate, up)
+
+
+class Attention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = attn_fwd(q, k, v, causal, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal, ctx.scale = causal, scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = attn_bwd(q, k, v, o, do.contiguous(), lse,
+                              ctx.causal, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+def attention(q, k, v, causal: bool = True, scale: float | None = None):
+    """q [B,Hq,S,D], k/v [B,Hkv,S,D] -> o [B,Hq,S,D]."""
+    if scale is None:
+        scale = 1.0 / (q.shape[-1] ** 0.5)
+    return Attention.apply(q, k, v, causal, scale)
+
+
+class CrossEntropy(torch.autograd.Function):
+    """Mean CE over non-ignored targets; logits [N,V], targets [N]."""
+
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        loss, lse = softmax_xent_fwd(logits, targets, ignore_index)
+        n_valid = (targets != ignore_index).sum().clamp(min=1)
+        ctx.save_for_backward(logits, targets, lse, n_valid)
+        ctx.ignore_index = ignore_index
+        return loss.sum() / n_valid.float()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, lse, n_valid = ctx.saved_tensors
+        scale = (dloss.float() / n_valid.float()).expand(logits.shape[0])
+        dlogits = softmax_xent_bwd(logits, targets, lse, scale.contiguous(),
+                                   ctx.ignore_index)
+        return dlogits, None, None
+
+
+def cross_entropy(logits, targets, ignore_index: int = -100):
+    return CrossEntropy.apply(logits, targets, ignore_index)
+
+
+class LoRALinear(torch.autograd.Function):
+    """y = x @ W^T + scale * (x @ A^T) @ B^T  with frozen W.
+
+    W: [N,K] (frozen base, no wgrad) — the base GEMM goes through
+    torch.matmul (hipBLASLt on ROCm); the low-rank path is the fused HIP
+    contract/expand pair. A: [r,K], B: [N,r] (HF PEFT adapter layout, the
+    checkpoint-compat contract — SURVEY.md §5 Checkpoint/resume).
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, a, b, scale):
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1])
+        y = x2 @ w.t()
+        t = lora_contract(x2, a)                 # [M,r] f32
+        lora_expand_add(y, t, b, scale)          # y += s * t @ B^T
+        ctx.save_for_backward(x2, w, a, b, t)
+        ctx.scale = scale
+        ctx.xshape = xs
+        return y.reshape(*xs[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, w, a, b, t = ctx.saved_tensors
+        s = ctx.scale
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = dy2 @ w                                  # base dgrad (hipBLASLt)
+        dt = lora_contract(dy2, b.t().contiguous())   # [M,r] = dy @ B
+        db = lora_wgrad(t, dy2, s)                    # [r,N] -> B grad is [N,r]
+        da = lora_wgrad(dt, x2, s)                    # [r,K]
+        lora_expand_add(dx, dt, a.t().contiguous(), s)  # dx += s * dt @ A
+        return (dx.reshape(ctx.xshape), None, da.to(a.dtype),
+                db.t().contiguous().to(b.dtype), None)
+
+
+def lora_linear(x, w, a, b, scale: float):
+    return LoRALinear.apply(x, w, a, b, scale)

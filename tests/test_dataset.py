@@ -1,0 +1,92 @@
+"""Preprocessing parity tests: template encode, -100 masking,
+proportional truncation (reference: cmd/tuning/train.py:58-135,
+template.py)."""
+
+import torch
+
+from datatunerx_amd.data.dataset import (IGNORE_INDEX, ByteTokenizer,
+                                         SFTDataset, collate,
+                                         preprocess_supervised_example,
+                                         read_csv_rows)
+from datatunerx_amd.data.templates import TEMPLATES, get_template
+
+
+def test_all_templates_registered():
+    for name in ["vanilla", "default", "llama2", "llama2_zh", "alpaca",
+                 "vicuna", "belle", "ziya", "aquila", "intern", "baichuan",
+                 "baichuan2", "starchat", "chatml", "chatglm2", "chatglm3",
+                 "openchat", "xverse"]:
+        assert name in TEMPLATES
+
+
+def test_llama2_template_encoding():
+    tok = ByteTokenizer()
+    t = get_template("llama2")
+    pairs = t.encode_multiturn(tok, "hi", "hello", system="sys")
+    assert len(pairs) == 1
+    src, tgt = pairs[0]
+    assert src[0] == tok.bos_token_id
+    assert tgt[-1] == tok.eos_token_id
+    text = tok.decode(src)
+    assert "[INST]" in text and "[/INST]" in text and "<<SYS>>" in text
+    assert tok.decode(tgt) == "hello"
+
+
+def test_multiturn_masking():
+    tok = ByteTokenizer()
+    ids, labels = preprocess_supervised_example(
+        tok, "default", "q2", "a2",
+        history=[("q1", "a1")], cutoff_len=512)
+    assert len(ids) == len(labels)
+    # source positions masked, target positions = ids
+    n_masked = sum(1 for l in labels if l == IGNORE_INDEX)
+    assert 0 < n_masked < len(labels)
+    for i, l in enumerate(labels):
+        if l != IGNORE_INDEX:
+            assert l == ids[i]
+    # both answers present unmasked
+    ans = [l for l in labels if l != IGNORE_INDEX]
+    decoded = tok.decode([a for a in ans if a >= 3])
+    assert "a1" in decoded and "a2" in decoded
+
+
+def test_proportional_truncation():
+    tok = ByteTokenizer()
+    long_src = "x" * 300
+    long_tgt = "y" * 100
+    ids, labels = preprocess_supervised_example(
+        tok, "vanilla", long_src, long_tgt, cutoff_len=100)
+    assert len(ids) <= 100
+    n_src = sum(1 for l in labels if l == IGNORE_INDEX)
+    n_tgt = len(labels) - n_src
+    # proportional: src gets ~3/4 of the budget
+    assert 60 <= n_src <= 80
+    assert 20 <= n_tgt <= 40
+
+
+def test_csv_column_mapping(tmp_path):
+    p = tmp_path / "d.csv"
+    p.write_text("col_a,col_b\nhello,world\nfoo,bar\n")
+    rows = read_csv_rows(str(p), {"instruction": "col_a",
+                                  "response": "col_b"})
+    assert rows[0] == {"instruction": "hello", "response": "world"}
+    ds = SFTDataset.from_rows(rows, ByteTokenizer(), "vanilla")
+    assert len(ds) == 2
+
+
+def test_collate_pads_to_multiple():
+    batch = [{"input_ids": [1, 2, 3], "labels": [1, 2, 3]},
+             {"input_ids": [1, 2, 3, 4, 5], "labels": [1, 2, 3, 4, 5]}]
+    out = collate(batch, pad_token_id=0, pad_to_multiple_of=4)
+    assert out["input_ids"].shape == (2, 8)
+    assert out["labels"][0, 3] == IGNORE_INDEX
+    assert out["input_ids"][0, 3] == 0
+
+
+def test_synthetic_dataset_shapes():
+    ds = SFTDataset.synthetic(10, 64, 512)
+    assert len(ds) == 10
+    ex = ds[0]
+    assert len(ex["input_ids"]) == 64
+    assert ex["labels"][0] == IGNORE_INDEX
+    assert ex["labels"][-1] == ex["input_ids"][-1]
