@@ -1,0 +1,289 @@
+// Torch bindings for the gfx950 HIP kernel library.
+// All tensors must be contiguous; bf16 compute dtype, fp32 reductions.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+// ---- launchers (defined in the .hip files) ----
+void launch_rmsnorm_fwd(const void*, const void*, void*, float*, int, int,
+                        float, hipStream_t);
+int rmsnorm_bwd_nblocks(int M);
+void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
+                        void*, float*, float*, int, int, hipStream_t);
+void launch_rope(const void*, const float*, const float*, void*, long, int,
+                 int, int, int, int, hipStream_t);
+void launch_swiglu_fwd(const void*, const void*, void*, long, hipStream_t);
+void launch_swiglu_bwd(const void*, const void*, const void*, void*, void*,
+                       long, hipStream_t);
+void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
+                     long, hipStream_t);
+void launch_xent_bwd(const void*, const long*, const float*, const float*,
+                     void*, long, int, long, hipStream_t);
+void launch_adamw(void*, float*, const float*, float*, float*, long, float,
+                  float, float, float, float, int, hipStream_t);
+void launch_l2_norm(const float*, float*, float*, long, hipStream_t);
+void launch_lora_contract(const void*, const void*, float*, long, int, int,
+                          hipStream_t);
+void launch_lora_expand_add(void*, const float*, const void*, long, int, int,
+                            float, hipStream_t);
+int lora_wgrad_splitm(int K);
+void launch_lora_wgrad(const float*, const void*, float*, float*, long, int,
+                       int, float, hipStream_t);
+void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
+                     int, int, int, int, int, int, float, int, hipStream_t);
+void launch_attn_delta(const void*, const void*, float*, long, int,
+                       hipStream_t);
+void launch_attn_bwd(const void*, const void*, const void*, const void*,
+                     const float*, const float*, void*, void*, void*, int,
+                     int, int, int, int, int, float, int, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check_bf16_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// ------------------------------------------------------------- RMSNorm
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int H = (int)x.size(-1);
+  const long M = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0, "H % 8");
+  auto y = torch::empty_like(x);
+  auto inv = torch::empty({M}, x.options().dtype(torch::kFloat));
+  launch_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                     inv.data_ptr<float>(), (int)M, H, (float)eps,
+                     cur_stream());
+  return {y, inv};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor inv) {
+  check_bf16_contig(dy, "dy");
+  check_bf16_contig(x, "x");
+  const int H = (int)x.size(-1);
+  const long M = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat));
+  const int nb = rmsnorm_bwd_nblocks((int)M);
+  auto part = torch::empty({nb, H}, x.options().dtype(torch::kFloat));
+  launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     inv.data_ptr<float>(), dx.data_ptr(),
+                     part.data_ptr<float>(), dw.data_ptr<float>(), (int)M, H,
+                     cur_stream());
+  return {dx, dw};
+}
+
+// ---------------------------------------------------------------- RoPE
+torch::Tensor rope(torch::Tensor x, torch::Tensor cosb, torch::Tensor sinb,
+                   long pos0, bool backward) {
+  check_bf16_contig(x, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be [B,S,H,D]");
+  TORCH_CHECK(cosb.scalar_type() == torch::kFloat && cosb.is_contiguous());
+  const int B = (int)x.size(0), S = (int)x.size(1), H = (int)x.size(2),
+            D = (int)x.size(3);
+  TORCH_CHECK(D % 8 == 0, "D % 8");
+  TORCH_CHECK(cosb.size(0) >= pos0 + S, "rope table too short");
+  auto y = torch::empty_like(x);
+  launch_rope(x.data_ptr(), cosb.data_ptr<float>(), sinb.data_ptr<float>(),
+              y.data_ptr(), B, S, H, D, (int)pos0, backward ? 1 : 0,
+              cur_stream());
+  return y;
+}
+
+// --------------------------------------------------------------- SwiGLU
+torch::Tensor swiglu_fwd(torch::Tensor gate, torch::Tensor up) {
+  check_bf16_contig(gate, "gate");
+  check_bf16_contig(up, "up");
+  TORCH_CHECK(gate.numel() % 8 == 0);
+  auto out = torch::empty_like(gate);
+  launch_swiglu_fwd(gate.data_ptr(), up.data_ptr(), out.data_ptr(),
+                    gate.numel(), cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dout, torch::Tensor gate,
+                                      torch::Tensor up) {
+  check_bf16_contig(dout, "dout");
+  auto dg = torch::empty_like(gate);
+  auto du = torch::empty_like(up);
+  launch_swiglu_bwd(dout.data_ptr(), gate.data_ptr(), up.data_ptr(),
+                    dg.data_ptr(), du.data_ptr(), gate.numel(),
+                    cur_stream());
+  return {dg, du};
+}
+
+// -------------------------------------------------------- cross entropy
+std::vector<torch::Tensor> xent_fwd(torch::Tensor logits,
+                                    torch::Tensor targets, long ignore) {
+  check_bf16_contig(logits, "logits");
+  TORCH_CHECK(targets.scalar_type() == torch::kLong);
+  const int V = (int)logits.size(-1);
+  const long N = logits.numel() / V;
+  TORCH_CHECK(V % 8 == 0, "V % 8");
+  auto loss = torch::empty({N}, logits.options().dtype(torch::kFloat));
+  auto lse = torch::empty({N}, logits.options().dtype(torch::kFloat));
+  launch_xent_fwd(logits.data_ptr(), targets.data_ptr<long>(),
+                  loss.data_ptr<float>(), lse.data_ptr<float>(), N, V,
+                  ignore, cur_stream());
+  return {loss, lse};
+}
+
+torch::Tensor xent_bwd(torch::Tensor logits, torch::Tensor targets,
+                       torch::Tensor lse, torch::Tensor dloss, long ignore) {
+  check_bf16_contig(logits, "logits");
+  const int V = (int)logits.size(-1);
+  const long N = logits.numel() / V;
+  auto dlogits = torch::empty_like(logits);
+  launch_xent_bwd(logits.data_ptr(), targets.data_ptr<long>(),
+                  lse.data_ptr<float>(), dloss.data_ptr<float>(),
+                  dlogits.data_ptr(), N, V, ignore, cur_stream());
+  return dlogits;
+}
+
+// ----------------------------------------------------------------- LoRA
+torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int K = (int)x.size(-1);
+  const long M = x.numel() / K;
+  const int r = (int)w.size(0);
+  TORCH_CHECK(w.size(1) == K, "w [r,K] mismatch");
+  TORCH_CHECK(r <= 64, "r <= 64");
+  TORCH_CHECK(K % 8 == 0);
+  auto t = torch::empty({M, r}, x.options().dtype(torch::kFloat));
+  launch_lora_contract(x.data_ptr(), w.data_ptr(), t.data_ptr<float>(), M,
+                       K, r, cur_stream());
+  return t;
+}
+
+void lora_expand_add(torch::Tensor y, torch::Tensor t, torch::Tensor w,
+                     double scale) {
+  check_bf16_contig(y, "y");
+  check_bf16_contig(w, "w");
+  const int N = (int)y.size(-1);
+  const long M = y.numel() / N;
+  const int r = (int)w.size(1);
+  TORCH_CHECK(w.size(0) == N, "w [N,r] mismatch");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat && t.is_contiguous());
+  launch_lora_expand_add(y.data_ptr(), t.data_ptr<float>(), w.data_ptr(),
+                         M, N, r, (float)scale, cur_stream());
+}
+
+torch::Tensor lora_wgrad(torch::Tensor t, torch::Tensor x, double scale) {
+  check_bf16_contig(x, "x");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat && t.is_contiguous());
+  const int K = (int)x.size(-1);
+  const long M = x.numel() / K;
+  const int r = (int)t.size(-1);
+  auto out = torch::empty({r, K}, x.options().dtype(torch::kFloat));
+  const int sm = lora_wgrad_splitm(K);
+  auto part = torch::empty({sm, r, K}, x.options().dtype(torch::kFloat));
+  launch_lora_wgrad(t.data_ptr<float>(), x.data_ptr(),
+                    part.data_ptr<float>(), out.data_ptr<float>(), M, K, r,
+                    (float)scale, cur_stream());
+  return out;
+}
+
+// ---------------------------------------------------------------- AdamW
+void adamw(torch::Tensor p, torch::Tensor master, torch::Tensor grad,
+           torch::Tensor m, torch::Tensor v, double lr, double b1, double b2,
+           double eps, double wd, long step) {
+  check_bf16_contig(p, "p");
+  TORCH_CHECK(master.scalar_type() == torch::kFloat);
+  TORCH_CHECK(grad.scalar_type() == torch::kFloat);
+  TORCH_CHECK(p.numel() % 4 == 0, "flat param buffer must be padded to 4");
+  launch_adamw(p.data_ptr(), master.data_ptr<float>(),
+               grad.data_ptr<float>(), m.data_ptr<float>(),
+               v.data_ptr<float>(), p.numel(), (float)lr, (float)b1,
+               (float)b2, (float)eps, (float)wd, (int)step, cur_stream());
+}
+
+torch::Tensor l2_norm(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat &&
+              x.is_contiguous());
+  auto ws = torch::empty({1024}, x.options());
+  auto out = torch::empty({1}, x.options());
+  launch_l2_norm(x.data_ptr<float>(), ws.data_ptr<float>(),
+                 out.data_ptr<float>(), x.numel(), cur_stream());
+  return out;
+}
+
+// ------------------------------------------------------------ attention
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal,
+                                    double scale) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Hkv = (int)k.size(1), Skv = (int)k.size(2);
+  TORCH_CHECK(D == 64 || D == 128, "D must be 64 or 128");
+  TORCH_CHECK(Hq % Hkv == 0, "GQA group");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat));
+  launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse.data_ptr<float>(), B, Hq, Hkv, S, Skv, D,
+                  (float)scale, causal ? 1 : 0, cur_stream());
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    torch::Tensor dO, torch::Tensor lse,
+                                    bool causal, double scale) {
+  check_bf16_contig(q, "q");
+  auto dO_c = dO.contiguous();
+  const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Hkv = (int)k.size(1), Skv = (int)k.size(2);
+  auto delta = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat));
+  launch_attn_delta(dO_c.data_ptr(), o.data_ptr(), delta.data_ptr<float>(),
+                    (long)B * Hq * S, D, cur_stream());
+  auto dq = torch::empty_like(q);
+  // per-Q-head partials; reduced below for GQA
+  auto dk_h = torch::empty({B, Hq, Skv, D}, q.options());
+  auto dv_h = torch::empty({B, Hq, Skv, D}, q.options());
+  launch_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), dO_c.data_ptr(),
+                  lse.data_ptr<float>(), delta.data_ptr<float>(),
+                  dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(), B, Hq,
+                  Hkv, S, Skv, D, (float)scale, causal ? 1 : 0,
+                  cur_stream());
+  torch::Tensor dk = dk_h, dv = dv_h;
+  if (Hq != Hkv) {
+    const long rep = Hq / Hkv;
+    dk = dk_h.view({B, Hkv, rep, Skv, D})
+             .to(torch::kFloat).sum(2).to(torch::kBFloat16);
+    dv = dv_h.view({B, Hkv, rep, Skv, D})
+             .to(torch::kFloat).sum(2).to(torch::kBFloat16);
+  }
+  return {dq, dk, dv};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope", &rope);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("xent_fwd", &xent_fwd);
+  m.def("xent_bwd", &xent_bwd);
+  m.def("lora_contract", &lora_contract);
+  m.def("lora_expand_add", &lora_expand_add);
+  m.def("lora_wgrad", &lora_wgrad);
+  m.def("adamw", &adamw);
+  m.def("l2_norm", &l2_norm);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+}

@@ -1,0 +1,78 @@
+// Shared helpers for the gfx950 (CDNA4) kernels.
+// Conventions: wave = 64 lanes; block = 256 threads unless stated;
+// bf16 handled as raw ushort bits, vectorized as short4/short8 (16 B/lane
+// loads — guide G13: hipcc does not auto-vectorize bf16 scalar loads).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#define WAVE 64
+#define DTX_BLOCK 256
+
+typedef __attribute__((ext_vector_type(2))) short short2v;
+typedef __attribute__((ext_vector_type(4))) short short4v;
+typedef __attribute__((ext_vector_type(8))) short short8v;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+typedef __attribute__((ext_vector_type(8))) float float8v;
+typedef __attribute__((ext_vector_type(16))) float float16v;
+
+__device__ __forceinline__ float bf2f(unsigned short u) {
+  union { float f; unsigned int i; } x;
+  x.i = ((unsigned int)u) << 16;
+  return x.f;
+}
+
+__device__ __forceinline__ unsigned short f2bf(float f) {
+  union { float f; unsigned int i; } x;
+  x.f = f;
+  unsigned int u = x.i;
+  if ((u & 0x7fffffffu) > 0x7f800000u) return 0x7fc0;  // NaN
+  unsigned int rounding = 0x7fffu + ((u >> 16) & 1u);
+  return (unsigned short)((u + rounding) >> 16);
+}
+
+// load 8 bf16 (16B) and convert to 8 floats
+__device__ __forceinline__ void load_bf16x8(const unsigned short* p,
+                                            float* out) {
+  short8v v = *reinterpret_cast<const short8v*>(p);
+#pragma unroll
+  for (int i = 0; i < 8; ++i) out[i] = bf2f((unsigned short)v[i]);
+}
+
+__device__ __forceinline__ void store_bf16x8(unsigned short* p,
+                                             const float* in) {
+  short8v v;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) v[i] = (short)f2bf(in[i]);
+  *reinterpret_cast<short8v*>(p) = v;
+}
+
+// ---- wave/block reductions (64-wide wave) ----
+__device__ __forceinline__ float wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    x += __shfl_xor(x, off, WAVE);
+  return x;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    x = fmaxf(x, __shfl_xor(x, off, WAVE));
+  return x;
+}
+
+// Block-wide sum for 256-thread blocks; `scratch` must hold >= 4 floats.
+// Result valid on all threads.
+__device__ __forceinline__ float block_reduce_sum(float x, float* scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  x = wave_reduce_sum(x);
+  if (lane == 0) scratch[wid] = x;
+  __syncthreads();
+  float r = (scratch[0] + scratch[1]) + (scratch[2] + scratch[3]);
+  __syncthreads();
+  return r;
+}
+
+#define DTX_CDIV(a, b) (((a) + (b) - 1) / (b))

@@ -75,13 +75,14 @@ class FlatAdamW:
         # deterministic order across ranks
         self.named = sorted(named_params, key=lambda kv: kv[0])
         self.params = [p for _, p in self.named]
-        self.numel = sum(p.numel() for p in self.params)
         self.offsets = []
         off = 0
         for p in self.params:
             self.offsets.append(off)
             off += p.numel()
-        self.master = torch.empty(self.numel, dtype=torch.float32,
+        # pad so the fused AdamW kernel can use 4-wide vectors
+        self.numel = (off + 63) // 64 * 64
+        self.master = torch.zeros(self.numel, dtype=torch.float32,
                                   device=device)
         for p, o in zip(self.params, self.offsets):
             self.master[o:o + p.numel()].copy_(

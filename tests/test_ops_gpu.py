@@ -1,0 +1,213 @@
+"""GPU numerics tests: every gfx950 HIP kernel vs the plain-PyTorch fp32
+reference of the same op (SURVEY.md §4(d) strategy). Run via gpurun:
+  python -m pytest tests/test_ops_gpu.py -m gpu -x -q
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from datatunerx_amd import ops
+    from datatunerx_amd.ops import reference as ref
+    DEV = torch.device("cuda:0")
+    assert ops.have_ext(), "HIP extension must be built for GPU tests"
+
+torch.manual_seed(0)
+
+
+def mk(*shape, dtype=torch.bfloat16, scale=1.0):
+    return (torch.randn(*shape, device=DEV, dtype=torch.float32)
+            .mul(scale).to(dtype))
+
+
+def assert_close(got, want, rtol=2e-2, name=""):
+    got, want = got.float(), want.float()
+    denom = want.abs().max().clamp(min=1e-3)
+    err = (got - want).abs().max() / denom
+    assert err < rtol, f"{name}: rel err {err:.4f} (max {denom:.3f})"
+
+
+# ---------------------------------------------------------------- rmsnorm
+@pytest.mark.parametrize("M,H", [(512, 4096), (33, 128), (2048, 5120)])
+def test_rmsnorm_fwd_bwd(M, H):
+    x, w = mk(M, H), mk(H)
+    y, inv = ops.rmsnorm_fwd(x, w, 1e-5)
+    y_ref, inv_ref = ref.rmsnorm_fwd(x.cpu(), w.cpu(), 1e-5)
+    assert_close(y.cpu(), y_ref, name="rmsnorm y")
+    assert_close(inv.cpu(), inv_ref, name="rmsnorm inv")
+    dy = mk(M, H)
+    dx, dw = ops.rmsnorm_bwd(dy, x, w, inv)
+    dx_ref, dw_ref = ref.rmsnorm_bwd(dy.cpu(), x.cpu(), w.cpu(), inv_ref)
+    assert_close(dx.cpu(), dx_ref, name="rmsnorm dx")
+    assert_close(dw.cpu(), dw_ref, rtol=3e-2, name="rmsnorm dw")
+
+
+# ------------------------------------------------------------------- rope
+@pytest.mark.parametrize("D", [128, 64])
+def test_rope(D):
+    B, S, H = 2, 130, 4
+    cos, sin = ref.rope_tables(256, D, device=DEV)
+    x = mk(B, S, H, D)
+    y = ops.rope_fwd(x, cos, sin, pos0=3)
+    y_ref = ref.rope_fwd(x.cpu(), cos.cpu(), sin.cpu(), 3)
+    assert_close(y.cpu(), y_ref, name="rope fwd")
+    dy = mk(B, S, H, D)
+    dx = ops.rope_bwd(dy, cos, sin, pos0=3)
+    dx_ref = ref.rope_bwd(dy.cpu(), cos.cpu(), sin.cpu(), 3)
+    assert_close(dx.cpu(), dx_ref, name="rope bwd")
+
+
+# ----------------------------------------------------------------- swiglu
+def test_swiglu():
+    g, u = mk(1024, 1408), mk(1024, 1408)
+    out = ops.swiglu_fwd(g, u)
+    assert_close(out.cpu(), ref.swiglu_fwd(g.cpu(), u.cpu()), name="swiglu")
+    d = mk(1024, 1408)
+    dg, du = ops.swiglu_bwd(d, g, u)
+    dg_r, du_r = ref.swiglu_bwd(d.cpu(), g.cpu(), u.cpu())
+    assert_close(dg.cpu(), dg_r, name="swiglu dg")
+    assert_close(du.cpu(), du_r, name="swiglu du")
+
+
+# ------------------------------------------------------------------- xent
+@pytest.mark.parametrize("N,V", [(512, 32000), (37, 512)])
+def test_xent(N, V):
+    logits = mk(N, V, scale=3.0)
+    targets = torch.randint(0, V, (N,), device=DEV)
+    targets[::7] = -100
+    loss, lse = ops.softmax_xent_fwd(logits, targets)
+    loss_r, lse_r = ref.softmax_xent_fwd(logits.cpu(), targets.cpu())
+    assert_close(loss.cpu(), loss_r, rtol=1e-2, name="xent loss")
+    assert_close(lse.cpu(), lse_r, rtol=1e-2, name="xent lse")
+    dloss = torch.rand(N, device=DEV)
+    dl = ops.softmax_xent_bwd(logits, targets, lse, dloss)
+    dl_r = ref.softmax_xent_bwd(logits.cpu(), targets.cpu(), lse_r,
+                                dloss.cpu())
+    assert_close(dl.cpu(), dl_r, rtol=3e-2, name="xent dlogits")
+
+
+# ------------------------------------------------------------------- lora
+@pytest.mark.parametrize("M,K,r", [(1024, 4096, 8), (513, 4096, 16),
+                                   (256, 11008, 8), (512, 2048, 64)])
+def test_lora_contract(M, K, r):
+    x, w = mk(M, K, scale=0.3), mk(r, K, scale=0.3)
+    t = ops.lora_contract(x, w)
+    t_ref = ref.lora_contract(x.cpu(), w.cpu())
+    assert_close(t.cpu(), t_ref, name="lora contract")
+
+
+@pytest.mark.parametrize("M,N,r", [(1024, 4096, 8), (511, 1024, 16)])
+def test_lora_expand_add(M, N, r):
+    y = mk(M, N)
+    y0 = y.clone()
+    t = torch.randn(M, r, device=DEV)
+    w = mk(N, r, scale=0.3)
+    ops.lora_expand_add(y, t, w, 0.5)
+    y_ref = ref.lora_expand_add(y0.cpu().clone(), t.cpu(), w.cpu(), 0.5)
+    assert_close(y.cpu(), y_ref, name="lora expand")
+
+
+@pytest.mark.parametrize("M,K,r", [(1024, 4096, 8), (512, 2048, 16)])
+def test_lora_wgrad(M, K, r):
+    t = torch.randn(M, r, device=DEV)
+    x = mk(M, K, scale=0.3)
+    dw = ops.lora_wgrad(t, x, 0.7)
+    dw_ref = ref.lora_wgrad(t.cpu(), x.cpu(), 0.7)
+    assert_close(dw.cpu(), dw_ref, name="lora wgrad")
+
+
+# ------------------------------------------------------------------ adamw
+def test_adamw():
+    n = 4096
+    master = torch.randn(n, device=DEV)
+    p = master.to(torch.bfloat16)
+    g = torch.randn(n, device=DEV)
+    m = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    mc, pc = master.cpu().clone(), p.cpu().clone()
+    gc, mmc, vvc = g.cpu(), m.cpu().clone(), v.cpu().clone()
+    for step in (1, 2, 3):
+        ops.adamw_step(p, master, g, m, v, 1e-3, 0.9, 0.999, 1e-8, 0.01,
+                       step)
+        ref.adamw_step(pc, mc, gc, mmc, vvc, 1e-3, 0.9, 0.999, 1e-8, 0.01,
+                       step)
+    assert_close(master.cpu(), mc, rtol=1e-5, name="adamw master")
+    assert_close(m.cpu(), mmc, rtol=1e-5, name="adamw m")
+
+
+def test_l2_norm():
+    x = torch.randn(1 << 20, device=DEV)
+    got = ops.l2_norm(x)
+    want = ref.l2_norm(x.cpu())
+    assert abs(float(got) - float(want)) / float(want) < 1e-4
+
+
+# -------------------------------------------------------------- attention
+@pytest.mark.parametrize("B,Hq,Hkv,S,Skv,D,causal", [
+    (2, 4, 4, 128, 128, 128, True),
+    (2, 4, 4, 128, 128, 128, False),
+    (1, 8, 2, 256, 256, 128, True),       # GQA
+    (2, 4, 4, 100, 100, 128, True),       # ragged S
+    (1, 2, 2, 64, 192, 128, True),        # Skv > S (cache decode shape)
+    (2, 4, 4, 128, 128, 64, True),        # D=64
+    (1, 4, 4, 1024, 1024, 128, True),     # training shape
+])
+def test_attn_fwd(B, Hq, Hkv, S, Skv, D, causal):
+    q = mk(B, Hq, S, D, scale=0.5)
+    k = mk(B, Hkv, Skv, D, scale=0.5)
+    v = mk(B, Hkv, Skv, D, scale=0.5)
+    o, lse = ops.attn_fwd(q, k, v, causal, 1.0 / math.sqrt(D))
+    o_ref, lse_ref = ref.attn_fwd(q.cpu(), k.cpu(), v.cpu(), causal)
+    assert_close(o.cpu(), o_ref, name="attn o")
+    assert_close(lse.cpu(), lse_ref, rtol=1e-2, name="attn lse")
+
+
+@pytest.mark.parametrize("B,Hq,Hkv,S,D,causal", [
+    (2, 4, 4, 128, 128, True),
+    (2, 4, 4, 128, 128, False),
+    (1, 8, 2, 256, 128, True),
+    (2, 2, 2, 100, 128, True),
+    (1, 4, 4, 512, 64, True),
+])
+def test_attn_bwd(B, Hq, Hkv, S, D, causal):
+    q = mk(B, Hq, S, D, scale=0.5)
+    k = mk(B, Hkv, S, D, scale=0.5)
+    v = mk(B, Hkv, S, D, scale=0.5)
+    o, lse = ops.attn_fwd(q, k, v, causal, 1.0 / math.sqrt(D))
+    do = mk(B, Hq, S, D, scale=0.5)
+    dq, dk, dv = ops.attn_bwd(q, k, v, o, do, lse, causal,
+                              1.0 / math.sqrt(D))
+    o_ref, lse_ref = ref.attn_fwd(q.cpu(), k.cpu(), v.cpu(), causal)
+    dq_r, dk_r, dv_r = ref.attn_bwd(q.cpu(), k.cpu(), v.cpu(), o_ref,
+                                    do.cpu(), lse_ref, causal)
+    assert_close(dq.cpu(), dq_r, rtol=3e-2, name="attn dq")
+    assert_close(dk.cpu(), dk_r, rtol=3e-2, name="attn dk")
+    assert_close(dv.cpu(), dv_r, rtol=3e-2, name="attn dv")
+
+
+# --------------------------------------------- end-to-end GPU train step
+def test_tiny_train_step_gpu():
+    from datatunerx_amd.data.dataset import SFTDataset
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.train.trainer import SFTTrainer, TrainerConfig
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=256,
+                      intermediate_size=512, num_hidden_layers=2,
+                      num_attention_heads=2, num_key_value_heads=2,
+                      max_position_embeddings=512)
+    with torch.device(DEV):
+        model = LlamaForCausalLM(cfg, lora=True, dtype=torch.bfloat16)
+    model.init_random()
+    ds = SFTDataset.synthetic(32, 256, cfg.vocab_size)
+    tr = SFTTrainer(model, ds,
+                    TrainerConfig(output_dir="gpurun_out/test_out",
+                                  max_steps=8, micro_batch_size=4,
+                                  logging_steps=0, learning_rate=1e-3),
+                    device=DEV)
+    it = iter(tr.train_loader)
+    losses = [tr.train_step([next(it)]) for _ in range(8)]
+    assert all(l == l for l in losses), f"NaN in {losses}"
+    assert sum(losses[-2:]) / 2 < sum(losses[:2]) / 2, losses
