@@ -1,0 +1,180 @@
+"""Inference engine for the inference-compare service.
+
+Replaces the reference's Ray Serve `LlamaDeployment` app (baked into the
+checkpoint image — finetunejob_controller.go:378-384):
+native PyTorch-ROCm decode through the same fused HIP kernels as
+training, with a preallocated KV cache. Serving is ephemeral per-job
+evaluation (the job controller tears it down after scoring), so the
+engine favors simplicity + correctness; the TP path for 13B is in
+parallel/tp.py.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+
+from ..data.dataset import IGNORE_INDEX, ByteTokenizer
+from ..data.templates import get_template
+from ..models import (GPT2Config, GPT2ForCausalLM, LlamaConfig,
+                      LlamaForCausalLM, load_adapter)
+
+
+class KVCache:
+    def __init__(self, B, Hkv, max_s, D, device, dtype):
+        self.k = torch.zeros(B, Hkv, max_s, D, device=device, dtype=dtype)
+        self.v = torch.zeros(B, Hkv, max_s, D, device=device, dtype=dtype)
+        self.cur = 0
+
+    def update(self, k, v):
+        S = k.shape[2]
+        self.k[:, :, self.cur:self.cur + S] = k
+        self.v[:, :, self.cur:self.cur + S] = v
+        self.cur += S
+        return (self.k[:, :, :self.cur].contiguous(),
+                self.v[:, :, :self.cur].contiguous())
+
+
+def build_model(name: str, device, adapter_dir: Optional[str] = None,
+                lora_kw: Optional[dict] = None):
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    lora_kw = dict(lora_kw or {})
+    if adapter_dir and not lora_kw:
+        # adapt the LoRA geometry to the checkpoint's adapter_config.json
+        import json
+        import os
+        cfg_path = os.path.join(adapter_dir, "adapter_config.json")
+        if os.path.exists(cfg_path):
+            with open(cfg_path) as f:
+                ac = json.load(f)
+            lora_kw = {"lora_r": int(ac.get("r", 8)),
+                       "lora_alpha": float(ac.get("lora_alpha", 32)),
+                       "lora_targets": tuple(ac.get("target_modules") or
+                                             ("q_proj", "v_proj"))}
+    with torch.device(device):
+        if name in ("llama2-7b", "llama-2-7b"):
+            model = LlamaForCausalLM(LlamaConfig.llama2_7b(**lora_kw),
+                                     dtype=dtype)
+        elif name in ("llama2-13b", "llama-2-13b"):
+            model = LlamaForCausalLM(LlamaConfig.llama2_13b(**lora_kw),
+                                     dtype=dtype)
+        elif name == "llama-tiny":
+            model = LlamaForCausalLM(LlamaConfig.tiny(**lora_kw),
+                                     dtype=dtype)
+        elif name in ("gpt2-small", "gpt2"):
+            model = GPT2ForCausalLM(GPT2Config.small(), dtype=dtype)
+        elif name == "gpt2-tiny":
+            model = GPT2ForCausalLM(GPT2Config.tiny(), dtype=dtype)
+        else:
+            raise ValueError(f"unknown model {name!r}")
+    model.init_random()          # no network: base weights random-init
+    if adapter_dir:
+        load_adapter(model, adapter_dir)
+    model.eval()
+    return model
+
+
+class InferenceEngine:
+    def __init__(self, model, tokenizer=None, template: str = "llama2",
+                 device=None):
+        self.model = model
+        self.tok = tokenizer or ByteTokenizer()
+        self.template = template
+        self.device = device or next(model.parameters()).device
+        self.is_llama = isinstance(model, LlamaForCausalLM)
+
+    # ------------------------------------------------------------ chat
+    def chat(self, messages: List[dict], max_tokens: int = 64,
+             temperature: float = 0.0, top_p: float = 1.0) -> str:
+        """messages: [{role, content}] -> completion text."""
+        system = ""
+        history = []
+        query = ""
+        pending_user = None
+        for m in messages:
+            if m["role"] == "system":
+                system = m["content"]
+            elif m["role"] == "user":
+                pending_user = m["content"]
+            elif m["role"] == "assistant" and pending_user is not None:
+                history.append((pending_user, m["content"]))
+                pending_user = None
+        query = pending_user or ""
+        t = get_template(self.template)
+        src, _ = t.encode_oneturn(self.tok, query, "", history, system)
+        ids = self.generate(src, max_tokens, temperature, top_p)
+        return self.tok.decode(ids)
+
+    @torch.no_grad()
+    def generate(self, prompt_ids: List[int], max_new_tokens: int = 64,
+                 temperature: float = 0.0, top_p: float = 1.0) -> List[int]:
+        ids = torch.tensor([prompt_ids], dtype=torch.long,
+                           device=self.device)
+        out = []
+        if self.is_llama:
+            cfg = self.model.cfg
+            max_s = min(cfg.max_position_embeddings,
+                        len(prompt_ids) + max_new_tokens + 8)
+            caches = [KVCache(1, cfg.num_key_value_heads, max_s,
+                              cfg.head_dim, self.device,
+                              next(self.model.parameters()).dtype)
+                      for _ in range(cfg.num_hidden_layers)]
+            pos = 0
+            cur = ids
+            for _ in range(max_new_tokens):
+                logits = self.model(cur, pos0=pos, kv_caches=caches)
+                pos += cur.shape[1]
+                if pos >= max_s:
+                    break
+                nxt = self._sample(logits[0, -1], temperature, top_p)
+                if nxt == self.tok.eos_token_id:
+                    break
+                out.append(nxt)
+                cur = torch.tensor([[nxt]], dtype=torch.long,
+                                   device=self.device)
+        else:
+            seq = list(prompt_ids)
+            for _ in range(max_new_tokens):
+                cur = torch.tensor([seq], dtype=torch.long,
+                                   device=self.device)
+                logits = self.model(cur)
+                nxt = self._sample(logits[0, -1], temperature, top_p)
+                if nxt == self.tok.eos_token_id:
+                    break
+                out.append(nxt)
+                seq.append(nxt)
+        return out
+
+    def _sample(self, logits: torch.Tensor, temperature: float,
+                top_p: float) -> int:
+        if temperature <= 0.0:
+            return int(logits.argmax())
+        probs = torch.softmax(logits.float() / temperature, dim=-1)
+        if top_p < 1.0:
+            sp, si = probs.sort(descending=True)
+            cum = sp.cumsum(0)
+            keep = cum <= top_p
+            keep[0] = True
+            probs = torch.zeros_like(probs).scatter_(0, si[keep], sp[keep])
+            probs /= probs.sum()
+        return int(torch.multinomial(probs, 1))
+
+    # ----------------------------------------------------------- score
+    @torch.no_grad()
+    def perplexity(self, texts: List[str]) -> float:
+        """Mean perplexity over texts (built-in Scoring metric)."""
+        losses = []
+        for t in texts:
+            ids = [self.tok.bos_token_id] + self.tok.encode(t)
+            ids = ids[:512]
+            if len(ids) < 2:
+                continue
+            x = torch.tensor([ids], dtype=torch.long, device=self.device)
+            loss = self.model(x, labels=x)
+            losses.append(float(loss))
+        if not losses:
+            return float("inf")
+        mean = sum(losses) / len(losses)
+        return math.exp(min(mean, 30.0))

@@ -1,0 +1,103 @@
+"""Inference-compare HTTP service: /chat/completions (the endpoint the
+reference's Scoring hits — finetunejob_controller.go:433), /v1/score
+(built-in scoring metric), /health.
+
+Std-lib ThreadingHTTPServer (no web-framework dependency): the service
+is an ephemeral per-job evaluation endpoint (torn down after scoring,
+SURVEY.md §3.4), so a single-model, low-QPS server is the right shape.
+Generation requests are serialized through an engine lock (one GPU, one
+model instance).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import threading
+import time
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+import torch
+
+
+def build_handler(engine):
+    lock = threading.Lock()
+
+    class Handler(BaseHTTPRequestHandler):
+        def log_message(self, *a):       # quiet
+            pass
+
+        def _send(self, code: int, obj: dict):
+            body = json.dumps(obj).encode()
+            self.send_response(code)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def do_GET(self):
+            if self.path == "/health":
+                self._send(200, {"status": "ok"})
+            else:
+                self._send(404, {"error": "not found"})
+
+        def do_POST(self):
+            try:
+                n = int(self.headers.get("Content-Length", "0"))
+                body = json.loads(self.rfile.read(n) or b"{}")
+            except Exception:
+                self._send(400, {"error": "bad json"})
+                return
+            try:
+                if self.path == "/chat/completions":
+                    with lock:
+                        text = engine.chat(
+                            body.get("messages", []),
+                            int(body.get("max_tokens", 64)),
+                            float(body.get("temperature", 0.0)),
+                            float(body.get("top_p", 1.0)))
+                    self._send(200, {
+                        "id": f"chatcmpl-{int(time.time()*1000)}",
+                        "object": "chat.completion",
+                        "model": body.get("model", "finetuned"),
+                        "choices": [{
+                            "index": 0,
+                            "message": {"role": "assistant",
+                                        "content": text},
+                            "finish_reason": "stop",
+                        }],
+                    })
+                elif self.path == "/v1/score":
+                    with lock:
+                        ppl = engine.perplexity(body.get("texts", []))
+                    self._send(200, {"perplexity": ppl})
+                else:
+                    self._send(404, {"error": "not found"})
+            except Exception as e:      # surface engine errors as 500s
+                self._send(500, {"error": f"{type(e).__name__}: {e}"})
+
+    return Handler
+
+
+def serve_forever(engine, host: str, port: int):
+    httpd = ThreadingHTTPServer((host, port), build_handler(engine))
+    httpd.serve_forever()
+
+
+def main(argv=None):
+    from .engine import InferenceEngine, build_model
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--model", default="llama2-7b")
+    ap.add_argument("--adapter", default=None)
+    ap.add_argument("--template", default="llama2")
+    args = ap.parse_args(argv)
+    device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+    model = build_model(args.model, device, adapter_dir=args.adapter)
+    engine = InferenceEngine(model, template=args.template, device=device)
+    serve_forever(engine, args.host, args.port)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,155 @@
+"""Trainer process entrypoint — the replacement for the reference's
+RayJob entrypoint `python /tuning/train.py ...` (finetune_controller.go:
+451-516 builds the flag list; train.py:308-390 is the driver).
+
+Launched by the Finetune controller as one process per GPU (torchrun-
+style env: RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT, HIP_VISIBLE_DEVICES
+pinned per rank). Writes a status JSON file the controller polls —
+replacing the reference's pod-exec of /home/ray/checkpoint_path
+(finetune_controller.go:278-305).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import sys
+import traceback
+
+
+def write_status(path: str, state: str, **kw):
+    if not path:
+        return
+    tmp = path + ".tmp"
+    with open(tmp, "w") as f:
+        json.dump({"state": state, **kw}, f)
+    os.replace(tmp, path)
+
+
+def main(argv=None):
+    import torch
+
+    from ..data.dataset import ByteTokenizer, SFTDataset
+    from ..models import (GPT2Config, GPT2ForCausalLM, LlamaConfig,
+                          LlamaForCausalLM)
+    from ..parallel.ddp import init_distributed, is_main
+    from .args import get_train_args
+    from .trainer import SFTTrainer, TrainerConfig
+
+    margs, fargs, dargs = get_train_args(argv)
+    status_file = os.environ.get("DTX_STATUS_FILE", "")
+    rank, world, local_rank, device = init_distributed()
+    try:
+        if rank == 0:
+            write_status(status_file, "Running")
+        torch.manual_seed(fargs.seed)
+        dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+        name = margs.model_name_or_path
+        lora = fargs.finetuning_type == "lora"
+        lora_kw = dict(lora_r=fargs.lora_rank, lora_alpha=fargs.lora_alpha,
+                       lora_dropout=fargs.lora_dropout,
+                       lora_targets=fargs.lora_targets)
+        with torch.device(device):
+            if name in ("llama2-7b", "llama-2-7b"):
+                model = LlamaForCausalLM(LlamaConfig.llama2_7b(
+                    gradient_checkpointing=fargs.gradient_checkpointing,
+                    **lora_kw), lora=lora, dtype=dtype)
+            elif name in ("llama2-13b", "llama-2-13b"):
+                model = LlamaForCausalLM(LlamaConfig.llama2_13b(
+                    gradient_checkpointing=fargs.gradient_checkpointing,
+                    **lora_kw), lora=lora, dtype=dtype)
+            elif name == "llama-tiny":
+                model = LlamaForCausalLM(LlamaConfig.tiny(**lora_kw),
+                                         lora=lora, dtype=dtype)
+            elif name in ("gpt2-small", "gpt2"):
+                model = GPT2ForCausalLM(GPT2Config.small(
+                    lora_r=fargs.lora_rank, lora_alpha=fargs.lora_alpha,
+                    lora_dropout=fargs.lora_dropout), dtype=dtype)
+            elif name == "gpt2-tiny":
+                model = GPT2ForCausalLM(GPT2Config.tiny(), dtype=dtype)
+            else:
+                raise SystemExit(f"unknown model {name!r}")
+        # no network: checkpoints load from local dirs, else random init
+        if margs.checkpoint_dir and os.path.isdir(margs.checkpoint_dir):
+            from safetensors.torch import load_file
+            sd = load_file(os.path.join(margs.checkpoint_dir,
+                                        "model.safetensors"))
+            model.load_state_dict(sd, strict=False)
+        else:
+            model.init_random(seed=fargs.seed)
+
+        tok = ByteTokenizer()
+        vocab = getattr(model.cfg, "vocab_size")
+        if dargs.dataset_path and os.path.exists(dargs.dataset_path):
+            ds = SFTDataset.from_csv(
+                dargs.dataset_path, tok,
+                column_map={"instruction": dargs.instruction_column,
+                            "response": dargs.response_column},
+                template_name=dargs.prompt_template,
+                cutoff_len=dargs.block_size)
+        else:
+            n = dargs.synthetic_examples or 256
+            ds = SFTDataset.synthetic(n, dargs.block_size, vocab,
+                                      seed=fargs.seed)
+        eval_ds = None
+        if dargs.eval_dataset_path and os.path.exists(dargs.eval_dataset_path):
+            eval_ds = SFTDataset.from_csv(
+                dargs.eval_dataset_path, tok,
+                column_map={"instruction": dargs.instruction_column,
+                            "response": dargs.response_column},
+                template_name=dargs.prompt_template,
+                cutoff_len=dargs.block_size)
+
+        tcfg = TrainerConfig(
+            output_dir=fargs.output_dir,
+            learning_rate=fargs.learning_rate,
+            weight_decay=fargs.weight_decay,
+            max_grad_norm=fargs.max_grad_norm,
+            warmup_ratio=fargs.warmup_ratio,
+            lr_scheduler_type=fargs.lr_scheduler_type,
+            num_train_epochs=fargs.num_train_epochs,
+            max_steps=fargs.max_steps,
+            micro_batch_size=fargs.per_device_train_batch_size,
+            gradient_accumulation_steps=fargs.gradient_accumulation_steps,
+            logging_steps=fargs.logging_steps,
+            eval_steps=fargs.eval_steps, save_steps=fargs.save_steps,
+            seed=fargs.seed,
+            metrics_export_address=fargs.metrics_export_address,
+            uid=fargs.uid, lora_r=fargs.lora_rank,
+            lora_alpha=fargs.lora_alpha, lora_dropout=fargs.lora_dropout,
+            lora_targets=fargs.lora_targets, base_model=name)
+        trainer = SFTTrainer(model, ds, tcfg, eval_dataset=eval_ds,
+                             device=device, rank=rank, world_size=world,
+                             pad_token_id=tok.pad_token_id)
+        final_loss = trainer.train()
+        ckpt_path = None
+        if is_main():
+            ckpt_path = trainer.save_checkpoint(
+                os.path.join(fargs.output_dir, "checkpoint"))
+            if fargs.storage_path:
+                # "upload": copy to the storage root (replaces S3 push,
+                # train.py:302-305)
+                import shutil
+                dst = os.path.join(fargs.storage_path, fargs.uid or "ckpt",
+                                   "checkpoint")
+                if os.path.abspath(dst) != os.path.abspath(ckpt_path):
+                    shutil.copytree(ckpt_path, dst, dirs_exist_ok=True)
+                    ckpt_path = dst
+            write_status(status_file, "Successful",
+                         checkpoint_path=ckpt_path,
+                         final_loss=final_loss,
+                         eval=trainer.evaluate() if eval_ds else None)
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.barrier()
+            dist.destroy_process_group()
+        return 0
+    except Exception:
+        traceback.print_exc()
+        if rank == 0:
+            write_status(status_file, "Failed", error=traceback.format_exc())
+        return 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -10,6 +10,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: needs an MI355X GPU (run via gpurun)")
+    config.addinivalue_line(
+        "markers", "slow: multi-process integration test (>30s)")
 
 
 def pytest_collection_modifyitems(config, items):
