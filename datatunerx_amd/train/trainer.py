@@ -49,6 +49,8 @@ class TrainerConfig:
     logging_steps: int = 10
     eval_steps: int = 0                    # 0 = eval at end only
     save_steps: int = 0                    # 0 = terminal save only
+    optimizer_mode: str = "auto"           # auto | flat | overlap | zero1
+    comm_bucket_bytes: int = 128 << 20
     seed: int = 42
     metrics_export_address: Optional[str] = None
     uid: str = ""
@@ -63,74 +65,199 @@ class TrainerConfig:
 class FlatAdamW:
     """Flat-buffer AdamW over the model's trainable params.
 
-    Buffers: master fp32 (authoritative weights), m, v, grad fp32.
-    step(): optional all-reduce -> clip -> fused adamw -> write back to
-    the (bf16) module params.
+    Layout: every trainable param's storage is re-pointed into ONE flat
+    bf16/fp32 buffer (`param_flat`), laid out in REVERSED registration
+    order — backward produces gradients roughly front-to-back, which
+    makes comm buckets contiguous. Master fp32 weights + m/v + fp32 grad
+    accumulator are flat too; the fused AdamW HIP kernel updates
+    master and writes the low-precision params in ONE pass, in place.
+
+    Gradient folding happens inside `post_accumulate_grad_hook`s (fp32
+    accumulation across the grad-accum window, p.grad freed
+    immediately).
+
+    Modes (cfg via ctor args):
+    - "flat":   one fused all-reduce of the whole fp32 grad at step()
+                (LoRA ~4.2M params: latency-bound over xGMI — optimal).
+    - "overlap": buckets of `bucket_bytes` all-reduced asynchronously as
+                the LAST micro-batch's backward fills them (full-param
+                models: comm hides behind compute).
+    - "zero1":  fp32 grad reduce-scattered across ranks; each rank runs
+                AdamW on its 1/world shard of master/m/v, then the bf16
+                params are all-gathered (DeepSpeed-ZeRO-1-style optimizer
+                state sharding for the 13B full-param config,
+                SURVEY.md §2.2).
     """
 
     def __init__(self, named_params, device, cfg: TrainerConfig,
-                 sync: Optional[GradSynchronizer] = None):
+                 sync: Optional[GradSynchronizer] = None,
+                 mode: str = "flat", bucket_bytes: int = 128 << 20,
+                 world_size: int = 1):
+        import torch.distributed as dist
         self.cfg = cfg
         self.sync = sync
-        # deterministic order across ranks
-        self.named = sorted(named_params, key=lambda kv: kv[0])
+        self.mode = mode if (world_size > 1 or mode == "flat") else "flat"
+        self.world = world_size
+        self.named = list(named_params)           # registration order
+        self.named.reverse()                      # ≈ backward order
         self.params = [p for _, p in self.named]
+        if not self.params:
+            raise ValueError("no trainable parameters")
         self.offsets = []
         off = 0
         for p in self.params:
             self.offsets.append(off)
             off += p.numel()
-        # pad so the fused AdamW kernel can use 4-wide vectors
-        self.numel = (off + 63) // 64 * 64
-        self.master = torch.zeros(self.numel, dtype=torch.float32,
-                                  device=device)
-        for p, o in zip(self.params, self.offsets):
-            self.master[o:o + p.numel()].copy_(
-                p.detach().reshape(-1).float())
-        self.m = torch.zeros_like(self.master)
-        self.v = torch.zeros_like(self.master)
-        self.grad = torch.zeros_like(self.master)
-        self.t = 0
-
-    def accumulate_grads_(self):
-        """Fold p.grad into the flat fp32 grad buffer and clear p.grad
-        (called once per micro-batch: fp32 accumulation across the
-        gradient-accumulation window)."""
-        for p, o in zip(self.params, self.offsets):
-            if p.grad is not None:
-                self.grad[o:o + p.numel()].add_(
-                    p.grad.detach().reshape(-1).float())
-                p.grad = None
-
-    def grad_norm_and_clip_(self) -> float:
-        norm = ops.l2_norm(self.grad)
-        max_norm = self.cfg.max_grad_norm
-        if max_norm and max_norm > 0:
-            scale = max_norm / (float(norm) + 1e-6)
-            if scale < 1.0:
-                self.grad.mul_(scale)
-        return float(norm)
-
-    def step(self, lr: float) -> float:
-        if self.sync is not None:
-            self.sync.allreduce_flat_(self.grad)
-        gnorm = self.grad_norm_and_clip_()
-        self.t += 1
-        # fused path wants flat bf16 param target; update master then
-        # scatter back per-tensor (cheap at LoRA size).
-        pbf = torch.empty_like(self.master,
-                               dtype=self.params[0].dtype)
-        ops.adamw_step(pbf, self.master, self.grad, self.m, self.v,
-                       lr, self.cfg.betas[0], self.cfg.betas[1],
-                       self.cfg.eps, self.cfg.weight_decay, self.t)
+        # pad so the fused AdamW kernel can use 4-wide vectors and the
+        # zero1 shard divides evenly
+        align = 64 * max(1, world_size)
+        self.numel = (off + align - 1) // align * align
+        pdtype = self.params[0].dtype
+        self.param_flat = torch.zeros(self.numel, dtype=pdtype,
+                                      device=device)
         with torch.no_grad():
             for p, o in zip(self.params, self.offsets):
-                p.copy_(pbf[o:o + p.numel()].view_as(p))
+                self.param_flat[o:o + p.numel()].copy_(
+                    p.detach().reshape(-1))
+                p.data = self.param_flat[o:o + p.numel()].view_as(p)
+        self.grad = torch.zeros(self.numel, dtype=torch.float32,
+                                device=device)
+        if self.mode == "zero1":
+            self.shard_n = self.numel // world_size
+            self.shard_off = self.shard_n * dist.get_rank() \
+                if dist.is_initialized() else 0
+            mslice = self.param_flat[
+                self.shard_off:self.shard_off + self.shard_n]
+        else:
+            self.shard_n, self.shard_off = self.numel, 0
+            mslice = self.param_flat
+        self.master = mslice.float()
+        self.m = torch.zeros_like(self.master)
+        self.v = torch.zeros_like(self.master)
+        self.t = 0
+        # ---- grad-fold hooks + overlap buckets
+        self._final_micro = False
+        self._handles = []
+        self._bucket_of = {}
+        if self.mode == "overlap":
+            elems = max(1, bucket_bytes // 4)
+            buckets = []      # list of [start, end, n_params, seen]
+            for p, o in zip(self.params, self.offsets):
+                if not buckets or o + p.numel() - buckets[-1][0] > elems:
+                    buckets.append([o, o + p.numel(), 0, 0])
+                else:
+                    buckets[-1][1] = o + p.numel()
+                buckets[-1][2] += 1
+                self._bucket_of[p] = buckets[-1]
+            buckets[-1][1] = self.numel
+            self.buckets = buckets
+        self._install_hooks()
+
+    def _install_hooks(self):
+        for p, o in zip(self.params, self.offsets):
+            n = p.numel()
+            dst = self.grad[o:o + n]
+
+            def hook(param, dst=dst, n=n):
+                g = param.grad
+                if g is None:
+                    return
+                dst.add_(g.detach().reshape(-1))
+                param.grad = None
+                if self._final_micro and self.mode == "overlap":
+                    b = self._bucket_of[param]
+                    b[3] += 1
+                    if b[3] == b[2]:
+                        self._launch_bucket(b)
+
+            p.register_post_accumulate_grad_hook(hook)
+
+    def _launch_bucket(self, b):
+        import torch.distributed as dist
+        b[3] = 0
+        if self.world <= 1 or not dist.is_initialized():
+            return
+        chunk = self.grad[b[0]:b[1]]
+        self._handles.append(
+            dist.all_reduce(chunk, op=dist.ReduceOp.AVG, async_op=True))
+
+    def mark_final_microbatch(self):
+        """Call before the LAST micro-batch's backward of the window so
+        overlap-mode buckets all-reduce as backward fills them."""
+        self._final_micro = True
+
+    def accumulate_grads_(self):
+        """Kept for API compatibility — folding is done by hooks."""
+
+    def _sync_grad(self):
+        import torch.distributed as dist
+        if self.world <= 1 or not dist.is_initialized():
+            return
+        if self.mode == "flat":
+            if self.sync is not None:
+                self.sync.allreduce_flat_(self.grad)
+            else:
+                dist.all_reduce(self.grad, op=dist.ReduceOp.AVG)
+        elif self.mode == "overlap":
+            for h in self._handles:
+                h.wait()
+            self._handles.clear()
+        elif self.mode == "zero1":
+            shard = torch.empty(self.shard_n, dtype=torch.float32,
+                                device=self.grad.device)
+            if dist.get_backend() == "gloo":
+                # gloo lacks reduce_scatter_tensor: all-reduce then slice
+                dist.all_reduce(self.grad, op=dist.ReduceOp.AVG)
+                shard.copy_(self.grad[self.shard_off:
+                                      self.shard_off + self.shard_n])
+            else:
+                dist.reduce_scatter_tensor(shard, self.grad,
+                                           op=dist.ReduceOp.AVG)
+            self._grad_shard = shard
+
+    def _grad_sq_sum(self) -> torch.Tensor:
+        import torch.distributed as dist
+        if self.mode == "zero1":
+            sq = self._grad_shard.float().pow(2).sum()
+            if dist.is_initialized():
+                dist.all_reduce(sq, op=dist.ReduceOp.SUM)
+            return sq
+        return self.grad.pow(2).sum()
+
+    def step(self, lr: float) -> float:
+        import torch.distributed as dist
+        self._sync_grad()
+        self._final_micro = False
+        gnorm = float(self._grad_sq_sum().sqrt())
+        g = self._grad_shard if self.mode == "zero1" else self.grad
+        max_norm = self.cfg.max_grad_norm
+        if max_norm and max_norm > 0:
+            scale = max_norm / (gnorm + 1e-6)
+            if scale < 1.0:
+                g.mul_(scale)
+        self.t += 1
+        pslice = self.param_flat[self.shard_off:
+                                 self.shard_off + self.shard_n]
+        ops.adamw_step(pslice, self.master, g, self.m, self.v,
+                       lr, self.cfg.betas[0], self.cfg.betas[1],
+                       self.cfg.eps, self.cfg.weight_decay, self.t)
+        if self.mode == "zero1" and dist.is_initialized():
+            if dist.get_backend() == "gloo":
+                shards = [torch.empty_like(pslice)
+                          for _ in range(self.world)]
+                dist.all_gather(shards, pslice.contiguous())
+                for i, s in enumerate(shards):
+                    self.param_flat[i * self.shard_n:
+                                    (i + 1) * self.shard_n].copy_(s)
+            else:
+                dist.all_gather_into_tensor(self.param_flat,
+                                            pslice.contiguous())
         self.grad.zero_()
         return gnorm
 
     def state_dict(self):
-        return {"master": self.master, "m": self.m, "v": self.v, "t": self.t}
+        return {"master": self.master, "m": self.m, "v": self.v,
+                "t": self.t, "mode": self.mode}
 
     def load_state_dict(self, sd):
         self.master.copy_(sd["master"])
@@ -138,8 +265,9 @@ class FlatAdamW:
         self.v.copy_(sd["v"])
         self.t = sd["t"]
         with torch.no_grad():
-            for p, o in zip(self.params, self.offsets):
-                p.copy_(self.master[o:o + p.numel()].view_as(p).to(p.dtype))
+            self.param_flat[self.shard_off:
+                            self.shard_off + self.shard_n].copy_(
+                self.master.to(self.param_flat.dtype))
 
 
 def lr_at(step: int, total: int, base_lr: float, warmup_ratio: float,
@@ -169,8 +297,16 @@ class SFTTrainer:
         self.eval_dataset = eval_dataset
         self.pad_token_id = pad_token_id
         sync = GradSynchronizer(world_size) if world_size > 1 else None
-        self.opt = FlatAdamW(model.trainable_parameters(), self.device,
-                             cfg, sync)
+        named = model.trainable_parameters()
+        mode = cfg.optimizer_mode
+        if mode == "auto":
+            # LoRA-size trainables (≤64M params): one fused latency-bound
+            # all-reduce. Full-param models: overlap buckets with backward.
+            n_train = sum(p.numel() for _, p in named)
+            mode = "flat" if n_train <= (64 << 20) else "overlap"
+        self.opt = FlatAdamW(named, self.device, cfg, sync, mode=mode,
+                             bucket_bytes=cfg.comm_bucket_bytes,
+                             world_size=world_size)
         steps_per_epoch = max(1, len(train_dataset) //
                               (cfg.micro_batch_size * world_size *
                                cfg.gradient_accumulation_steps))
@@ -212,11 +348,12 @@ class SFTTrainer:
         cfg = self.cfg
         acc = cfg.gradient_accumulation_steps
         total = 0.0
-        for mb in micro_batches:
+        for i, mb in enumerate(micro_batches):
+            if i == len(micro_batches) - 1:
+                self.opt.mark_final_microbatch()
             loss = self.model(mb["input_ids"], labels=mb["labels"])
             (loss / acc).backward()
             total += float(loss.detach())
-            self.opt.accumulate_grads_()
         lr = lr_at(self.global_step, self.total_steps, cfg.learning_rate,
                    cfg.warmup_ratio, cfg.lr_scheduler_type)
         self.opt.step(lr)

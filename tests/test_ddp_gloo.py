@@ -31,18 +31,26 @@ ds = SFTDataset.synthetic(32, 32, 512, seed=0)
 tr = SFTTrainer(model, ds,
                 TrainerConfig(output_dir=os.environ["DTX_OUT"] + f"/r{rank}",
                               max_steps=3, micro_batch_size=2,
+                              gradient_accumulation_steps=int(
+                                  os.environ.get("DTX_ACC", "1")),
+                              optimizer_mode=os.environ.get(
+                                  "DTX_OPT_MODE", "auto"),
+                              comm_bucket_bytes=1 << 12,
                               logging_steps=0, learning_rate=1e-3,
                               lora_dropout=0.0),
                 device=device, rank=rank, world_size=world)
 tr.train()
-out = {"master": tr.opt.master.tolist()[:64],
-       "master_sum": float(tr.opt.master.abs().sum())}
+out = {"params": tr.opt.param_flat.tolist()[:64],
+       "param_sum": float(tr.opt.param_flat.abs().sum())}
 with open(os.environ["DTX_OUT"] + f"/rank{rank}.json", "w") as f:
     json.dump(out, f)
+import torch.distributed as dist
+if dist.is_initialized():
+    dist.destroy_process_group()
 """
 
 
-def run_workers(nproc, out_dir, port):
+def run_workers(nproc, out_dir, port, extra_env=None):
     script = os.path.join(out_dir, "worker.py")
     with open(script, "w") as f:
         f.write(WORKER)
@@ -53,6 +61,7 @@ def run_workers(nproc, out_dir, port):
                     "LOCAL_RANK": str(rank),
                     "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
                     "DTX_ROOT": ROOT, "DTX_OUT": out_dir})
+        env.update(extra_env or {})
         procs.append(subprocess.Popen([sys.executable, script], env=env))
     for p in procs:
         assert p.wait(timeout=300) == 0
@@ -61,11 +70,29 @@ def run_workers(nproc, out_dir, port):
 
 
 def test_ddp_two_ranks_converge_identically(tmp_path):
-    """Both ranks must hold identical optimizer state after sync steps."""
+    """Both ranks must hold identical trained params after sync steps."""
     outs = run_workers(2, str(tmp_path), 29712)
-    assert outs[0]["master"] == pytest.approx(outs[1]["master"], abs=1e-7)
-    assert outs[0]["master_sum"] == pytest.approx(outs[1]["master_sum"],
-                                                  rel=1e-6)
+    assert outs[0]["params"] == pytest.approx(outs[1]["params"], abs=1e-7)
+    assert outs[0]["param_sum"] == pytest.approx(outs[1]["param_sum"],
+                                                 rel=1e-6)
+
+
+@pytest.mark.parametrize("mode,port", [("overlap", 29716), ("zero1", 29718)])
+def test_ddp_modes_match_flat(tmp_path, mode, port):
+    """Overlapped bucketed all-reduce and ZeRO-1 sharding must produce the
+    same trained parameters as the flat fused all-reduce (grad-accum 2
+    exercises the final-microbatch hook path)."""
+    flat_dir = tmp_path / "flat"
+    mode_dir = tmp_path / mode
+    flat_dir.mkdir(), mode_dir.mkdir()
+    base = run_workers(2, str(flat_dir), port,
+                       {"DTX_OPT_MODE": "flat", "DTX_ACC": "2"})
+    outs = run_workers(2, str(mode_dir), port + 40,
+                       {"DTX_OPT_MODE": mode, "DTX_ACC": "2"})
+    assert outs[0]["params"] == pytest.approx(outs[1]["params"], abs=1e-7)
+    assert outs[0]["params"] == pytest.approx(base[0]["params"], abs=1e-5)
+    assert outs[0]["param_sum"] == pytest.approx(base[0]["param_sum"],
+                                                 rel=1e-4)
 
 
 def _allreduce_worker(rank, world, port, q):
