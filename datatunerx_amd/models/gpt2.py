@@ -76,11 +76,11 @@ class GPT2Block(nn.Module):
         B, S, _ = x.shape
         H, D = self.cfg.num_attention_heads, self.cfg.head_dim
         hx = self.ln_1(x)
-        q = self.q_proj(hx).view(B, S, H, D).transpose(1, 2)
-        k = self.k_proj(hx).view(B, S, H, D).transpose(1, 2)
-        v = self.v_proj(hx).view(B, S, H, D).transpose(1, 2)
-        o = attention(q, k, v, causal=True)
-        o = o.transpose(1, 2).reshape(B, S, H * D)
+        q = self.q_proj(hx).view(B, S, H, D)
+        k = self.k_proj(hx).view(B, S, H, D)
+        v = self.v_proj(hx).view(B, S, H, D)
+        o = attention(q, k, v, causal=True)      # BSHD in/out
+        o = o.reshape(B, S, H * D)
         x = x + self.o_proj(o)
         hx = self.ln_2(x)
         return x + self.mlp_proj(F.gelu(self.mlp_fc(hx)))

@@ -93,14 +93,10 @@ class LlamaAttention(nn.Module):
         v = self.v_proj(x).view(B, S, Hkv, D)
         q = rope(q, cos, sin, pos0)
         k = rope(k, cos, sin, pos0)
-        q = q.transpose(1, 2)                    # [B,H,S,D]
-        k = k.transpose(1, 2)
-        v = v.transpose(1, 2)
         if kv_cache is not None:
-            k, v = kv_cache.update(k, v)         # serving path
-        o = attention(q, k, v, causal=True)
-        o = o.transpose(1, 2).reshape(B, S, H * D)
-        return self.o_proj(o)
+            k, v = kv_cache.update(k, v)         # serving path (BSHD)
+        o = attention(q, k, v, causal=True)      # BSHD in/out, no copies
+        return self.o_proj(o.reshape(B, S, H * D))
 
 
 class LlamaMLP(nn.Module):

@@ -35,7 +35,9 @@ class FrozenLinear(nn.Module):
             requires_grad=False)
 
     def forward(self, x):
-        return x @ self.weight.t()
+        # F.linear routes to addmm/hipBLASLt: +18% over x @ w.t() on the
+        # [16K,4096]x[4096,4096] shape (measured, tools/bench_gemm.py)
+        return F.linear(x, self.weight)
 
 
 class LoRAFunctionWithDropout(torch.autograd.Function):
@@ -49,7 +51,7 @@ class LoRAFunctionWithDropout(torch.autograd.Function):
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
         xd = x2 * mask if mask is not None else x2
-        y = x2 @ w.t()
+        y = F.linear(x2, w)
         t = lora_contract(xd, a)
         lora_expand_add(y, t, b, scale)
         ctx.save_for_backward(x2, w, a, b, t,

@@ -106,10 +106,12 @@ def softmax_xent_bwd(logits, targets, lse, dloss, ignore_index: int = -100):
 
 # --------------------------------------------------------- attention
 def attn_fwd(q, k, v, causal: bool = True, scale: float | None = None):
+    """BSHD: q [B,S,Hq,D], k/v [B,Skv,Hkv,D] -> (o [B,S,Hq,D], lse)."""
     if scale is None:
         scale = 1.0 / (q.shape[-1] ** 0.5)
     if _gpu(q):
-        return _EXT.attn_fwd(q, k, v, causal, scale)
+        vt = _EXT.transpose_sd(v)        # [B,Hkv,D,Skv] for the PV tiles
+        return _EXT.attn_fwd(q, k, vt, causal, scale)
     return ref.attn_fwd(q, k, v, causal, scale)
 
 

@@ -66,6 +66,8 @@ def swiglu(gate, up):
 
 
 class Attention(torch.autograd.Function):
+    """BSHD flash attention (no transpose copies around the kernel)."""
+
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
@@ -83,7 +85,7 @@ class Attention(torch.autograd.Function):
 
 
 def attention(q, k, v, causal: bool = True, scale: float | None = None):
-    """q [B,Hq,S,D], k/v [B,Hkv,S,D] -> o [B,Hq,S,D]."""
+    """q [B,S,Hq,D], k/v [B,Skv,Hkv,D] -> o [B,S,Hq,D]."""
     if scale is None:
         scale = 1.0 / (q.shape[-1] ** 0.5)
     return Attention.apply(q, k, v, causal, scale)
@@ -126,7 +128,7 @@ class LoRALinear(torch.autograd.Function):
     def forward(ctx, x, w, a, b, scale):
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
-        y = x2 @ w.t()
+        y = torch.nn.functional.linear(x2, w)
         t = lora_contract(x2, a)                 # [M,r] f32
         lora_expand_add(y, t, b, scale)          # y += s * t @ B^T
         ctx.save_for_backward(x2, w, a, b, t)

@@ -1,382 +1,521 @@
-// Flash-attention backward for gfx950, FA2-style recompute from saved
-// lse. Three kernels:
-//   1. preprocess: delta = rowsum(dO * O)
-//   2. dK/dV: per KV-tile block, loop q-tiles; S^T recomputed as
-//      mfma(K, Q^T) so P^T lands directly in the contraction layout for
-//      dV = P^T dO and dK = dS^T Q (no cross-operand transposes beyond
-//      the LDS staging of Q/dO both row-major and transposed).
-//   3. dQ: per Q-tile block, loop kv-tiles; dQ = dS K.
-// GQA: kernel 2 writes per-Q-head partials [B,Hq,Skv,D]; the host sums
-// over the group (deterministic; no atomics anywhere).
+// Flash-attention backward for gfx950 (CDNA4), bf16 I/O, fp32 math.
+//
+// v2 — 8-wave 32x32-MFMA structure matching attn_fwd.hip, BSHD layout.
+// Recompute strategy: P = exp2(S*scale*log2e - lse*log2e) from the saved
+// lse (no S x S materialization). Host pre-transposes Q, K, dO once per
+// call with transpose_sd ([B,S,H,D] -> [B,H,D,S]) so every LDS staging
+// load is a coalesced 16-byte row read.
+//
+// dkdv kernel (one block = 256 kv rows, 8 waves x 32):
+//   wave-resident K fragments; V re-read from LDS; per q-tile (32 rows):
+//     S  [q][kv] = mfma(A=Q-rows,  B=K-frag)      (K-frag doubles as B)
+//     dP [q][kv] = mfma(A=dO-rows, B=V-frag)
+//     dS = P o (dP - delta) * scale
+//     dV[kv][d] += mfma(A=conv(P),  B=dO^T-rows)  (conv = cvt_pk+permlane
+//     dK[kv][d] += mfma(A=conv(dS), B=Q^T-rows)    C-layout -> A-frag)
+// dq kernel (one block = 256 q rows, 8 waves x 32):
+//   wave-resident Q/dO fragments; per kv-tile (64 rows):
+//     S^T, dP^T as in the forward (swapped), then
+//     dQ[q][d] += mfma(A=conv(dS^T), B=K^T-rows)
+//
 // Numerics contract: ops/reference.py attn_bwd.
 #include "dtx_common.h"
 
-typedef __attribute__((ext_vector_type(4))) float f32x4;
-#define MFMA_B16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
-#define NEG_INF (-3.0e38f)
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
-__device__ __forceinline__ short8v s8load_or_zero(
-    const unsigned short* p, bool ok) {
-  short8v v;
-  if (ok) {
-    v = *reinterpret_cast<const short8v*>(p);
-  } else {
-#pragma unroll
-    for (int i = 0; i < 8; ++i) v[i] = 0;
-  }
-  return v;
+#define MFMA32(a, b, c) __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
+#define NEG_INF (-3.0e38f)
+#define LOG2E 1.4426950408889634f
+
+__device__ __forceinline__ unsigned bw_cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+               : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
 }
 
-// ------------------------------------------------------------ preprocess
-// delta[rows] = sum_d dO[row,d] * O[row,d]; wave per row, D%32==0.
+__device__ __forceinline__ int bw_crow(int r, int hi) {
+  return (r & 3) + 8 * (r >> 2) + 4 * hi;
+}
+
+// C-layout f32x16 (rows R on regs, cols on lanes) -> two bf16 A/B
+// fragments with k = R (frag0: R 0..15, frag1: R 16..31).
+__device__ __forceinline__ void conv_c_to_frag(const f32x16& p,
+                                               short8v& f0, short8v& f1) {
+  unsigned c0 = bw_cvt_pk_bf16(p[0], p[1]);
+  unsigned c1 = bw_cvt_pk_bf16(p[2], p[3]);
+  unsigned c2 = bw_cvt_pk_bf16(p[4], p[5]);
+  unsigned c3 = bw_cvt_pk_bf16(p[6], p[7]);
+  unsigned c4 = bw_cvt_pk_bf16(p[8], p[9]);
+  unsigned c5 = bw_cvt_pk_bf16(p[10], p[11]);
+  unsigned c6 = bw_cvt_pk_bf16(p[12], p[13]);
+  unsigned c7 = bw_cvt_pk_bf16(p[14], p[15]);
+  auto r02 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
+  auto r13 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
+  auto r46 = __builtin_amdgcn_permlane32_swap(c4, c6, false, false);
+  auto r57 = __builtin_amdgcn_permlane32_swap(c5, c7, false, false);
+  u32x4 lo{(unsigned)r02[0], (unsigned)r13[0],
+           (unsigned)r02[1], (unsigned)r13[1]};
+  u32x4 hi4{(unsigned)r46[0], (unsigned)r57[0],
+            (unsigned)r46[1], (unsigned)r57[1]};
+  f0 = *reinterpret_cast<short8v*>(&lo);
+  f1 = *reinterpret_cast<short8v*>(&hi4);
+}
+
+// ------------------------------------------------------------- delta
+// delta[b,h,s] = sum_d dO[b,s,h,d] * O[b,s,h,d]   (BSHD in, [B,H,S] out)
 __global__ __launch_bounds__(DTX_BLOCK)
-void attn_delta_kernel(const unsigned short* __restrict__ dO,
-                       const unsigned short* __restrict__ O,
-                       float* __restrict__ delta, long rows, int D) {
+void attn_delta2_kernel(const unsigned short* __restrict__ dO,
+                        const unsigned short* __restrict__ O,
+                        float* __restrict__ delta,
+                        long nrows, int H, int S, int D) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const long wstride = (long)gridDim.x * 4;
-  for (long row = (long)blockIdx.x * 4 + wid; row < rows; row += wstride) {
+  for (long row = blockIdx.x * 4 + wid; row < nrows;
+       row += (long)gridDim.x * 4) {
+    // row = (b*S + s)*H + h
+    const long base = row * D;
     float acc = 0.f;
     for (int d = lane * 2; d < D; d += 128) {
-      float a0 = bf2f(dO[row * D + d]), a1 = bf2f(dO[row * D + d + 1]);
-      float b0 = bf2f(O[row * D + d]), b1 = bf2f(O[row * D + d + 1]);
-      acc += a0 * b0 + a1 * b1;
+      acc += bf2f(dO[base + d]) * bf2f(O[base + d]) +
+             bf2f(dO[base + d + 1]) * bf2f(O[base + d + 1]);
     }
     acc = wave_reduce_sum(acc);
-    if (lane == 0) delta[row] = acc;
+    if (lane == 0) {
+      const int h = (int)(row % H);
+      const long bs = row / H;
+      const int s = (int)(bs % S);
+      const long b = bs / S;
+      delta[((long)b * H + h) * S + s] = acc;
+    }
   }
 }
 
-// --------------------------------------------------------------- dK / dV
+// ------------------------------------------------------------- dk/dv
+// 4 waves (256 threads), one wave per SIMD: the 512-register budget
+// holds K, V, dK, dV and the softmax tiles with zero spill.
 template <int D>
-struct BwdKVLds {
-  unsigned short Qr[32][D + 8];     // Q rows (B-frag for S^T)
-  unsigned short dOr[32][D + 8];    // dO rows (B-frag for dP^T)
-  unsigned short QT[D][32 + 8];     // Q^T (B-frag for dK)
-  unsigned short dOT[D][32 + 8];    // dO^T (B-frag for dV)
-  unsigned short PT[4][16][32 + 8];   // per-wave P^T (A-frag stage)
-  unsigned short DST[4][16][32 + 8];  // per-wave dS^T (A-frag stage)
+struct DkdvLds {
+  unsigned short Qr[32][D + 8];     // q-tile rows
+  unsigned short dOr[32][D + 8];
+  unsigned short QT[D][40];         // q-tile columns (from QT_g)
+  unsigned short dOT[D][40];
   float lse[32];
-  float delta[32];
+  float dlt[32];
 };
 
 template <int D>
-__global__ __launch_bounds__(256, 2)
-void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ Q,
-                          const unsigned short* __restrict__ K,
-                          const unsigned short* __restrict__ V,
-                          const unsigned short* __restrict__ dO,
-                          const float* __restrict__ lse,
-                          const float* __restrict__ delta,
-                          unsigned short* __restrict__ dKout,  // [B,Hq,Skv,D]
-                          unsigned short* __restrict__ dVout,  // [B,Hq,Skv,D]
-                          int B, int Hq, int Hkv, int S, int Skv,
-                          float scale, int causal) {
-  constexpr int DC = D / 32;
-  constexpr int NC2 = D / 16;
-  __shared__ BwdKVLds<D> lds;
+__global__ __launch_bounds__(256, 1)
+void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
+                           const unsigned short* __restrict__ QTg,
+                           const unsigned short* __restrict__ Kp,
+                           const unsigned short* __restrict__ Vp,
+                           const unsigned short* __restrict__ dO,
+                           const unsigned short* __restrict__ dOTg,
+                           const float* __restrict__ lse_in,
+                           const float* __restrict__ delta_in,
+                           unsigned short* __restrict__ dK,
+                           unsigned short* __restrict__ dV,
+                           int B, int Hq, int Hkv, int S, int Skv,
+                           float scale, int causal) {
+  constexpr int DC16 = D / 16;
+  constexpr int ND32 = D / 32;
+  __shared__ DkdvLds<D> lds;
+
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int l15 = lane & 15, l4 = lane >> 4;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+  const int rep = Hq / Hkv;
 
   const int bh = blockIdx.y;
-  const int b = bh / Hq, hq = bh % Hq;
-  const int hkv = hq / (Hq / Hkv);
-  const int kv0 = blockIdx.x * 64;
-  const int kvw = kv0 + wid * 16;          // wave's kv rows
-  const long qbase = (((long)b * Hq + hq) * S) * D;
-  const long kbase = (((long)b * Hkv + hkv) * Skv) * D;
-  const long obase = (((long)b * Hq + hq) * Skv) * D;  // dK/dV per-hq
-  const long lbase = ((long)b * Hq + hq) * S;
-  const int diag = Skv - S;
+  const int b = bh / Hkv, hkv = bh % Hkv;
+  const int kv0b = blockIdx.x * 128;
+  const int kw = kv0b + wid * 32;          // wave's first kv row
 
-  // resident K and V fragments (A-layout): row kvw+l15, k = kc*32+l4*8
-  short8v kfrag[DC], vfrag[DC];
+  const int qrowstr = Hq * D, krowstr = Hkv * D;
+  const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
+  const int diag = Skv - S;
+  const float kscale = scale * LOG2E;
+
+  // wave-resident K and V fragments: lane holds X[kw+l31][kc*16+hi*8+j]
+  short8v kfrag[DC16], vfrag[DC16];
   {
-    const int kr = kvw + l15;
-    const bool ok = kr < Skv;
+    const int krow = kw + l31;
 #pragma unroll
-    for (int kc = 0; kc < DC; ++kc) {
-      kfrag[kc] = s8load_or_zero(K + kbase + (long)kr * D + kc * 32 + l4 * 8, ok);
-      vfrag[kc] = s8load_or_zero(V + kbase + (long)kr * D + kc * 32 + l4 * 8, ok);
+    for (int kc = 0; kc < DC16; ++kc) {
+      if (krow < Skv) {
+        const long off = kbase + (long)krow * krowstr + kc * 16 + hi * 8;
+        kfrag[kc] = *reinterpret_cast<const short8v*>(Kp + off);
+        vfrag[kc] = *reinterpret_cast<const short8v*>(Vp + off);
+      } else {
+        kfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
+        vfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
+      }
     }
   }
-  f32x4 dk_acc[NC2], dv_acc[NC2];
-#pragma unroll
-  for (int c = 0; c < NC2; ++c) {
-    dk_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
-    dv_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
-  }
 
-  const int q_lo = causal ? max(0, kv0 - diag) : 0;
-  const int qt0 = (q_lo / 32) * 32;
-
-  for (int qt = qt0; qt < S; qt += 32) {
-    // ---- stage Q/dO rows + transposes + lse/delta
-    {
-      const int gpr = D / 8;
-      for (int idx = threadIdx.x; idx < 32 * gpr; idx += 256) {
-        const int row = idx / gpr, g = idx - row * gpr;
-        const bool ok = qt + row < S;
-        short8v qv = s8load_or_zero(Q + qbase + (long)(qt + row) * D + g * 8, ok);
-        short8v dv = s8load_or_zero(dO + qbase + (long)(qt + row) * D + g * 8, ok);
-        *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = qv;
-        *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) = dv;
+  f32x16 dk_acc[ND32], dv_acc[ND32];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          lds.QT[g * 8 + i][row] = (unsigned short)qv[i];
-          lds.dOT[g * 8 + i][row] = (unsigned short)dv[i];
+  for (int c = 0; c < ND32; ++c)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) { dk_acc[c][r] = 0.f; dv_acc[c][r] = 0.f; }
+
+  for (int gi = 0; gi < rep; ++gi) {
+    const int hq = hkv * rep + gi;
+    const long qbase = (long)b * S * qrowstr + (long)hq * D;
+    const long qtbase = ((long)b * Hq + hq) * (long)D * S;
+    const long lbase = ((long)b * Hq + hq) * S;
+
+    // q rows that can see this block's kv rows: q + diag >= kv0b
+    const int qt_lo = causal ? max(0, (kv0b - diag) / 32) : 0;
+    const int qt_hi = (S + 31) / 32;
+    for (int qt = qt_lo; qt < qt_hi; ++qt) {
+      const int q0 = qt * 32;
+      __syncthreads();
+      // ---- stage q-tile: Q/dO rows, QT/dOT columns, lse, delta
+      for (int idx = threadIdx.x; idx < 32 * (D / 8); idx += 256) {
+        const int row = idx / (D / 8), g = idx % (D / 8);
+        short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (q0 + row < S) {
+          const long off = qbase + (long)(q0 + row) * qrowstr + g * 8;
+          q8 = *reinterpret_cast<const short8v*>(Q + off);
+          d8 = *reinterpret_cast<const short8v*>(dO + off);
         }
+        *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = q8;
+        *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) = d8;
+      }
+      for (int idx = threadIdx.x; idx < D * 4; idx += 256) {
+        const int dd = idx / 4, g = idx % 4;       // 4 groups of 8 q-cols
+        short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        const int s0 = q0 + g * 8;
+        if (s0 + 8 <= S) {
+          q8 = *reinterpret_cast<const short8v*>(QTg + qtbase +
+                                                 (long)dd * S + s0);
+          d8 = *reinterpret_cast<const short8v*>(dOTg + qtbase +
+                                                 (long)dd * S + s0);
+        } else if (s0 < S) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (s0 + j < S) {
+              q8[j] = (short)QTg[qtbase + (long)dd * S + s0 + j];
+              d8[j] = (short)dOTg[qtbase + (long)dd * S + s0 + j];
+            }
+        }
+        *reinterpret_cast<short8v*>(&lds.QT[dd][g * 8]) = q8;
+        *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 8]) = d8;
       }
       if (threadIdx.x < 32) {
-        const int row = threadIdx.x;
-        const bool ok = qt + row < S;
-        lds.lse[row] = ok ? lse[lbase + qt + row] : 0.f;
-        lds.delta[row] = ok ? delta[lbase + qt + row] : 0.f;
+        const int qg = q0 + threadIdx.x;
+        lds.lse[threadIdx.x] = qg < S ? lse_in[lbase + qg] : 0.f;
+        lds.dlt[threadIdx.x] = qg < S ? delta_in[lbase + qg] : 0.f;
+      }
+      __syncthreads();
+
+      // wave skip: its kv rows all above this q-tile's diagonal
+      if (causal && (q0 + 31 + diag < kw)) continue;
+
+      // ---- S[q][kv] and dP[q][kv] (C-layout: q rows on regs, kv on
+      // lanes = the wave's kv rows kw + l31)
+      f32x16 sv, dpv;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { sv[r] = 0.f; dpv[r] = 0.f; }
+#pragma unroll
+      for (int kc = 0; kc < DC16; ++kc) {
+        short8v qa = *reinterpret_cast<const short8v*>(
+            &lds.Qr[l31][kc * 16 + hi * 8]);
+        short8v da = *reinterpret_cast<const short8v*>(
+            &lds.dOr[l31][kc * 16 + hi * 8]);
+        sv = MFMA32(qa, kfrag[kc], sv);
+        dpv = MFMA32(da, vfrag[kc], dpv);
+      }
+
+      // ---- P = exp2(s*kscale - lse2); dS = P o (dP - delta) * scale
+      const int kvg = kw + l31;
+      const bool interior = (kv0b + 128 <= Skv) && (q0 + 32 <= S) &&
+                            (!causal || (kw + 31 <= q0 + diag));
+      // P overwrites sv; dS overwrites dpv (keeps the VGPR count spill-free)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qg = q0 + bw_crow(r, hi);
+        const float lse2 = lds.lse[bw_crow(r, hi)] * LOG2E;
+        const float dlt = lds.dlt[bw_crow(r, hi)];
+        float p;
+        if (interior) {
+          p = __builtin_exp2f(sv[r] * kscale - lse2);
+        } else {
+          const bool dead = (kvg >= Skv) | (qg >= S) |
+                            (causal && (kvg > qg + diag));
+          p = dead ? 0.f : __builtin_exp2f(sv[r] * kscale - lse2);
+        }
+        sv[r] = p;
+        dpv[r] = p * (dpv[r] - dlt) * scale;
+      }
+
+      // ---- fragments (k = q) and accumulate dV, dK
+      short8v pf0, pf1, df0, df1;
+      conv_c_to_frag(sv, pf0, pf1);
+      conv_c_to_frag(dpv, df0, df1);
+#pragma unroll
+      for (int c = 0; c < ND32; ++c) {
+        short8v dot0 = *reinterpret_cast<const short8v*>(
+            &lds.dOT[c * 32 + l31][hi * 8]);
+        short8v dot1 = *reinterpret_cast<const short8v*>(
+            &lds.dOT[c * 32 + l31][16 + hi * 8]);
+        dv_acc[c] = MFMA32(pf0, dot0, dv_acc[c]);
+        dv_acc[c] = MFMA32(pf1, dot1, dv_acc[c]);
+        short8v qt0 = *reinterpret_cast<const short8v*>(
+            &lds.QT[c * 32 + l31][hi * 8]);
+        short8v qt1 = *reinterpret_cast<const short8v*>(
+            &lds.QT[c * 32 + l31][16 + hi * 8]);
+        dk_acc[c] = MFMA32(df0, qt0, dk_acc[c]);
+        dk_acc[c] = MFMA32(df1, qt1, dk_acc[c]);
       }
     }
-    __syncthreads();
-
-    // ---- S^T = K Q^T ; dP^T = V dO^T   (C-layout: kv=(l4*4+r), q=c*16+l15)
-    f32x4 st_acc[2], dpt_acc[2];
-#pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      st_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
-      dpt_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
-    }
-#pragma unroll
-    for (int kc = 0; kc < DC; ++kc) {
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        short8v qf = *reinterpret_cast<const short8v*>(
-            &lds.Qr[c * 16 + l15][kc * 32 + l4 * 8]);
-        short8v df = *reinterpret_cast<const short8v*>(
-            &lds.dOr[c * 16 + l15][kc * 32 + l4 * 8]);
-        st_acc[c] = MFMA_B16(kfrag[kc], qf, st_acc[c]);
-        dpt_acc[c] = MFMA_B16(vfrag[kc], df, dpt_acc[c]);
-      }
-    }
-
-    // ---- P^T and dS^T (elementwise in C-layout), stage to LDS
-#pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      const int qg = qt + c * 16 + l15;
-      const float l_q = lds.lse[c * 16 + l15];
-      const float d_q = lds.delta[c * 16 + l15];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int kvg = kvw + l4 * 4 + r;
-        bool dead = (kvg >= Skv) | (qg >= S) |
-                    (causal && (kvg > qg + diag));
-        float pt = dead ? 0.f : __expf(st_acc[c][r] * scale - l_q);
-        float dst = pt * (dpt_acc[c][r] - d_q) * scale;
-        lds.PT[wid][l4 * 4 + r][c * 16 + l15] = f2bf(pt);
-        lds.DST[wid][l4 * 4 + r][c * 16 + l15] = f2bf(dst);
-      }
-    }
-
-    // ---- dV += P^T dO ; dK += dS^T Q  (contraction over 32 q)
-    short8v ptf = *reinterpret_cast<const short8v*>(
-        &lds.PT[wid][l15][l4 * 8]);
-    short8v dstf = *reinterpret_cast<const short8v*>(
-        &lds.DST[wid][l15][l4 * 8]);
-#pragma unroll
-    for (int c2 = 0; c2 < NC2; ++c2) {
-      short8v dotf = *reinterpret_cast<const short8v*>(
-          &lds.dOT[c2 * 16 + l15][l4 * 8]);
-      short8v qtf = *reinterpret_cast<const short8v*>(
-          &lds.QT[c2 * 16 + l15][l4 * 8]);
-      dv_acc[c2] = MFMA_B16(ptf, dotf, dv_acc[c2]);
-      dk_acc[c2] = MFMA_B16(dstf, qtf, dk_acc[c2]);
-    }
-    __syncthreads();
   }
 
-  // ---- store dK/dV (per-hq layout [B,Hq,Skv,D]; host reduces GQA groups)
+  // ---- epilogue: dK/dV C-layout [kv regs][d lanes] -> BSHD stores
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int kvg = kvw + l4 * 4 + r;
+  for (int r = 0; r < 16; ++r) {
+    const int kvg = kw + bw_crow(r, hi);
     if (kvg < Skv) {
-      unsigned short* dkrow = dKout + obase + (long)kvg * D;
-      unsigned short* dvrow = dVout + obase + (long)kvg * D;
+      unsigned short* krow = dK + kbase + (long)kvg * krowstr;
+      unsigned short* vrow = dV + kbase + (long)kvg * krowstr;
 #pragma unroll
-      for (int c2 = 0; c2 < NC2; ++c2) {
-        dkrow[c2 * 16 + l15] = f2bf(dk_acc[c2][r]);
-        dvrow[c2 * 16 + l15] = f2bf(dv_acc[c2][r]);
+      for (int c = 0; c < ND32; ++c) {
+        krow[c * 32 + l31] = f2bf(dk_acc[c][r]);
+        vrow[c * 32 + l31] = f2bf(dv_acc[c][r]);
       }
     }
   }
 }
 
-// -------------------------------------------------------------------- dQ
+// ---------------------------------------------------------------- dq
 template <int D>
-struct BwdQLds {
-  unsigned short Kr[32][D + 8];       // K rows (B-frag for S)
-  unsigned short Vr[32][D + 8];       // V rows (B-frag for dP)
-  unsigned short KT[D][32 + 8];       // K^T (B-frag for dQ)
-  unsigned short DS[4][16][32 + 8];   // per-wave dS (A-frag stage)
+struct DqLds {
+  unsigned short K[64][D + 8];
+  unsigned short V[64][D + 8];
+  unsigned short KT[D][64 + 8];
 };
 
 template <int D>
-__global__ __launch_bounds__(256, 2)
-void attn_bwd_dq_kernel(const unsigned short* __restrict__ Q,
-                        const unsigned short* __restrict__ K,
-                        const unsigned short* __restrict__ V,
-                        const unsigned short* __restrict__ dO,
-                        const float* __restrict__ lse,
-                        const float* __restrict__ delta,
-                        unsigned short* __restrict__ dQout,
-                        int B, int Hq, int Hkv, int S, int Skv,
-                        float scale, int causal) {
-  constexpr int DC = D / 32;
-  constexpr int NC2 = D / 16;
-  __shared__ BwdQLds<D> lds;
+__global__ __launch_bounds__(512, 1)
+void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
+                         const unsigned short* __restrict__ Kp,
+                         const unsigned short* __restrict__ KTg,
+                         const unsigned short* __restrict__ Vp,
+                         const unsigned short* __restrict__ dO,
+                         const float* __restrict__ lse_in,
+                         const float* __restrict__ delta_in,
+                         unsigned short* __restrict__ dQ,
+                         int B, int Hq, int Hkv, int S, int Skv,
+                         float scale, int causal) {
+  constexpr int KVB = 64;
+  constexpr int DC16 = D / 16;
+  constexpr int ND32 = D / 32;
+  __shared__ DqLds<D> lds;
+
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int l15 = lane & 15, l4 = lane >> 4;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
 
   const int bh = blockIdx.y;
   const int b = bh / Hq, hq = bh % Hq;
   const int hkv = hq / (Hq / Hkv);
-  const int q0 = blockIdx.x * 64;
-  const int qw = q0 + wid * 16;
-  const long qbase = (((long)b * Hq + hq) * S) * D;
-  const long kbase = (((long)b * Hkv + hkv) * Skv) * D;
+  const int q0 = blockIdx.x * 256;
+  const int qw = q0 + wid * 32;
+
+  const int qrowstr = Hq * D, krowstr = Hkv * D;
+  const long qbase = (long)b * S * qrowstr + (long)hq * D;
+  const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
+  const long ktbase = ((long)b * Hkv + hkv) * (long)D * Skv;
   const long lbase = ((long)b * Hq + hq) * S;
   const int diag = Skv - S;
+  const float kscale = scale * LOG2E;
 
-  // resident Q and dO fragments (A-layout) + per-row lse/delta
-  short8v qfrag[DC], dofrag[DC];
+  // wave-resident Q and dO fragments (B-layout: lane q = qw + l31)
+  short8v qfrag[DC16], dofrag[DC16];
+  float lse2 = 0.f, dlt = 0.f;
   {
-    const int qr = qw + l15;
-    const bool ok = qr < S;
+    const int qrow = qw + l31;
+    if (qrow < S) {
 #pragma unroll
-    for (int kc = 0; kc < DC; ++kc) {
-      qfrag[kc] = s8load_or_zero(Q + qbase + (long)qr * D + kc * 32 + l4 * 8, ok);
-      dofrag[kc] = s8load_or_zero(dO + qbase + (long)qr * D + kc * 32 + l4 * 8, ok);
+      for (int kc = 0; kc < DC16; ++kc) {
+        const long off = qbase + (long)qrow * qrowstr + kc * 16 + hi * 8;
+        qfrag[kc] = *reinterpret_cast<const short8v*>(Q + off);
+        dofrag[kc] = *reinterpret_cast<const short8v*>(dO + off);
+      }
+      lse2 = lse_in[lbase + qrow] * LOG2E;
+      dlt = delta_in[lbase + qrow];
+    } else {
+#pragma unroll
+      for (int kc = 0; kc < DC16; ++kc) {
+        qfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
+        dofrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
+      }
     }
   }
-  float lse_r[4], delta_r[4];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qg = qw + l4 * 4 + r;
-    lse_r[r] = (qg < S) ? lse[lbase + qg] : 0.f;
-    delta_r[r] = (qg < S) ? delta[lbase + qg] : 0.f;
-  }
-  f32x4 dq_acc[NC2];
-#pragma unroll
-  for (int c = 0; c < NC2; ++c) dq_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int q_hi = min(q0 + 63, S - 1);
-  const int kv_hi = causal ? min(Skv - 1, q_hi + diag) : (Skv - 1);
-
-  for (int kt = 0; kt <= kv_hi; kt += 32) {
-    // ---- stage K/V rows + K^T
-    {
-      const int gpr = D / 8;
-      for (int idx = threadIdx.x; idx < 32 * gpr; idx += 256) {
-        const int row = idx / gpr, g = idx - row * gpr;
-        const bool ok = kt + row < Skv;
-        short8v kv8 = s8load_or_zero(K + kbase + (long)(kt + row) * D + g * 8, ok);
-        short8v vv8 = s8load_or_zero(V + kbase + (long)(kt + row) * D + g * 8, ok);
-        *reinterpret_cast<short8v*>(&lds.Kr[row][g * 8]) = kv8;
-        *reinterpret_cast<short8v*>(&lds.Vr[row][g * 8]) = vv8;
+  f32x16 dq_acc[ND32];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) lds.KT[g * 8 + i][row] = (unsigned short)kv8[i];
+  for (int c = 0; c < ND32; ++c)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[c][r] = 0.f;
+
+  const int q_hi_blk = min(q0 + 255, S - 1);
+  const int kv_hi = causal ? min(Skv - 1, q_hi_blk + diag) : (Skv - 1);
+  const int ntiles = kv_hi / KVB + 1;
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int kv0 = t * KVB;
+    // ---- stage K rows, V rows, KT columns
+    for (int idx = threadIdx.x; idx < KVB * (D / 8); idx += 512) {
+      const int row = idx / (D / 8), g = idx % (D / 8);
+      short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (kv0 + row < Skv) {
+        const long off = kbase + (long)(kv0 + row) * krowstr + g * 8;
+        k8 = *reinterpret_cast<const short8v*>(Kp + off);
+        v8 = *reinterpret_cast<const short8v*>(Vp + off);
       }
+      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = k8;
+      *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = v8;
+    }
+    for (int idx = threadIdx.x; idx < D * (KVB / 8); idx += 512) {
+      const int dd = idx / (KVB / 8), g = idx % (KVB / 8);
+      short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int s0 = kv0 + g * 8;
+      if (s0 + 8 <= Skv) {
+        k8 = *reinterpret_cast<const short8v*>(KTg + ktbase +
+                                               (long)dd * Skv + s0);
+      } else if (s0 < Skv) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (s0 + j < Skv)
+            k8[j] = (short)KTg[ktbase + (long)dd * Skv + s0 + j];
+      }
+      *reinterpret_cast<short8v*>(&lds.KT[dd][g * 8]) = k8;
     }
     __syncthreads();
 
-    // ---- S = Q K^T ; dP = dO V^T  (C-layout: q=(l4*4+r), kv=c*16+l15)
-    f32x4 s_acc[2], dp_acc[2];
+    const bool wave_dead = causal && (kv0 > qw + 31 + diag);
+    if (!wave_dead) {
+      const int qg = qw + l31;
+      const bool interior = (kv0 + KVB <= Skv) &&
+                            (!causal || (kv0 + KVB - 1 <= qw + diag));
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      s_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
-      dp_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
-    }
+      for (int ss = 0; ss < 2; ++ss) {
+        // S^T and dP^T for kv subtile ss (C-layout: kv rows on regs,
+        // q on lanes)
+        f32x16 st, dpt;
 #pragma unroll
-    for (int kc = 0; kc < DC; ++kc) {
+        for (int r = 0; r < 16; ++r) { st[r] = 0.f; dpt[r] = 0.f; }
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        short8v kf = *reinterpret_cast<const short8v*>(
-            &lds.Kr[c * 16 + l15][kc * 32 + l4 * 8]);
-        short8v vf = *reinterpret_cast<const short8v*>(
-            &lds.Vr[c * 16 + l15][kc * 32 + l4 * 8]);
-        s_acc[c] = MFMA_B16(qfrag[kc], kf, s_acc[c]);
-        dp_acc[c] = MFMA_B16(dofrag[kc], vf, dp_acc[c]);
+        for (int kc = 0; kc < DC16; ++kc) {
+          short8v ka = *reinterpret_cast<const short8v*>(
+              &lds.K[ss * 32 + l31][kc * 16 + hi * 8]);
+          short8v va = *reinterpret_cast<const short8v*>(
+              &lds.V[ss * 32 + l31][kc * 16 + hi * 8]);
+          st = MFMA32(ka, qfrag[kc], st);
+          dpt = MFMA32(va, dofrag[kc], dpt);
+        }
+        f32x16 dst;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float p;
+          if (interior) {
+            p = __builtin_exp2f(st[r] * kscale - lse2);
+          } else {
+            const int kvg = kv0 + ss * 32 + bw_crow(r, hi);
+            const bool dead = (kvg >= Skv) | (qg >= S) |
+                              (causal && (kvg > qg + diag));
+            p = dead ? 0.f : __builtin_exp2f(st[r] * kscale - lse2);
+          }
+          dst[r] = p * (dpt[r] - dlt) * scale;
+        }
+        short8v f0, f1;
+        conv_c_to_frag(dst, f0, f1);
+#pragma unroll
+        for (int c = 0; c < ND32; ++c) {
+          short8v kt0 = *reinterpret_cast<const short8v*>(
+              &lds.KT[c * 32 + l31][ss * 32 + hi * 8]);
+          short8v kt1 = *reinterpret_cast<const short8v*>(
+              &lds.KT[c * 32 + l31][ss * 32 + 16 + hi * 8]);
+          dq_acc[c] = MFMA32(f0, kt0, dq_acc[c]);
+          dq_acc[c] = MFMA32(f1, kt1, dq_acc[c]);
+        }
       }
-    }
-
-    // ---- dS = P (dP - delta) scale, stage per-wave
-#pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      const int kvg = kt + c * 16 + l15;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qg = qw + l4 * 4 + r;
-        bool dead = (kvg >= Skv) | (qg >= S) |
-                    (causal && (kvg > qg + diag));
-        float p = dead ? 0.f : __expf(s_acc[c][r] * scale - lse_r[r]);
-        float ds = p * (dp_acc[c][r] - delta_r[r]) * scale;
-        lds.DS[wid][l4 * 4 + r][c * 16 + l15] = f2bf(ds);
-      }
-    }
-
-    // ---- dQ += dS K  (contraction over 32 kv)
-    short8v dsf = *reinterpret_cast<const short8v*>(
-        &lds.DS[wid][l15][l4 * 8]);
-#pragma unroll
-    for (int c2 = 0; c2 < NC2; ++c2) {
-      short8v ktf = *reinterpret_cast<const short8v*>(
-          &lds.KT[c2 * 16 + l15][l4 * 8]);
-      dq_acc[c2] = MFMA_B16(dsf, ktf, dq_acc[c2]);
     }
     __syncthreads();
   }
 
-  // ---- store dQ
+  // ---- epilogue: dQ C-layout [q regs][d lanes] -> BSHD stores
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qg = qw + l4 * 4 + r;
+  for (int r = 0; r < 16; ++r) {
+    const int qg = qw + bw_crow(r, hi);
     if (qg < S) {
-      unsigned short* dqrow = dQout + qbase + (long)qg * D;
+      unsigned short* qrow = dQ + qbase + (long)qg * qrowstr;
 #pragma unroll
-      for (int c2 = 0; c2 < NC2; ++c2)
-        dqrow[c2 * 16 + l15] = f2bf(dq_acc[c2][r]);
+      for (int c = 0; c < ND32; ++c)
+        qrow[c * 32 + l31] = f2bf(dq_acc[c][r]);
     }
   }
 }
 
 // ------------------------------------------------------------- launchers
 void launch_attn_delta(const void* dO, const void* O, float* delta,
-                       long rows, int D, hipStream_t s) {
-  long w = DTX_CDIV(rows, 4);
-  int grid = (int)(w < 2048 ? (w < 1 ? 1 : w) : 2048);
-  attn_delta_kernel<<<grid, DTX_BLOCK, 0, s>>>(
-      (const unsigned short*)dO, (const unsigned short*)O, delta, rows, D);
+                       long nrows, int H, int S, int D, hipStream_t st) {
+  long gw = DTX_CDIV(nrows, 4);
+  int grid = (int)(gw < 2048 ? (gw < 1 ? 1 : gw) : 2048);
+  attn_delta2_kernel<<<grid, DTX_BLOCK, 0, st>>>(
+      (const unsigned short*)dO, (const unsigned short*)O, delta,
+      nrows, H, S, D);
 }
 
-void launch_attn_bwd(const void* q, const void* k, const void* v,
-                     const void* dO, const float* lse, const float* delta,
-                     void* dq, void* dk, void* dv,
-                     int B, int Hq, int Hkv, int S, int Skv, int D,
-                     float scale, int causal, hipStream_t st) {
-#define LAUNCH(DD)                                                        \
-  do {                                                                    \
-    dim3 gkv(DTX_CDIV(Skv, 64), B * Hq);                                  \
-    attn_bwd_dkdv_kernel<DD><<<gkv, 256, 0, st>>>(                        \
-        (const unsigned short*)q, (const unsigned short*)k,               \
-        (const unsigned short*)v, (const unsigned short*)dO, lse, delta,  \
-        (unsigned short*)dk, (unsigned short*)dv,                         \
-        B, Hq, Hkv, S, Skv, scale, causal);                               \
-    dim3 gq(DTX_CDIV(S, 64), B * Hq);                                     \
-    attn_bwd_dq_kernel<DD><<<gq, 256, 0, st>>>(                           \
-        (const unsigned short*)q, (const unsigned short*)k,               \
-        (const unsigned short*)v, (const unsigned short*)dO, lse, delta,  \
-        (unsigned short*)dq, B, Hq, Hkv, S, Skv, scale, causal);          \
-  } while (0)
-  if (D == 128) LAUNCH(128);
-  else if (D == 64) LAUNCH(64);
-#undef LAUNCH
+void launch_attn_bwd_dkdv(const void* q, const void* qt, const void* k,
+                          const void* v, const void* dO, const void* dOT,
+                          const float* lse, const float* delta, void* dk,
+                          void* dv, int B, int Hq, int Hkv, int S, int Skv,
+                          int D, float scale, int causal, hipStream_t st) {
+  dim3 grid(DTX_CDIV(Skv, 128), B * Hkv);
+  if (D == 128) {
+    attn_bwd_dkdv2_kernel<128><<<grid, 256, 0, st>>>(
+        (const unsigned short*)q, (const unsigned short*)qt,
+        (const unsigned short*)k, (const unsigned short*)v,
+        (const unsigned short*)dO, (const unsigned short*)dOT,
+        lse, delta, (unsigned short*)dk, (unsigned short*)dv,
+        B, Hq, Hkv, S, Skv, scale, causal);
+  } else if (D == 64) {
+    attn_bwd_dkdv2_kernel<64><<<grid, 256, 0, st>>>(
+        (const unsigned short*)q, (const unsigned short*)qt,
+        (const unsigned short*)k, (const unsigned short*)v,
+        (const unsigned short*)dO, (const unsigned short*)dOT,
+        lse, delta, (unsigned short*)dk, (unsigned short*)dv,
+        B, Hq, Hkv, S, Skv, scale, causal);
+  }
+}
+
+void launch_attn_bwd_dq(const void* q, const void* k, const void* kt,
+                        const void* v, const void* dO, const float* lse,
+                        const float* delta, void* dq, int B, int Hq,
+                        int Hkv, int S, int Skv, int D, float scale,
+                        int causal, hipStream_t st) {
+  dim3 grid(DTX_CDIV(S, 256), B * Hq);
+  if (D == 128) {
+    attn_bwd_dq2_kernel<128><<<grid, 512, 0, st>>>(
+        (const unsigned short*)q, (const unsigned short*)k,
+        (const unsigned short*)kt, (const unsigned short*)v,
+        (const unsigned short*)dO, lse, delta, (unsigned short*)dq,
+        B, Hq, Hkv, S, Skv, scale, causal);
+  } else if (D == 64) {
+    attn_bwd_dq2_kernel<64><<<grid, 512, 0, st>>>(
+        (const unsigned short*)q, (const unsigned short*)k,
+        (const unsigned short*)kt, (const unsigned short*)v,
+        (const unsigned short*)dO, lse, delta, (unsigned short*)dq,
+        B, Hq, Hkv, S, Skv, scale, causal);
+  }
 }

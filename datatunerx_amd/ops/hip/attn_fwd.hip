@@ -1,223 +1,351 @@
 // Flash-attention forward for gfx950 (CDNA4), bf16 I/O, fp32 softmax.
 //
-// Structure (v1, correctness-first with MFMA throughput):
-//   grid = (ceil(S/64), B*Hq); block = 256 threads = 4 waves.
-//   Each wave owns 16 q-rows (A-fragment resident in VGPRs, pre-loaded
-//   once); KV tiles of 64 staged in LDS per block: K row-major (padded),
-//   V transposed (so the PV B-fragment is a contiguous ds_read_b128).
-//   S-tile = mfma_f32_16x16x32_bf16(Q, K^T) over D; online softmax in
-//   the MFMA C-layout (row = (lane>>4)*4+reg, col = lane&15) with the
-//   16-lane xor-shuffle row reduce; P staged per-wave in LDS to convert
-//   C-layout -> A-layout for the PV mfma.
-// Causal + GQA + ragged S handled by masking. lse (= m + log l) saved
-// for the backward. Numerics contract: ops/reference.py attn_fwd.
+// v2 — 8-wave 32x32-MFMA structure (guide Appendix B "8-warp 32x32
+// ladder"): block = 512 threads = 8 waves, each wave owns 32 q-rows
+// (QBLK=32, block covers 256), KV tiles of 64 staged in LDS.
+//
+//  - Layout is BSHD (q/k: [B,S,H,D]) — no transpose copies in the model;
+//    V arrives PRE-TRANSPOSED as VT [B,Hkv,D,Skv] (transpose_sd kernel).
+//  - Swapped QK^T: S^T[kv][q] = mfma(A=K-frag, B=Q-frag) puts a whole
+//    P-row (over kv) in each lane's registers -> softmax (row max, exp2,
+//    row sum) is lane-local; the cross-half reduce is one shfl_xor(32).
+//  - P(C-layout) -> PV A-fragments via v_cvt_pk_bf16_f32 +
+//    permlane32_swap (16 cvt + 8 swaps per tile, no LDS round trip).
+//  - PV unswapped: O[q][d] = mfma(A=P, B=V^T-frag from the VT tile); the
+//    per-q-row rescale factor is fetched with one ds_bpermute per
+//    accumulator row.
+//  - K LDS tile [64][D+8] and VT tile [D][64+8]: +8 padding makes the
+//    column-subtile ds_read_b128 fragment reads bank-conflict-free
+//    (row strides 68/36 dwords; distinct (4r mod 64) starts per group).
+//
+// Numerics contract: ops/reference.py attn_fwd (fp32 softmax, exp2
+// domain internally, natural-log lse out). Causal + GQA + ragged S by
+// masking; interior tiles skip the mask; waves fully above the causal
+// diagonal skip compute.
 #include "dtx_common.h"
 
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
-#define MFMA_B16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
+#define MFMA32(a, b, c) __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
 #define NEG_INF (-3.0e38f)
+#define LOG2E 1.4426950408889634f
+#define LN2 0.6931471805599453f
+
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  // s_nop 1 covers the VALU-write -> v_permlane read hazard window
+  // (guide T21) for the swap that consumes this value next.
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+               : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// C-layout row index for f32x16 accumulator register r (32x32 MFMA).
+__device__ __forceinline__ int crow(int r, int hi) {
+  return (r & 3) + 8 * (r >> 2) + 4 * hi;
+}
 
 template <int D>
 struct AttnFwdLds {
   unsigned short K[64][D + 8];
   unsigned short VT[D][64 + 8];
-  unsigned short P[4][16][64 + 8];
 };
 
 template <int D>
-__global__ __launch_bounds__(256, 2)
-void attn_fwd_kernel(const unsigned short* __restrict__ Q,
-                     const unsigned short* __restrict__ Kp,
-                     const unsigned short* __restrict__ Vp,
-                     unsigned short* __restrict__ O,
-                     float* __restrict__ lse_out,
-                     int B, int Hq, int Hkv, int S, int Skv,
-                     float scale, int causal) {
+__global__ __launch_bounds__(512, 1)
+void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
+                      const unsigned short* __restrict__ Kp,
+                      const unsigned short* __restrict__ VTp,
+                      unsigned short* __restrict__ O,
+                      float* __restrict__ lse_out,
+                      int B, int Hq, int Hkv, int S, int Skv,
+                      float scale, int causal) {
   constexpr int KVB = 64;
-  constexpr int DC = D / 32;       // QK^T k-chunks
-  constexpr int NC2 = D / 16;      // PV output column tiles
+  constexpr int DC16 = D / 16;       // QK^T d-chunks (k-dim 16 each)
+  constexpr int ND32 = D / 32;       // PV output d-subtiles
   __shared__ AttnFwdLds<D> lds;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int l15 = lane & 15;
-  const int l4 = lane >> 4;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
 
   const int bh = blockIdx.y;
   const int b = bh / Hq, hq = bh % Hq;
   const int hkv = hq / (Hq / Hkv);
-  const int q0 = blockIdx.x * 64;          // block's first q row
-  const int qw = q0 + wid * 16;            // wave's first q row
+  const int q0 = blockIdx.x * 256;
+  const int qw = q0 + wid * 32;            // wave's first q row
 
-  const long qbase = (((long)b * Hq + hq) * S) * D;
-  const long kbase = (((long)b * Hkv + hkv) * Skv) * D;
+  const int qrowstr = Hq * D;              // BSHD row stride
+  const int krowstr = Hkv * D;
+  const long qbase = (long)b * S * qrowstr + (long)hq * D;
+  const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
+  const long vtbase = ((long)b * Hkv + hkv) * (long)D * Skv;
+  const long lbase = ((long)b * Hq + hq) * S;
   const int diag = Skv - S;                // causal diagonal offset
 
-  // ---- load Q fragments (A-layout): row l15, k = kc*32 + l4*8 + j
-  short8v qfrag[DC];
+  // ---- Q fragments (B-layout for swapped QK^T): lane holds
+  // Q[qw+l31][kc*16 + hi*8 + j]
+  short8v qfrag[DC16];
   {
-    const int qrow = qw + l15;
+    const int qrow = qw + l31;
 #pragma unroll
-    for (int kc = 0; kc < DC; ++kc) {
+    for (int kc = 0; kc < DC16; ++kc) {
       if (qrow < S) {
         qfrag[kc] = *reinterpret_cast<const short8v*>(
-            Q + qbase + (long)qrow * D + kc * 32 + l4 * 8);
+            Q + qbase + (long)qrow * qrowstr + kc * 16 + hi * 8);
       } else {
-#pragma unroll
-        for (int i = 0; i < 8; ++i) qfrag[kc][i] = 0;
+        qfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
   }
 
-  f32x4 o_acc[NC2];
+  f32x16 o_acc[ND32];
 #pragma unroll
-  for (int c = 0; c < NC2; ++c) o_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
-  float m_run[4], l_run[4];
+  for (int c = 0; c < ND32; ++c)
 #pragma unroll
-  for (int r = 0; r < 4; ++r) { m_run[r] = NEG_INF; l_run[r] = 0.f; }
+    for (int r = 0; r < 16; ++r) o_acc[c][r] = 0.f;
+  float m_run = NEG_INF, l_run = 0.f;      // per-lane: q-row qw + l31
+  const float kscale = scale * LOG2E;      // softmax in exp2 domain
 
-  // kv tiles this block must visit
-  const int q_hi = min(q0 + 63, S - 1);
-  const int kv_hi = causal ? min(Skv - 1, q_hi + diag) : (Skv - 1);
+  const int q_hi_blk = min(q0 + 255, S - 1);
+  const int kv_hi = causal ? min(Skv - 1, q_hi_blk + diag) : (Skv - 1);
   const int ntiles = kv_hi / KVB + 1;
 
   for (int t = 0; t < ntiles; ++t) {
     const int kv0 = t * KVB;
-    // ---- stage K tile [64][D] and V^T tile [D][64]
+    // ---- stage K [64][D] rows and VT [D][64] rows (coalesced 16B)
     {
-      const int gpr = D / 8;                      // 16B groups per row
-      for (int idx = threadIdx.x; idx < KVB * gpr; idx += 256) {
-        const int row = idx / gpr, g = idx - row * gpr;
-        short8v kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        short8v vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (kv0 + row < Skv) {
-          kv8 = *reinterpret_cast<const short8v*>(
-              Kp + kbase + (long)(kv0 + row) * D + g * 8);
-          vv8 = *reinterpret_cast<const short8v*>(
-              Vp + kbase + (long)(kv0 + row) * D + g * 8);
-        }
-        *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = kv8;
+      constexpr int KGROUPS = KVB * D / 8;     // 16B groups in the K tile
+      for (int idx = threadIdx.x; idx < KGROUPS; idx += 512) {
+        const int row = idx / (D / 8), g = idx % (D / 8);
+        short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (kv0 + row < Skv)
+          k8 = *reinterpret_cast<const short8v*>(
+              Kp + kbase + (long)(kv0 + row) * krowstr + g * 8);
+        *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = k8;
+      }
+      constexpr int VGROUPS = D * KVB / 8;
+      for (int idx = threadIdx.x; idx < VGROUPS; idx += 512) {
+        const int dd = idx / (KVB / 8), g = idx % (KVB / 8);
+        short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        const int s0 = kv0 + g * 8;
+        const long vrow = vtbase + (long)dd * Skv;
+        if (s0 + 8 <= Skv) {
+          v8 = *reinterpret_cast<const short8v*>(VTp + vrow + s0);
+        } else if (s0 < Skv) {
 #pragma unroll
-        for (int i = 0; i < 8; ++i)
-          lds.VT[g * 8 + i][row] = (unsigned short)vv8[i];
+          for (int j = 0; j < 8; ++j)
+            if (s0 + j < Skv) v8[j] = (short)VTp[vrow + s0 + j];
+        }
+        *reinterpret_cast<short8v*>(&lds.VT[dd][g * 8]) = v8;
       }
     }
     __syncthreads();
 
-    // ---- S tile: s_acc[c] = Q @ K^T (C-layout)
-    f32x4 s_acc[4];
+    const bool wave_dead = causal && (kv0 > qw + 31 + diag);
+    if (!wave_dead) {
+      // ---- S^T tiles: s[ss] = K[kv0+ss*32..][*] x Q^T  (C-layout:
+      // row kv = crow(r,hi)+32*ss, col q = qw+l31)
+      f32x16 s0v, s1v;
 #pragma unroll
-    for (int c = 0; c < 4; ++c) s_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
+      for (int r = 0; r < 16; ++r) { s0v[r] = 0.f; s1v[r] = 0.f; }
 #pragma unroll
-    for (int kc = 0; kc < DC; ++kc) {
-#pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        short8v kf = *reinterpret_cast<const short8v*>(
-            &lds.K[c * 16 + l15][kc * 32 + l4 * 8]);
-        s_acc[c] = MFMA_B16(qfrag[kc], kf, s_acc[c]);
+      for (int kc = 0; kc < DC16; ++kc) {
+        short8v k0 = *reinterpret_cast<const short8v*>(
+            &lds.K[l31][kc * 16 + hi * 8]);
+        short8v k1 = *reinterpret_cast<const short8v*>(
+            &lds.K[32 + l31][kc * 16 + hi * 8]);
+        s0v = MFMA32(k0, qfrag[kc], s0v);
+        s1v = MFMA32(k1, qfrag[kc], s1v);
       }
-    }
 
-    // ---- scale + mask (value at q-row (l4*4+reg), kv-col (c*16+l15))
+      // ---- scale (+ mask on boundary tiles)
+      const bool interior =
+          (kv0 + KVB <= Skv) &&
+          (!causal || (kv0 + KVB - 1 <= qw + diag));
+      const int qg = qw + l31;
+      if (interior) {
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      const int kvg = kv0 + c * 16 + l15;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qg = qw + l4 * 4 + r;
-        bool dead = (kvg >= Skv) | (qg >= S) |
-                    (causal && (kvg > qg + diag));
-        s_acc[c][r] = dead ? NEG_INF : s_acc[c][r] * scale;
-      }
-    }
-
-    // ---- online softmax (per q-row r owned at reg r)
-    float p[4][4];
-    float alpha[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float mx = fmaxf(fmaxf(s_acc[0][r], s_acc[1][r]),
-                       fmaxf(s_acc[2][r], s_acc[3][r]));
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
-      const float m_new = fmaxf(m_run[r], mx);
-      alpha[r] = (m_new == m_run[r]) ? 1.f : __expf(m_run[r] - m_new);
-      float rowsum = 0.f;
-      if (m_new == NEG_INF) {          // fully-masked row (padding)
-#pragma unroll
-        for (int c = 0; c < 4; ++c) p[c][r] = 0.f;
+        for (int r = 0; r < 16; ++r) {
+          s0v[r] *= kscale;
+          s1v[r] *= kscale;
+        }
       } else {
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {
-          p[c][r] = __expf(s_acc[c][r] - m_new);
-          rowsum += p[c][r];
+        for (int r = 0; r < 16; ++r) {
+          const int kv_a = kv0 + crow(r, hi);
+          const int kv_b = kv_a + 32;
+          bool dead_a = (kv_a >= Skv) | (causal && (kv_a > qg + diag));
+          bool dead_b = (kv_b >= Skv) | (causal && (kv_b > qg + diag));
+          s0v[r] = dead_a ? NEG_INF : s0v[r] * kscale;
+          s1v[r] = dead_b ? NEG_INF : s1v[r] * kscale;
         }
       }
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        rowsum += __shfl_xor(rowsum, off, 64);
-      l_run[r] = l_run[r] * alpha[r] + rowsum;
-      m_run[r] = m_new;
-    }
-    // rescale O accumulator
-#pragma unroll
-    for (int c2 = 0; c2 < NC2; ++c2)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[c2][r] *= alpha[r];
 
-    // ---- stage P (C-layout -> LDS), then PV with A-layout reads
+      // ---- online softmax, lane-local over 32 regs + one cross-half
+      float mx = s0v[0];
 #pragma unroll
-    for (int c = 0; c < 4; ++c)
+      for (int r = 1; r < 16; ++r) mx = fmaxf(mx, s0v[r]);
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        lds.P[wid][l4 * 4 + r][c * 16 + l15] = f2bf(p[c][r]);
-    // wave-private LDS: no __syncthreads needed (compiler orders ds ops)
+      for (int r = 0; r < 16; ++r) mx = fmaxf(mx, s1v[r]);
+      mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+      const float m_new = fmaxf(m_run, mx);
+      const float alpha =
+          (m_new == NEG_INF) ? 1.f : __builtin_exp2f(m_run - m_new);
+      float rowsum = 0.f;
+      if (m_new == NEG_INF) {
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      short8v pf = *reinterpret_cast<const short8v*>(
-          &lds.P[wid][l15][ks * 32 + l4 * 8]);
+        for (int r = 0; r < 16; ++r) { s0v[r] = 0.f; s1v[r] = 0.f; }
+      } else {
 #pragma unroll
-      for (int c2 = 0; c2 < NC2; ++c2) {
-        short8v vf = *reinterpret_cast<const short8v*>(
-            &lds.VT[c2 * 16 + l15][ks * 32 + l4 * 8]);
-        o_acc[c2] = MFMA_B16(pf, vf, o_acc[c2]);
+        for (int r = 0; r < 16; ++r) {
+          s0v[r] = __builtin_exp2f(s0v[r] - m_new);
+          s1v[r] = __builtin_exp2f(s1v[r] - m_new);
+          rowsum += s0v[r] + s1v[r];
+        }
+      }
+      rowsum += __shfl_xor(rowsum, 32, 64);
+      l_run = l_run * alpha + rowsum;
+      m_run = m_new;
+
+      // ---- P (C-layout f32) -> bf16 A-fragments [q][kv-slice]
+      short8v pa[4];
+#pragma unroll
+      for (int ss = 0; ss < 2; ++ss) {
+        const f32x16& p = ss ? s1v : s0v;
+        unsigned c0 = cvt_pk_bf16(p[0], p[1]);
+        unsigned c1 = cvt_pk_bf16(p[2], p[3]);
+        unsigned c2 = cvt_pk_bf16(p[4], p[5]);
+        unsigned c3 = cvt_pk_bf16(p[6], p[7]);
+        unsigned c4 = cvt_pk_bf16(p[8], p[9]);
+        unsigned c5 = cvt_pk_bf16(p[10], p[11]);
+        unsigned c6 = cvt_pk_bf16(p[12], p[13]);
+        unsigned c7 = cvt_pk_bf16(p[14], p[15]);
+        auto r02 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
+        auto r13 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
+        auto r46 = __builtin_amdgcn_permlane32_swap(c4, c6, false, false);
+        auto r57 = __builtin_amdgcn_permlane32_swap(c5, c7, false, false);
+        u32x4 lo{(unsigned)r02[0], (unsigned)r13[0],
+                 (unsigned)r02[1], (unsigned)r13[1]};
+        u32x4 hi4{(unsigned)r46[0], (unsigned)r57[0],
+                  (unsigned)r46[1], (unsigned)r57[1]};
+        pa[2 * ss + 0] = *reinterpret_cast<short8v*>(&lo);
+        pa[2 * ss + 1] = *reinterpret_cast<short8v*>(&hi4);
+      }
+
+      // ---- rescale O rows by alpha[q-row] (one bpermute per row)
+      if (t > 0) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float a_r = __shfl(alpha, crow(r, hi), 64);
+#pragma unroll
+          for (int c = 0; c < ND32; ++c) o_acc[c][r] *= a_r;
+        }
+      }
+
+      // ---- PV: O[q][d] += P[q][kv] x V^T-frag
+#pragma unroll
+      for (int c = 0; c < ND32; ++c) {
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          short8v vf = *reinterpret_cast<const short8v*>(
+              &lds.VT[c * 32 + l31][ks * 16 + hi * 8]);
+          o_acc[c] = MFMA32(pa[ks], vf, o_acc[c]);
+        }
       }
     }
     __syncthreads();
   }
 
-  // ---- epilogue: normalize, store O (bf16) and lse (f32)
+  // ---- epilogue: normalize rows, store O (bf16, coalesced over d) + lse
+  const float rcp = l_run > 0.f ? 1.f / l_run : 0.f;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qg = qw + l4 * 4 + r;
+  for (int r = 0; r < 16; ++r) {
+    const int qg = qw + crow(r, hi);
+    const float n_r = __shfl(rcp, crow(r, hi), 64);
     if (qg < S) {
-      const float rcp = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
-      unsigned short* orow = O + qbase + (long)qg * D;
+      unsigned short* orow = O + qbase + (long)qg * qrowstr;
 #pragma unroll
-      for (int c2 = 0; c2 < NC2; ++c2)
-        orow[c2 * 16 + l15] = f2bf(o_acc[c2][r] * rcp);
-      if (l15 == 0)
-        lse_out[((long)b * Hq + hq) * S + qg] =
-            m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+      for (int c = 0; c < ND32; ++c)
+        orow[c * 32 + l31] = f2bf(o_acc[c][r] * n_r);
+    }
+  }
+  if (hi == 0) {
+    const int qg = qw + l31;
+    if (qg < S)
+      lse_out[lbase + qg] =
+          m_run * LN2 + __logf(fmaxf(l_run, 1e-30f));
+  }
+}
+
+// ---------------------------------------------------------------- V^T
+// [B,S,H,D] -> [B,H,D,S] tile transpose (64x64 tiles through LDS).
+__global__ __launch_bounds__(256)
+void transpose_sd_kernel(const unsigned short* __restrict__ X,
+                         unsigned short* __restrict__ XT,
+                         int B, int S, int H, int D) {
+  __shared__ unsigned short tile[64][72];
+  const int s0 = blockIdx.x * 64;
+  const int d0 = blockIdx.y * 64;
+  const int bh = blockIdx.z;
+  const int b = bh / H, h = bh % H;
+  const long xbase = (long)b * S * H * D + (long)h * D;
+  const long tbase = ((long)b * H + h) * (long)D * S;
+  const int rowstr = H * D;
+
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int row = (threadIdx.x >> 3) + i * 32;     // s offset
+    const int g = threadIdx.x & 7;                   // d group
+    short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (s0 + row < S)
+      v8 = *reinterpret_cast<const short8v*>(
+          X + xbase + (long)(s0 + row) * rowstr + d0 + g * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) tile[g * 8 + j][row] = (unsigned short)v8[j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int drow = (threadIdx.x >> 3) + i * 32;    // d offset
+    const int g = threadIdx.x & 7;                   // s group
+    if (s0 + g * 8 + 8 <= S) {
+      *reinterpret_cast<short8v*>(XT + tbase + (long)(d0 + drow) * S +
+                                  s0 + g * 8) =
+          *reinterpret_cast<const short8v*>(&tile[drow][g * 8]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (s0 + g * 8 + j < S)
+          XT[tbase + (long)(d0 + drow) * S + s0 + g * 8 + j] =
+              tile[drow][g * 8 + j];
     }
   }
 }
 
-void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
+void launch_transpose_sd(const void* x, void* xt, int B, int S, int H,
+                         int D, hipStream_t st) {
+  dim3 grid(DTX_CDIV(S, 64), DTX_CDIV(D, 64), B * H);
+  transpose_sd_kernel<<<grid, 256, 0, st>>>(
+      (const unsigned short*)x, (unsigned short*)xt, B, S, H, D);
+}
+
+void launch_attn_fwd(const void* q, const void* k, const void* vt, void* o,
                      float* lse, int B, int Hq, int Hkv, int S, int Skv,
                      int D, float scale, int causal, hipStream_t st) {
-  dim3 grid(DTX_CDIV(S, 64), B * Hq);
+  dim3 grid(DTX_CDIV(S, 256), B * Hq);
   if (D == 128) {
-    attn_fwd_kernel<128><<<grid, 256, 0, st>>>(
+    attn_fwd2_kernel<128><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)v, (unsigned short*)o, lse,
+        (const unsigned short*)vt, (unsigned short*)o, lse,
         B, Hq, Hkv, S, Skv, scale, causal);
   } else if (D == 64) {
-    attn_fwd_kernel<64><<<grid, 256, 0, st>>>(
+    attn_fwd2_kernel<64><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)v, (unsigned short*)o, lse,
+        (const unsigned short*)vt, (unsigned short*)o, lse,
         B, Hq, Hkv, S, Skv, scale, causal);
   }
 }

@@ -31,11 +31,17 @@ void launch_lora_wgrad(const float*, const void*, float*, float*, long, int,
                        int, float, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, int, float, int, hipStream_t);
-void launch_attn_delta(const void*, const void*, float*, long, int,
-                       hipStream_t);
-void launch_attn_bwd(const void*, const void*, const void*, const void*,
-                     const float*, const float*, void*, void*, void*, int,
-                     int, int, int, int, int, float, int, hipStream_t);
+void launch_transpose_sd(const void*, void*, int, int, int, int,
+                         hipStream_t);
+void launch_attn_delta(const void*, const void*, float*, long, int, int,
+                       int, hipStream_t);
+void launch_attn_bwd_dkdv(const void*, const void*, const void*,
+                          const void*, const void*, const void*,
+                          const float*, const float*, void*, void*, int,
+                          int, int, int, int, int, float, int, hipStream_t);
+void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
+                        const void*, const float*, const float*, void*, int,
+                        int, int, int, int, int, float, int, hipStream_t);
 
 namespace {
 
@@ -218,54 +224,76 @@ torch::Tensor l2_norm(torch::Tensor x) {
 }
 
 // ------------------------------------------------------------ attention
+// [B,S,H,D] -> [B,H,D,S]
+torch::Tensor transpose_sd(torch::Tensor x) {
+  check_bf16_contig(x, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be [B,S,H,D]");
+  const int B = (int)x.size(0), S = (int)x.size(1), H = (int)x.size(2),
+            D = (int)x.size(3);
+  auto xt = torch::empty({B, H, D, S}, x.options());
+  launch_transpose_sd(x.data_ptr(), xt.data_ptr(), B, S, H, D,
+                      cur_stream());
+  return xt;
+}
+
+// q,k: [B,S,H,D] BSHD; vt: [B,Hkv,D,Skv] (pre-transposed V).
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, bool causal,
+                                    torch::Tensor vt, bool causal,
                                     double scale) {
   check_bf16_contig(q, "q");
   check_bf16_contig(k, "k");
-  check_bf16_contig(v, "v");
-  const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2),
+  check_bf16_contig(vt, "vt");
+  const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2),
             D = (int)q.size(3);
-  const int Hkv = (int)k.size(1), Skv = (int)k.size(2);
+  const int Skv = (int)k.size(1), Hkv = (int)k.size(2);
+  TORCH_CHECK(vt.size(1) == Hkv && vt.size(2) == D && vt.size(3) == Skv,
+              "vt must be [B,Hkv,D,Skv]");
   TORCH_CHECK(D == 64 || D == 128, "D must be 64 or 128");
   TORCH_CHECK(Hq % Hkv == 0, "GQA group");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat));
-  launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+  launch_attn_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(),
                   lse.data_ptr<float>(), B, Hq, Hkv, S, Skv, D,
                   (float)scale, causal ? 1 : 0, cur_stream());
   return {o, lse};
 }
 
+// All BSHD; v untransposed. Pre-transposes Q/K/dO internally.
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
                                     torch::Tensor dO, torch::Tensor lse,
                                     bool causal, double scale) {
   check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
   auto dO_c = dO.contiguous();
-  const int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2),
+  const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2),
             D = (int)q.size(3);
-  const int Hkv = (int)k.size(1), Skv = (int)k.size(2);
+  const int Skv = (int)k.size(1), Hkv = (int)k.size(2);
   auto delta = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat));
   launch_attn_delta(dO_c.data_ptr(), o.data_ptr(), delta.data_ptr<float>(),
-                    (long)B * Hq * S, D, cur_stream());
+                    (long)B * S * Hq, Hq, S, D, cur_stream());
+  auto qt = torch::empty({B, Hq, D, S}, q.options());
+  auto dot = torch::empty({B, Hq, D, S}, q.options());
+  auto kt = torch::empty({B, Hkv, D, Skv}, q.options());
+  launch_transpose_sd(q.data_ptr(), qt.data_ptr(), B, S, Hq, D,
+                      cur_stream());
+  launch_transpose_sd(dO_c.data_ptr(), dot.data_ptr(), B, S, Hq, D,
+                      cur_stream());
+  launch_transpose_sd(k.data_ptr(), kt.data_ptr(), B, Skv, Hkv, D,
+                      cur_stream());
   auto dq = torch::empty_like(q);
-  // per-Q-head partials; reduced below for GQA
-  auto dk_h = torch::empty({B, Hq, Skv, D}, q.options());
-  auto dv_h = torch::empty({B, Hq, Skv, D}, q.options());
-  launch_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), dO_c.data_ptr(),
-                  lse.data_ptr<float>(), delta.data_ptr<float>(),
-                  dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(), B, Hq,
-                  Hkv, S, Skv, D, (float)scale, causal ? 1 : 0,
-                  cur_stream());
-  torch::Tensor dk = dk_h, dv = dv_h;
-  if (Hq != Hkv) {
-    const long rep = Hq / Hkv;
-    dk = dk_h.view({B, Hkv, rep, Skv, D})
-             .to(torch::kFloat).sum(2).to(torch::kBFloat16);
-    dv = dv_h.view({B, Hkv, rep, Skv, D})
-             .to(torch::kFloat).sum(2).to(torch::kBFloat16);
-  }
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  launch_attn_bwd_dkdv(q.data_ptr(), qt.data_ptr(), k.data_ptr(),
+                       v.data_ptr(), dO_c.data_ptr(), dot.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       dk.data_ptr(), dv.data_ptr(), B, Hq, Hkv, S, Skv, D,
+                       (float)scale, causal ? 1 : 0, cur_stream());
+  launch_attn_bwd_dq(q.data_ptr(), k.data_ptr(), kt.data_ptr(),
+                     v.data_ptr(), dO_c.data_ptr(), lse.data_ptr<float>(),
+                     delta.data_ptr<float>(), dq.data_ptr(), B, Hq, Hkv, S,
+                     Skv, D, (float)scale, causal ? 1 : 0, cur_stream());
   return {dq, dk, dv};
 }
 
@@ -286,4 +314,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2_norm", &l2_norm);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("transpose_sd", &transpose_sd);
 }

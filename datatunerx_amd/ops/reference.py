@@ -123,6 +123,25 @@ def softmax_xent_bwd(logits: torch.Tensor, targets: torch.Tensor,
 # -------------------------------------------------------- flash attention
 def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
              causal: bool = True, scale: float | None = None):
+    """BSHD public contract: q [B,S,Hq,D], k/v [B,Skv,Hkv,D].
+
+    Returns (o [B,S,Hq,D], lse [f32, B,Hq,S])."""
+    o, lse = _attn_fwd_bhsd(q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3),
+                            v.permute(0, 2, 1, 3), causal, scale)
+    return o.permute(0, 2, 1, 3).contiguous(), lse
+
+
+def attn_bwd(q, k, v, o, do, lse, causal: bool = True,
+             scale: float | None = None):
+    """BSHD public contract; returns (dq, dk, dv) in BSHD."""
+    p = lambda t: t.permute(0, 2, 1, 3)
+    dq, dk, dv = _attn_bwd_bhsd(p(q), p(k), p(v), p(o), p(do), lse,
+                                causal, scale)
+    return (p(dq).contiguous(), p(dk).contiguous(), p(dv).contiguous())
+
+
+def _attn_fwd_bhsd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   causal: bool = True, scale: float | None = None):
     """q: [B,Hq,S,D], k/v: [B,Hkv,S,D] (GQA by head repeat).
 
     Returns (o [B,Hq,S,D], lse [f32, B,Hq,S]). fp32 math.
@@ -147,8 +166,8 @@ def attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return o.to(q.dtype), lse
 
 
-def attn_bwd(q, k, v, o, do, lse, causal: bool = True,
-             scale: float | None = None):
+def _attn_bwd_bhsd(q, k, v, o, do, lse, causal: bool = True,
+                   scale: float | None = None):
     """Returns (dq, dk, dv) with GQA reduction over repeated heads."""
     B, Hq, S, D = q.shape
     Hkv = k.shape[1]

@@ -23,18 +23,21 @@ from ..models import (GPT2Config, GPT2ForCausalLM, LlamaConfig,
 
 
 class KVCache:
+    """BSHD cache ([B, max_s, Hkv, D]) matching the attention kernels'
+    native layout — appends along dim 1."""
+
     def __init__(self, B, Hkv, max_s, D, device, dtype):
-        self.k = torch.zeros(B, Hkv, max_s, D, device=device, dtype=dtype)
-        self.v = torch.zeros(B, Hkv, max_s, D, device=device, dtype=dtype)
+        self.k = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
+        self.v = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
         self.cur = 0
 
     def update(self, k, v):
-        S = k.shape[2]
-        self.k[:, :, self.cur:self.cur + S] = k
-        self.v[:, :, self.cur:self.cur + S] = v
+        S = k.shape[1]
+        self.k[:, self.cur:self.cur + S] = k
+        self.v[:, self.cur:self.cur + S] = v
         self.cur += S
-        return (self.k[:, :, :self.cur].contiguous(),
-                self.v[:, :, :self.cur].contiguous())
+        return (self.k[:, :self.cur].contiguous(),
+                self.v[:, :self.cur].contiguous())
 
 
 def build_model(name: str, device, adapter_dir: Optional[str] = None,

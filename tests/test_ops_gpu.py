@@ -157,9 +157,9 @@ def test_l2_norm():
     (1, 4, 4, 1024, 1024, 128, True),     # training shape
 ])
 def test_attn_fwd(B, Hq, Hkv, S, Skv, D, causal):
-    q = mk(B, Hq, S, D, scale=0.5)
-    k = mk(B, Hkv, Skv, D, scale=0.5)
-    v = mk(B, Hkv, Skv, D, scale=0.5)
+    q = mk(B, S, Hq, D, scale=0.5)
+    k = mk(B, Skv, Hkv, D, scale=0.5)
+    v = mk(B, Skv, Hkv, D, scale=0.5)
     o, lse = ops.attn_fwd(q, k, v, causal, 1.0 / math.sqrt(D))
     o_ref, lse_ref = ref.attn_fwd(q.cpu(), k.cpu(), v.cpu(), causal)
     assert_close(o.cpu(), o_ref, name="attn o")
@@ -172,13 +172,14 @@ def test_attn_fwd(B, Hq, Hkv, S, Skv, D, causal):
     (1, 8, 2, 256, 128, True),
     (2, 2, 2, 100, 128, True),
     (1, 4, 4, 512, 64, True),
+    (1, 2, 2, 1024, 128, True),           # training shape
 ])
 def test_attn_bwd(B, Hq, Hkv, S, D, causal):
-    q = mk(B, Hq, S, D, scale=0.5)
-    k = mk(B, Hkv, S, D, scale=0.5)
-    v = mk(B, Hkv, S, D, scale=0.5)
+    q = mk(B, S, Hq, D, scale=0.5)
+    k = mk(B, S, Hkv, D, scale=0.5)
+    v = mk(B, S, Hkv, D, scale=0.5)
     o, lse = ops.attn_fwd(q, k, v, causal, 1.0 / math.sqrt(D))
-    do = mk(B, Hq, S, D, scale=0.5)
+    do = mk(B, S, Hq, D, scale=0.5)
     dq, dk, dv = ops.attn_bwd(q, k, v, o, do, lse, causal,
                               1.0 / math.sqrt(D))
     o_ref, lse_ref = ref.attn_fwd(q.cpu(), k.cpu(), v.cpu(), causal)
@@ -187,6 +188,15 @@ def test_attn_bwd(B, Hq, Hkv, S, D, causal):
     assert_close(dq.cpu(), dq_r, rtol=3e-2, name="attn dq")
     assert_close(dk.cpu(), dk_r, rtol=3e-2, name="attn dk")
     assert_close(dv.cpu(), dv_r, rtol=3e-2, name="attn dv")
+
+
+@pytest.mark.parametrize("B,S,H,D", [(2, 128, 4, 128), (1, 100, 2, 64)])
+def test_transpose_sd(B, S, H, D):
+    from datatunerx_amd.ops import _dtx_hip
+    x = mk(B, S, H, D)
+    xt = _dtx_hip.transpose_sd(x)
+    want = x.permute(0, 2, 3, 1).contiguous()
+    assert torch.equal(xt, want)
 
 
 # --------------------------------------------- end-to-end GPU train step

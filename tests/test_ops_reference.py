@@ -88,17 +88,20 @@ def test_cross_entropy_all_ignored_is_finite():
 @pytest.mark.parametrize("causal", [True, False])
 @pytest.mark.parametrize("hkv", [4, 2])
 def test_attention_matches_sdpa(causal, hkv):
+    """BSHD public contract vs torch SDPA (which wants BHSD)."""
     B, Hq, S, D = 2, 4, 32, 16
-    q = torch.randn(B, Hq, S, D, requires_grad=True)
-    k = torch.randn(B, hkv, S, D, requires_grad=True)
-    v = torch.randn(B, hkv, S, D, requires_grad=True)
+    q = torch.randn(B, S, Hq, D, requires_grad=True)
+    k = torch.randn(B, S, hkv, D, requires_grad=True)
+    v = torch.randn(B, S, hkv, D, requires_grad=True)
     o = attention(q, k, v, causal=causal)
     qr = q.detach().clone().requires_grad_(True)
     kr = k.detach().clone().requires_grad_(True)
     vr = v.detach().clone().requires_grad_(True)
-    kk = kr.repeat_interleave(Hq // hkv, 1) if hkv != Hq else kr
-    vv = vr.repeat_interleave(Hq // hkv, 1) if hkv != Hq else vr
-    orr = F.scaled_dot_product_attention(qr, kk, vv, is_causal=causal)
+    kk = kr.repeat_interleave(Hq // hkv, 2) if hkv != Hq else kr
+    vv = vr.repeat_interleave(Hq // hkv, 2) if hkv != Hq else vr
+    orr = F.scaled_dot_product_attention(
+        qr.permute(0, 2, 1, 3), kk.permute(0, 2, 1, 3),
+        vv.permute(0, 2, 1, 3), is_causal=causal).permute(0, 2, 1, 3)
     assert torch.allclose(o, orr, atol=1e-5)
     g = torch.randn_like(o)
     o.backward(g)
