@@ -3,61 +3,80 @@
 // these kernels fuse everything low-rank around it).
 //
 // Layouts (HF PEFT adapter contract): A [r,K], B [N,r].
-//   contract:   t[M,r]  = X[M,K] @ W[r,K]^T           (W LDS-resident)
-//   expand_add: Y[M,N] += s * T[M,r] @ W[N,r]^T       (W transposed into
-//               LDS; fused in-place epilogue on Y)
-//   wgrad:      dW[r,K] = s * T[M,r]^T @ X[M,K]       (deterministic
-//               split-M partials + reduce, no atomics)
-// All skinny/memory-bound: stream X/Y at HBM rate, stage the tiny
-// operand in LDS. R is a compile-time bound so accumulators stay in
-// VGPRs (runtime-indexed arrays spill to scratch — guide §5.4 rule 20).
+//   contract:   t[M,r]  = X[M,K] @ W[r,K]^T     (MFMA 32x32x16 over a
+//               zero-padded 32-col r tile; split-K partials + reduce)
+//   expand_add: Y[M,N] += s * T[M,r] @ W[N,r]^T (W^T LDS-resident; one
+//               wave per row, 2-chunk ILP; fused in-place epilogue on Y)
+//   wgrad:      dW[r,K] = s * T[M,r]^T @ X[M,K] (deterministic split-M
+//               partials + reduce, no atomics)
+// All skinny/memory-bound: the roofline is streaming X/Y at HBM rate.
+// R is a compile-time bound so accumulators stay in VGPRs (guide §5.4
+// rule 20).
 #include "dtx_common.h"
 
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+#define MFMA32L(a, b, c) __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
+
 // ---------------------------------------------------------------- contract
-// Block: 4 waves; wave handles one row m (grid-stride). W chunked in LDS
-// ([R][chunk] bf16); multi-chunk accumulates into fp32 out.
-template <int R>
+// One wave = one 32-row m-tile x 32-col r-tile (cols >= r zero-padded).
+// A-frag: lane streams X[mw+l31][kc*16 + hi*8 + j] straight from HBM
+// (each row read once; 8 kc loads/lane = 128 contiguous bytes).
+// B-frag: lane reads the LDS copy of W at row l31 (zero rows >= r).
+// Split-K across blockIdx.y; fp32 partials reduced by reduce_partials.
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_contract_kernel(const unsigned short* __restrict__ X,
                           const unsigned short* __restrict__ W,
-                          float* __restrict__ out,
-                          long M, int K, int r, int chunk) {
-  extern __shared__ __attribute__((aligned(16))) unsigned short wlds[];
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wid = threadIdx.x / WAVE;
-  for (int k0 = 0; k0 < K; k0 += chunk) {
-    const int kc = min(chunk, K - k0);
-    for (int idx = threadIdx.x * 8; idx < r * kc; idx += DTX_BLOCK * 8) {
-      int j = idx / kc, k = idx - j * kc;
-      *reinterpret_cast<short8v*>(&wlds[j * chunk + k]) =
-          *reinterpret_cast<const short8v*>(&W[(long)j * K + k0 + k]);
+                          float* __restrict__ part,
+                          long M, int K, int r, int kspan) {
+  __shared__ unsigned short wlds[32][1024 + 8];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+  const int k0 = blockIdx.y * kspan;
+  const int kend = min(K, k0 + kspan);
+
+  float* pout = part + (long)blockIdx.y * M * r;
+
+  for (int kc0 = k0; kc0 < kend; kc0 += 1024) {
+    const int kc_n = min(1024, kend - kc0);
+    // stage W rows (zero-padded to 32) for this k-chunk
+    for (int idx = threadIdx.x; idx < 32 * (1024 / 8); idx += DTX_BLOCK) {
+      const int row = idx / 128, g = idx % 128;
+      short8v w8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (row < r && g * 8 < kc_n)
+        w8 = *reinterpret_cast<const short8v*>(&W[(long)row * K + kc0 + g * 8]);
+      *reinterpret_cast<short8v*>(&wlds[row][g * 8]) = w8;
     }
     __syncthreads();
-    for (long m = blockIdx.x * 4 + wid; m < M; m += (long)gridDim.x * 4) {
-      const unsigned short* xr = X + m * K + k0;
-      float acc[R];
+
+    for (long mw = (long)(blockIdx.x * 4 + wid) * 32; mw < M;
+         mw += (long)gridDim.x * 4 * 32) {
+      f32x16 acc;
 #pragma unroll
-      for (int j = 0; j < R; ++j) acc[j] = 0.f;
-      for (int k = lane * 8; k < kc; k += WAVE * 8) {
-        float xv[8];
-        load_bf16x8(xr + k, xv);
-#pragma unroll
-        for (int j = 0; j < R; ++j) {
-          if (j < r) {
-            float wv[8];
-            load_bf16x8(&wlds[j * chunk + k], wv);
-#pragma unroll
-            for (int i = 0; i < 8; ++i) acc[j] += xv[i] * wv[i];
-          }
-        }
+      for (int q = 0; q < 16; ++q) acc[q] = 0.f;
+      const long xrow = mw + l31;
+      const unsigned short* xp =
+          xrow < M ? X + xrow * K + kc0 : X;     // row<M guard, safe base
+      for (int kc = 0; kc < kc_n; kc += 16) {
+        short8v xf = xrow < M
+            ? *reinterpret_cast<const short8v*>(xp + kc + hi * 8)
+            : short8v{0, 0, 0, 0, 0, 0, 0, 0};
+        short8v wf = *reinterpret_cast<const short8v*>(
+            &wlds[l31][kc + hi * 8]);
+        acc = MFMA32L(xf, wf, acc);
       }
+      // C-layout: col = r-index = l31, row m = crow(q,hi)
+      if (l31 < r) {
 #pragma unroll
-      for (int j = 0; j < R; ++j) {
-        if (j < r) {
-          float tot = wave_reduce_sum(acc[j]);
-          if (lane == 0) {
-            if (k0 == 0) out[m * r + j] = tot;
-            else out[m * r + j] += tot;
+        for (int q = 0; q < 16; ++q) {
+          const long m = mw + (q & 3) + 8 * (q >> 2) + 4 * hi;
+          if (m < M) {
+            if (kc0 == k0)
+              pout[m * r + l31] = acc[q];
+            else
+              pout[m * r + l31] += acc[q];
           }
         }
       }
@@ -67,7 +86,8 @@ void lora_contract_kernel(const unsigned short* __restrict__ X,
 }
 
 // -------------------------------------------------------------- expand_add
-// Y[M,N] += s * T[M,r] @ W[N,r]^T. W transposed into LDS [r][chunk of N].
+// Y[M,N] += s * T[M,r] @ W[N,r]^T. W transposed into LDS [r][N-chunk];
+// one wave per row with two 8-element n-chunks in flight (ILP).
 template <int R>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_expand_add_kernel(unsigned short* __restrict__ Y,
@@ -92,19 +112,27 @@ void lora_expand_add_kernel(unsigned short* __restrict__ Y,
 #pragma unroll
       for (int j = 0; j < R; ++j) tv[j] = (j < r) ? s * tr[j] : 0.f;
       unsigned short* yr = Y + m * N + n0;
-      for (int n = lane * 8; n < nc; n += WAVE * 8) {
-        float y[8];
-        load_bf16x8(yr + n, y);
+      // two chunks in flight per iteration
+      for (int n = lane * 8; n < nc; n += WAVE * 16) {
+        const int n2 = n + WAVE * 8;
+        float y0[8], y1[8];
+        load_bf16x8(yr + n, y0);
+        if (n2 < nc) load_bf16x8(yr + n2, y1);
 #pragma unroll
         for (int j = 0; j < R; ++j) {
           if (j < r) {
-            float wv[8];
-            load_bf16x8(&wlds[j * chunk + n], wv);
+            float w0[8], w1[8];
+            load_bf16x8(&wlds[j * chunk + n], w0);
+            if (n2 < nc) load_bf16x8(&wlds[j * chunk + n2], w1);
 #pragma unroll
-            for (int i = 0; i < 8; ++i) y[i] += tv[j] * wv[i];
+            for (int i = 0; i < 8; ++i) {
+              y0[i] += tv[j] * w0[i];
+              if (n2 < nc) y1[i] += tv[j] * w1[i];
+            }
           }
         }
-        store_bf16x8(yr + n, y);
+        store_bf16x8(yr + n, y0);
+        if (n2 < nc) store_bf16x8(yr + n2, y1);
       }
     }
     __syncthreads();
@@ -112,43 +140,63 @@ void lora_expand_add_kernel(unsigned short* __restrict__ Y,
 }
 
 // ------------------------------------------------------------------ wgrad
-// part[ms][j][K] = s * sum_{m in split ms} T[m,j] * X[m,k]
+// part[ms][j][k0..k0+512) = s * sum_{m in split ms} T[m,j] * X[m,k]
+// 4 waves cover interleaved m; LDS-reduce to one partial per block.
 template <int RCH>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_wgrad_kernel(const float* __restrict__ T,
                        const unsigned short* __restrict__ X,
                        float* __restrict__ part,
                        long M, int K, int r, int j0, int splitm, float s) {
-  const int col = blockIdx.x * 2048 + threadIdx.x * 8;
+  __shared__ float red[3][RCH > 8 ? 8 : RCH][512];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int col = blockIdx.x * 512 + lane * 8;
   const int ms = blockIdx.y;
-  if (col >= K) return;
-  float acc[RCH][8];
+  constexpr int RC = RCH > 8 ? 8 : RCH;
+  float acc[RC][8];
 #pragma unroll
-  for (int j = 0; j < RCH; ++j)
+  for (int j = 0; j < RC; ++j)
 #pragma unroll
     for (int i = 0; i < 8; ++i) acc[j][i] = 0.f;
   const long m_begin = (M * ms) / splitm;
   const long m_end = (M * (ms + 1)) / splitm;
-  for (long m = m_begin; m < m_end; ++m) {
-    float xv[8];
-    load_bf16x8(X + m * K + col, xv);
-    const float* tr = T + m * r + j0;
+  if (col < K) {
+    for (long m = m_begin + wid; m < m_end; m += 4) {
+      float xv[8];
+      load_bf16x8(X + m * K + col, xv);
+      const float* tr = T + m * r + j0;
 #pragma unroll
-    for (int j = 0; j < RCH; ++j) {
-      if (j0 + j < r) {
-        float t = tr[j];
+      for (int j = 0; j < RC; ++j) {
+        if (j0 + j < r) {
+          float t = tr[j];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) acc[j][i] += t * xv[i];
+          for (int i = 0; i < 8; ++i) acc[j][i] += t * xv[i];
+        }
       }
     }
   }
-  float* pb = part + ((long)ms * r) * K;
+  // cross-wave reduce through LDS (waves 1..3 park, wave 0 sums)
+  if (wid > 0) {
 #pragma unroll
-  for (int j = 0; j < RCH; ++j) {
-    if (j0 + j < r) {
+    for (int j = 0; j < RC; ++j)
 #pragma unroll
       for (int i = 0; i < 8; ++i)
-        pb[(long)(j0 + j) * K + col + i] = s * acc[j][i];
+        red[wid - 1][j][lane * 8 + i] = acc[j][i];
+  }
+  __syncthreads();
+  if (wid == 0 && col < K) {
+    float* pb = part + ((long)ms * r) * K;
+#pragma unroll
+    for (int j = 0; j < RC; ++j) {
+      if (j0 + j < r) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float v = acc[j][i] + red[0][j][lane * 8 + i] +
+                    red[1][j][lane * 8 + i] + red[2][j][lane * 8 + i];
+          pb[(long)(j0 + j) * K + col + i] = s * v;
+        }
+      }
     }
   }
 }
@@ -157,32 +205,33 @@ void lora_wgrad_kernel(const float* __restrict__ T,
 void launch_reduce_partials(const float* part, float* out, int P, long L,
                             hipStream_t s);
 
-static int lora_chunk(int Kdim, int r) {
-  int budget = (32768 / r) & ~7;           // 64 KiB of bf16 LDS
-  return Kdim < budget ? Kdim : budget;
+int lora_contract_ksplit(int K) {
+  // aim for >= 512 blocks: grid.x is capped at 128 (x4 waves = 512 rows)
+  int ks = DTX_CDIV(K, 4096) * 1024;     // 1..4 chunks of 1024 per block
+  int nsplit = DTX_CDIV(K, ks);
+  return nsplit;
 }
 
-void launch_lora_contract(const void* X, const void* W, float* out, long M,
-                          int K, int r, hipStream_t s) {
-  const int chunk = lora_chunk(K, r);
-  size_t lds = (size_t)r * chunk * 2;
-  long gw = DTX_CDIV(M, 4);
-  int grid = (int)(gw < 1024 ? (gw < 1 ? 1 : gw) : 1024);
-#define CASE(RR) lora_contract_kernel<RR><<<grid, DTX_BLOCK, lds, s>>>( \
-      (const unsigned short*)X, (const unsigned short*)W, out, M, K, r, chunk)
-  if (r <= 8) CASE(8);
-  else if (r <= 16) CASE(16);
-  else if (r <= 32) CASE(32);
-  else CASE(64);
-#undef CASE
+void launch_lora_contract(const void* X, const void* W, float* part,
+                          float* out, long M, int K, int r, hipStream_t s) {
+  const int nsplit = lora_contract_ksplit(K);
+  const int kspan = DTX_CDIV(K, nsplit);
+  long gw = DTX_CDIV(M, 128);
+  dim3 grid((int)(gw < 128 ? (gw < 1 ? 1 : gw) : 128), nsplit);
+  lora_contract_kernel<<<grid, DTX_BLOCK, 0, s>>>(
+      (const unsigned short*)X, (const unsigned short*)W,
+      nsplit > 1 ? part : out, M, K, r, kspan);
+  if (nsplit > 1)
+    launch_reduce_partials(part, out, nsplit, M * r, s);
 }
 
 void launch_lora_expand_add(void* Y, const float* T, const void* W, long M,
                             int N, int r, float scale, hipStream_t s) {
-  const int chunk = lora_chunk(N, r);
+  int chunk = (32768 / r) & ~7;          // 64 KiB of bf16 LDS
+  if (N < chunk) chunk = N;
   size_t lds = (size_t)r * chunk * 2;
   long gw = DTX_CDIV(M, 4);
-  int grid = (int)(gw < 1024 ? (gw < 1 ? 1 : gw) : 1024);
+  int grid = (int)(gw < 2048 ? (gw < 1 ? 1 : gw) : 2048);
 #define CASE(RR) lora_expand_add_kernel<RR><<<grid, DTX_BLOCK, lds, s>>>( \
       (unsigned short*)Y, T, (const unsigned short*)W, M, N, r, scale, chunk)
   if (r <= 8) CASE(8);
@@ -193,23 +242,21 @@ void launch_lora_expand_add(void* Y, const float* T, const void* W, long M,
 }
 
 int lora_wgrad_splitm(int K) {
-  int kblocks = DTX_CDIV(K, 2048);
-  int sm = 512 / kblocks;
-  return sm < 1 ? 1 : sm;
+  int kblocks = DTX_CDIV(K, 512);
+  int sm = 256 / kblocks;
+  return sm < 1 ? 1 : (sm > 32 ? 32 : sm);
 }
 
 void launch_lora_wgrad(const float* T, const void* X, float* part,
                        float* out, long M, int K, int r, float s,
                        hipStream_t st) {
   const int splitm = lora_wgrad_splitm(K);
-  dim3 grid(DTX_CDIV(K, 2048), splitm);
-  for (int j0 = 0; j0 < r; j0 += 16) {
-    int rch = r - j0;
+  dim3 grid(DTX_CDIV(K, 512), splitm);
+  for (int j0 = 0; j0 < r; j0 += 8) {
 #define CASE(RC) lora_wgrad_kernel<RC><<<grid, DTX_BLOCK, 0, st>>>( \
         T, (const unsigned short*)X, part, M, K, r, j0, splitm, s)
-    if (rch <= 4) CASE(4);
-    else if (rch <= 8) CASE(8);
-    else CASE(16);
+    if (r - j0 <= 4) CASE(4);
+    else CASE(8);
 #undef CASE
   }
   launch_reduce_partials(part, out, splitm, (long)r * K, st);
