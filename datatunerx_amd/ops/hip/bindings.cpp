@@ -22,8 +22,9 @@ void launch_xent_bwd(const void*, const long*, const float*, const float*,
 void launch_adamw(void*, float*, const float*, float*, float*, long, float,
                   float, float, float, float, int, hipStream_t);
 void launch_l2_norm(const float*, float*, float*, long, hipStream_t);
-void launch_lora_contract(const void*, const void*, float*, long, int, int,
-                          hipStream_t);
+int lora_contract_ksplit(int K);
+void launch_lora_contract(const void*, const void*, float*, float*, long,
+                          int, int, hipStream_t);
 void launch_lora_expand_add(void*, const float*, const void*, long, int, int,
                             float, hipStream_t);
 int lora_wgrad_splitm(int K);
@@ -166,8 +167,15 @@ torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(r <= 64, "r <= 64");
   TORCH_CHECK(K % 8 == 0);
   auto t = torch::empty({M, r}, x.options().dtype(torch::kFloat));
-  launch_lora_contract(x.data_ptr(), w.data_ptr(), t.data_ptr<float>(), M,
-                       K, r, cur_stream());
+  const int nsplit = lora_contract_ksplit(K);
+  torch::Tensor part;
+  float* part_ptr = nullptr;
+  if (nsplit > 1) {
+    part = torch::empty({nsplit, M, r}, x.options().dtype(torch::kFloat));
+    part_ptr = part.data_ptr<float>();
+  }
+  launch_lora_contract(x.data_ptr(), w.data_ptr(), part_ptr,
+                       t.data_ptr<float>(), M, K, r, cur_stream());
   return t;
 }
 

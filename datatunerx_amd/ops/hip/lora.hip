@@ -41,54 +41,57 @@ void lora_contract_kernel(const unsigned short* __restrict__ X,
 
   for (int kc0 = k0; kc0 < kend; kc0 += 1024) {
     const int kc_n = min(1024, kend - kc0);
-    // stage W rows (zero-padded to 32) for this k-chunk
-    for (int idx = threadIdx.x; idx < 32 * (1024 / 8); idx += DTX_BLOCK) {
-      const int row = idx / 128, g = idx % 128;
-      short8v w8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (row < r && g * 8 < kc_n)
-        w8 = *reinterpret_cast<const short8v*>(&W[(long)row * K + kc0 + g * 8]);
-      *reinterpret_cast<short8v*>(&wlds[row][g * 8]) = w8;
-    }
-    __syncthreads();
-
-    for (long mw = (long)(blockIdx.x * 4 + wid) * 32; mw < M;
-         mw += (long)gridDim.x * 4 * 32) {
-      f32x16 acc;
-#pragma unroll
-      for (int q = 0; q < 16; ++q) acc[q] = 0.f;
-      const long xrow = mw + l31;
-      const unsigned short* xp =
-          xrow < M ? X + xrow * K + kc0 : X;     // row<M guard, safe base
-      for (int kc = 0; kc < kc_n; kc += 16) {
-        short8v xf = xrow < M
-            ? *reinterpret_cast<const short8v*>(xp + kc + hi * 8)
-            : short8v{0, 0, 0, 0, 0, 0, 0, 0};
-        short8v wf = *reinterpret_cast<const short8v*>(
-            &wlds[l31][kc + hi * 8]);
-        acc = MFMA32L(xf, wf, acc);
+    for (int rt = 0; rt < r; rt += 32) {
+      // stage 32 W rows (zero-padded) for this (r-tile, k-chunk)
+      for (int idx = threadIdx.x; idx < 32 * (1024 / 8); idx += DTX_BLOCK) {
+        const int row = idx / 128, g = idx % 128;
+        short8v w8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (rt + row < r && g * 8 < kc_n)
+          w8 = *reinterpret_cast<const short8v*>(
+              &W[(long)(rt + row) * K + kc0 + g * 8]);
+        *reinterpret_cast<short8v*>(&wlds[row][g * 8]) = w8;
       }
-      // C-layout: col = r-index = l31, row m = crow(q,hi)
-      if (l31 < r) {
+      __syncthreads();
+
+      for (long mw = (long)(blockIdx.x * 4 + wid) * 32; mw < M;
+           mw += (long)gridDim.x * 4 * 32) {
+        f32x16 acc;
 #pragma unroll
-        for (int q = 0; q < 16; ++q) {
-          const long m = mw + (q & 3) + 8 * (q >> 2) + 4 * hi;
-          if (m < M) {
-            if (kc0 == k0)
-              pout[m * r + l31] = acc[q];
-            else
-              pout[m * r + l31] += acc[q];
+        for (int q = 0; q < 16; ++q) acc[q] = 0.f;
+        const long xrow = mw + l31;
+        const unsigned short* xp =
+            xrow < M ? X + xrow * K + kc0 : X;   // row<M guard, safe base
+        for (int kc = 0; kc < kc_n; kc += 16) {
+          short8v xf = xrow < M
+              ? *reinterpret_cast<const short8v*>(xp + kc + hi * 8)
+              : short8v{0, 0, 0, 0, 0, 0, 0, 0};
+          short8v wf = *reinterpret_cast<const short8v*>(
+              &wlds[l31][kc + hi * 8]);
+          acc = MFMA32L(xf, wf, acc);
+        }
+        // C-layout: col = r-index = rt + l31, row m = crow(q,hi)
+        if (rt + l31 < r) {
+#pragma unroll
+          for (int q = 0; q < 16; ++q) {
+            const long m = mw + (q & 3) + 8 * (q >> 2) + 4 * hi;
+            if (m < M) {
+              if (kc0 == k0)
+                pout[m * r + rt + l31] = acc[q];
+              else
+                pout[m * r + rt + l31] += acc[q];
+            }
           }
         }
       }
+      __syncthreads();
     }
-    __syncthreads();
   }
 }
 
 // -------------------------------------------------------------- expand_add
 // Y[M,N] += s * T[M,r] @ W[N,r]^T. W transposed into LDS [r][N-chunk];
-// one wave per row with two 8-element n-chunks in flight (ILP).
-template <int R>
+// one wave per row, two 8-element n-chunks in flight (ILP), j blocked by
+// 8 so register pressure is independent of r (no template on r).
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_expand_add_kernel(unsigned short* __restrict__ Y,
                             const float* __restrict__ T,
@@ -101,38 +104,44 @@ void lora_expand_add_kernel(unsigned short* __restrict__ Y,
     const int nc = min(chunk, N - n0);
     for (int n = threadIdx.x; n < nc; n += DTX_BLOCK) {
       const unsigned short* wr = W + (long)(n0 + n) * r;
-#pragma unroll
-      for (int j = 0; j < R; ++j)
-        if (j < r) wlds[j * chunk + n] = wr[j];
+      for (int j = 0; j < r; ++j) wlds[j * chunk + n] = wr[j];
     }
     __syncthreads();
     for (long m = blockIdx.x * 4 + wid; m < M; m += (long)gridDim.x * 4) {
-      float tv[R];
       const float* tr = T + m * r;
-#pragma unroll
-      for (int j = 0; j < R; ++j) tv[j] = (j < r) ? s * tr[j] : 0.f;
       unsigned short* yr = Y + m * N + n0;
-      // two chunks in flight per iteration
       for (int n = lane * 8; n < nc; n += WAVE * 16) {
-        const int n2 = n + WAVE * 8;
-        float y0[8], y1[8];
-        load_bf16x8(yr + n, y0);
-        if (n2 < nc) load_bf16x8(yr + n2, y1);
+        float y[2][8];
+        bool live[2];
 #pragma unroll
-        for (int j = 0; j < R; ++j) {
-          if (j < r) {
-            float w0[8], w1[8];
-            load_bf16x8(&wlds[j * chunk + n], w0);
-            if (n2 < nc) load_bf16x8(&wlds[j * chunk + n2], w1);
+        for (int c = 0; c < 2; ++c) {
+          live[c] = n + c * WAVE * 8 < nc;
+          if (live[c]) load_bf16x8(yr + n + c * WAVE * 8, y[c]);
+        }
+        for (int jb = 0; jb < r; jb += 8) {
+          float tv[8];
 #pragma unroll
-            for (int i = 0; i < 8; ++i) {
-              y0[i] += tv[j] * w0[i];
-              if (n2 < nc) y1[i] += tv[j] * w1[i];
+          for (int t = 0; t < 8; ++t)
+            tv[t] = (jb + t < r) ? s * tr[jb + t] : 0.f;
+#pragma unroll
+          for (int t = 0; t < 8; ++t) {
+            if (jb + t < r) {
+#pragma unroll
+              for (int c = 0; c < 2; ++c) {
+                if (live[c]) {
+                  float w[8];
+                  load_bf16x8(&wlds[(jb + t) * chunk + n + c * WAVE * 8],
+                              w);
+#pragma unroll
+                  for (int i = 0; i < 8; ++i) y[c][i] += tv[t] * w[i];
+                }
+              }
             }
           }
         }
-        store_bf16x8(yr + n, y0);
-        if (n2 < nc) store_bf16x8(yr + n2, y1);
+#pragma unroll
+        for (int c = 0; c < 2; ++c)
+          if (live[c]) store_bf16x8(yr + n + c * WAVE * 8, y[c]);
       }
     }
     __syncthreads();
@@ -232,13 +241,9 @@ void launch_lora_expand_add(void* Y, const float* T, const void* W, long M,
   size_t lds = (size_t)r * chunk * 2;
   long gw = DTX_CDIV(M, 4);
   int grid = (int)(gw < 2048 ? (gw < 1 ? 1 : gw) : 2048);
-#define CASE(RR) lora_expand_add_kernel<RR><<<grid, DTX_BLOCK, lds, s>>>( \
-      (unsigned short*)Y, T, (const unsigned short*)W, M, N, r, scale, chunk)
-  if (r <= 8) CASE(8);
-  else if (r <= 16) CASE(16);
-  else if (r <= 32) CASE(32);
-  else CASE(64);
-#undef CASE
+  lora_expand_add_kernel<<<grid, DTX_BLOCK, lds, s>>>(
+      (unsigned short*)Y, T, (const unsigned short*)W, M, N, r, scale,
+      chunk);
 }
 
 int lora_wgrad_splitm(int K) {
