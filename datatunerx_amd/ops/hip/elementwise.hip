@@ -120,16 +120,37 @@ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
   }
 }
 
-// Deterministic partial reduce: out[l] = sum_p part[p*L + l]
+// Deterministic partial reduce: out[l] = sum_p part[p*L + l].
+// float4 loads, 4 planes in flight (the scalar version was
+// latency-bound at ~0.5 TB/s; this one streams).
 __global__ __launch_bounds__(DTX_BLOCK)
 void reduce_partials_kernel(const float* __restrict__ part,
                             float* __restrict__ out, int P, long L) {
+  typedef __attribute__((ext_vector_type(4))) float f4;
+  const long nvec = L / 4;
   long i = (long)blockIdx.x * DTX_BLOCK + threadIdx.x;
-  long stride = (long)gridDim.x * DTX_BLOCK;
-  for (; i < L; i += stride) {
+  const long stride = (long)gridDim.x * DTX_BLOCK;
+  for (; i < nvec; i += stride) {
+    f4 s = {0.f, 0.f, 0.f, 0.f};
+    int p = 0;
+    for (; p + 4 <= P; p += 4) {
+      f4 a = *reinterpret_cast<const f4*>(&part[(long)p * L + i * 4]);
+      f4 b = *reinterpret_cast<const f4*>(&part[(long)(p + 1) * L + i * 4]);
+      f4 c = *reinterpret_cast<const f4*>(&part[(long)(p + 2) * L + i * 4]);
+      f4 d = *reinterpret_cast<const f4*>(&part[(long)(p + 3) * L + i * 4]);
+      s += (a + b) + (c + d);
+    }
+    for (; p < P; ++p)
+      s += *reinterpret_cast<const f4*>(&part[(long)p * L + i * 4]);
+    *reinterpret_cast<f4*>(&out[i * 4]) = s;
+  }
+  // ragged tail (L % 4)
+  const long tail0 = nvec * 4;
+  for (long t = tail0 + blockIdx.x * DTX_BLOCK + threadIdx.x; t < L;
+       t += stride) {
     float s = 0.f;
-    for (int p = 0; p < P; ++p) s += part[(long)p * L + i];
-    out[i] = s;
+    for (int p = 0; p < P; ++p) s += part[(long)p * L + t];
+    out[t] = s;
   }
 }
 

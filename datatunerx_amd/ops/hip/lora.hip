@@ -149,63 +149,45 @@ void lora_expand_add_kernel(unsigned short* __restrict__ Y,
 }
 
 // ------------------------------------------------------------------ wgrad
-// part[ms][j][k0..k0+512) = s * sum_{m in split ms} T[m,j] * X[m,k]
-// 4 waves cover interleaved m; LDS-reduce to one partial per block.
+// part[ms][j][K] = s * sum_{m in split ms} T[m,j] * X[m,k]
+// (k-span 2048/block keeps X re-reads at ceil(K/2048); the partials are
+// streamed back by the vectorized reduce_partials kernel)
 template <int RCH>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_wgrad_kernel(const float* __restrict__ T,
                        const unsigned short* __restrict__ X,
                        float* __restrict__ part,
                        long M, int K, int r, int j0, int splitm, float s) {
-  __shared__ float red[3][RCH > 8 ? 8 : RCH][512];
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
-  const int col = blockIdx.x * 512 + lane * 8;
+  const int col = blockIdx.x * 2048 + threadIdx.x * 8;
   const int ms = blockIdx.y;
-  constexpr int RC = RCH > 8 ? 8 : RCH;
-  float acc[RC][8];
+  if (col >= K) return;
+  float acc[RCH][8];
 #pragma unroll
-  for (int j = 0; j < RC; ++j)
+  for (int j = 0; j < RCH; ++j)
 #pragma unroll
     for (int i = 0; i < 8; ++i) acc[j][i] = 0.f;
   const long m_begin = (M * ms) / splitm;
   const long m_end = (M * (ms + 1)) / splitm;
-  if (col < K) {
-    for (long m = m_begin + wid; m < m_end; m += 4) {
-      float xv[8];
-      load_bf16x8(X + m * K + col, xv);
-      const float* tr = T + m * r + j0;
+  for (long m = m_begin; m < m_end; ++m) {
+    float xv[8];
+    load_bf16x8(X + m * K + col, xv);
+    const float* tr = T + m * r + j0;
 #pragma unroll
-      for (int j = 0; j < RC; ++j) {
-        if (j0 + j < r) {
-          float t = tr[j];
+    for (int j = 0; j < RCH; ++j) {
+      if (j0 + j < r) {
+        float t = tr[j];
 #pragma unroll
-          for (int i = 0; i < 8; ++i) acc[j][i] += t * xv[i];
-        }
+        for (int i = 0; i < 8; ++i) acc[j][i] += t * xv[i];
       }
     }
   }
-  // cross-wave reduce through LDS (waves 1..3 park, wave 0 sums)
-  if (wid > 0) {
+  float* pb = part + ((long)ms * r) * K;
 #pragma unroll
-    for (int j = 0; j < RC; ++j)
+  for (int j = 0; j < RCH; ++j) {
+    if (j0 + j < r) {
 #pragma unroll
       for (int i = 0; i < 8; ++i)
-        red[wid - 1][j][lane * 8 + i] = acc[j][i];
-  }
-  __syncthreads();
-  if (wid == 0 && col < K) {
-    float* pb = part + ((long)ms * r) * K;
-#pragma unroll
-    for (int j = 0; j < RC; ++j) {
-      if (j0 + j < r) {
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          float v = acc[j][i] + red[0][j][lane * 8 + i] +
-                    red[1][j][lane * 8 + i] + red[2][j][lane * 8 + i];
-          pb[(long)(j0 + j) * K + col + i] = s * v;
-        }
-      }
+        pb[(long)(j0 + j) * K + col + i] = s * acc[j][i];
     }
   }
 }
@@ -247,21 +229,23 @@ void launch_lora_expand_add(void* Y, const float* T, const void* W, long M,
 }
 
 int lora_wgrad_splitm(int K) {
-  int kblocks = DTX_CDIV(K, 512);
-  int sm = 256 / kblocks;
-  return sm < 1 ? 1 : (sm > 32 ? 32 : sm);
+  int kblocks = DTX_CDIV(K, 2048);
+  int sm = 512 / kblocks;
+  return sm < 1 ? 1 : (sm > 128 ? 128 : sm);
 }
 
 void launch_lora_wgrad(const float* T, const void* X, float* part,
                        float* out, long M, int K, int r, float s,
                        hipStream_t st) {
   const int splitm = lora_wgrad_splitm(K);
-  dim3 grid(DTX_CDIV(K, 512), splitm);
-  for (int j0 = 0; j0 < r; j0 += 8) {
+  dim3 grid(DTX_CDIV(K, 2048), splitm);
+  for (int j0 = 0; j0 < r; j0 += 16) {
+    int rch = r - j0;
 #define CASE(RC) lora_wgrad_kernel<RC><<<grid, DTX_BLOCK, 0, st>>>( \
         T, (const unsigned short*)X, part, M, K, r, j0, splitm, s)
-    if (r - j0 <= 4) CASE(4);
-    else CASE(8);
+    if (rch <= 4) CASE(4);
+    else if (rch <= 8) CASE(8);
+    else CASE(16);
 #undef CASE
   }
   launch_reduce_partials(part, out, splitm, (long)r * K, st);
