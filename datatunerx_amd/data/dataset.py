@@ -40,8 +40,10 @@ class ByteTokenizer:
         return ids
 
     def decode(self, ids):
-        return bytes(max(0, i - 3) for i in ids
-                     if i >= 3).decode("utf-8", errors="replace")
+        # ids past the byte range (models with a larger vocab emit
+        # them on random weights) are dropped, not an error
+        return bytes(i - 3 for i in ids
+                     if 3 <= i < 259).decode("utf-8", errors="replace")
 
 
 def read_csv_rows(path: str, column_map: Optional[Dict[str, str]] = None

@@ -37,9 +37,14 @@ class LlamaConfig:
     lora_dropout: float = 0.0
     lora_targets: tuple = ("q_proj", "v_proj")
     gradient_checkpointing: bool = False
+    # set when heads are tensor-parallel-sharded (head_dim no longer
+    # derivable from hidden_size / local heads)
+    head_dim_override: int = 0
 
     @property
     def head_dim(self) -> int:
+        if self.head_dim_override:
+            return self.head_dim_override
         return self.hidden_size // self.num_attention_heads
 
     @classmethod

@@ -63,15 +63,19 @@ def build_tp_llama(cfg: LlamaConfig, tp_rank: int, tp_ws: int,
     assert cfg.vocab_size % tp_ws == 0, "vocab % tp"
     device = device or torch.device("cpu")
 
-    # rank-local config: fewer heads / smaller ffn
+    # rank-local config: fewer heads / smaller ffn, same head_dim
     import dataclasses
     local = dataclasses.replace(
         cfg,
         num_attention_heads=cfg.num_attention_heads // tp_ws,
         num_key_value_heads=cfg.num_key_value_heads // tp_ws,
-        intermediate_size=cfg.intermediate_size // tp_ws)
+        intermediate_size=cfg.intermediate_size // tp_ws,
+        head_dim_override=cfg.head_dim)
     with torch.device(device):
         model = LlamaForCausalLM(local, lora=lora, dtype=dtype)
+        # vocab-parallel lm_head shard (embed stays replicated)
+        model.lm_head = FrozenLinear(cfg.hidden_size,
+                                     cfg.vocab_size // tp_ws, dtype=dtype)
     model.cfg_full = cfg
     model.tp_rank, model.tp_ws = tp_rank, tp_ws
 
@@ -191,6 +195,33 @@ def _init_tp_weights(model, cfg: LlamaConfig, rank: int, ws: int,
         else:                             # unexpected 2D param: replicate
             p.copy_(draw(tuple(p.shape)).to(p.dtype))
     return model
+
+
+def load_adapter_tp(model, adapter_dir: str, cfg: LlamaConfig, rank: int,
+                    ws: int, prefix: str = "base_model.model."):
+    """Load a full HF-PEFT adapter checkpoint into a TP shard."""
+    import os
+
+    from safetensors.torch import load_file
+    sd = load_file(os.path.join(adapter_dir, "adapter_model.safetensors"))
+    sd = shard_adapter_state(sd, cfg, rank, ws)
+    mods = {n: m for n, m in model.named_modules()
+            if isinstance(m, LoRALinearModule)}
+    n_loaded = 0
+    for key, tensor in sd.items():
+        if not key.startswith(prefix):
+            continue
+        rest = key[len(prefix):]
+        for suffix, attr in ((".lora_A.weight", "lora_A"),
+                             (".lora_B.weight", "lora_B")):
+            if rest.endswith(suffix):
+                mod_name = rest[: -len(suffix)]
+                # strip the _RowParallelLinear wrapper path if present
+                if mod_name in mods:
+                    getattr(mods[mod_name], attr).data.copy_(
+                        tensor.to(getattr(mods[mod_name], attr).dtype))
+                    n_loaded += 1
+    return n_loaded
 
 
 def shard_adapter_state(sd: dict, cfg: LlamaConfig, rank: int, ws: int):

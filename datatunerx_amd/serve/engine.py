@@ -152,17 +152,32 @@ class InferenceEngine:
 
     def _sample(self, logits: torch.Tensor, temperature: float,
                 top_p: float) -> int:
+        import torch.distributed as dist
+        tp = dist.is_initialized() and dist.get_world_size() > 1
+        comm_dev = (self.device if tp and dist.get_backend() == "nccl"
+                    else torch.device("cpu"))
+        if tp and dist.get_rank() != 0:
+            # follower ranks take rank 0's choice (RNG is per-process)
+            t = torch.zeros(1, dtype=torch.long, device=comm_dev)
+            dist.broadcast(t, src=0)
+            return int(t)
         if temperature <= 0.0:
-            return int(logits.argmax())
-        probs = torch.softmax(logits.float() / temperature, dim=-1)
-        if top_p < 1.0:
-            sp, si = probs.sort(descending=True)
-            cum = sp.cumsum(0)
-            keep = cum <= top_p
-            keep[0] = True
-            probs = torch.zeros_like(probs).scatter_(0, si[keep], sp[keep])
-            probs /= probs.sum()
-        return int(torch.multinomial(probs, 1))
+            nxt = int(logits.argmax())
+        else:
+            probs = torch.softmax(logits.float() / temperature, dim=-1)
+            if top_p < 1.0:
+                sp, si = probs.sort(descending=True)
+                cum = sp.cumsum(0)
+                keep = cum <= top_p
+                keep[0] = True
+                probs = torch.zeros_like(probs).scatter_(0, si[keep],
+                                                         sp[keep])
+                probs /= probs.sum()
+            nxt = int(torch.multinomial(probs, 1))
+        if tp:
+            dist.broadcast(torch.tensor([nxt], dtype=torch.long,
+                                        device=comm_dev), src=0)
+        return nxt
 
     # ----------------------------------------------------------- score
     @torch.no_grad()

@@ -84,7 +84,57 @@ def serve_forever(engine, host: str, port: int):
     httpd.serve_forever()
 
 
+class TPFrontEngine:
+    """Rank-0 wrapper: broadcasts each request to the follower ranks so
+    the TP group steps through the model collectives in lockstep."""
+
+    def __init__(self, engine):
+        self.engine = engine
+
+    def _bcast(self, req: dict):
+        import torch.distributed as dist
+        dist.broadcast_object_list([req], src=0)
+
+    def chat(self, messages, max_tokens, temperature, top_p):
+        self._bcast({"op": "chat", "messages": messages,
+                     "max_tokens": max_tokens, "temperature": temperature,
+                     "top_p": top_p})
+        return self.engine.chat(messages, max_tokens, temperature, top_p)
+
+    def perplexity(self, texts):
+        self._bcast({"op": "ppl", "texts": texts})
+        return self.engine.perplexity(texts)
+
+
+def tp_follower_loop(engine):
+    import torch.distributed as dist
+    while True:
+        obj = [None]
+        dist.broadcast_object_list(obj, src=0)
+        req = obj[0]
+        if req["op"] == "chat":
+            engine.chat(req["messages"], req["max_tokens"],
+                        req["temperature"], req["top_p"])
+        elif req["op"] == "ppl":
+            engine.perplexity(req["texts"])
+        elif req["op"] == "stop":
+            return
+
+
+def _llama_config(name: str):
+    from ..models import LlamaConfig
+    if name in ("llama2-7b", "llama-2-7b"):
+        return LlamaConfig.llama2_7b()
+    if name in ("llama2-13b", "llama-2-13b"):
+        return LlamaConfig.llama2_13b()
+    if name == "llama-tiny":
+        return LlamaConfig.tiny()
+    raise ValueError(f"TP serving supports llama models, not {name!r}")
+
+
 def main(argv=None):
+    import os
+
     from .engine import InferenceEngine, build_model
     ap = argparse.ArgumentParser()
     ap.add_argument("--port", type=int, default=8000)
@@ -93,6 +143,29 @@ def main(argv=None):
     ap.add_argument("--adapter", default=None)
     ap.add_argument("--template", default="llama2")
     args = ap.parse_args(argv)
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        # tensor-parallel service (launched under torchrun, 1 rank/GPU):
+        # rank 0 serves HTTP; followers run the collectives in lockstep.
+        import torch.distributed as dist
+
+        from ..parallel.ddp import init_distributed
+        from ..parallel.tp import build_tp_llama, load_adapter_tp
+        rank, world, local_rank, device = init_distributed()
+        cfg = _llama_config(args.model)
+        dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+        model = build_tp_llama(cfg, rank, world, lora=bool(args.adapter),
+                               dtype=dtype, device=device)
+        if args.adapter:
+            load_adapter_tp(model, args.adapter, cfg, rank, world)
+        model.eval()
+        engine = InferenceEngine(model, template=args.template,
+                                 device=device)
+        if rank == 0:
+            serve_forever(TPFrontEngine(engine), args.host, args.port)
+        else:
+            tp_follower_loop(engine)
+        return
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     model = build_model(args.model, device, adapter_dir=args.adapter)
     engine = InferenceEngine(model, template=args.template, device=device)
