@@ -104,9 +104,11 @@ class LoRALinearModule(nn.Module):
         mask = None
         if self.training and self.dropout > 0.0:
             keep = 1.0 - self.dropout
-            x2 = x.reshape(-1, x.shape[-1])
-            mask = (torch.rand_like(x2, dtype=torch.float32) < keep)
-            mask = mask.to(x.dtype) / keep
+            # bf16 bernoulli directly (no fp32 rand materialization)
+            mask = torch.empty(
+                (x.numel() // x.shape[-1], x.shape[-1]),
+                device=x.device, dtype=x.dtype)
+            mask.bernoulli_(keep).mul_(1.0 / keep)
         if mask is None:
             return _FusedLoRAFn.apply(x, self.weight, self.lora_A,
                                       self.lora_B, self.scale)
