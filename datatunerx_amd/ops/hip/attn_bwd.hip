@@ -97,12 +97,12 @@ void attn_delta2_kernel(const unsigned short* __restrict__ dO,
 // holds K, V, dK, dV and the softmax tiles with zero spill.
 template <int D>
 struct DkdvLds {
-  unsigned short Qr[32][D + 8];     // q-tile rows
-  unsigned short dOr[32][D + 8];
-  unsigned short QT[D][40];         // q-tile columns (from QT_g)
-  unsigned short dOT[D][40];
-  float lse[32];
-  float dlt[32];
+  unsigned short Qr[64][D + 8];     // staged q rows (two 32-row tiles)
+  unsigned short dOr[64][D + 8];
+  unsigned short QT[D][72];         // staged q columns (from QT_g)
+  unsigned short dOT[D][72];
+  float lse[64];
+  float dlt[64];
 };
 
 template <int D>
@@ -169,29 +169,29 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
     const long lbase = ((long)b * Hq + hq) * S;
 
     // q rows that can see this block's kv rows: q + diag >= kv0b
-    const int qt_lo = causal ? max(0, (kv0b - diag) / 32) : 0;
-    const int qt_hi = (S + 31) / 32;
-    for (int qt = qt_lo; qt < qt_hi; ++qt) {
-      const int q0 = qt * 32;
+    const int qs_lo = causal ? max(0, (kv0b - diag) / 64) : 0;
+    const int qs_hi = (S + 63) / 64;
+    for (int qs = qs_lo; qs < qs_hi; ++qs) {
+      const int q0s = qs * 64;
       __syncthreads();
-      // ---- stage q-tile: Q/dO rows, QT/dOT columns, lse, delta
-      for (int idx = threadIdx.x; idx < 32 * (D / 8); idx += 256) {
+      // ---- stage 64 q rows: Q/dO rows, QT/dOT columns, lse, delta
+      for (int idx = threadIdx.x; idx < 64 * (D / 8); idx += 256) {
         const int row = idx / (D / 8), g = idx % (D / 8);
         short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
         short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (q0 + row < S) {
-          const long off = qbase + (long)(q0 + row) * qrowstr + g * 8;
+        if (q0s + row < S) {
+          const long off = qbase + (long)(q0s + row) * qrowstr + g * 8;
           q8 = *reinterpret_cast<const short8v*>(Q + off);
           d8 = *reinterpret_cast<const short8v*>(dO + off);
         }
         *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = q8;
         *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) = d8;
       }
-      for (int idx = threadIdx.x; idx < D * 4; idx += 256) {
-        const int dd = idx / 4, g = idx % 4;       // 4 groups of 8 q-cols
+      for (int idx = threadIdx.x; idx < D * 8; idx += 256) {
+        const int dd = idx / 8, g = idx % 8;       // 8 groups of 8 q-cols
         short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
         short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        const int s0 = q0 + g * 8;
+        const int s0 = q0s + g * 8;
         if (s0 + 8 <= S) {
           q8 = *reinterpret_cast<const short8v*>(QTg + qtbase +
                                                  (long)dd * S + s0);
@@ -208,15 +208,18 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         *reinterpret_cast<short8v*>(&lds.QT[dd][g * 8]) = q8;
         *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 8]) = d8;
       }
-      if (threadIdx.x < 32) {
-        const int qg = q0 + threadIdx.x;
+      if (threadIdx.x < 64) {
+        const int qg = q0s + threadIdx.x;
         lds.lse[threadIdx.x] = qg < S ? lse_in[lbase + qg] : 0.f;
         lds.dlt[threadIdx.x] = qg < S ? delta_in[lbase + qg] : 0.f;
       }
       __syncthreads();
 
+      for (int qh = 0; qh < 2; ++qh) {
+      const int q0 = q0s + qh * 32;
+      const int qoff = qh * 32;                  // LDS row offset
       // wave skip: its kv rows all above this q-tile's diagonal
-      if (causal && (q0 + 31 + diag < kw)) continue;
+      if (q0 >= S || (causal && (q0 + 31 + diag < kw))) continue;
 
       // ---- S[q][kv] and dP[q][kv] (C-layout: q rows on regs, kv on
       // lanes = the wave's kv rows kw + l31)
@@ -226,9 +229,9 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int kc = 0; kc < DC16; ++kc) {
         short8v qa = *reinterpret_cast<const short8v*>(
-            &lds.Qr[l31][kc * 16 + hi * 8]);
+            &lds.Qr[qoff + l31][kc * 16 + hi * 8]);
         short8v da = *reinterpret_cast<const short8v*>(
-            &lds.dOr[l31][kc * 16 + hi * 8]);
+            &lds.dOr[qoff + l31][kc * 16 + hi * 8]);
         sv = MFMA32(qa, kfrag[kc], sv);
         dpv = MFMA32(da, vfrag[kc], dpv);
       }
@@ -241,8 +244,8 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qg = q0 + bw_crow(r, hi);
-        const float lse2 = lds.lse[bw_crow(r, hi)] * LOG2E;
-        const float dlt = lds.dlt[bw_crow(r, hi)];
+        const float lse2 = lds.lse[qoff + bw_crow(r, hi)] * LOG2E;
+        const float dlt = lds.dlt[qoff + bw_crow(r, hi)];
         float p;
         if (interior) {
           p = __builtin_exp2f(sv[r] * kscale - lse2);
@@ -262,18 +265,19 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int c = 0; c < ND32; ++c) {
         short8v dot0 = *reinterpret_cast<const short8v*>(
-            &lds.dOT[c * 32 + l31][hi * 8]);
+            &lds.dOT[c * 32 + l31][qoff + hi * 8]);
         short8v dot1 = *reinterpret_cast<const short8v*>(
-            &lds.dOT[c * 32 + l31][16 + hi * 8]);
+            &lds.dOT[c * 32 + l31][qoff + 16 + hi * 8]);
         dv_acc[c] = MFMA32(pf0, dot0, dv_acc[c]);
         dv_acc[c] = MFMA32(pf1, dot1, dv_acc[c]);
         short8v qt0 = *reinterpret_cast<const short8v*>(
-            &lds.QT[c * 32 + l31][hi * 8]);
+            &lds.QT[c * 32 + l31][qoff + hi * 8]);
         short8v qt1 = *reinterpret_cast<const short8v*>(
-            &lds.QT[c * 32 + l31][16 + hi * 8]);
+            &lds.QT[c * 32 + l31][qoff + 16 + hi * 8]);
         dk_acc[c] = MFMA32(df0, qt0, dk_acc[c]);
         dk_acc[c] = MFMA32(df1, qt1, dk_acc[c]);
       }
+      }  // qh
     }
   }
 
@@ -296,9 +300,9 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 // ---------------------------------------------------------------- dq
 template <int D>
 struct DqLds {
-  unsigned short K[64][D + 8];
-  unsigned short V[64][D + 8];
-  unsigned short KT[D][64 + 8];
+  unsigned short K[128][D + 8];
+  unsigned short V[128][D + 8];
+  unsigned short KT[D][128 + 8];
 };
 
 template <int D>
@@ -368,27 +372,27 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
 
   const int q_hi_blk = min(q0 + 255, S - 1);
   const int kv_hi = causal ? min(Skv - 1, q_hi_blk + diag) : (Skv - 1);
-  const int ntiles = kv_hi / KVB + 1;
+  const int nstages = kv_hi / 128 + 1;     // 128 kv rows per stage
 
-  for (int t = 0; t < ntiles; ++t) {
-    const int kv0 = t * KVB;
-    // ---- stage K rows, V rows, KT columns
-    for (int idx = threadIdx.x; idx < KVB * (D / 8); idx += 512) {
+  for (int st2 = 0; st2 < nstages; ++st2) {
+    const int kvs = st2 * 128;
+    // ---- stage 128 K rows, V rows, KT columns
+    for (int idx = threadIdx.x; idx < 128 * (D / 8); idx += 512) {
       const int row = idx / (D / 8), g = idx % (D / 8);
       short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
       short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (kv0 + row < Skv) {
-        const long off = kbase + (long)(kv0 + row) * krowstr + g * 8;
+      if (kvs + row < Skv) {
+        const long off = kbase + (long)(kvs + row) * krowstr + g * 8;
         k8 = *reinterpret_cast<const short8v*>(Kp + off);
         v8 = *reinterpret_cast<const short8v*>(Vp + off);
       }
       *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = k8;
       *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = v8;
     }
-    for (int idx = threadIdx.x; idx < D * (KVB / 8); idx += 512) {
-      const int dd = idx / (KVB / 8), g = idx % (KVB / 8);
+    for (int idx = threadIdx.x; idx < D * (128 / 8); idx += 512) {
+      const int dd = idx / (128 / 8), g = idx % (128 / 8);
       short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      const int s0 = kv0 + g * 8;
+      const int s0 = kvs + g * 8;
       if (s0 + 8 <= Skv) {
         k8 = *reinterpret_cast<const short8v*>(KTg + ktbase +
                                                (long)dd * Skv + s0);
@@ -402,7 +406,11 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
     }
     __syncthreads();
 
-    const bool wave_dead = causal && (kv0 > qw + 31 + diag);
+    for (int kh = 0; kh < 2; ++kh) {
+    const int kv0 = kvs + kh * KVB;
+    const int koff = kh * KVB;               // LDS row/col offset
+    const bool wave_dead = (kv0 > kv_hi) ||
+                           (causal && (kv0 > qw + 31 + diag));
     if (!wave_dead) {
       const int qg = qw + l31;
       const bool interior = (kv0 + KVB <= Skv) &&
@@ -417,9 +425,9 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
         for (int kc = 0; kc < DC16; ++kc) {
           short8v ka = *reinterpret_cast<const short8v*>(
-              &lds.K[ss * 32 + l31][kc * 16 + hi * 8]);
+              &lds.K[koff + ss * 32 + l31][kc * 16 + hi * 8]);
           short8v va = *reinterpret_cast<const short8v*>(
-              &lds.V[ss * 32 + l31][kc * 16 + hi * 8]);
+              &lds.V[koff + ss * 32 + l31][kc * 16 + hi * 8]);
           st = MFMA32(ka, qfrag[kc], st);
           dpt = MFMA32(va, dofrag[kc], dpt);
         }
@@ -442,14 +450,15 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
         for (int c = 0; c < ND32; ++c) {
           short8v kt0 = *reinterpret_cast<const short8v*>(
-              &lds.KT[c * 32 + l31][ss * 32 + hi * 8]);
+              &lds.KT[c * 32 + l31][koff + ss * 32 + hi * 8]);
           short8v kt1 = *reinterpret_cast<const short8v*>(
-              &lds.KT[c * 32 + l31][ss * 32 + 16 + hi * 8]);
+              &lds.KT[c * 32 + l31][koff + ss * 32 + 16 + hi * 8]);
           dq_acc[c] = MFMA32(f0, kt0, dq_acc[c]);
           dq_acc[c] = MFMA32(f1, kt1, dq_acc[c]);
         }
       }
     }
+    }  // kh
     __syncthreads();
   }
 

@@ -49,8 +49,8 @@ __device__ __forceinline__ int crow(int r, int hi) {
 
 template <int D>
 struct AttnFwdLds {
-  unsigned short K[64][D + 8];
-  unsigned short VT[D][64 + 8];
+  unsigned short K[128][D + 8];
+  unsigned short VT[D][128 + 8];
 };
 
 template <int D>
@@ -112,26 +112,26 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
 
   const int q_hi_blk = min(q0 + 255, S - 1);
   const int kv_hi = causal ? min(Skv - 1, q_hi_blk + diag) : (Skv - 1);
-  const int ntiles = kv_hi / KVB + 1;
+  const int nstages = kv_hi / 128 + 1;     // 128 kv rows per stage
 
-  for (int t = 0; t < ntiles; ++t) {
-    const int kv0 = t * KVB;
-    // ---- stage K [64][D] rows and VT [D][64] rows (coalesced 16B)
+  for (int st2 = 0; st2 < nstages; ++st2) {
+    const int kvs = st2 * 128;
+    // ---- stage K [128][D] rows and VT [D][128] rows (coalesced 16B)
     {
-      constexpr int KGROUPS = KVB * D / 8;     // 16B groups in the K tile
+      constexpr int KGROUPS = 128 * D / 8;     // 16B groups in the K tile
       for (int idx = threadIdx.x; idx < KGROUPS; idx += 512) {
         const int row = idx / (D / 8), g = idx % (D / 8);
         short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (kv0 + row < Skv)
+        if (kvs + row < Skv)
           k8 = *reinterpret_cast<const short8v*>(
-              Kp + kbase + (long)(kv0 + row) * krowstr + g * 8);
+              Kp + kbase + (long)(kvs + row) * krowstr + g * 8);
         *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = k8;
       }
-      constexpr int VGROUPS = D * KVB / 8;
+      constexpr int VGROUPS = D * 128 / 8;
       for (int idx = threadIdx.x; idx < VGROUPS; idx += 512) {
-        const int dd = idx / (KVB / 8), g = idx % (KVB / 8);
+        const int dd = idx / (128 / 8), g = idx % (128 / 8);
         short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        const int s0 = kv0 + g * 8;
+        const int s0 = kvs + g * 8;
         const long vrow = vtbase + (long)dd * Skv;
         if (s0 + 8 <= Skv) {
           v8 = *reinterpret_cast<const short8v*>(VTp + vrow + s0);
@@ -145,7 +145,11 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
     }
     __syncthreads();
 
-    const bool wave_dead = causal && (kv0 > qw + 31 + diag);
+    for (int half = 0; half < 2; ++half) {
+    const int kv0 = kvs + half * KVB;
+    const int koff = half * KVB;             // LDS row offset
+    const bool wave_dead = (kv0 > kv_hi) ||
+                           (causal && (kv0 > qw + 31 + diag));
     if (!wave_dead) {
       // ---- S^T tiles: s[ss] = K[kv0+ss*32..][*] x Q^T  (C-layout:
       // row kv = crow(r,hi)+32*ss, col q = qw+l31)
@@ -155,9 +159,9 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int kc = 0; kc < DC16; ++kc) {
         short8v k0 = *reinterpret_cast<const short8v*>(
-            &lds.K[l31][kc * 16 + hi * 8]);
+            &lds.K[koff + l31][kc * 16 + hi * 8]);
         short8v k1 = *reinterpret_cast<const short8v*>(
-            &lds.K[32 + l31][kc * 16 + hi * 8]);
+            &lds.K[koff + 32 + l31][kc * 16 + hi * 8]);
         s0v = MFMA32(k0, qfrag[kc], s0v);
         s1v = MFMA32(k1, qfrag[kc], s1v);
       }
@@ -237,7 +241,7 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       }
 
       // ---- rescale O rows by alpha[q-row] (one bpermute per row)
-      if (t > 0) {
+      if (st2 > 0 || half > 0) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const float a_r = __shfl(alpha, crow(r, hi), 64);
@@ -252,11 +256,12 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
           short8v vf = *reinterpret_cast<const short8v*>(
-              &lds.VT[c * 32 + l31][ks * 16 + hi * 8]);
+              &lds.VT[c * 32 + l31][koff + ks * 16 + hi * 8]);
           o_acc[c] = MFMA32(pa[ks], vf, o_acc[c]);
         }
       }
     }
+    }  // half
     __syncthreads();
   }
 
