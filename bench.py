@@ -33,7 +33,8 @@ from datatunerx_amd.parallel.ddp import init_distributed  # noqa: E402
 from datatunerx_amd.train.trainer import SFTTrainer, TrainerConfig  # noqa: E402
 
 
-def build_model(name: str, device, lora_dropout: float):
+def build_model(name: str, device, lora_dropout: float,
+                full_param: bool = False, grad_ckpt: bool = False):
     if name == "llama2-7b":
         cfg = LlamaConfig.llama2_7b(lora_dropout=lora_dropout)
     elif name == "llama2-13b":
@@ -42,9 +43,11 @@ def build_model(name: str, device, lora_dropout: float):
         cfg = LlamaConfig.tiny(lora_dropout=lora_dropout)
     else:
         raise SystemExit(f"unknown --model {name}")
+    cfg.gradient_checkpointing = grad_ckpt
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     with torch.device(device):
-        model = LlamaForCausalLM(cfg, lora=True, dtype=dtype)
+        model = LlamaForCausalLM(cfg, lora=not full_param, dtype=dtype,
+                                 train_base=full_param)
     model.init_random(seed=1234)
     return model, cfg
 
@@ -60,6 +63,11 @@ def main():
                     help="0 = auto (16 on GPU, 2 on CPU)")
     ap.add_argument("--grad-accum", type=int, default=1)
     ap.add_argument("--lora-dropout", type=float, default=0.1)
+    ap.add_argument("--full-param", action="store_true",
+                    help="full-parameter SFT (configs[4]) instead of LoRA")
+    ap.add_argument("--grad-ckpt", action="store_true")
+    ap.add_argument("--optimizer-mode", default="auto",
+                    help="auto | flat | overlap | zero1")
     args = ap.parse_args()
 
     rank, world, local_rank, device = init_distributed()
@@ -70,7 +78,8 @@ def main():
         args.seq_len = min(args.seq_len, 128)
 
     mb = args.micro_batch or (16 if device.type == "cuda" else 2)
-    model, cfg = build_model(args.model, device, args.lora_dropout)
+    model, cfg = build_model(args.model, device, args.lora_dropout,
+                             args.full_param, args.grad_ckpt)
 
     n_examples = mb * world * max(2, args.grad_accum) * 4
     ds = SFTDataset.synthetic(n_examples, args.seq_len, cfg.vocab_size,
@@ -80,6 +89,7 @@ def main():
         max_steps=10 ** 9, micro_batch_size=mb,
         gradient_accumulation_steps=args.grad_accum,
         logging_steps=0, lora_dropout=args.lora_dropout,
+        optimizer_mode=args.optimizer_mode,
         lora_r=cfg.lora_r, lora_alpha=cfg.lora_alpha,
         lora_targets=cfg.lora_targets, base_model=args.model)
     trainer = SFTTrainer(model, ds, tcfg, device=device, rank=rank,
@@ -136,9 +146,11 @@ def main():
                 "global_batch": mb * args.grad_accum * world,
                 "seq_len": args.seq_len,
                 "parallelism": f"dp{world}",
-                "lora": {"r": cfg.lora_r, "alpha": cfg.lora_alpha,
-                          "dropout": args.lora_dropout,
-                          "targets": list(cfg.lora_targets)},
+                "finetuning_type": "full" if args.full_param else "lora",
+                "lora": None if args.full_param else {
+                    "r": cfg.lora_r, "alpha": cfg.lora_alpha,
+                    "dropout": args.lora_dropout,
+                    "targets": list(cfg.lora_targets)},
             },
         }))
     if dist.is_initialized():

@@ -137,8 +137,12 @@ class LlamaDecoderLayer(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
+    """`lora=True` trains only adapters (reference's finetuning_type
+    lora); `train_base=True` makes every weight trainable (full-param
+    SFT, the reference's finetuning_type=full — parser.py:131-137)."""
+
     def __init__(self, cfg: LlamaConfig, lora: bool = True,
-                 dtype=torch.bfloat16):
+                 dtype=torch.bfloat16, train_base: bool = False):
         super().__init__()
         self.cfg = cfg
         self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
@@ -155,6 +159,9 @@ class LlamaForCausalLM(nn.Module):
                                cfg.rope_theta, dtype=torch.float32)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+        if train_base:
+            for p in self.parameters():
+                p.requires_grad_(True)
 
     @torch.no_grad()
     def init_random(self, std: float = 0.02, seed: int = 0):

@@ -36,6 +36,30 @@ def test_loss_decreases(tmp_path):
     assert sum(last_losses[-3:]) / 3 < sum(first_losses) / 3
 
 
+def test_full_param_loss_decreases(tmp_path):
+    """Full-parameter SFT path (reference finetuning_type=full): every
+    weight trainable, overlap-bucketed optimizer auto-selected."""
+    cfg = LlamaConfig.tiny()
+    model = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32,
+                             train_base=True).init_random()
+    n_train = sum(p.numel() for _, p in model.trainable_parameters())
+    n_all = sum(p.numel() for p in model.parameters())
+    assert n_train == n_all
+    ds = SFTDataset.synthetic(64, 48, 512, seed=0)
+    tr = SFTTrainer(model, ds,
+                    TrainerConfig(output_dir=str(tmp_path), max_steps=12,
+                                  micro_batch_size=4, logging_steps=0,
+                                  learning_rate=5e-4))
+    losses = []
+    it = iter(tr.train_loader)
+    for _ in range(12):
+        losses.append(tr.train_step([next(it)]))
+    assert sum(losses[-3:]) / 3 < sum(losses[:3]) / 3
+    # full (non-LoRA) checkpoint writes a plain safetensors state dict
+    out = tr.save_checkpoint(str(tmp_path / "ck"))
+    assert os.path.exists(os.path.join(out, "model.safetensors"))
+
+
 def test_jsonl_and_eval_metrics(tmp_path):
     tr = _tiny_trainer(tmp_path, steps=4)
     tr.train()
