@@ -41,21 +41,20 @@ class FrozenLinear(nn.Module):
 
 
 class LoRAFunctionWithDropout(torch.autograd.Function):
-    """Fused LoRA linear with PEFT-style input dropout on the low-rank path.
-
-    mask is pre-scaled (0 or 1/keep) or None.
+    """Fused LoRA linear with PEFT-style input dropout on the low-rank
+    path. The pre-scaled mask (0 or 1/keep, bf16) is applied INSIDE the
+    contract/wgrad/expand kernels — no x*mask materialization, no
+    separate addcmul in backward.
     """
 
     @staticmethod
     def forward(ctx, x, w, a, b, scale, mask):
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
-        xd = x2 * mask if mask is not None else x2
         y = F.linear(x2, w)
-        t = lora_contract(xd, a)
+        t = lora_contract(x2, a, mask)           # (x o mask) @ A^T
         lora_expand_add(y, t, b, scale)
-        ctx.save_for_backward(x2, w, a, b, t,
-                              mask if mask is not None else torch.empty(0))
+        ctx.save_for_backward(x2, w, a, b, t, mask)
         ctx.scale, ctx.xshape = scale, xs
         return y.reshape(*xs[:-1], w.shape[0])
 
@@ -64,19 +63,13 @@ class LoRAFunctionWithDropout(torch.autograd.Function):
         from ..ops import lora_wgrad
         x2, w, a, b, t, mask = ctx.saved_tensors
         s = ctx.scale
-        has_mask = mask.numel() > 0
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = dy2 @ w
         dt = lora_contract(dy2, b.t().contiguous())       # [M,r] = dy @ B
-        xd = x2 * mask if has_mask else x2
-        da = lora_wgrad(dt, xd, s)
+        da = lora_wgrad(dt, x2, s, mask)         # dt^T @ (x o mask)
         db = lora_wgrad(t, dy2, s).t().contiguous()
-        if has_mask:
-            dxl = torch.zeros_like(dx)
-            lora_expand_add(dxl, dt, a.t().contiguous(), s)
-            dx = dx.addcmul_(dxl, mask)
-        else:
-            lora_expand_add(dx, dt, a.t().contiguous(), s)
+        # dx += mask o (s * dt @ A)
+        lora_expand_add(dx, dt, a.t().contiguous(), s, mask)
         return (dx.reshape(ctx.xshape), None, da.to(a.dtype),
                 db.to(b.dtype), None, None)
 

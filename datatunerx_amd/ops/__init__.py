@@ -125,23 +125,25 @@ def attn_bwd(q, k, v, o, do, lse, causal: bool = True,
 
 
 # -------------------------------------------------------------- LoRA
-def lora_contract(x, w):
+# `mask` (optional, same shape/dtype as x / y) fuses the PEFT input
+# dropout into the kernels.
+def lora_contract(x, w, mask=None):
     if _gpu(x):
-        return _EXT.lora_contract(x, w)
-    return ref.lora_contract(x, w)
+        return _EXT.lora_contract(x, w, mask)
+    return ref.lora_contract(x, w, mask)
 
 
-def lora_expand_add(y, t, w, scale: float):
+def lora_expand_add(y, t, w, scale: float, mask=None):
     if _gpu(y):
-        _EXT.lora_expand_add(y, t, w, scale)
+        _EXT.lora_expand_add(y, t, w, scale, mask)
         return y
-    return ref.lora_expand_add(y, t, w, scale)
+    return ref.lora_expand_add(y, t, w, scale, mask)
 
 
-def lora_wgrad(t, x, scale: float = 1.0):
+def lora_wgrad(t, x, scale: float = 1.0, mask=None):
     if _gpu(x):
-        return _EXT.lora_wgrad(t, x, scale)
-    return ref.lora_wgrad(t, x, scale)
+        return _EXT.lora_wgrad(t, x, scale, mask)
+    return ref.lora_wgrad(t, x, scale, mask)
 
 
 # ------------------------------------------------------------- AdamW

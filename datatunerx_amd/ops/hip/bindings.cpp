@@ -23,13 +23,13 @@ void launch_adamw(void*, float*, const float*, float*, float*, long, float,
                   float, float, float, float, int, hipStream_t);
 void launch_l2_norm(const float*, float*, float*, long, hipStream_t);
 int lora_contract_ksplit(int K);
-void launch_lora_contract(const void*, const void*, float*, float*, long,
-                          int, int, hipStream_t);
-void launch_lora_expand_add(void*, const float*, const void*, long, int, int,
-                            float, hipStream_t);
+void launch_lora_contract(const void*, const void*, const void*, float*,
+                          float*, long, int, int, hipStream_t);
+void launch_lora_expand_add(void*, const float*, const void*, const void*,
+                            long, int, int, float, hipStream_t);
 int lora_wgrad_splitm(int K);
-void launch_lora_wgrad(const float*, const void*, float*, float*, long, int,
-                       int, float, hipStream_t);
+void launch_lora_wgrad(const float*, const void*, const void*, float*,
+                       float*, long, int, int, float, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, int, float, int, hipStream_t);
 void launch_transpose_sd(const void*, void*, int, int, int, int,
@@ -157,7 +157,18 @@ torch::Tensor xent_bwd(torch::Tensor logits, torch::Tensor targets,
 }
 
 // ----------------------------------------------------------------- LoRA
-torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w) {
+// Optional `mask` (bf16, same shape as x / y) fuses the PEFT-style
+// input dropout into the kernels (no mask-multiply materialization).
+static const void* opt_mask(const c10::optional<torch::Tensor>& m,
+                            long numel, const char* name) {
+  if (!m.has_value()) return nullptr;
+  check_bf16_contig(*m, name);
+  TORCH_CHECK(m->numel() == numel, name, " shape mismatch");
+  return m->data_ptr();
+}
+
+torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w,
+                            c10::optional<torch::Tensor> mask) {
   check_bf16_contig(x, "x");
   check_bf16_contig(w, "w");
   const int K = (int)x.size(-1);
@@ -174,13 +185,14 @@ torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w) {
     part = torch::empty({nsplit, M, r}, x.options().dtype(torch::kFloat));
     part_ptr = part.data_ptr<float>();
   }
-  launch_lora_contract(x.data_ptr(), w.data_ptr(), part_ptr,
+  launch_lora_contract(x.data_ptr(), w.data_ptr(),
+                       opt_mask(mask, x.numel(), "mask"), part_ptr,
                        t.data_ptr<float>(), M, K, r, cur_stream());
   return t;
 }
 
 void lora_expand_add(torch::Tensor y, torch::Tensor t, torch::Tensor w,
-                     double scale) {
+                     double scale, c10::optional<torch::Tensor> mask) {
   check_bf16_contig(y, "y");
   check_bf16_contig(w, "w");
   const int N = (int)y.size(-1);
@@ -188,11 +200,14 @@ void lora_expand_add(torch::Tensor y, torch::Tensor t, torch::Tensor w,
   const int r = (int)w.size(1);
   TORCH_CHECK(w.size(0) == N, "w [N,r] mismatch");
   TORCH_CHECK(t.scalar_type() == torch::kFloat && t.is_contiguous());
-  launch_lora_expand_add(y.data_ptr(), t.data_ptr<float>(), w.data_ptr(),
-                         M, N, r, (float)scale, cur_stream());
+  auto wt = w.t().contiguous();          // [r,N] for coalesced rows
+  launch_lora_expand_add(y.data_ptr(), t.data_ptr<float>(), wt.data_ptr(),
+                         opt_mask(mask, y.numel(), "mask"), M, N, r,
+                         (float)scale, cur_stream());
 }
 
-torch::Tensor lora_wgrad(torch::Tensor t, torch::Tensor x, double scale) {
+torch::Tensor lora_wgrad(torch::Tensor t, torch::Tensor x, double scale,
+                         c10::optional<torch::Tensor> mask) {
   check_bf16_contig(x, "x");
   TORCH_CHECK(t.scalar_type() == torch::kFloat && t.is_contiguous());
   const int K = (int)x.size(-1);
@@ -202,6 +217,7 @@ torch::Tensor lora_wgrad(torch::Tensor t, torch::Tensor x, double scale) {
   const int sm = lora_wgrad_splitm(K);
   auto part = torch::empty({sm, r, K}, x.options().dtype(torch::kFloat));
   launch_lora_wgrad(t.data_ptr<float>(), x.data_ptr(),
+                    opt_mask(mask, x.numel(), "mask"),
                     part.data_ptr<float>(), out.data_ptr<float>(), M, K, r,
                     (float)scale, cur_stream());
   return out;
@@ -315,9 +331,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("xent_fwd", &xent_fwd);
   m.def("xent_bwd", &xent_bwd);
-  m.def("lora_contract", &lora_contract);
-  m.def("lora_expand_add", &lora_expand_add);
-  m.def("lora_wgrad", &lora_wgrad);
+  m.def("lora_contract", &lora_contract, py::arg("x"), py::arg("w"),
+        py::arg("mask") = py::none());
+  m.def("lora_expand_add", &lora_expand_add, py::arg("y"), py::arg("t"),
+        py::arg("w"), py::arg("scale"), py::arg("mask") = py::none());
+  m.def("lora_wgrad", &lora_wgrad, py::arg("t"), py::arg("x"),
+        py::arg("scale"), py::arg("mask") = py::none());
   m.def("adamw", &adamw);
   m.def("l2_norm", &l2_norm);
   m.def("attn_fwd", &attn_fwd);

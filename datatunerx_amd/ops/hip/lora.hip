@@ -24,9 +24,11 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 // (each row read once; 8 kc loads/lane = 128 contiguous bytes).
 // B-frag: lane reads the LDS copy of W at row l31 (zero rows >= r).
 // Split-K across blockIdx.y; fp32 partials reduced by reduce_partials.
+template <bool MASKED>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_contract_kernel(const unsigned short* __restrict__ X,
                           const unsigned short* __restrict__ W,
+                          const unsigned short* __restrict__ Mk,
                           float* __restrict__ part,
                           long M, int K, int r, int kspan) {
   __shared__ unsigned short wlds[32][1024 + 8];
@@ -59,12 +61,22 @@ void lora_contract_kernel(const unsigned short* __restrict__ X,
 #pragma unroll
         for (int q = 0; q < 16; ++q) acc[q] = 0.f;
         const long xrow = mw + l31;
-        const unsigned short* xp =
-            xrow < M ? X + xrow * K + kc0 : X;   // row<M guard, safe base
+        const long rbase = xrow * K + kc0;
+        const unsigned short* xp = xrow < M ? X + rbase : X;
+        const unsigned short* mp =
+            (MASKED && xrow < M) ? Mk + rbase : nullptr;
         for (int kc = 0; kc < kc_n; kc += 16) {
           short8v xf = xrow < M
               ? *reinterpret_cast<const short8v*>(xp + kc + hi * 8)
               : short8v{0, 0, 0, 0, 0, 0, 0, 0};
+          if (MASKED && mp) {
+            short8v mf = *reinterpret_cast<const short8v*>(
+                mp + kc + hi * 8);
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              xf[e] = (short)f2bf(bf2f((unsigned short)xf[e]) *
+                                  bf2f((unsigned short)mf[e]));
+          }
           short8v wf = *reinterpret_cast<const short8v*>(
               &wlds[l31][kc + hi * 8]);
           acc = MFMA32L(xf, wf, acc);
@@ -89,62 +101,70 @@ void lora_contract_kernel(const unsigned short* __restrict__ X,
 }
 
 // -------------------------------------------------------------- expand_add
-// Y[M,N] += s * T[M,r] @ W[N,r]^T. W transposed into LDS [r][N-chunk];
-// one wave per row, two 8-element n-chunks in flight (ILP), j blocked by
-// 8 so register pressure is independent of r (no template on r).
+// Y[M,N] += (mask o) s * T[M,r] @ WT[r,N].  WT is the PRE-TRANSPOSED
+// adapter (binding does w.t().contiguous(): 64 KB, L1/L2-resident), so
+// every access here is a coalesced 16-byte load — no LDS staging pass.
+// j blocked by 8 so register pressure is independent of r.
+template <bool MASKED>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_expand_add_kernel(unsigned short* __restrict__ Y,
                             const float* __restrict__ T,
-                            const unsigned short* __restrict__ W,
-                            long M, int N, int r, float s, int chunk) {
-  extern __shared__ __attribute__((aligned(16))) unsigned short wlds[];
+                            const unsigned short* __restrict__ WT,
+                            const unsigned short* __restrict__ Mk,
+                            long M, int N, int r, float s) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  for (int n0 = 0; n0 < N; n0 += chunk) {
-    const int nc = min(chunk, N - n0);
-    for (int n = threadIdx.x; n < nc; n += DTX_BLOCK) {
-      const unsigned short* wr = W + (long)(n0 + n) * r;
-      for (int j = 0; j < r; ++j) wlds[j * chunk + n] = wr[j];
-    }
-    __syncthreads();
-    for (long m = blockIdx.x * 4 + wid; m < M; m += (long)gridDim.x * 4) {
-      const float* tr = T + m * r;
-      unsigned short* yr = Y + m * N + n0;
-      for (int n = lane * 8; n < nc; n += WAVE * 16) {
-        float y[2][8];
-        bool live[2];
+  for (long m = blockIdx.x * 4 + wid; m < M; m += (long)gridDim.x * 4) {
+    const float* tr = T + m * r;
+    unsigned short* yr = Y + m * N;
+    const unsigned short* mr = MASKED ? Mk + m * N : nullptr;
+    for (int n = lane * 8; n < N; n += WAVE * 16) {
+      const int n2 = n + WAVE * 8;
+      const bool l2 = n2 < N;
+      float a0[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      float a1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      for (int jb = 0; jb < r; jb += 8) {
+        float tv[8];
 #pragma unroll
-        for (int c = 0; c < 2; ++c) {
-          live[c] = n + c * WAVE * 8 < nc;
-          if (live[c]) load_bf16x8(yr + n + c * WAVE * 8, y[c]);
-        }
-        for (int jb = 0; jb < r; jb += 8) {
-          float tv[8];
+        for (int t = 0; t < 8; ++t)
+          tv[t] = (jb + t < r) ? s * tr[jb + t] : 0.f;
 #pragma unroll
-          for (int t = 0; t < 8; ++t)
-            tv[t] = (jb + t < r) ? s * tr[jb + t] : 0.f;
+        for (int t = 0; t < 8; ++t) {
+          if (jb + t < r) {
+            const unsigned short* wrow = WT + (long)(jb + t) * N;
+            float w0[8], w1[8];
+            load_bf16x8(wrow + n, w0);
+            if (l2) load_bf16x8(wrow + n2, w1);
 #pragma unroll
-          for (int t = 0; t < 8; ++t) {
-            if (jb + t < r) {
-#pragma unroll
-              for (int c = 0; c < 2; ++c) {
-                if (live[c]) {
-                  float w[8];
-                  load_bf16x8(&wlds[(jb + t) * chunk + n + c * WAVE * 8],
-                              w);
-#pragma unroll
-                  for (int i = 0; i < 8; ++i) y[c][i] += tv[t] * w[i];
-                }
-              }
+            for (int i = 0; i < 8; ++i) {
+              a0[i] += tv[t] * w0[i];
+              if (l2) a1[i] += tv[t] * w1[i];
             }
           }
         }
-#pragma unroll
-        for (int c = 0; c < 2; ++c)
-          if (live[c]) store_bf16x8(yr + n + c * WAVE * 8, y[c]);
       }
+      float y0[8], y1[8];
+      load_bf16x8(yr + n, y0);
+      if (l2) load_bf16x8(yr + n2, y1);
+      if (MASKED) {
+        float m0[8], m1[8];
+        load_bf16x8(mr + n, m0);
+        if (l2) load_bf16x8(mr + n2, m1);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          y0[i] += m0[i] * a0[i];
+          if (l2) y1[i] += m1[i] * a1[i];
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          y0[i] += a0[i];
+          if (l2) y1[i] += a1[i];
+        }
+      }
+      store_bf16x8(yr + n, y0);
+      if (l2) store_bf16x8(yr + n2, y1);
     }
-    __syncthreads();
   }
 }
 
@@ -152,10 +172,11 @@ void lora_expand_add_kernel(unsigned short* __restrict__ Y,
 // part[ms][j][K] = s * sum_{m in split ms} T[m,j] * X[m,k]
 // (k-span 2048/block keeps X re-reads at ceil(K/2048); the partials are
 // streamed back by the vectorized reduce_partials kernel)
-template <int RCH>
+template <int RCH, bool MASKED>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_wgrad_kernel(const float* __restrict__ T,
                        const unsigned short* __restrict__ X,
+                       const unsigned short* __restrict__ Mk,
                        float* __restrict__ part,
                        long M, int K, int r, int j0, int splitm, float s) {
   const int col = blockIdx.x * 2048 + threadIdx.x * 8;
@@ -171,6 +192,13 @@ void lora_wgrad_kernel(const float* __restrict__ T,
   for (long m = m_begin; m < m_end; ++m) {
     float xv[8];
     load_bf16x8(X + m * K + col, xv);
+    if (MASKED) {
+      float mv[8];
+      load_bf16x8(Mk + m * K + col, mv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        xv[i] = bf2f(f2bf(xv[i] * mv[i]));   // match bf16 x*mask numerics
+    }
     const float* tr = T + m * r + j0;
 #pragma unroll
     for (int j = 0; j < RCH; ++j) {
@@ -203,29 +231,41 @@ int lora_contract_ksplit(int K) {
   return nsplit;
 }
 
-void launch_lora_contract(const void* X, const void* W, float* part,
-                          float* out, long M, int K, int r, hipStream_t s) {
+void launch_lora_contract(const void* X, const void* W, const void* Mk,
+                          float* part, float* out, long M, int K, int r,
+                          hipStream_t s) {
   const int nsplit = lora_contract_ksplit(K);
   const int kspan = DTX_CDIV(K, nsplit);
   long gw = DTX_CDIV(M, 128);
   dim3 grid((int)(gw < 128 ? (gw < 1 ? 1 : gw) : 128), nsplit);
-  lora_contract_kernel<<<grid, DTX_BLOCK, 0, s>>>(
-      (const unsigned short*)X, (const unsigned short*)W,
-      nsplit > 1 ? part : out, M, K, r, kspan);
+  float* dst = nsplit > 1 ? part : out;
+  if (Mk) {
+    lora_contract_kernel<true><<<grid, DTX_BLOCK, 0, s>>>(
+        (const unsigned short*)X, (const unsigned short*)W,
+        (const unsigned short*)Mk, dst, M, K, r, kspan);
+  } else {
+    lora_contract_kernel<false><<<grid, DTX_BLOCK, 0, s>>>(
+        (const unsigned short*)X, (const unsigned short*)W, nullptr, dst,
+        M, K, r, kspan);
+  }
   if (nsplit > 1)
     launch_reduce_partials(part, out, nsplit, M * r, s);
 }
 
-void launch_lora_expand_add(void* Y, const float* T, const void* W, long M,
-                            int N, int r, float scale, hipStream_t s) {
-  int chunk = (32768 / r) & ~7;          // 64 KiB of bf16 LDS
-  if (N < chunk) chunk = N;
-  size_t lds = (size_t)r * chunk * 2;
+void launch_lora_expand_add(void* Y, const float* T, const void* WT,
+                            const void* Mk, long M, int N, int r,
+                            float scale, hipStream_t s) {
   long gw = DTX_CDIV(M, 4);
   int grid = (int)(gw < 2048 ? (gw < 1 ? 1 : gw) : 2048);
-  lora_expand_add_kernel<<<grid, DTX_BLOCK, lds, s>>>(
-      (unsigned short*)Y, T, (const unsigned short*)W, M, N, r, scale,
-      chunk);
+  if (Mk) {
+    lora_expand_add_kernel<true><<<grid, DTX_BLOCK, 0, s>>>(
+        (unsigned short*)Y, T, (const unsigned short*)WT,
+        (const unsigned short*)Mk, M, N, r, scale);
+  } else {
+    lora_expand_add_kernel<false><<<grid, DTX_BLOCK, 0, s>>>(
+        (unsigned short*)Y, T, (const unsigned short*)WT, nullptr, M, N,
+        r, scale);
+  }
 }
 
 int lora_wgrad_splitm(int K) {
@@ -234,15 +274,24 @@ int lora_wgrad_splitm(int K) {
   return sm < 1 ? 1 : (sm > 128 ? 128 : sm);
 }
 
-void launch_lora_wgrad(const float* T, const void* X, float* part,
-                       float* out, long M, int K, int r, float s,
-                       hipStream_t st) {
+void launch_lora_wgrad(const float* T, const void* X, const void* Mk,
+                       float* part, float* out, long M, int K, int r,
+                       float s, hipStream_t st) {
   const int splitm = lora_wgrad_splitm(K);
   dim3 grid(DTX_CDIV(K, 2048), splitm);
   for (int j0 = 0; j0 < r; j0 += 16) {
     int rch = r - j0;
-#define CASE(RC) lora_wgrad_kernel<RC><<<grid, DTX_BLOCK, 0, st>>>( \
-        T, (const unsigned short*)X, part, M, K, r, j0, splitm, s)
+#define CASE(RC)                                                          \
+    do {                                                                  \
+      if (Mk)                                                             \
+        lora_wgrad_kernel<RC, true><<<grid, DTX_BLOCK, 0, st>>>(          \
+            T, (const unsigned short*)X, (const unsigned short*)Mk,       \
+            part, M, K, r, j0, splitm, s);                                \
+      else                                                                \
+        lora_wgrad_kernel<RC, false><<<grid, DTX_BLOCK, 0, st>>>(         \
+            T, (const unsigned short*)X, nullptr, part, M, K, r, j0,      \
+            splitm, s);                                                   \
+    } while (0)
     if (rch <= 4) CASE(4);
     else if (rch <= 8) CASE(8);
     else CASE(16);

@@ -197,23 +197,35 @@ def _attn_bwd_bhsd(q, k, v, o, do, lse, causal: bool = True,
 
 
 # ------------------------------------------------------------- LoRA pieces
-def lora_contract(x: torch.Tensor, w: torch.Tensor):
-    """t[M,r] = x[M,K] @ w[r,K]^T  (fp32 out). w is HF lora_A layout [r,K],
-    or for the dgrad pass w = lora_B [N,r] transposed to [r,N] by caller."""
-    return x.float() @ w.float().t()
+# Optional `mask` mirrors the fused-dropout GPU kernels: the masked
+# element enters the product in the INPUT dtype (x*mask rounded once).
+def _masked(x, mask):
+    if mask is None:
+        return x
+    return (x * mask).to(x.dtype)
+
+
+def lora_contract(x: torch.Tensor, w: torch.Tensor, mask=None):
+    """t[M,r] = (x o mask)[M,K] @ w[r,K]^T  (fp32 out). w is HF lora_A
+    layout [r,K], or for the dgrad pass w = lora_B transposed."""
+    return _masked(x, mask).float() @ w.float().t()
 
 
 def lora_expand_add(y: torch.Tensor, t: torch.Tensor, w: torch.Tensor,
-                    scale: float):
-    """y[M,N] += scale * t[M,r] @ w[N,r]^T  (in-place on y). HF lora_B
-    layout is [N,r]."""
-    y.add_((scale * (t.float() @ w.float().t())).to(y.dtype))
+                    scale: float, mask=None):
+    """y[M,N] += (mask o) scale * t[M,r] @ w[N,r]^T  (in-place on y).
+    HF lora_B layout is [N,r]."""
+    d = scale * (t.float() @ w.float().t())
+    if mask is not None:
+        d = d * mask.float()
+    y.add_(d.to(y.dtype))
     return y
 
 
-def lora_wgrad(t: torch.Tensor, x: torch.Tensor, scale: float = 1.0):
-    """dW[r,K] = scale * t[M,r]^T @ x[M,K]  (fp32)."""
-    return scale * (t.float().t() @ x.float())
+def lora_wgrad(t: torch.Tensor, x: torch.Tensor, scale: float = 1.0,
+               mask=None):
+    """dW[r,K] = scale * t[M,r]^T @ (x o mask)[M,K]  (fp32)."""
+    return scale * (t.float().t() @ _masked(x, mask).float())
 
 
 # ------------------------------------------------------------ fused AdamW

@@ -100,6 +100,38 @@ def test_lora_contract(M, K, r):
     assert_close(t.cpu(), t_ref, name="lora contract")
 
 
+def _mk_mask(M, K, keep=0.9):
+    m = torch.empty(M, K, device=DEV, dtype=torch.bfloat16)
+    m.bernoulli_(keep).mul_(1.0 / keep)
+    return m
+
+
+def test_lora_masked_fused_dropout():
+    """Fused-dropout variants (mask inside contract/expand/wgrad) match
+    the reference with the same mask."""
+    M, K, N, r = 512, 2048, 1024, 8
+    x, a = mk(M, K, scale=0.3), mk(r, K, scale=0.3)
+    mask = _mk_mask(M, K)
+    t = ops.lora_contract(x, a, mask)
+    t_ref = ref.lora_contract(x.cpu(), a.cpu(), mask.cpu())
+    assert_close(t.cpu(), t_ref, name="masked contract")
+
+    dt = torch.randn(M, r, device=DEV)
+    dw = ops.lora_wgrad(dt, x, 0.5, mask)
+    dw_ref = ref.lora_wgrad(dt.cpu(), x.cpu(), 0.5, mask.cpu())
+    assert_close(dw.cpu(), dw_ref, name="masked wgrad")
+
+    y = mk(M, N)
+    y0 = y.clone()
+    b = mk(N, r, scale=0.3)
+    t2 = torch.randn(M, r, device=DEV)
+    mask_y = _mk_mask(M, N)
+    ops.lora_expand_add(y, t2, b, 0.5, mask_y)
+    y_ref = ref.lora_expand_add(y0.cpu().clone(), t2.cpu(), b.cpu(), 0.5,
+                                mask_y.cpu())
+    assert_close(y.cpu(), y_ref, name="masked expand")
+
+
 @pytest.mark.parametrize("M,N,r", [(1024, 4096, 8), (511, 1024, 16)])
 def test_lora_expand_add(M, N, r):
     y = mk(M, N)
