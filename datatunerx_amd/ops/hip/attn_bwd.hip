@@ -101,8 +101,6 @@ struct DkdvLds {
   unsigned short dOr[64][D + 8];
   unsigned short QT[D][72];         // staged q columns (from QT_g)
   unsigned short dOT[D][72];
-  float lse[64];
-  float dlt[64];
 };
 
 template <int D>
@@ -208,18 +206,16 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         *reinterpret_cast<short8v*>(&lds.QT[dd][g * 8]) = q8;
         *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 8]) = d8;
       }
-      if (threadIdx.x < 64) {
-        const int qg = q0s + threadIdx.x;
-        lds.lse[threadIdx.x] = qg < S ? lse_in[lbase + qg] : 0.f;
-        lds.dlt[threadIdx.x] = qg < S ? delta_in[lbase + qg] : 0.f;
-      }
       __syncthreads();
 
+#pragma unroll
       for (int qh = 0; qh < 2; ++qh) {
       const int q0 = q0s + qh * 32;
       const int qoff = qh * 32;                  // LDS row offset
       // wave skip: its kv rows all above this q-tile's diagonal
-      if (q0 >= S || (causal && (q0 + 31 + diag < kw))) continue;
+      const bool live_tile =
+          !(q0 >= S || (causal && (q0 + 31 + diag < kw)));
+      if (live_tile) {
 
       // ---- S[q][kv] and dP[q][kv] (C-layout: q rows on regs, kv on
       // lanes = the wave's kv rows kw + l31)
@@ -244,8 +240,11 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qg = q0 + bw_crow(r, hi);
-        const float lse2 = lds.lse[qoff + bw_crow(r, hi)] * LOG2E;
-        const float dlt = lds.dlt[qoff + bw_crow(r, hi)];
+        const bool qv = qg < S;
+        // L1/L2-hot scalar loads (the lse/dlt LDS arrays used to make
+        // hipcc scalarize every b128 fragment read in this kernel)
+        const float lse2 = (qv ? lse_in[lbase + qg] : 0.f) * LOG2E;
+        const float dlt = qv ? delta_in[lbase + qg] : 0.f;
         float p;
         if (interior) {
           p = __builtin_exp2f(sv[r] * kscale - lse2);
@@ -277,6 +276,7 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         dk_acc[c] = MFMA32(df0, qt0, dk_acc[c]);
         dk_acc[c] = MFMA32(df1, qt1, dk_acc[c]);
       }
+      }  // live_tile
       }  // qh
     }
   }

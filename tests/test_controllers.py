@@ -187,6 +187,55 @@ def test_full_job_pipeline_with_serve_and_scoring(tmp_path):
         pytest.fail("serve process still alive")
 
 
+def test_experiment_fanout_best_version(tmp_path):
+    """configs[3] shape: a FinetuneExperiment fans out concurrent jobs
+    (lr sweep via hyperparameter overrides), aggregates their statuses
+    and picks bestVersion by descending score
+    (finetuneexperiment_controller.go:123-216)."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store)
+    spec_a = finetune_spec()
+    spec_b = finetune_spec()
+    spec_b["hyperparameter"]["overrides"] = {"learningRate": "5e-4"}
+    exp = FinetuneExperiment(name="exp2", spec={
+        "finetuneJobs": [
+            {"name": "e2-j1", "spec": {"fineTune":
+                                       {"finetuneSpec": spec_a}}},
+            {"name": "e2-j2", "spec": {"fineTune":
+                                       {"finetuneSpec": spec_b}}},
+        ]})
+    mgr.store.create(exp)
+    deadline = time.time() + 300
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        cur = mgr.store.get(FinetuneExperiment, "default", "exp2")
+        if cur.status.get("state") in ("Success", "Failed"):
+            break
+        time.sleep(0.3)
+    cur = mgr.store.get(FinetuneExperiment, "default", "exp2")
+    assert cur.status.get("state") == "Success", cur.status
+    assert len(cur.status["jobsStatus"]) == 2
+    best = cur.status.get("bestVersion")
+    assert best and best.get("llm") == "llama-tiny"
+    scores = [_int_score(js["finetuneJobStatus"].get("result", {})
+                         .get("score"))
+              for js in cur.status["jobsStatus"]
+              if js["finetuneJobStatus"].get("state") == "Successful"]
+    assert _int_score(best["score"]) == max(scores)
+    # jobs were gang-scheduled through the shared GPU inventory: both
+    # jobs' finetunes reached Successful
+    for name in ("e2-j1", "e2-j2"):
+        st = mgr.store.get(FinetuneJob, "default", name).status
+        assert st.get("state") in ("Successful", "Failed")
+
+
+def _int_score(s):
+    try:
+        return int(float(s))
+    except (TypeError, ValueError):
+        return -1
+
+
 def test_experiment_pending_pause(tmp_path):
     mgr = mk_manager(tmp_path)
     seed_resources(mgr.store)
