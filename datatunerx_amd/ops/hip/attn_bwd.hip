@@ -169,12 +169,15 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
     // q rows that can see this block's kv rows: q + diag >= kv0b
     const int qs_lo = causal ? max(0, (kv0b - diag) / 64) : 0;
     const int qs_hi = (S + 63) / 64;
-    for (int qs = qs_lo; qs < qs_hi; ++qs) {
-      const int q0s = qs * 64;
-      __syncthreads();
-      // ---- stage 64 q rows: Q/dO rows, QT/dOT columns, lse, delta
-      for (int idx = threadIdx.x; idx < 64 * (D / 8); idx += 256) {
+    // T14 async-stage split: stage qs+1's global loads are issued while
+    // stage qs computes; the LDS write happens after the barrier.
+    short8v stg[16];
+    auto issue_stage = [&](int qs) {
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        const int idx = threadIdx.x + it * 256;
         const int row = idx / (D / 8), g = idx % (D / 8);
+        const int q0s = qs * 64;
         short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
         short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
         if (q0s + row < S) {
@@ -182,31 +185,70 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
           q8 = *reinterpret_cast<const short8v*>(Q + off);
           d8 = *reinterpret_cast<const short8v*>(dO + off);
         }
-        *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = q8;
-        *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) = d8;
+        stg[it * 2] = q8;
+        stg[it * 2 + 1] = d8;
       }
-      for (int idx = threadIdx.x; idx < D * 8; idx += 256) {
-        const int dd = idx / 8, g = idx % 8;       // 8 groups of 8 q-cols
-        short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        const int s0 = q0s + g * 8;
-        if (s0 + 8 <= S) {
-          q8 = *reinterpret_cast<const short8v*>(QTg + qtbase +
-                                                 (long)dd * S + s0);
-          d8 = *reinterpret_cast<const short8v*>(dOTg + qtbase +
-                                                 (long)dd * S + s0);
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        const int idx = threadIdx.x + it * 256;
+        const int dd = idx / 4, g = idx % 4;     // 4 groups of 16 q-cols
+        const int s0 = qs * 64 + g * 16;
+        short8v q8a = {0, 0, 0, 0, 0, 0, 0, 0};
+        short8v d8a = {0, 0, 0, 0, 0, 0, 0, 0};
+        short8v q8b = {0, 0, 0, 0, 0, 0, 0, 0};
+        short8v d8b = {0, 0, 0, 0, 0, 0, 0, 0};
+        const long tb = qtbase + (long)dd * S;
+        if (s0 + 16 <= S) {
+          q8a = *reinterpret_cast<const short8v*>(QTg + tb + s0);
+          d8a = *reinterpret_cast<const short8v*>(dOTg + tb + s0);
+          q8b = *reinterpret_cast<const short8v*>(QTg + tb + s0 + 8);
+          d8b = *reinterpret_cast<const short8v*>(dOTg + tb + s0 + 8);
         } else if (s0 < S) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
+          for (int j = 0; j < 16; ++j)
             if (s0 + j < S) {
-              q8[j] = (short)QTg[qtbase + (long)dd * S + s0 + j];
-              d8[j] = (short)dOTg[qtbase + (long)dd * S + s0 + j];
+              short qv = (short)QTg[tb + s0 + j];
+              short dv = (short)dOTg[tb + s0 + j];
+              if (j < 8) { q8a[j] = qv; d8a[j] = dv; }
+              else { q8b[j - 8] = qv; d8b[j - 8] = dv; }
             }
         }
-        *reinterpret_cast<short8v*>(&lds.QT[dd][g * 8]) = q8;
-        *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 8]) = d8;
+        stg[8 + it * 4] = q8a;
+        stg[8 + it * 4 + 1] = q8b;
+        stg[8 + it * 4 + 2] = d8a;
+        stg[8 + it * 4 + 3] = d8b;
       }
+    };
+    auto write_stage = [&]() {
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        const int idx = threadIdx.x + it * 256;
+        const int row = idx / (D / 8), g = idx % (D / 8);
+        *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = stg[it * 2];
+        *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) =
+            stg[it * 2 + 1];
+      }
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        const int idx = threadIdx.x + it * 256;
+        const int dd = idx / 4, g = idx % 4;
+        *reinterpret_cast<short8v*>(&lds.QT[dd][g * 16]) =
+            stg[8 + it * 4];
+        *reinterpret_cast<short8v*>(&lds.QT[dd][g * 16 + 8]) =
+            stg[8 + it * 4 + 1];
+        *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 16]) =
+            stg[8 + it * 4 + 2];
+        *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 16 + 8]) =
+            stg[8 + it * 4 + 3];
+      }
+    };
+    if (qs_lo < qs_hi) issue_stage(qs_lo);
+    for (int qs = qs_lo; qs < qs_hi; ++qs) {
+      const int q0s = qs * 64;
       __syncthreads();
+      write_stage();
+      __syncthreads();
+      if (qs + 1 < qs_hi) issue_stage(qs + 1);
 
 #pragma unroll
       for (int qh = 0; qh < 2; ++qh) {
