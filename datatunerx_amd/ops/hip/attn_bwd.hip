@@ -171,10 +171,12 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
     const int qs_hi = (S + 63) / 64;
     // T14 async-stage split: stage qs+1's global loads are issued while
     // stage qs computes; the LDS write happens after the barrier.
-    short8v stg[16];
+    constexpr int RIT = (64 * (D / 8)) / 256;   // row-staging iters
+    constexpr int TIT = (D * 4) / 256;          // transposed-staging iters
+    short8v stg[RIT * 2 + TIT * 4];
     auto issue_stage = [&](int qs) {
 #pragma unroll
-      for (int it = 0; it < 4; ++it) {
+      for (int it = 0; it < RIT; ++it) {
         const int idx = threadIdx.x + it * 256;
         const int row = idx / (D / 8), g = idx % (D / 8);
         const int q0s = qs * 64;
@@ -189,7 +191,7 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         stg[it * 2 + 1] = d8;
       }
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < TIT; ++it) {
         const int idx = threadIdx.x + it * 256;
         const int dd = idx / 4, g = idx % 4;     // 4 groups of 16 q-cols
         const int s0 = qs * 64 + g * 16;
@@ -213,15 +215,15 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
               else { q8b[j - 8] = qv; d8b[j - 8] = dv; }
             }
         }
-        stg[8 + it * 4] = q8a;
-        stg[8 + it * 4 + 1] = q8b;
-        stg[8 + it * 4 + 2] = d8a;
-        stg[8 + it * 4 + 3] = d8b;
+        stg[RIT * 2 + it * 4] = q8a;
+        stg[RIT * 2 + it * 4 + 1] = q8b;
+        stg[RIT * 2 + it * 4 + 2] = d8a;
+        stg[RIT * 2 + it * 4 + 3] = d8b;
       }
     };
     auto write_stage = [&]() {
 #pragma unroll
-      for (int it = 0; it < 4; ++it) {
+      for (int it = 0; it < RIT; ++it) {
         const int idx = threadIdx.x + it * 256;
         const int row = idx / (D / 8), g = idx % (D / 8);
         *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = stg[it * 2];
@@ -229,17 +231,17 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
             stg[it * 2 + 1];
       }
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < TIT; ++it) {
         const int idx = threadIdx.x + it * 256;
         const int dd = idx / 4, g = idx % 4;
         *reinterpret_cast<short8v*>(&lds.QT[dd][g * 16]) =
-            stg[8 + it * 4];
+            stg[RIT * 2 + it * 4];
         *reinterpret_cast<short8v*>(&lds.QT[dd][g * 16 + 8]) =
-            stg[8 + it * 4 + 1];
+            stg[RIT * 2 + it * 4 + 1];
         *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 16]) =
-            stg[8 + it * 4 + 2];
+            stg[RIT * 2 + it * 4 + 2];
         *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 16 + 8]) =
-            stg[8 + it * 4 + 3];
+            stg[RIT * 2 + it * 4 + 3];
       }
     };
     if (qs_lo < qs_hi) issue_stage(qs_lo);
