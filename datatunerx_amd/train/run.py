@@ -46,6 +46,7 @@ def main(argv=None):
         dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
         name = margs.model_name_or_path
         lora = fargs.finetuning_type == "lora"
+        full = fargs.finetuning_type == "full"
         lora_kw = dict(lora_r=fargs.lora_rank, lora_alpha=fargs.lora_alpha,
                        lora_dropout=fargs.lora_dropout,
                        lora_targets=fargs.lora_targets)
@@ -53,14 +54,15 @@ def main(argv=None):
             if name in ("llama2-7b", "llama-2-7b"):
                 model = LlamaForCausalLM(LlamaConfig.llama2_7b(
                     gradient_checkpointing=fargs.gradient_checkpointing,
-                    **lora_kw), lora=lora, dtype=dtype)
+                    **lora_kw), lora=lora, dtype=dtype, train_base=full)
             elif name in ("llama2-13b", "llama-2-13b"):
                 model = LlamaForCausalLM(LlamaConfig.llama2_13b(
                     gradient_checkpointing=fargs.gradient_checkpointing,
-                    **lora_kw), lora=lora, dtype=dtype)
+                    **lora_kw), lora=lora, dtype=dtype, train_base=full)
             elif name == "llama-tiny":
                 model = LlamaForCausalLM(LlamaConfig.tiny(**lora_kw),
-                                         lora=lora, dtype=dtype)
+                                         lora=lora, dtype=dtype,
+                                         train_base=full)
             elif name in ("gpt2-small", "gpt2"):
                 model = GPT2ForCausalLM(GPT2Config.small(
                     lora_r=fargs.lora_rank, lora_alpha=fargs.lora_alpha,
