@@ -114,36 +114,62 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
   const int kv_hi = causal ? min(Skv - 1, q_hi_blk + diag) : (Skv - 1);
   const int nstages = kv_hi / 128 + 1;     // 128 kv rows per stage
 
-  for (int st2 = 0; st2 < nstages; ++st2) {
+  // T14 async-stage split: the next stage's K/VT global loads are
+  // issued while the current stage computes; LDS writes after the
+  // barrier.  KIT/VIT iterations cover the 128-row K and VT tiles.
+  constexpr int KIT = (128 * D / 8) / 512;
+  constexpr int VIT = (D * 128 / 8) / 512;
+  short8v stg[KIT + VIT];
+  auto issue_stage = [&](int st2) {
     const int kvs = st2 * 128;
-    // ---- stage K [128][D] rows and VT [D][128] rows (coalesced 16B)
-    {
-      constexpr int KGROUPS = 128 * D / 8;     // 16B groups in the K tile
-      for (int idx = threadIdx.x; idx < KGROUPS; idx += 512) {
-        const int row = idx / (D / 8), g = idx % (D / 8);
-        short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (kvs + row < Skv)
-          k8 = *reinterpret_cast<const short8v*>(
-              Kp + kbase + (long)(kvs + row) * krowstr + g * 8);
-        *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = k8;
-      }
-      constexpr int VGROUPS = D * 128 / 8;
-      for (int idx = threadIdx.x; idx < VGROUPS; idx += 512) {
-        const int dd = idx / (128 / 8), g = idx % (128 / 8);
-        short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        const int s0 = kvs + g * 8;
-        const long vrow = vtbase + (long)dd * Skv;
-        if (s0 + 8 <= Skv) {
-          v8 = *reinterpret_cast<const short8v*>(VTp + vrow + s0);
-        } else if (s0 < Skv) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (s0 + j < Skv) v8[j] = (short)VTp[vrow + s0 + j];
-        }
-        *reinterpret_cast<short8v*>(&lds.VT[dd][g * 8]) = v8;
-      }
+    for (int it = 0; it < KIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
+      const int row = idx / (D / 8), g = idx % (D / 8);
+      short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (kvs + row < Skv)
+        k8 = *reinterpret_cast<const short8v*>(
+            Kp + kbase + (long)(kvs + row) * krowstr + g * 8);
+      stg[it] = k8;
     }
+#pragma unroll
+    for (int it = 0; it < VIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
+      const int dd = idx / (128 / 8), g = idx % (128 / 8);
+      short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int s0 = kvs + g * 8;
+      const long vrow = vtbase + (long)dd * Skv;
+      if (s0 + 8 <= Skv) {
+        v8 = *reinterpret_cast<const short8v*>(VTp + vrow + s0);
+      } else if (s0 < Skv) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (s0 + j < Skv) v8[j] = (short)VTp[vrow + s0 + j];
+      }
+      stg[KIT + it] = v8;
+    }
+  };
+  auto write_stage = [&]() {
+#pragma unroll
+    for (int it = 0; it < KIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
+      const int row = idx / (D / 8), g = idx % (D / 8);
+      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it];
+    }
+#pragma unroll
+    for (int it = 0; it < VIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
+      const int dd = idx / (128 / 8), g = idx % (128 / 8);
+      *reinterpret_cast<short8v*>(&lds.VT[dd][g * 8]) = stg[KIT + it];
+    }
+  };
+  issue_stage(0);
+  for (int st2 = 0; st2 < nstages; ++st2) {
     __syncthreads();
+    write_stage();
+    __syncthreads();
+    if (st2 + 1 < nstages) issue_stage(st2 + 1);
+    const int kvs = st2 * 128;
 
     for (int half = 0; half < 2; ++half) {
     const int kv0 = kvs + half * KVB;
@@ -262,7 +288,6 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       }
     }
     }  // half
-    __syncthreads();
   }
 
   // ---- epilogue: normalize rows, store O (bf16, coalesced over d) + lse
