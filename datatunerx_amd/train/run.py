@@ -79,6 +79,14 @@ def main(argv=None):
             model.load_state_dict(sd, strict=False)
         else:
             model.init_random(seed=fargs.seed)
+        if margs.quantization in ("int4", "int8"):
+            from ..models.quant import quantize_model_
+            n_q = quantize_model_(model,
+                                  bits=4 if margs.quantization == "int4"
+                                  else 8)
+            if rank == 0:
+                print(f"quantized {n_q} frozen base layers to "
+                      f"{margs.quantization}")
 
         tok = ByteTokenizer()
         vocab = getattr(model.cfg, "vocab_size")
