@@ -418,10 +418,15 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
   const int kv_hi = causal ? min(Skv - 1, q_hi_blk + diag) : (Skv - 1);
   const int nstages = kv_hi / 128 + 1;     // 128 kv rows per stage
 
-  for (int st2 = 0; st2 < nstages; ++st2) {
+  // T14 async-stage split (as in fwd/dkdv)
+  constexpr int KIT = (128 * (D / 8)) / 512;
+  constexpr int TIT = (D * (128 / 8)) / 512;
+  short8v stg[KIT * 2 + TIT];
+  auto issue_stage = [&](int st2) {
     const int kvs = st2 * 128;
-    // ---- stage 128 K rows, V rows, KT columns
-    for (int idx = threadIdx.x; idx < 128 * (D / 8); idx += 512) {
+#pragma unroll
+    for (int it = 0; it < KIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
       const int row = idx / (D / 8), g = idx % (D / 8);
       short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
       short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -430,10 +435,12 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
         k8 = *reinterpret_cast<const short8v*>(Kp + off);
         v8 = *reinterpret_cast<const short8v*>(Vp + off);
       }
-      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = k8;
-      *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = v8;
+      stg[it * 2] = k8;
+      stg[it * 2 + 1] = v8;
     }
-    for (int idx = threadIdx.x; idx < D * (128 / 8); idx += 512) {
+#pragma unroll
+    for (int it = 0; it < TIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
       const int dd = idx / (128 / 8), g = idx % (128 / 8);
       short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
       const int s0 = kvs + g * 8;
@@ -446,9 +453,31 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
           if (s0 + j < Skv)
             k8[j] = (short)KTg[ktbase + (long)dd * Skv + s0 + j];
       }
-      *reinterpret_cast<short8v*>(&lds.KT[dd][g * 8]) = k8;
+      stg[KIT * 2 + it] = k8;
     }
+  };
+  auto write_stage = [&]() {
+#pragma unroll
+    for (int it = 0; it < KIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
+      const int row = idx / (D / 8), g = idx % (D / 8);
+      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it * 2];
+      *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = stg[it * 2 + 1];
+    }
+#pragma unroll
+    for (int it = 0; it < TIT; ++it) {
+      const int idx = threadIdx.x + it * 512;
+      const int dd = idx / (128 / 8), g = idx % (128 / 8);
+      *reinterpret_cast<short8v*>(&lds.KT[dd][g * 8]) = stg[KIT * 2 + it];
+    }
+  };
+  issue_stage(0);
+  for (int st2 = 0; st2 < nstages; ++st2) {
     __syncthreads();
+    write_stage();
+    __syncthreads();
+    if (st2 + 1 < nstages) issue_stage(st2 + 1);
+    const int kvs = st2 * 128;
 
     for (int kh = 0; kh < 2; ++kh) {
     const int kv0 = kvs + kh * KVB;
@@ -503,7 +532,6 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
       }
     }
     }  // kh
-    __syncthreads();
   }
 
   // ---- epilogue: dQ C-layout [q regs][d lanes] -> BSHD stores
