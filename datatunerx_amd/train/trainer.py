@@ -417,6 +417,51 @@ class SFTTrainer:
         self._log(metrics, kind="eval")
         return metrics
 
+    # ------------------------------------------------------------ predict
+    @torch.no_grad()
+    def predict(self, dataset, max_new_tokens: int = 64,
+                out_file: Optional[str] = None):
+        """Greedy generation from each example's prompt (the -100-masked
+        prefix), mirroring the reference's prediction_step +
+        save_predictions jsonl (cmd/tuning/trainer.py:405-489)."""
+        from ..data.dataset import IGNORE_INDEX
+        self.model.eval()
+        results = []
+        for i in range(len(dataset)):
+            ex = dataset[i]
+            ids, labels = list(ex["input_ids"]), list(ex["labels"])
+            plen = 0
+            for lb in labels:
+                if lb == IGNORE_INDEX:
+                    plen += 1
+                else:
+                    break
+            prompt = ids[:plen] if plen > 0 else ids
+            cur = torch.tensor([prompt], dtype=torch.long,
+                               device=self.device)
+            out = []
+            for _ in range(max_new_tokens):
+                logits = self.model(cur)
+                nxt = int(logits[0, -1].argmax())
+                out.append(nxt)
+                cur = torch.cat(
+                    [cur, torch.tensor([[nxt]], dtype=torch.long,
+                                       device=self.device)], dim=1)
+            results.append({
+                "prompt_ids": prompt,
+                "predict_ids": out,
+                "label_ids": [lb for lb in labels if lb != IGNORE_INDEX],
+            })
+        if out_file is None:
+            out_file = os.path.join(self.cfg.output_dir,
+                                    "generated_predictions.jsonl")
+        if is_main():
+            os.makedirs(os.path.dirname(out_file) or ".", exist_ok=True)
+            with open(out_file, "w") as f:
+                for r in results:
+                    f.write(json.dumps(r) + "\n")
+        return results
+
     # -------------------------------------------------------- checkpoint
     def save_checkpoint(self, out_dir: str):
         """HF-adapter layout for LoRA models; full trainable state dict

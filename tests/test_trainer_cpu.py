@@ -195,3 +195,17 @@ def test_quantized_base_training(tmp_path):
     with torch.no_grad():
         l4 = m4(ids)
     assert torch.isfinite(l4).all()
+
+
+def test_predict_writes_jsonl(tmp_path):
+    """predict() greedy-generates from the -100-masked prompt prefix and
+    writes generated_predictions.jsonl (trainer.py:405-489 parity)."""
+    tr = _tiny_trainer(tmp_path, steps=1)
+    tr.train()
+    ds = SFTDataset.synthetic(3, 24, 512, seed=1)
+    res = tr.predict(ds, max_new_tokens=4)
+    assert len(res) == 3
+    assert all(len(r["predict_ids"]) == 4 for r in res)
+    out = os.path.join(str(tmp_path), "generated_predictions.jsonl")
+    rows = [json.loads(l) for l in open(out)]
+    assert len(rows) == 3 and "prompt_ids" in rows[0]
