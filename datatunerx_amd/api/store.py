@@ -40,6 +40,11 @@ class Store:
 
     # ------------------------------------------------------------- CRUD
     def create(self, obj: ApiObject) -> ApiObject:
+        # admission: defaulting then validating webhooks (native parity
+        # with controller_manager.go:112-135)
+        from .validation import default_, validate_
+        default_(obj)
+        validate_(obj)
         with self._lock:
             p = self._path(obj.kind, obj.namespace, obj.name)
             if os.path.exists(p):

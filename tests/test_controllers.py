@@ -89,7 +89,7 @@ def test_store_crud_and_gc(tmp_path):
     store.create(llm)
     got = store.get(LLM, "default", "m1")
     assert got.spec == {"a": 1}
-    child = Finetune(name="c1", spec={})
+    child = Finetune(name="c1", spec=finetune_spec())
     child.set_owner(llm)
     store.create(child)
     store.delete(LLM, "default", "m1")
@@ -263,3 +263,31 @@ def test_cli_apply_and_get(tmp_path, capsys):
     cli(["--state-dir", str(tmp_path / "s"), "get", "llm"])
     out = capsys.readouterr().out
     assert "m2" in out
+
+
+def test_admission_validation(tmp_path):
+    """Webhook parity: invalid objects rejected at create with every
+    problem listed; defaults filled (controller_manager.go:112-135)."""
+    from datatunerx_amd.api.store import Store
+    from datatunerx_amd.api.validation import ValidationError
+    store = Store(str(tmp_path / "s"))
+    with pytest.raises(ValidationError, match="llm is required"):
+        store.create(FinetuneJob(name="bad", spec={
+            "fineTune": {"finetuneSpec": {"dataset": "d",
+                                          "hyperparameter":
+                                          {"hyperparameterRef": "h"}}}}))
+    with pytest.raises(ValidationError, match="DNS-1123"):
+        store.create(LLM(name="Bad_Name", spec={}))
+    with pytest.raises(ValidationError, match="mutually exclusive"):
+        store.create(Hyperparameter(name="hpx", spec={
+            "parameters": {"int4": True, "int8": True}}))
+    with pytest.raises(ValidationError, match="non-empty"):
+        store.create(FinetuneExperiment(name="e", spec={
+            "finetuneJobs": []}))
+    # defaults: node filled, scoring config defaulted
+    job = FinetuneJob(name="ok", spec={
+        "fineTune": {"finetuneSpec": finetune_spec()}})
+    store.create(job)
+    got = store.get(FinetuneJob, "default", "ok")
+    assert got.spec["scoringPluginConfig"]["name"] == "builtin"
+    assert got.spec["fineTune"]["finetuneSpec"]["node"] == 1
