@@ -18,3 +18,17 @@ payload) as an MI355X-first stack:
 """
 
 __version__ = "0.1.0"
+
+# hipBLASLt/rocBLAS algorithm selections pre-tuned on MI355X (PyTorch
+# TunableOp, profiles/tunableop_gfx950.csv — ~2% end-to-end on the 7B
+# LoRA step). Read-only; explicit PYTORCH_TUNABLEOP_* env wins.
+import os as _os
+
+_tune = _os.path.join(_os.path.dirname(_os.path.dirname(
+    _os.path.abspath(__file__))), "profiles", "tunableop_gfx950.csv")
+if (_os.path.exists(_tune)
+        and "PYTORCH_TUNABLEOP_ENABLED" not in _os.environ):
+    _os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    _os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    _os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _tune
+del _os

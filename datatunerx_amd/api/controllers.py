@@ -447,7 +447,8 @@ class FinetuneJobController:
             plugin = job.spec.get("scoringPluginConfig") or {}
             sc = Scoring(name=scname, namespace=job.namespace, spec={
                 "inferenceService": endpoint,
-                "plugin": {"loadPlugin": bool(plugin.get("name")),
+                "plugin": {"loadPlugin": bool(plugin.get("name")) and
+                           plugin.get("name") != "builtin",
                            "name": plugin.get("name"),
                            "parameters": plugin.get("parameters")},
             })
