@@ -39,6 +39,8 @@ class Manager:
         ]
         self._not_before: Dict[str, float] = {}
         self.stop_event = threading.Event()
+        self.counters: Dict[str, int] = {"reconcile_total": 0,
+                                         "reconcile_errors_total": 0}
 
     def reconcile_once(self) -> int:
         """One pass over every object; returns number reconciled."""
@@ -57,7 +59,9 @@ class Manager:
                     import traceback
                     traceback.print_exc()
                     requeue = REQUEUE_ERROR
+                    self.counters["reconcile_errors_total"] += 1
                 n += 1
+                self.counters["reconcile_total"] += 1
                 if requeue:
                     self._not_before[key] = now + requeue
                 else:
@@ -69,6 +73,50 @@ class Manager:
         while not self.stop_event.is_set():
             self.reconcile_once()
             self.stop_event.wait(poll_interval)
+
+    def serve_metrics(self, host: str = "127.0.0.1", port: int = 8080):
+        """Prometheus text endpoint (controller-runtime's :8080 metrics
+        parity — options.go:12) + /healthz, /readyz. Runs in a daemon
+        thread; returns the server."""
+        from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+        mgr = self
+
+        class H(BaseHTTPRequestHandler):
+            def log_message(self, *a):
+                pass
+
+            def do_GET(self):
+                if self.path in ("/healthz", "/readyz"):
+                    body = b"ok"
+                elif self.path == "/metrics":
+                    lines = []
+                    for k, v in mgr.counters.items():
+                        lines.append(f"# TYPE dtx_{k} counter")
+                        lines.append(f"dtx_{k} {v}")
+                    for cls, _ in mgr.controllers:
+                        objs = mgr.store.list(cls)
+                        states: Dict[str, int] = {}
+                        for o in objs:
+                            st = o.status.get("state", "") or "none"
+                            states[st] = states.get(st, 0) + 1
+                        for st, c in states.items():
+                            lines.append(
+                                f'dtx_objects{{kind="{cls.kind}",'
+                                f'state="{st}"}} {c}')
+                    body = ("\n".join(lines) + "\n").encode()
+                else:
+                    self.send_response(404)
+                    self.end_headers()
+                    return
+                self.send_response(200)
+                self.send_header("Content-Length", str(len(body)))
+                self.end_headers()
+                self.wfile.write(body)
+
+        srv = ThreadingHTTPServer((host, port), H)
+        t = threading.Thread(target=srv.serve_forever, daemon=True)
+        t.start()
+        return srv
 
     def run_until_settled(self, timeout: float = 300.0,
                           poll_interval: float = 0.2) -> bool:
@@ -101,11 +149,16 @@ def main(argv=None):
     ap.add_argument("--storage-path", default="")
     ap.add_argument("--metrics-export-address", default="")
     ap.add_argument("--poll-interval", type=float, default=1.0)
+    ap.add_argument("--metrics-port", type=int, default=8080,
+                    help="0 disables the /metrics endpoint")
     args = ap.parse_args(argv)
     cfg = ManagerConfig(state_dir=args.state_dir, work_dir=args.work_dir,
                         n_gpus=args.n_gpus, storage_path=args.storage_path,
                         metrics_export_address=args.metrics_export_address)
-    Manager(cfg).run(args.poll_interval)
+    mgr = Manager(cfg)
+    if args.metrics_port:
+        mgr.serve_metrics(port=args.metrics_port)
+    mgr.run(args.poll_interval)
 
 
 if __name__ == "__main__":

@@ -291,3 +291,21 @@ def test_admission_validation(tmp_path):
     got = store.get(FinetuneJob, "default", "ok")
     assert got.spec["scoringPluginConfig"]["name"] == "builtin"
     assert got.spec["fineTune"]["finetuneSpec"]["node"] == 1
+
+
+def test_manager_metrics_endpoint(tmp_path):
+    """/metrics + /healthz (controller-runtime :8080 parity)."""
+    import urllib.request
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store)
+    mgr.reconcile_once()
+    srv = mgr.serve_metrics(port=_PORT[0] + 49)
+    try:
+        url = f"http://127.0.0.1:{_PORT[0] + 49}"
+        assert urllib.request.urlopen(url + "/healthz",
+                                      timeout=10).read() == b"ok"
+        text = urllib.request.urlopen(url + "/metrics",
+                                      timeout=10).read().decode()
+        assert "dtx_reconcile_total" in text
+    finally:
+        srv.shutdown()
