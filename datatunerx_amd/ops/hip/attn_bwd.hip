@@ -120,6 +120,9 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
   constexpr int DC16 = D / 16;
   constexpr int ND32 = D / 32;
   __shared__ DkdvLds<D> lds;
+  // separate object: float arrays INSIDE DkdvLds made hipcc scalarize
+  // every b128 fragment read of the struct (see profiles/README.md)
+  __shared__ float lsed[64 + 64];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -249,6 +252,11 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
       const int q0s = qs * 64;
       __syncthreads();
       write_stage();
+      if (threadIdx.x < 64) {
+        const int qg = q0s + threadIdx.x;
+        lsed[threadIdx.x] = qg < S ? lse_in[lbase + qg] : 0.f;
+        lsed[64 + threadIdx.x] = qg < S ? delta_in[lbase + qg] : 0.f;
+      }
       __syncthreads();
       if (qs + 1 < qs_hi) issue_stage(qs + 1);
 
@@ -284,11 +292,8 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qg = q0 + bw_crow(r, hi);
-        const bool qv = qg < S;
-        // L1/L2-hot scalar loads (the lse/dlt LDS arrays used to make
-        // hipcc scalarize every b128 fragment read in this kernel)
-        const float lse2 = (qv ? lse_in[lbase + qg] : 0.f) * LOG2E;
-        const float dlt = qv ? delta_in[lbase + qg] : 0.f;
+        const float lse2 = lsed[qoff + bw_crow(r, hi)] * LOG2E;
+        const float dlt = lsed[64 + qoff + bw_crow(r, hi)];
         float p;
         if (interior) {
           p = __builtin_exp2f(sv[r] * kscale - lse2);
