@@ -534,6 +534,9 @@ class FinetuneExperimentController:
                 job = FinetuneJob(name=js["name"], namespace=exp.namespace,
                                   spec=js.get("spec", {}))
                 job.set_owner(exp)
+                from .labels import generate_instance_label
+                job.metadata.labels.update(
+                    generate_instance_label(exp.name))
                 self.store.create(job)
         # aggregate (:154-197)
         jobs_status = []
