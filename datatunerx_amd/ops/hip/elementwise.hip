@@ -131,18 +131,25 @@ void reduce_partials_kernel(const float* __restrict__ part,
   long i = (long)blockIdx.x * DTX_BLOCK + threadIdx.x;
   const long stride = (long)gridDim.x * DTX_BLOCK;
   for (; i < nvec; i += stride) {
-    f4 s = {0.f, 0.f, 0.f, 0.f};
+    f4 s0 = {0.f, 0.f, 0.f, 0.f};
+    f4 s1 = s0, s2 = s0, s3 = s0;
+    f4 s4 = s0, s5 = s0, s6 = s0, s7 = s0;
     int p = 0;
-    for (; p + 4 <= P; p += 4) {
-      f4 a = *reinterpret_cast<const f4*>(&part[(long)p * L + i * 4]);
-      f4 b = *reinterpret_cast<const f4*>(&part[(long)(p + 1) * L + i * 4]);
-      f4 c = *reinterpret_cast<const f4*>(&part[(long)(p + 2) * L + i * 4]);
-      f4 d = *reinterpret_cast<const f4*>(&part[(long)(p + 3) * L + i * 4]);
-      s += (a + b) + (c + d);
+    for (; p + 8 <= P; p += 8) {       // 8 planes in flight: the
+      const long b = i * 4;            // plane-major layout makes each
+      s0 += *reinterpret_cast<const f4*>(&part[(long)p * L + b]);
+      s1 += *reinterpret_cast<const f4*>(&part[(long)(p + 1) * L + b]);
+      s2 += *reinterpret_cast<const f4*>(&part[(long)(p + 2) * L + b]);
+      s3 += *reinterpret_cast<const f4*>(&part[(long)(p + 3) * L + b]);
+      s4 += *reinterpret_cast<const f4*>(&part[(long)(p + 4) * L + b]);
+      s5 += *reinterpret_cast<const f4*>(&part[(long)(p + 5) * L + b]);
+      s6 += *reinterpret_cast<const f4*>(&part[(long)(p + 6) * L + b]);
+      s7 += *reinterpret_cast<const f4*>(&part[(long)(p + 7) * L + b]);
     }
     for (; p < P; ++p)
-      s += *reinterpret_cast<const f4*>(&part[(long)p * L + i * 4]);
-    *reinterpret_cast<f4*>(&out[i * 4]) = s;
+      s0 += *reinterpret_cast<const f4*>(&part[(long)p * L + i * 4]);
+    *reinterpret_cast<f4*>(&out[i * 4]) =
+        ((s0 + s1) + (s2 + s3)) + ((s4 + s5) + (s6 + s7));
   }
   // ragged tail (L % 4)
   const long tail0 = nvec * 4;
