@@ -146,6 +146,13 @@ void reduce_partials_kernel(const float* __restrict__ part,
       s6 += *reinterpret_cast<const f4*>(&part[(long)(p + 6) * L + b]);
       s7 += *reinterpret_cast<const f4*>(&part[(long)(p + 7) * L + b]);
     }
+    for (; p + 4 <= P; p += 4) {
+      const long b = i * 4;
+      s0 += *reinterpret_cast<const f4*>(&part[(long)p * L + b]);
+      s1 += *reinterpret_cast<const f4*>(&part[(long)(p + 1) * L + b]);
+      s2 += *reinterpret_cast<const f4*>(&part[(long)(p + 2) * L + b]);
+      s3 += *reinterpret_cast<const f4*>(&part[(long)(p + 3) * L + b]);
+    }
     for (; p < P; ++p)
       s0 += *reinterpret_cast<const f4*>(&part[(long)p * L + i * 4]);
     *reinterpret_cast<f4*>(&out[i * 4]) =
