@@ -110,6 +110,10 @@ def attn_fwd(q, k, v, causal: bool = True, scale: float | None = None):
     if scale is None:
         scale = 1.0 / (q.shape[-1] ** 0.5)
     if _gpu(q):
+        if q.shape[1] == 1:
+            # serving decode: flash-decoding split-KV path (with S=1
+            # every cached position is visible, causal or not)
+            return _EXT.attn_decode(q, k, v, scale)
         vt = _EXT.transpose_sd(v)        # [B,Hkv,D,Skv] for the PV tiles
         return _EXT.attn_fwd(q, k, vt, causal, scale)
     return ref.attn_fwd(q, k, v, causal, scale)

@@ -379,3 +379,173 @@ void launch_attn_fwd(const void* q, const void* k, const void* vt, void* o,
         B, Hq, Hkv, S, Skv, scale, causal);
   }
 }
+
+// ------------------------------------------------------- decode (S=1)
+// Flash-decoding for the serving engine: one token's attention against
+// a long KV cache. Split-KV partials (m, l, o[D]) per 512-row chunk,
+// combined by a second small kernel. The training kernel would leave
+// 255/256 of each block idle at S=1; this one streams K/V at the HBM
+// rate with every lane busy.
+//   grid (NS, B*Hq), block 256 = 4 waves; 16 lanes per kv row (8 els
+//   each), 4 row slots per wave.
+template <int D>
+__global__ __launch_bounds__(DTX_BLOCK)
+void attn_decode_kernel(const unsigned short* __restrict__ Q,
+                        const unsigned short* __restrict__ Kp,
+                        const unsigned short* __restrict__ Vp,
+                        float* __restrict__ part,   // [B*Hq, NS, D+2]
+                        int B, int Hq, int Hkv, int Skv, int chunk,
+                        float scale) {
+  constexpr int LPR = D / 8;                   // lanes per row
+  constexpr int SLOTS = 64 / LPR;              // row slots per wave
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int slot = lane / LPR;
+  const int g = lane % LPR;
+
+  const int bh = blockIdx.y;
+  const int b = bh / Hq, hq = bh % Hq;
+  const int hkv = hq / (Hq / Hkv);
+  const int krowstr = Hkv * D;
+  const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
+  const int kv0 = blockIdx.x * chunk;
+  const int kv_end = min(Skv, kv0 + chunk);
+
+  // q slice for this lane (f32)
+  float qv[8];
+  {
+    const long qoff = ((long)b * Hq + hq) * D + g * 8;
+    load_bf16x8(Q + qoff, qv);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) qv[i] *= scale;
+  }
+
+  float m_run = NEG_INF, l_run = 0.f;
+  float o_acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) o_acc[i] = 0.f;
+
+  const int rows_per_iter = 4 * SLOTS;         // per block
+  for (int r = kv0 + wid * SLOTS + slot; r < kv_end; r += rows_per_iter) {
+    const long off = kbase + (long)r * krowstr + g * 8;
+    float kv8[8];
+    load_bf16x8(Kp + off, kv8);
+    float dot = 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) dot += qv[i] * kv8[i];
+#pragma unroll
+    for (int o = LPR / 2; o > 0; o >>= 1)
+      dot += __shfl_xor(dot, o, 64);           // reduce within the row
+    const float m_new = fmaxf(m_run, dot);
+    const float alpha = __builtin_exp2f((m_run - m_new) * LOG2E);
+    const float pr = __builtin_exp2f((dot - m_new) * LOG2E);
+    l_run = l_run * alpha + pr;
+    m_run = m_new;
+    float vv[8];
+    load_bf16x8(Vp + off, vv);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) o_acc[i] = o_acc[i] * alpha + pr * vv[i];
+  }
+
+  // combine the SLOTS row-streams of this wave, then the 4 waves (LDS)
+  __shared__ float red[4 * SLOTS][D + 2];
+  const int my = wid * SLOTS + slot;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) red[my][g * 8 + i] = o_acc[i];
+  if (g == 0) {
+    red[my][D] = m_run;
+    red[my][D + 1] = l_run;
+  }
+  __syncthreads();
+  if (threadIdx.x < 64) {                      // one wave combines
+    const int gg = threadIdx.x % LPR;
+    float m_all = NEG_INF;
+    for (int s = 0; s < 4 * SLOTS; ++s) m_all = fmaxf(m_all, red[s][D]);
+    float l_all = 0.f;
+    float oc[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) oc[i] = 0.f;
+    for (int s = 0; s < 4 * SLOTS; ++s) {
+      const float w = (red[s][D] == NEG_INF)
+          ? 0.f : __builtin_exp2f((red[s][D] - m_all) * LOG2E);
+      l_all += w * red[s][D + 1];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) oc[i] += w * red[s][gg * 8 + i];
+    }
+    if (threadIdx.x < LPR) {                   // lanes 0..LPR-1 write
+      float* pb = part + ((long)bh * gridDim.x + blockIdx.x) * (D + 2);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) pb[gg * 8 + i] = oc[i];
+      if (gg == 0) {
+        pb[D] = m_all;
+        pb[D + 1] = l_all;
+      }
+    }
+  }
+}
+
+// combine NS chunk partials -> o (bf16 BSHD S=1) + lse
+template <int D>
+__global__ __launch_bounds__(64)
+void attn_decode_combine_kernel(const float* __restrict__ part,
+                                unsigned short* __restrict__ O,
+                                float* __restrict__ lse_out,
+                                int B, int Hq, int NS) {
+  constexpr int LPR = D / 8;
+  const int bh = blockIdx.x;
+  const int lane = threadIdx.x;
+  const float* pb = part + (long)bh * NS * (D + 2);
+  float m_all = NEG_INF;
+  for (int s = 0; s < NS; ++s) m_all = fmaxf(m_all, pb[s * (D + 2) + D]);
+  float l_all = 0.f;
+  float oc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) oc[i] = 0.f;
+  const int g = lane % LPR;
+  for (int s = 0; s < NS; ++s) {
+    const float ms = pb[s * (D + 2) + D];
+    const float w = (ms == NEG_INF)
+        ? 0.f : __builtin_exp2f((ms - m_all) * LOG2E);
+    l_all += w * pb[s * (D + 2) + D + 1];
+    if (lane < LPR) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        oc[i] += w * pb[s * (D + 2) + g * 8 + i];
+    }
+  }
+  if (lane < LPR) {
+    const float rcp = l_all > 0.f ? 1.f / l_all : 0.f;
+    unsigned short* orow = O + (long)bh * D;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) orow[g * 8 + i] = f2bf(oc[i] * rcp);
+    if (g == 0)
+      lse_out[bh] = m_all + __logf(fmaxf(l_all, 1e-30f));
+  }
+}
+
+int attn_decode_nsplit(int Skv) {
+  int ns = DTX_CDIV(Skv, 512);
+  return ns < 1 ? 1 : ns;
+}
+
+void launch_attn_decode(const void* q, const void* k, const void* v,
+                        float* part, void* o, float* lse, int B, int Hq,
+                        int Hkv, int Skv, int D, float scale,
+                        hipStream_t st) {
+  const int ns = attn_decode_nsplit(Skv);
+  const int chunk = DTX_CDIV(Skv, ns);
+  dim3 grid(ns, B * Hq);
+  if (D == 128) {
+    attn_decode_kernel<128><<<grid, DTX_BLOCK, 0, st>>>(
+        (const unsigned short*)q, (const unsigned short*)k,
+        (const unsigned short*)v, part, B, Hq, Hkv, Skv, chunk, scale);
+    attn_decode_combine_kernel<128><<<B * Hq, 64, 0, st>>>(
+        part, (unsigned short*)o, lse, B, Hq, ns);
+  } else if (D == 64) {
+    attn_decode_kernel<64><<<grid, DTX_BLOCK, 0, st>>>(
+        (const unsigned short*)q, (const unsigned short*)k,
+        (const unsigned short*)v, part, B, Hq, Hkv, Skv, chunk, scale);
+    attn_decode_combine_kernel<64><<<B * Hq, 64, 0, st>>>(
+        part, (unsigned short*)o, lse, B, Hq, ns);
+  }
+}

@@ -34,6 +34,10 @@ void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, int, float, int, hipStream_t);
 void launch_transpose_sd(const void*, void*, int, int, int, int,
                          hipStream_t);
+int attn_decode_nsplit(int Skv);
+void launch_attn_decode(const void*, const void*, const void*, float*,
+                        void*, float*, int, int, int, int, int, float,
+                        hipStream_t);
 void launch_attn_delta(const void*, const void*, float*, long, int, int,
                        int, hipStream_t);
 void launch_attn_bwd_dkdv(const void*, const void*, const void*,
@@ -282,6 +286,29 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   return {o, lse};
 }
 
+// Decode fast path: q [B,1,Hq,D] against the KV cache (flash-decoding
+// split-KV partials; no V transpose needed).
+std::vector<torch::Tensor> attn_decode(torch::Tensor q, torch::Tensor k,
+                                       torch::Tensor v, double scale) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  const int B = (int)q.size(0), Hq = (int)q.size(2), D = (int)q.size(3);
+  const int Skv = (int)k.size(1), Hkv = (int)k.size(2);
+  TORCH_CHECK(q.size(1) == 1, "decode path wants S == 1");
+  TORCH_CHECK(D == 64 || D == 128, "D must be 64 or 128");
+  const int ns = attn_decode_nsplit(Skv);
+  auto part = torch::empty({(long)B * Hq, ns, D + 2},
+                           q.options().dtype(torch::kFloat));
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, Hq, 1}, q.options().dtype(torch::kFloat));
+  launch_attn_decode(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                     part.data_ptr<float>(), o.data_ptr(),
+                     lse.data_ptr<float>(), B, Hq, Hkv, Skv, D,
+                     (float)scale, cur_stream());
+  return {o, lse};
+}
+
 // All BSHD; v untransposed. Pre-transposes Q/K/dO internally.
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
@@ -342,4 +369,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("transpose_sd", &transpose_sd);
+  m.def("attn_decode", &attn_decode);
 }
