@@ -70,7 +70,11 @@ class Attention(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        q = q.contiguous()
+        if not (q.is_cuda and q.shape[1] == 1):
+            # decode (S=1) accepts dense KV-cache prefix views; every
+            # other path wants contiguous k/v
+            k, v = k.contiguous(), v.contiguous()
         o, lse = attn_fwd(q, k, v, causal, scale)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal, ctx.scale = causal, scale

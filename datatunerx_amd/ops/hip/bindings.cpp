@@ -291,10 +291,20 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
 std::vector<torch::Tensor> attn_decode(torch::Tensor q, torch::Tensor k,
                                        torch::Tensor v, double scale) {
   check_bf16_contig(q, "q");
-  check_bf16_contig(k, "k");
-  check_bf16_contig(v, "v");
   const int B = (int)q.size(0), Hq = (int)q.size(2), D = (int)q.size(3);
   const int Skv = (int)k.size(1), Hkv = (int)k.size(2);
+  // accept KV-cache VIEWS (prefix of a preallocated cache): inner
+  // strides must be dense; the batch stride only matters for B > 1
+  auto dense_kv = [&](const torch::Tensor& t, const char* n) {
+    TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16,
+                n, " must be bf16 GPU");
+    TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == D &&
+                t.stride(1) == (long)Hkv * D,
+                n, " must be s/h/d-dense");
+    if (B > 1) TORCH_CHECK(t.is_contiguous(), n, " batch>1 needs contig");
+  };
+  dense_kv(k, "k");
+  dense_kv(v, "v");
   TORCH_CHECK(q.size(1) == 1, "decode path wants S == 1");
   TORCH_CHECK(D == 64 || D == 128, "D must be 64 or 128");
   const int ns = attn_decode_nsplit(Skv);

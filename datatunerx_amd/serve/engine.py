@@ -36,8 +36,12 @@ class KVCache:
         self.k[:, self.cur:self.cur + S] = k
         self.v[:, self.cur:self.cur + S] = v
         self.cur += S
-        return (self.k[:, :self.cur].contiguous(),
-                self.v[:, :self.cur].contiguous())
+        kk, vv = self.k[:, :self.cur], self.v[:, :self.cur]
+        if self.k.shape[0] > 1 or not k.is_cuda:
+            # batch>1 / CPU paths want contiguous; the GPU decode kernel
+            # takes the dense prefix view directly (no per-token copy)
+            kk, vv = kk.contiguous(), vv.contiguous()
+        return kk, vv
 
 
 def build_model(name: str, device, adapter_dir: Optional[str] = None,
