@@ -227,6 +227,18 @@ class FinetuneController:
                 if _alive(pid):
                     self.sup.terminate(pid)
             self.inv.release_owner(f"ft/{ft.namespace}/{ft.name}")
+            # bounded restart (an improvement over the reference, which
+            # only propagates failure — SURVEY.md §5 failure detection):
+            # spec.restartPolicy.maxRetries relaunches from scratch
+            retries = ft.status.get("restarts", 0)
+            max_r = int(((ft.spec.get("restartPolicy") or {})
+                         .get("maxRetries") or 0))
+            if retries < max_r:
+                ft.status["restarts"] = retries + 1
+                ft.status["trainJobInfo"] = None
+                ft.status["state"] = "Pending"
+                self.store.update(ft)
+                return REQUEUE_RECALIBRATE
             ft.status["state"] = "Failed"
             self.store.update(ft)
             return None

@@ -309,3 +309,27 @@ def test_manager_metrics_endpoint(tmp_path):
         assert "dtx_reconcile_total" in text
     finally:
         srv.shutdown()
+
+
+def test_finetune_restart_policy(tmp_path):
+    """spec.restartPolicy.maxRetries relaunches a crashed trainer before
+    declaring Failed (bounded elasticity; improvement over the
+    reference's propagate-only failure handling)."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store)
+    spec = finetune_spec()
+    spec["llm"] = "nonexistent-model"         # trainer exits nonzero
+    spec["restartPolicy"] = {"maxRetries": 2}
+    ft = Finetune(name="ftretry", spec=spec)
+    mgr.store.create(ft)
+    deadline = time.time() + 180
+    while time.time() < deadline:
+        mgr._not_before.clear()
+        mgr.reconcile_once()
+        cur = mgr.store.get(Finetune, "default", "ftretry")
+        if cur.status.get("state") == "Failed":
+            break
+        time.sleep(0.2)
+    cur = mgr.store.get(Finetune, "default", "ftretry")
+    assert cur.status.get("state") == "Failed"
+    assert cur.status.get("restarts") == 2
