@@ -46,6 +46,15 @@ def main(argv=None):
     p_mgr.add_argument("--once", action="store_true",
                        help="reconcile until settled, then exit")
 
+    p_run = sub.add_parser(
+        "run", help="apply manifest(s), reconcile to completion, print "
+                    "final statuses (dtx-ctl-style one-shot)")
+    p_run.add_argument("-f", "--filename", required=True, action="append")
+    p_run.add_argument("--work-dir", default="./dtx-work")
+    p_run.add_argument("--n-gpus", type=int, default=8)
+    p_run.add_argument("--storage-path", default="")
+    p_run.add_argument("--timeout", type=float, default=1800.0)
+
     args = ap.parse_args(argv)
     store = Store(args.state_dir)
 
@@ -85,6 +94,29 @@ def main(argv=None):
             ok = mgr.run_until_settled()
             sys.exit(0 if ok else 1)
         mgr.run()
+    elif args.cmd == "run":
+        from .api.controllers import ManagerConfig
+        from .api.manager import Manager
+        from .api.types import (FinetuneExperiment, FinetuneJob, Finetune)
+        for fn in args.filename:
+            text = sys.stdin.read() if fn == "-" else open(fn).read()
+            for obj in store.apply_manifest(text):
+                print(f"{obj.kind.lower()}/{obj.name} applied")
+        cfg = ManagerConfig(state_dir=args.state_dir,
+                            work_dir=args.work_dir, n_gpus=args.n_gpus,
+                            storage_path=args.storage_path)
+        mgr = Manager(cfg)
+        ok = mgr.run_until_settled(timeout=args.timeout)
+        for cls in (FinetuneExperiment, FinetuneJob, Finetune):
+            for o in store.list(cls):
+                extra = ""
+                if cls is FinetuneExperiment and o.status.get("bestVersion"):
+                    bv = o.status["bestVersion"]
+                    extra = (f"  bestVersion: score={bv.get('score')} "
+                             f"image={bv.get('image')}")
+                print(f"{o.kind.lower()}/{o.name}: "
+                      f"{o.status.get('state', '')}{extra}")
+        sys.exit(0 if ok else 1)
 
 
 def _resolve_kind(k: str) -> str:

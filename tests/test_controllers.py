@@ -333,3 +333,27 @@ def test_finetune_restart_policy(tmp_path):
     cur = mgr.store.get(Finetune, "default", "ftretry")
     assert cur.status.get("state") == "Failed"
     assert cur.status.get("restarts") == 2
+
+
+def test_cli_run_one_shot(tmp_path, capsys):
+    """`dtx run -f ...` applies manifests, reconciles to completion and
+    prints final states (dtx-ctl-style one-shot)."""
+    from datatunerx_amd.cli import main as cli
+    _PORT[0] += 50
+    st = str(tmp_path / "s")
+    store = Store(st)
+    seed_resources(store)
+    man = {
+        "apiVersion": "finetune.datatunerx.io/v1beta1",
+        "kind": "FinetuneJob",
+        "metadata": {"name": "clijob", "namespace": "default"},
+        "spec": {"fineTune": {"finetuneSpec": finetune_spec()}}}
+    f = tmp_path / "job.yaml"
+    f.write_text(yaml.safe_dump(man))
+    import datatunerx_amd.api.controllers as C
+    with pytest.raises(SystemExit) as e:
+        cli(["--state-dir", st, "run", "-f", str(f),
+             "--work-dir", str(tmp_path / "w"), "--timeout", "300"])
+    assert e.value.code == 0
+    out = capsys.readouterr().out
+    assert "finetunejob/clijob: Successful" in out
