@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import os
 import threading
-from typing import Dict, List, Optional, Type
+from typing import List, Optional
 
 import yaml
 

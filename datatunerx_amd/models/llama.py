@@ -10,7 +10,7 @@ for the loss. No HF dependency in the training hot path.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 import torch.nn as nn

@@ -16,7 +16,7 @@ from typing import List, Optional
 
 import torch
 
-from ..data.dataset import IGNORE_INDEX, ByteTokenizer
+from ..data.dataset import ByteTokenizer
 from ..data.templates import get_template
 from ..models import (GPT2Config, GPT2ForCausalLM, LlamaConfig,
                       LlamaForCausalLM, load_adapter)

@@ -21,13 +21,13 @@ import json
 import math
 import os
 import time
-from dataclasses import asdict, dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import torch
 
 from .. import ops
-from ..data.dataset import ShardedLoader, collate
+from ..data.dataset import ShardedLoader
 from ..models.lora import LoRALinearModule, save_adapter
 from ..parallel.ddp import GradSynchronizer, is_main, sync_scalar_mean
 
