@@ -24,7 +24,7 @@
 // diagonal skip compute.
 #include "dtx_common.h"
 
-typedef __attribute__((ext_vector_type(4))) float f32x4;
+
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
