@@ -455,11 +455,17 @@ class SFTTrainer:
         if out_file is None:
             out_file = os.path.join(self.cfg.output_dir,
                                     "generated_predictions.jsonl")
+        from .gen_metrics import generation_metrics
+        metrics = generation_metrics(
+            [r["predict_ids"] for r in results],
+            [r["label_ids"] for r in results])
         if is_main():
             os.makedirs(os.path.dirname(out_file) or ".", exist_ok=True)
             with open(out_file, "w") as f:
                 for r in results:
                     f.write(json.dumps(r) + "\n")
+            self._log({**metrics, "current_steps": self.global_step,
+                       "total_steps": self.total_steps}, kind="eval")
         return results
 
     # -------------------------------------------------------- checkpoint

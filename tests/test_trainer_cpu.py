@@ -209,3 +209,14 @@ def test_predict_writes_jsonl(tmp_path):
     out = os.path.join(str(tmp_path), "generated_predictions.jsonl")
     rows = [json.loads(l) for l in open(out)]
     assert len(rows) == 3 and "prompt_ids" in rows[0]
+
+
+def test_generation_metrics():
+    from datatunerx_amd.train.gen_metrics import (bleu, generation_metrics,
+                                                  rouge_l)
+    assert rouge_l([1, 2, 3], [1, 2, 3]) == 1.0
+    assert rouge_l([1, 2, 3], [4, 5, 6]) == 0.0
+    assert bleu([1, 2, 3, 4], [1, 2, 3, 4]) > 0.99
+    assert bleu([9, 9, 9, 9], [1, 2, 3, 4]) < 0.01
+    m = generation_metrics([[1, 2, 3]], [[1, 2, 4]])
+    assert 0 < m["predict_rouge-l"] < 1
