@@ -89,18 +89,20 @@ class LlamaAttention(nn.Module):
         self.v_proj = _proj(cfg, "v_proj", cfg.hidden_size, Hkv * D, lora, dtype)
         self.o_proj = _proj(cfg, "o_proj", H * D, cfg.hidden_size, lora, dtype)
 
-    def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None):
+    def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None,
+                pos_dev=None):
         B, S, _ = x.shape
         cfg = self.cfg
         H, Hkv, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
         q = self.q_proj(x).view(B, S, H, D)
         k = self.k_proj(x).view(B, S, Hkv, D)
         v = self.v_proj(x).view(B, S, Hkv, D)
-        q = rope(q, cos, sin, pos0)
-        k = rope(k, cos, sin, pos0)
+        q = rope(q, cos, sin, pos0, pos_dev)
+        k = rope(k, cos, sin, pos0, pos_dev)
+        len_dev = None
         if kv_cache is not None:
-            k, v = kv_cache.update(k, v)         # serving path (BSHD)
-        o = attention(q, k, v, causal=True)      # BSHD in/out, no copies
+            k, v, len_dev = kv_cache.update(k, v)   # serving path (BSHD)
+        o = attention(q, k, v, causal=True, len_dev=len_dev)
         return self.o_proj(o.reshape(B, S, H * D))
 
 
@@ -129,9 +131,10 @@ class LlamaDecoderLayer(nn.Module):
         self.self_attn = LlamaAttention(cfg, lora, dtype)
         self.mlp = LlamaMLP(cfg, lora, dtype)
 
-    def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None):
+    def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None,
+                pos_dev=None):
         h = rmsnorm(x, self.input_layernorm, self.cfg.rms_norm_eps)
-        x = x + self.self_attn(h, cos, sin, pos0, kv_cache)
+        x = x + self.self_attn(h, cos, sin, pos0, kv_cache, pos_dev)
         h = rmsnorm(x, self.post_attention_layernorm, self.cfg.rms_norm_eps)
         return x + self.mlp(h)
 
@@ -179,7 +182,8 @@ class LlamaForCausalLM(nn.Module):
                 p.zero_()
         return self
 
-    def hidden_states(self, input_ids, pos0: int = 0, kv_caches=None):
+    def hidden_states(self, input_ids, pos0: int = 0, kv_caches=None,
+                      pos_dev=None):
         x = self.embed_tokens(input_ids)
         cos, sin = self.rope_cos, self.rope_sin
         for i, layer in enumerate(self.layers):
@@ -188,13 +192,14 @@ class LlamaForCausalLM(nn.Module):
                 x = torch.utils.checkpoint.checkpoint(
                     layer, x, cos, sin, pos0, cache, use_reentrant=False)
             else:
-                x = layer(x, cos, sin, pos0, cache)
+                x = layer(x, cos, sin, pos0, cache, pos_dev)
         return rmsnorm(x, self.norm, self.cfg.rms_norm_eps)
 
-    def forward(self, input_ids, labels=None, pos0: int = 0, kv_caches=None):
+    def forward(self, input_ids, labels=None, pos0: int = 0,
+                kv_caches=None, pos_dev=None):
         """Returns loss (if labels given, shifted CE with -100 ignore)
         else logits."""
-        h = self.hidden_states(input_ids, pos0, kv_caches)
+        h = self.hidden_states(input_ids, pos0, kv_caches, pos_dev)
         if labels is None:
             return self.lm_head(h)
         # shift: predict token t+1 from position t

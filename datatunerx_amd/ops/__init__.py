@@ -66,9 +66,11 @@ def rmsnorm_bwd(dy, x, w, inv):
 rope_tables = ref.rope_tables
 
 
-def rope_fwd(x, cos, sin, pos0: int = 0):
+def rope_fwd(x, cos, sin, pos0: int = 0, pos_dev=None):
     if _gpu(x):
-        return _EXT.rope(x, cos, sin, pos0, False)
+        return _EXT.rope(x, cos, sin, pos0, False, pos_dev)
+    if pos_dev is not None:
+        pos0 = pos0 + int(pos_dev)
     return ref.rope_fwd(x, cos, sin, pos0)
 
 
@@ -105,17 +107,22 @@ def softmax_xent_bwd(logits, targets, lse, dloss, ignore_index: int = -100):
 
 
 # --------------------------------------------------------- attention
-def attn_fwd(q, k, v, causal: bool = True, scale: float | None = None):
+def attn_fwd(q, k, v, causal: bool = True, scale: float | None = None,
+             len_dev=None):
     """BSHD: q [B,S,Hq,D], k/v [B,Skv,Hkv,D] -> (o [B,S,Hq,D], lse)."""
     if scale is None:
         scale = 1.0 / (q.shape[-1] ** 0.5)
     if _gpu(q):
         if q.shape[1] == 1:
             # serving decode: flash-decoding split-KV path (with S=1
-            # every cached position is visible, causal or not)
-            return _EXT.attn_decode(q, k, v, scale)
+            # every cached position is visible, causal or not).
+            # len_dev: device-resident cache length (hipGraph decode).
+            return _EXT.attn_decode(q, k, v, scale, len_dev)
         vt = _EXT.transpose_sd(v)        # [B,Hkv,D,Skv] for the PV tiles
         return _EXT.attn_fwd(q, k, vt, causal, scale)
+    if len_dev is not None:
+        k = k[:, :int(len_dev)]
+        v = v[:, :int(len_dev)]
     return ref.attn_fwd(q, k, v, causal, scale)
 
 

@@ -33,19 +33,20 @@ def rmsnorm(x, w, eps: float = 1e-5):
 
 class Rope(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, cos, sin, pos0):
+    def forward(ctx, x, cos, sin, pos0, pos_dev):
         ctx.save_for_backward(cos, sin)
         ctx.pos0 = pos0
-        return rope_fwd(x, cos, sin, pos0)
+        return rope_fwd(x, cos, sin, pos0, pos_dev)
 
     @staticmethod
     def backward(ctx, dy):
         cos, sin = ctx.saved_tensors
-        return rope_bwd(dy.contiguous(), cos, sin, ctx.pos0), None, None, None
+        return (rope_bwd(dy.contiguous(), cos, sin, ctx.pos0),
+                None, None, None, None)
 
 
-def rope(x, cos, sin, pos0: int = 0):
-    return Rope.apply(x, cos, sin, pos0)
+def rope(x, cos, sin, pos0: int = 0, pos_dev=None):
+    return Rope.apply(x, cos, sin, pos0, pos_dev)
 
 
 class SwiGLU(torch.autograd.Function):
@@ -69,13 +70,13 @@ class Attention(torch.autograd.Function):
     """BSHD flash attention (no transpose copies around the kernel)."""
 
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale):
+    def forward(ctx, q, k, v, causal, scale, len_dev):
         q = q.contiguous()
         if not (q.is_cuda and q.shape[1] == 1):
             # decode (S=1) accepts dense KV-cache prefix views; every
             # other path wants contiguous k/v
             k, v = k.contiguous(), v.contiguous()
-        o, lse = attn_fwd(q, k, v, causal, scale)
+        o, lse = attn_fwd(q, k, v, causal, scale, len_dev)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal, ctx.scale = causal, scale
         return o
@@ -85,14 +86,15 @@ class Attention(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         dq, dk, dv = attn_bwd(q, k, v, o, do.contiguous(), lse,
                               ctx.causal, ctx.scale)
-        return dq, dk, dv, None, None
+        return dq, dk, dv, None, None, None
 
 
-def attention(q, k, v, causal: bool = True, scale: float | None = None):
+def attention(q, k, v, causal: bool = True, scale: float | None = None,
+              len_dev=None):
     """q [B,S,Hq,D], k/v [B,Skv,Hkv,D] -> o [B,S,Hq,D]."""
     if scale is None:
         scale = 1.0 / (q.shape[-1] ** 0.5)
-    return Attention.apply(q, k, v, causal, scale)
+    return Attention.apply(q, k, v, causal, scale, len_dev)
 
 
 class CrossEntropy(torch.autograd.Function):

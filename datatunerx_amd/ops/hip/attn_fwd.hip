@@ -388,11 +388,15 @@ void launch_attn_fwd(const void* q, const void* k, const void* vt, void* o,
 // rate with every lane busy.
 //   grid (NS, B*Hq), block 256 = 4 waves; 16 lanes per kv row (8 els
 //   each), 4 row slots per wave.
+// len_dev (optional): current cache length read on DEVICE so a
+// hipGraph-captured decode step can grow the window without re-capture
+// (Skv stays the CAPACITY for addressing; len_dev bounds the rows).
 template <int D>
 __global__ __launch_bounds__(DTX_BLOCK)
 void attn_decode_kernel(const unsigned short* __restrict__ Q,
                         const unsigned short* __restrict__ Kp,
                         const unsigned short* __restrict__ Vp,
+                        const int* __restrict__ len_dev,
                         float* __restrict__ part,   // [B*Hq, NS, D+2]
                         int B, int Hq, int Hkv, int Skv, int chunk,
                         float scale) {
@@ -408,8 +412,9 @@ void attn_decode_kernel(const unsigned short* __restrict__ Q,
   const int hkv = hq / (Hq / Hkv);
   const int krowstr = Hkv * D;
   const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
+  const int len = len_dev ? *len_dev : Skv;
   const int kv0 = blockIdx.x * chunk;
-  const int kv_end = min(Skv, kv0 + chunk);
+  const int kv_end = min(len, kv0 + chunk);
 
   // q slice for this lane (f32)
   float qv[8];
@@ -529,22 +534,24 @@ int attn_decode_nsplit(int Skv) {
 }
 
 void launch_attn_decode(const void* q, const void* k, const void* v,
-                        float* part, void* o, float* lse, int B, int Hq,
-                        int Hkv, int Skv, int D, float scale,
-                        hipStream_t st) {
+                        const int* len_dev, float* part, void* o,
+                        float* lse, int B, int Hq, int Hkv, int Skv,
+                        int D, float scale, hipStream_t st) {
   const int ns = attn_decode_nsplit(Skv);
   const int chunk = DTX_CDIV(Skv, ns);
   dim3 grid(ns, B * Hq);
   if (D == 128) {
     attn_decode_kernel<128><<<grid, DTX_BLOCK, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)v, part, B, Hq, Hkv, Skv, chunk, scale);
+        (const unsigned short*)v, len_dev, part, B, Hq, Hkv, Skv, chunk,
+        scale);
     attn_decode_combine_kernel<128><<<B * Hq, 64, 0, st>>>(
         part, (unsigned short*)o, lse, B, Hq, ns);
   } else if (D == 64) {
     attn_decode_kernel<64><<<grid, DTX_BLOCK, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)v, part, B, Hq, Hkv, Skv, chunk, scale);
+        (const unsigned short*)v, len_dev, part, B, Hq, Hkv, Skv, chunk,
+        scale);
     attn_decode_combine_kernel<64><<<B * Hq, 64, 0, st>>>(
         part, (unsigned short*)o, lse, B, Hq, ns);
   }

@@ -11,7 +11,7 @@ int rmsnorm_bwd_nblocks(int M);
 void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
                         void*, float*, float*, int, int, hipStream_t);
 void launch_rope(const void*, const float*, const float*, void*, long, int,
-                 int, int, int, int, hipStream_t);
+                 int, int, int, int, const int*, hipStream_t);
 void launch_swiglu_fwd(const void*, const void*, void*, long, hipStream_t);
 void launch_swiglu_bwd(const void*, const void*, const void*, void*, void*,
                        long, hipStream_t);
@@ -35,9 +35,9 @@ void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
 void launch_transpose_sd(const void*, void*, int, int, int, int,
                          hipStream_t);
 int attn_decode_nsplit(int Skv);
-void launch_attn_decode(const void*, const void*, const void*, float*,
-                        void*, float*, int, int, int, int, int, float,
-                        hipStream_t);
+void launch_attn_decode(const void*, const void*, const void*,
+                        const int*, float*, void*, float*, int, int, int,
+                        int, int, float, hipStream_t);
 void launch_attn_delta(const void*, const void*, float*, long, int, int,
                        int, hipStream_t);
 void launch_attn_bwd_dkdv(const void*, const void*, const void*,
@@ -95,17 +95,26 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
 
 // ---------------------------------------------------------------- RoPE
 torch::Tensor rope(torch::Tensor x, torch::Tensor cosb, torch::Tensor sinb,
-                   long pos0, bool backward) {
+                   long pos0, bool backward,
+                   c10::optional<torch::Tensor> pos_dev) {
   check_bf16_contig(x, "x");
   TORCH_CHECK(x.dim() == 4, "x must be [B,S,H,D]");
   TORCH_CHECK(cosb.scalar_type() == torch::kFloat && cosb.is_contiguous());
   const int B = (int)x.size(0), S = (int)x.size(1), H = (int)x.size(2),
             D = (int)x.size(3);
   TORCH_CHECK(D % 8 == 0, "D % 8");
-  TORCH_CHECK(cosb.size(0) >= pos0 + S, "rope table too short");
+  const int* pd = nullptr;
+  if (pos_dev.has_value()) {
+    TORCH_CHECK(pos_dev->scalar_type() == torch::kInt &&
+                pos_dev->is_cuda(), "pos_dev must be int32 on GPU");
+    pd = pos_dev->data_ptr<int>();
+    TORCH_CHECK(cosb.size(0) >= S, "rope table too short");
+  } else {
+    TORCH_CHECK(cosb.size(0) >= pos0 + S, "rope table too short");
+  }
   auto y = torch::empty_like(x);
   launch_rope(x.data_ptr(), cosb.data_ptr<float>(), sinb.data_ptr<float>(),
-              y.data_ptr(), B, S, H, D, (int)pos0, backward ? 1 : 0,
+              y.data_ptr(), B, S, H, D, (int)pos0, backward ? 1 : 0, pd,
               cur_stream());
   return y;
 }
@@ -289,7 +298,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
 // Decode fast path: q [B,1,Hq,D] against the KV cache (flash-decoding
 // split-KV partials; no V transpose needed).
 std::vector<torch::Tensor> attn_decode(torch::Tensor q, torch::Tensor k,
-                                       torch::Tensor v, double scale) {
+                                       torch::Tensor v, double scale,
+                                       c10::optional<torch::Tensor>
+                                           len_dev) {
   check_bf16_contig(q, "q");
   const int B = (int)q.size(0), Hq = (int)q.size(2), D = (int)q.size(3);
   const int Skv = (int)k.size(1), Hkv = (int)k.size(2);
@@ -312,7 +323,13 @@ std::vector<torch::Tensor> attn_decode(torch::Tensor q, torch::Tensor k,
                            q.options().dtype(torch::kFloat));
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, Hq, 1}, q.options().dtype(torch::kFloat));
-  launch_attn_decode(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+  const int* ld = nullptr;
+  if (len_dev.has_value()) {
+    TORCH_CHECK(len_dev->scalar_type() == torch::kInt &&
+                len_dev->is_cuda(), "len_dev must be int32 on GPU");
+    ld = len_dev->data_ptr<int>();
+  }
+  launch_attn_decode(q.data_ptr(), k.data_ptr(), v.data_ptr(), ld,
                      part.data_ptr<float>(), o.data_ptr(),
                      lse.data_ptr<float>(), B, Hq, Hkv, Skv, D,
                      (float)scale, cur_stream());
@@ -363,7 +380,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
-  m.def("rope", &rope);
+  m.def("rope", &rope, py::arg("x"), py::arg("cosb"), py::arg("sinb"),
+        py::arg("pos0"), py::arg("backward"),
+        py::arg("pos_dev") = py::none());
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("xent_fwd", &xent_fwd);
@@ -379,5 +398,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("transpose_sd", &transpose_sd);
-  m.def("attn_decode", &attn_decode);
+  m.def("attn_decode", &attn_decode, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("scale"), py::arg("len_dev") = py::none());
 }

@@ -176,8 +176,10 @@ void rope_kernel(const unsigned short* __restrict__ x,
                  const float* __restrict__ cosb,
                  const float* __restrict__ sinb,
                  unsigned short* __restrict__ y,
+                 const int* __restrict__ pos_dev,  // graph-mode position
                  long total_quads,  // B*S*H*(D/2/4)
                  int S, int H, int D, int pos0, int backward) {
+  if (pos_dev) pos0 += *pos_dev;
   const int qpr = D / 8;                      // 4-pair groups per head-row
   long idx = (long)blockIdx.x * DTX_BLOCK + threadIdx.x;
   long stride = (long)gridDim.x * DTX_BLOCK;
@@ -307,10 +309,10 @@ void launch_reduce_partials(const float* part, float* out, int P, long L,
 
 void launch_rope(const void* x, const float* cosb, const float* sinb,
                  void* y, long B, int S, int H, int D, int pos0,
-                 int backward, hipStream_t s) {
+                 int backward, const int* pos_dev, hipStream_t s) {
   long quads = B * S * H * (D / 8);
   rope_kernel<<<ew_grid(quads), DTX_BLOCK, 0, s>>>(
-      (const unsigned short*)x, cosb, sinb, (unsigned short*)y, quads,
+      (const unsigned short*)x, cosb, sinb, (unsigned short*)y, pos_dev, quads,
       S, H, D, pos0, backward);
 }
 

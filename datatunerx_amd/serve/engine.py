@@ -12,6 +12,7 @@ parallel/tp.py.
 from __future__ import annotations
 
 import math
+import os
 from typing import List, Optional
 
 import torch
@@ -24,15 +25,28 @@ from ..models import (GPT2Config, GPT2ForCausalLM, LlamaConfig,
 
 class KVCache:
     """BSHD cache ([B, max_s, Hkv, D]) matching the attention kernels'
-    native layout — appends along dim 1."""
+    native layout — appends along dim 1.
 
-    def __init__(self, B, Hkv, max_s, D, device, dtype):
+    update() returns (k, v, len_dev): len_dev is None on the eager path
+    (the returned tensors are length-exact prefixes) and a device int32
+    scalar in graph mode (full-capacity tensors whose live length the
+    decode kernel reads on device — hipGraph-replayable)."""
+
+    def __init__(self, B, Hkv, max_s, D, device, dtype,
+                 graph_pos=None, graph_len=None):
         self.k = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
         self.v = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
         self.cur = 0
+        self.graph_pos = graph_pos       # int64 [1]: write index
+        self.graph_len = graph_len       # int32 [1]: length incl. new tok
 
     def update(self, k, v):
         S = k.shape[1]
+        if self.graph_pos is not None and S == 1 and k.is_cuda:
+            # graph-safe: device-indexed write, device-read length
+            self.k.index_copy_(1, self.graph_pos, k)
+            self.v.index_copy_(1, self.graph_pos, v)
+            return self.k, self.v, self.graph_len
         self.k[:, self.cur:self.cur + S] = k
         self.v[:, self.cur:self.cur + S] = v
         self.cur += S
@@ -41,7 +55,7 @@ class KVCache:
             # batch>1 / CPU paths want contiguous; the GPU decode kernel
             # takes the dense prefix view directly (no per-token copy)
             kk, vv = kk.contiguous(), vv.contiguous()
-        return kk, vv
+        return kk, vv, None
 
 
 def build_model(name: str, device, adapter_dir: Optional[str] = None,
@@ -83,14 +97,86 @@ def build_model(name: str, device, adapter_dir: Optional[str] = None,
     return model
 
 
+class _GraphedDecoder:
+    """hipGraph-captured decode step (torch.cuda.CUDAGraph): the whole
+    32-layer token forward replays as one graph — position, cache length
+    and the KV write index live in device tensors so the replay needs no
+    re-capture as the sequence grows. Built once per engine at fixed
+    cache capacity; counters reset per request."""
+
+    def __init__(self, model, cfg, device, max_s):
+        self.model, self.max_s = model, max_s
+        dtype = next(model.parameters()).dtype
+        self.pos64 = torch.zeros(1, dtype=torch.long, device=device)
+        self.pos32 = torch.zeros(1, dtype=torch.int32, device=device)
+        self.len32 = torch.ones(1, dtype=torch.int32, device=device)
+        self.caches = [
+            KVCache(1, cfg.num_key_value_heads, max_s, cfg.head_dim,
+                    device, dtype, graph_pos=self.pos64,
+                    graph_len=self.len32)
+            for _ in range(cfg.num_hidden_layers)]
+        self.input_id = torch.zeros(1, 1, dtype=torch.long, device=device)
+        # warmup (allocator + kernels), then capture
+        with torch.no_grad():
+            for _ in range(3):
+                self.model(self.input_id, pos_dev=self.pos32,
+                           kv_caches=self.caches)
+            torch.cuda.synchronize()
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self.logits = self.model(self.input_id,
+                                         pos_dev=self.pos32,
+                                         kv_caches=self.caches)
+
+    def reset(self):
+        self.pos64.zero_()
+        self.pos32.zero_()
+        self.len32.fill_(1)
+        for c in self.caches:
+            c.cur = 0
+
+    def after_prefill(self, n: int):
+        """counters for the first graphed token (written at index n)."""
+        self.pos64.fill_(n)
+        self.pos32.fill_(n)
+        self.len32.fill_(n + 1)
+        for c in self.caches:
+            c.cur = n
+
+    def step(self, token: int):
+        self.input_id[0, 0] = token
+        self.graph.replay()
+        self.pos64 += 1
+        self.pos32 += 1
+        self.len32 += 1
+        return self.logits
+
+
 class InferenceEngine:
     def __init__(self, model, tokenizer=None, template: str = "llama2",
-                 device=None):
+                 device=None, graph_decode: bool = True):
         self.model = model
         self.tok = tokenizer or ByteTokenizer()
         self.template = template
         self.device = device or next(model.parameters()).device
         self.is_llama = isinstance(model, LlamaForCausalLM)
+        import torch.distributed as dist
+        self._graphed = None
+        self._graph_ok = (graph_decode and self.is_llama and
+                          self.device.type == "cuda" and
+                          not dist.is_initialized() and
+                          os.environ.get("DTX_NO_GRAPH") != "1")
+
+    def _get_graphed(self):
+        if self._graphed is None and self._graph_ok:
+            try:
+                cfg = self.model.cfg
+                self._graphed = _GraphedDecoder(
+                    self.model, cfg, self.device,
+                    cfg.max_position_embeddings)
+            except Exception:
+                self._graph_ok = False
+        return self._graphed
 
     # ------------------------------------------------------------ chat
     def chat(self, messages: List[dict], max_tokens: int = 64,
@@ -120,6 +206,24 @@ class InferenceEngine:
         ids = torch.tensor([prompt_ids], dtype=torch.long,
                            device=self.device)
         out = []
+        gd = self._get_graphed() if self.is_llama else None
+        if gd is not None and len(prompt_ids) + max_new_tokens + 1 < \
+                gd.max_s:
+            # hipGraph-replayed decode: prefill eagerly into the graph's
+            # caches, then one graph replay per token
+            gd.reset()
+            logits = self.model(ids, pos0=0, kv_caches=gd.caches)
+            gd.after_prefill(len(prompt_ids))
+            nxt = self._sample(logits[0, -1], temperature, top_p)
+            for _ in range(max_new_tokens):
+                if nxt == self.tok.eos_token_id:
+                    break
+                out.append(nxt)
+                if len(prompt_ids) + len(out) + 1 >= gd.max_s:
+                    break
+                logits = gd.step(nxt)
+                nxt = self._sample(logits[0, -1], temperature, top_p)
+            return out
         if self.is_llama:
             cfg = self.model.cfg
             max_s = min(cfg.max_position_embeddings,

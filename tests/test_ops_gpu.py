@@ -269,11 +269,16 @@ def test_serve_engine_decode_gpu():
         model = LlamaForCausalLM(cfg, lora=False, dtype=torch.bfloat16)
     model.init_random(seed=3)
     model.eval()
-    eng = InferenceEngine(model, device=DEV)
+    eng = InferenceEngine(model, device=DEV)          # hipGraph decode
     out1 = eng.generate(list(range(4, 20)), max_new_tokens=8)
     out2 = eng.generate(list(range(4, 20)), max_new_tokens=8)
     assert out1 == out2
     assert 0 < len(out1) <= 8
+    assert eng._graphed is not None, "graph decode did not engage"
+    # graph replay must match the eager decode path token-for-token
+    eager = InferenceEngine(model, device=DEV, graph_decode=False)
+    out3 = eager.generate(list(range(4, 20)), max_new_tokens=8)
+    assert out1 == out3
     ppl = eng.perplexity(["hello world", "the quick brown fox"])
     assert 0 < ppl < float("inf")
 
