@@ -35,6 +35,9 @@ class FrozenLinear(nn.Module):
             requires_grad=False)
 
     def forward(self, x):
+        if x.is_cuda and x.numel() == x.shape[-1]:
+            from ..ops import gemv
+            return gemv(x, self.weight)      # decode: streaming GEMV
         # F.linear routes to addmm/hipBLASLt: +18% over x @ w.t() on the
         # [16K,4096]x[4096,4096] shape (measured, tools/bench_gemm.py)
         return F.linear(x, self.weight)

@@ -135,6 +135,13 @@ def attn_bwd(q, k, v, o, do, lse, causal: bool = True,
     return ref.attn_bwd(q, k, v, o, do, lse, causal, scale)
 
 
+def gemv(x, w):
+    """y[...,1,N] = x[...,1,K] @ w[N,K]^T — decode GEMV (single row)."""
+    if _gpu(x):
+        return _EXT.gemv(x, w)
+    return torch.nn.functional.linear(x, w)
+
+
 # -------------------------------------------------------------- LoRA
 # `mask` (optional, same shape/dtype as x / y) fuses the PEFT input
 # dropout into the kernels.

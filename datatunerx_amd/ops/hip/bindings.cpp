@@ -35,6 +35,8 @@ void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
 void launch_transpose_sd(const void*, void*, int, int, int, int,
                          hipStream_t);
 int attn_decode_nsplit(int Skv);
+void launch_gemv_bf16(const void*, const void*, void*, int, int,
+                      hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*,
                         const int*, float*, void*, float*, int, int, int,
                         int, int, float, hipStream_t);
@@ -295,6 +297,21 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   return {o, lse};
 }
 
+// y[1,N] = x[1,K] @ W[N,K]^T (decode GEMV; fp32 accumulate)
+torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int K = (int)w.size(1), N = (int)w.size(0);
+  TORCH_CHECK(x.numel() == K, "gemv wants a single row");
+  TORCH_CHECK(K % 8 == 0, "K % 8");
+  auto sizes = x.sizes().vec();
+  sizes[sizes.size() - 1] = N;
+  auto y = torch::empty(sizes, x.options());
+  launch_gemv_bf16(x.data_ptr(), w.data_ptr(), y.data_ptr(), N, K,
+                   cur_stream());
+  return y;
+}
+
 // Decode fast path: q [B,1,Hq,D] against the KV cache (flash-decoding
 // split-KV partials; no V transpose needed).
 std::vector<torch::Tensor> attn_decode(torch::Tensor q, torch::Tensor k,
@@ -398,6 +415,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("transpose_sd", &transpose_sd);
+  m.def("gemv", &gemv);
   m.def("attn_decode", &attn_decode, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("scale"), py::arg("len_dev") = py::none());
 }

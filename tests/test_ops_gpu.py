@@ -298,3 +298,14 @@ def test_attn_decode(B, Hq, Hkv, Skv, D):
     assert_close(o.cpu(), o_ref, name="decode o")
     assert_close(lse.cpu().reshape(-1), lse_ref.reshape(-1), rtol=1e-2,
                  name="decode lse")
+
+
+@pytest.mark.parametrize("N,K", [(4096, 4096), (32000, 4096),
+                                 (11008, 4096), (120, 64)])
+def test_gemv(N, K):
+    from datatunerx_amd.ops import _dtx_hip
+    x = mk(1, K, scale=0.3)
+    w = mk(N, K, scale=0.3)
+    y = _dtx_hip.gemv(x, w)
+    want = torch.nn.functional.linear(x.float(), w.float())
+    assert_close(y.cpu(), want.cpu(), name="gemv")
