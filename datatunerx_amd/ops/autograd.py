@@ -134,7 +134,11 @@ class LoRALinear(torch.autograd.Function):
     def forward(ctx, x, w, a, b, scale):
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
-        y = torch.nn.functional.linear(x2, w)
+        if x2.is_cuda and x2.shape[0] == 1:
+            from . import gemv
+            y = gemv(x2, w)                      # decode: streaming GEMV
+        else:
+            y = torch.nn.functional.linear(x2, w)
         t = lora_contract(x2, a)                 # [M,r] f32
         lora_expand_add(y, t, b, scale)          # y += s * t @ B^T
         ctx.save_for_backward(x2, w, a, b, t)
