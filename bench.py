@@ -39,6 +39,8 @@ def build_model(name: str, device, lora_dropout: float,
         cfg = LlamaConfig.llama2_7b(lora_dropout=lora_dropout)
     elif name == "llama2-13b":
         cfg = LlamaConfig.llama2_13b(lora_dropout=lora_dropout)
+    elif name == "llama3-8b":
+        cfg = LlamaConfig.llama3_8b(lora_dropout=lora_dropout)
     elif name == "llama-tiny":
         cfg = LlamaConfig.tiny(lora_dropout=lora_dropout)
     else:

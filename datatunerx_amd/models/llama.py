@@ -60,6 +60,16 @@ class LlamaConfig:
         return cls(**d)
 
     @classmethod
+    def llama3_8b(cls, **kw):
+        """GQA config (32 q heads / 8 kv heads), Llama-3-8B geometry."""
+        d = dict(vocab_size=128256, hidden_size=4096,
+                 intermediate_size=14336, num_hidden_layers=32,
+                 num_attention_heads=32, num_key_value_heads=8,
+                 max_position_embeddings=8192, rope_theta=500000.0)
+        d.update(kw)
+        return cls(**d)
+
+    @classmethod
     def tiny(cls, **kw):
         """CPU-testable config."""
         d = dict(vocab_size=512, hidden_size=128, intermediate_size=256,

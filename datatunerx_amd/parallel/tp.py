@@ -205,7 +205,9 @@ def load_adapter_tp(model, adapter_dir: str, cfg: LlamaConfig, rank: int,
     from safetensors.torch import load_file
     sd = load_file(os.path.join(adapter_dir, "adapter_model.safetensors"))
     sd = shard_adapter_state(sd, cfg, rank, ws)
-    mods = {n: m for n, m in model.named_modules()
+    # _RowParallelLinear wraps o_proj/down_proj: strip the ".inner" hop
+    # so checkpoint module paths still resolve
+    mods = {n.replace(".inner", ""): m for n, m in model.named_modules()
             if isinstance(m, LoRALinearModule)}
     n_loaded = 0
     for key, tensor in sd.items():

@@ -59,6 +59,10 @@ def main(argv=None):
                 model = LlamaForCausalLM(LlamaConfig.llama2_13b(
                     gradient_checkpointing=fargs.gradient_checkpointing,
                     **lora_kw), lora=lora, dtype=dtype, train_base=full)
+            elif name in ("llama3-8b", "llama-3-8b"):
+                model = LlamaForCausalLM(LlamaConfig.llama3_8b(
+                    gradient_checkpointing=fargs.gradient_checkpointing,
+                    **lora_kw), lora=lora, dtype=dtype, train_base=full)
             elif name == "llama-tiny":
                 model = LlamaForCausalLM(LlamaConfig.tiny(**lora_kw),
                                          lora=lora, dtype=dtype,

@@ -147,3 +147,69 @@ def test_shard_adapter_state_roundtrip():
     assert torch.equal(torch.cat([s0[qb], s1[qb]], 0), sd[qb])
     assert torch.equal(torch.cat([s0[oa], s1[oa]], 1), sd[oa])
     assert torch.equal(s0[qa], sd[qa])
+
+
+def test_tp_adapter_loading(tmp_path):
+    """A full HF-PEFT adapter loads into TP shards and reproduces the
+    single-process adapter-model logits (2 ranks, gloo)."""
+    worker = r"""
+import json, os, sys
+import torch
+import torch.distributed as dist
+sys.path.insert(0, os.environ["DTX_ROOT"])
+from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+from datatunerx_amd.models.lora import save_adapter, load_adapter
+from datatunerx_amd.parallel.ddp import init_distributed
+from datatunerx_amd.parallel.tp import build_tp_llama, load_adapter_tp
+
+rank, world, local, device = init_distributed(backend="gloo")
+cfg = LlamaConfig.tiny(lora_targets=("q_proj", "v_proj", "o_proj"))
+out = os.environ["DTX_OUT"]
+if rank == 0:
+    full = LlamaForCausalLM(cfg, lora=True,
+                            dtype=torch.float32).init_random(seed=5)
+    torch.manual_seed(3)
+    for n, p in full.named_parameters():
+        if "lora" in n:
+            with torch.no_grad():
+                p.add_(torch.randn_like(p) * 0.05)
+    save_adapter(full, out + "/ad", r=cfg.lora_r, alpha=cfg.lora_alpha,
+                 dropout=0.0, target_modules=list(cfg.lora_targets))
+dist.barrier()
+tp = build_tp_llama(cfg, rank, world, lora=True, dtype=torch.float32,
+                    device=device, seed=5)
+n = load_adapter_tp(tp, out + "/ad", cfg, rank, world)
+assert n > 0, "no adapter tensors loaded"
+tp.eval()
+torch.manual_seed(11)
+ids = torch.randint(0, cfg.vocab_size, (1, 16))
+dist.broadcast(ids, src=0)
+with torch.no_grad():
+    lt = tp(ids)
+if rank == 0:
+    full2 = LlamaForCausalLM(cfg, lora=True,
+                             dtype=torch.float32).init_random(seed=5)
+    load_adapter(full2, out + "/ad")
+    full2.eval()
+    with torch.no_grad():
+        lf = full2(ids)
+    err = (lt - lf).abs().max().item()
+    with open(out + "/ad_out.json", "w") as f:
+        json.dump({"err": err, "ref": lf.abs().max().item()}, f)
+dist.destroy_process_group()
+"""
+    script = tmp_path / "w2.py"
+    script.write_text(worker)
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({"RANK": str(rank), "WORLD_SIZE": "2",
+                    "LOCAL_RANK": str(rank),
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29781",
+                    "DTX_ROOT": ROOT, "DTX_OUT": str(tmp_path)})
+        procs.append(subprocess.Popen([sys.executable, str(script)],
+                                      env=env))
+    for p in procs:
+        assert p.wait(timeout=300) == 0
+    out = json.load(open(tmp_path / "ad_out.json"))
+    assert out["err"] < 1e-4 * max(1.0, out["ref"])
