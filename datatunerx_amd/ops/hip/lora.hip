@@ -224,18 +224,24 @@ void lora_wgrad_kernel(const float* __restrict__ T,
 void launch_reduce_partials(const float* part, float* out, int P, long L,
                             hipStream_t s);
 
+// split plan: spans are MULTIPLES OF 1024 elements so every staged
+// 16-byte load stays aligned (K=5120 used to produce a 1707-element
+// span -> misaligned b128 loads -> GPU memory fault on the 13B shapes).
+static int lora_contract_kspan(int K) {
+  int units = DTX_CDIV(K, 1024);
+  int nsplit = units < 4 ? units : 4;
+  return DTX_CDIV(units, nsplit) * 1024;
+}
+
 int lora_contract_ksplit(int K) {
-  // aim for >= 512 blocks: grid.x is capped at 128 (x4 waves = 512 rows)
-  int ks = DTX_CDIV(K, 4096) * 1024;     // 1..4 chunks of 1024 per block
-  int nsplit = DTX_CDIV(K, ks);
-  return nsplit;
+  return DTX_CDIV(K, lora_contract_kspan(K));
 }
 
 void launch_lora_contract(const void* X, const void* W, const void* Mk,
                           float* part, float* out, long M, int K, int r,
                           hipStream_t s) {
   const int nsplit = lora_contract_ksplit(K);
-  const int kspan = DTX_CDIV(K, nsplit);
+  const int kspan = lora_contract_kspan(K);
   long gw = DTX_CDIV(M, 128);
   dim3 grid((int)(gw < 128 ? (gw < 1 ? 1 : gw) : 128), nsplit);
   float* dst = nsplit > 1 ? part : out;

@@ -92,7 +92,8 @@ def test_xent(N, V):
 
 # ------------------------------------------------------------------- lora
 @pytest.mark.parametrize("M,K,r", [(1024, 4096, 8), (513, 4096, 16),
-                                   (256, 11008, 8), (512, 2048, 64)])
+                                   (256, 11008, 8), (512, 2048, 64),
+                                   (256, 5120, 8)])   # 13B K (odd split)
 def test_lora_contract(M, K, r):
     x, w = mk(M, K, scale=0.3), mk(r, K, scale=0.3)
     t = ops.lora_contract(x, w)
