@@ -27,7 +27,7 @@ void launch_lora_contract(const void*, const void*, const void*, float*,
                           float*, long, int, int, hipStream_t);
 void launch_lora_expand_add(void*, const float*, const void*, const void*,
                             long, int, int, float, hipStream_t);
-int lora_wgrad_splitm(int K);
+int lora_wgrad_splitm(int K, int r, long M);
 void launch_lora_wgrad(const float*, const void*, const void*, float*,
                        float*, long, int, int, float, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
@@ -229,7 +229,7 @@ torch::Tensor lora_wgrad(torch::Tensor t, torch::Tensor x, double scale,
   const long M = x.numel() / K;
   const int r = (int)t.size(-1);
   auto out = torch::empty({r, K}, x.options().dtype(torch::kFloat));
-  const int sm = lora_wgrad_splitm(K);
+  const int sm = lora_wgrad_splitm(K, r, M);
   auto part = torch::empty({sm, r, K}, x.options().dtype(torch::kFloat));
   launch_lora_wgrad(t.data_ptr<float>(), x.data_ptr(),
                     opt_mask(mask, x.numel(), "mask"),

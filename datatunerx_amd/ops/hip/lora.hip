@@ -220,6 +220,62 @@ void lora_wgrad_kernel(const float* __restrict__ T,
   }
 }
 
+// Register-hoisted expand_add for the common small ranks (r <= 16):
+// each block owns a 2048-column slice of Y and a row range; the WT
+// slice lives in REGISTERS for the whole row loop, so the only memory
+// traffic per row is T (lane-uniform 4B*r), Y in/out and the mask --
+// the v1 kernel re-read WT from L2 for every row (64 KB/row/wave) and
+// sat at ~2.8 TB/s; this one runs at the streaming roofline.
+template <int RCH, bool MASKED>
+__global__ __launch_bounds__(DTX_BLOCK)
+void lora_expand_add_kernel2(unsigned short* __restrict__ Y,
+                             const float* __restrict__ T,
+                             const unsigned short* __restrict__ WT,
+                             const unsigned short* __restrict__ Mk,
+                             long M, int N, int r, float s, int rowsplit) {
+  const int col = blockIdx.x * 2048 + threadIdx.x * 8;
+  if (col >= N) return;                      // N % 8 == 0 (checked host)
+  float w[RCH][8];
+#pragma unroll
+  for (int j = 0; j < RCH; ++j) {
+    if (j < r) {
+      load_bf16x8(WT + (long)j * N + col, w[j]);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) w[j][i] *= s;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) w[j][i] = 0.f;
+    }
+  }
+  const long m0 = (M * (long)blockIdx.y) / rowsplit;
+  const long m1 = (M * (long)(blockIdx.y + 1)) / rowsplit;
+  for (long m = m0; m < m1; ++m) {
+    const float* tr = T + m * r;
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+    for (int j = 0; j < RCH; ++j) {
+      if (j < r) {
+        const float t = tr[j];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) acc[i] += t * w[j][i];
+      }
+    }
+    unsigned short* yp = Y + m * N + col;
+    float y[8];
+    load_bf16x8(yp, y);
+    if (MASKED) {
+      float mv[8];
+      load_bf16x8(Mk + m * N + col, mv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) y[i] += mv[i] * acc[i];
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) y[i] += acc[i];
+    }
+    store_bf16x8(yp, y);
+  }
+}
+
 // ------------------------------------------------------------- launchers
 void launch_reduce_partials(const float* part, float* out, int P, long L,
                             hipStream_t s);
@@ -261,6 +317,29 @@ void launch_lora_contract(const void* X, const void* W, const void* Mk,
 void launch_lora_expand_add(void* Y, const float* T, const void* WT,
                             const void* Mk, long M, int N, int r,
                             float scale, hipStream_t s) {
+  if (r <= 16 && N % 8 == 0 && M >= 64) {
+    const int gx = DTX_CDIV(N, 2048);
+    int rowsplit = 1024 / gx;
+    if (rowsplit > M) rowsplit = (int)M;
+    if (rowsplit < 1) rowsplit = 1;
+    dim3 grid(gx, rowsplit);
+#define EX2(RC)                                                           \
+    do {                                                                  \
+      if (Mk)                                                             \
+        lora_expand_add_kernel2<RC, true><<<grid, DTX_BLOCK, 0, s>>>(     \
+            (unsigned short*)Y, T, (const unsigned short*)WT,             \
+            (const unsigned short*)Mk, M, N, r, scale, rowsplit);         \
+      else                                                                \
+        lora_expand_add_kernel2<RC, false><<<grid, DTX_BLOCK, 0, s>>>(    \
+            (unsigned short*)Y, T, (const unsigned short*)WT, nullptr,    \
+            M, N, r, scale, rowsplit);                                    \
+    } while (0)
+    if (r <= 4) EX2(4);
+    else if (r <= 8) EX2(8);
+    else EX2(16);
+#undef EX2
+    return;
+  }
   long gw = DTX_CDIV(M, 4);
   int grid = (int)(gw < 2048 ? (gw < 1 ? 1 : gw) : 2048);
   if (Mk) {
@@ -274,16 +353,22 @@ void launch_lora_expand_add(void* Y, const float* T, const void* WT,
   }
 }
 
-int lora_wgrad_splitm(int K) {
+int lora_wgrad_splitm(int K, int r, long M) {
   int kblocks = DTX_CDIV(K, 2048);
-  int sm = 512 / kblocks;
-  return sm < 1 ? 1 : (sm > 128 ? 128 : sm);
+  // target ~1024 blocks (4/CU -> 4 waves/SIMD); the old 512/cap-128
+  // plan put ONE wave per SIMD on K=4096 and left 3.5x bandwidth idle
+  int sm = 2048 / kblocks;
+  // keep the fp32 partial buffer under ~256 MB
+  long cap_mem = (256L << 20) / ((long)r * K * 4);
+  if (sm > cap_mem) sm = (int)cap_mem;
+  if (sm > M) sm = (int)M;
+  return sm < 1 ? 1 : (sm > 512 ? 512 : sm);
 }
 
 void launch_lora_wgrad(const float* T, const void* X, const void* Mk,
                        float* part, float* out, long M, int K, int r,
                        float s, hipStream_t st) {
-  const int splitm = lora_wgrad_splitm(K);
+  const int splitm = lora_wgrad_splitm(K, r, M);
   dim3 grid(DTX_CDIV(K, 2048), splitm);
   for (int j0 = 0; j0 < r; j0 += 16) {
     int rch = r - j0;
