@@ -47,6 +47,15 @@ __device__ __forceinline__ void store_bf16x8(unsigned short* p,
   *reinterpret_cast<short8v*>(p) = v;
 }
 
+// round two floats to bf16 (RNE) with one v_cvt_pk_bf16_f32 — ~3x
+// cheaper than the scalar f2bf path (no branch, no integer rounding)
+__device__ __forceinline__ unsigned dtx_cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+               : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
 // ---- wave/block reductions (64-wide wave) ----
 __device__ __forceinline__ float wave_reduce_sum(float x) {
 #pragma unroll

@@ -72,10 +72,15 @@ void lora_contract_kernel(const unsigned short* __restrict__ X,
           if (MASKED && mp) {
             short8v mf = *reinterpret_cast<const short8v*>(
                 mp + kc + hi * 8);
+            unsigned* xu = reinterpret_cast<unsigned*>(&xf);
 #pragma unroll
-            for (int e = 0; e < 8; ++e)
-              xf[e] = (short)f2bf(bf2f((unsigned short)xf[e]) *
-                                  bf2f((unsigned short)mf[e]));
+            for (int e = 0; e < 8; e += 2) {
+              float p0 = bf2f((unsigned short)xf[e]) *
+                         bf2f((unsigned short)mf[e]);
+              float p1 = bf2f((unsigned short)xf[e + 1]) *
+                         bf2f((unsigned short)mf[e + 1]);
+              xu[e / 2] = dtx_cvt_pk_bf16(p0, p1);
+            }
           }
           short8v wf = *reinterpret_cast<const short8v*>(
               &wlds[l31][kc + hi * 8]);
@@ -196,8 +201,11 @@ void lora_wgrad_kernel(const float* __restrict__ T,
       float mv[8];
       load_bf16x8(Mk + m * K + col, mv);
 #pragma unroll
-      for (int i = 0; i < 8; ++i)
-        xv[i] = bf2f(f2bf(xv[i] * mv[i]));   // match bf16 x*mask numerics
+      for (int i = 0; i < 8; i += 2) {       // match bf16 x*mask numerics
+        unsigned pk = dtx_cvt_pk_bf16(xv[i] * mv[i], xv[i + 1] * mv[i + 1]);
+        xv[i] = bf2f((unsigned short)(pk & 0xffffu));
+        xv[i + 1] = bf2f((unsigned short)(pk >> 16));
+      }
     }
     const float* tr = T + m * r + j0;
 #pragma unroll

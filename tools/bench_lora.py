@@ -3,8 +3,12 @@
 Run on the GPU box: python tools/bench_lora.py
 """
 
+import os
+import sys
+
 import torch
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from datatunerx_amd.ops import lora_contract, lora_expand_add, lora_wgrad
 
 
