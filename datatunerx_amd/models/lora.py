@@ -45,36 +45,39 @@ class FrozenLinear(nn.Module):
 
 class LoRAFunctionWithDropout(torch.autograd.Function):
     """Fused LoRA linear with PEFT-style input dropout on the low-rank
-    path. The pre-scaled mask (0 or 1/keep, bf16) is applied INSIDE the
-    contract/wgrad/expand kernels — no x*mask materialization, no
-    separate addcmul in backward.
+    path. The dropout mask is applied INSIDE the contract/wgrad/expand
+    kernels; with `seed`-mode (r <= 16, the default ranks) it is
+    RE-GENERATED from a counter-based RNG in each kernel and never
+    touches HBM at all — no bernoulli launch, no mask reads. `mask` is
+    the materialized-tensor fallback (r > 16 / CPU-provided masks).
     """
 
     @staticmethod
-    def forward(ctx, x, w, a, b, scale, mask):
+    def forward(ctx, x, w, a, b, scale, mask, seed, keep):
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
         y = F.linear(x2, w)
-        t = lora_contract(x2, a, mask)           # (x o mask) @ A^T
+        t = lora_contract(x2, a, mask, seed, keep)   # (x o mask) @ A^T
         lora_expand_add(y, t, b, scale)
         ctx.save_for_backward(x2, w, a, b, t, mask)
         ctx.scale, ctx.xshape = scale, xs
+        ctx.seed, ctx.keep = seed, keep
         return y.reshape(*xs[:-1], w.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
         from ..ops import lora_wgrad
         x2, w, a, b, t, mask = ctx.saved_tensors
-        s = ctx.scale
+        s, seed, keep = ctx.scale, ctx.seed, ctx.keep
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = dy2 @ w
         dt = lora_contract(dy2, b.t().contiguous())       # [M,r] = dy @ B
-        da = lora_wgrad(dt, x2, s, mask)         # dt^T @ (x o mask)
+        da = lora_wgrad(dt, x2, s, mask, seed, keep)  # dt^T @ (x o mask)
         db = lora_wgrad(t, dy2, s).t().contiguous()
         # dx += mask o (s * dt @ A)
-        lora_expand_add(dx, dt, a.t().contiguous(), s, mask)
+        lora_expand_add(dx, dt, a.t().contiguous(), s, mask, seed, keep)
         return (dx.reshape(ctx.xshape), None, da.to(a.dtype),
-                db.to(b.dtype), None, None)
+                db.to(b.dtype), None, None, None, None)
 
 
 class LoRALinearModule(nn.Module):
@@ -97,19 +100,24 @@ class LoRALinearModule(nn.Module):
         nn.init.kaiming_uniform_(self.lora_A, a=math.sqrt(5))
 
     def forward(self, x):
-        mask = None
-        if self.training and self.dropout > 0.0:
-            keep = 1.0 - self.dropout
-            # bf16 bernoulli directly (no fp32 rand materialization)
-            mask = torch.empty(
-                (x.numel() // x.shape[-1], x.shape[-1]),
-                device=x.device, dtype=x.dtype)
-            mask.bernoulli_(keep).mul_(1.0 / keep)
-        if mask is None:
+        if not (self.training and self.dropout > 0.0):
             return _FusedLoRAFn.apply(x, self.weight, self.lora_A,
                                       self.lora_B, self.scale)
+        keep = 1.0 - self.dropout
+        # draw the per-call RNG seed from torch's generator so dropout
+        # stays reproducible under torch.manual_seed
+        seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        if self.r <= 16 and self.in_features % 8 == 0:
+            # fused counter-based RNG: mask never materialized
+            return LoRAFunctionWithDropout.apply(
+                x, self.weight, self.lora_A, self.lora_B, self.scale,
+                None, seed, keep)
+        from ..ops import dropout_mask
+        mask = dropout_mask(x.numel() // x.shape[-1], x.shape[-1], seed,
+                            keep, x).to(x.dtype)
         return LoRAFunctionWithDropout.apply(x, self.weight, self.lora_A,
-                                             self.lora_B, self.scale, mask)
+                                             self.lora_B, self.scale,
+                                             mask, 0, 1.0)
 
     def merged_weight(self):
         """W + s·B@A — used by the serving engine (no adapter overhead)."""

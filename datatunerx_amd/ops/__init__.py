@@ -143,25 +143,40 @@ def gemv(x, w):
 
 
 # -------------------------------------------------------------- LoRA
-# `mask` (optional, same shape/dtype as x / y) fuses the PEFT input
-# dropout into the kernels.
-def lora_contract(x, w, mask=None):
+# Dropout enters the kernels one of two ways:
+#   - `mask`: a materialized bf16 mask tensor (same shape as x / y)
+#   - `seed` + `keep` < 1: counter-based RNG computed INSIDE the
+#     kernels (splitmix64 on the element offset) — the mask is never
+#     written to or read from HBM. The same (seed, offset) yields the
+#     same bits in contract/wgrad/expand, so fwd and bwd agree.
+def lora_contract(x, w, mask=None, seed: int = 0, keep: float = 1.0):
     if _gpu(x):
-        return _EXT.lora_contract(x, w, mask)
-    return ref.lora_contract(x, w, mask)
+        return _EXT.lora_contract(x, w, mask, seed, keep)
+    return ref.lora_contract(x, w, mask, seed, keep)
 
 
-def lora_expand_add(y, t, w, scale: float, mask=None):
+def lora_expand_add(y, t, w, scale: float, mask=None, seed: int = 0,
+                    keep: float = 1.0):
     if _gpu(y):
-        _EXT.lora_expand_add(y, t, w, scale, mask)
+        _EXT.lora_expand_add(y, t, w, scale, mask, seed, keep)
         return y
-    return ref.lora_expand_add(y, t, w, scale, mask)
+    return ref.lora_expand_add(y, t, w, scale, mask, seed, keep)
 
 
-def lora_wgrad(t, x, scale: float = 1.0, mask=None):
+def lora_wgrad(t, x, scale: float = 1.0, mask=None, seed: int = 0,
+               keep: float = 1.0):
     if _gpu(x):
-        return _EXT.lora_wgrad(t, x, scale, mask)
-    return ref.lora_wgrad(t, x, scale, mask)
+        return _EXT.lora_wgrad(t, x, scale, mask, seed, keep)
+    return ref.lora_wgrad(t, x, scale, mask, seed, keep)
+
+
+def dropout_mask(M: int, K: int, seed: int, keep: float, like):
+    """Materialize the RNG mask (bit-identical to the fused kernels'
+    in-kernel bits) — tests and the r>16 fallback path."""
+    if like.is_cuda and _EXT is not None:
+        return _EXT.dropout_mask(M, K, seed, keep, like)
+    return ref.dropout_mask(M, K, seed, keep, device=like.device,
+                            dtype=torch.bfloat16)
 
 
 # ------------------------------------------------------------- AdamW

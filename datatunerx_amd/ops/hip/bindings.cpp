@@ -23,13 +23,18 @@ void launch_adamw(void*, float*, const float*, float*, float*, long, float,
                   float, float, float, float, int, hipStream_t);
 void launch_l2_norm(const float*, float*, float*, long, hipStream_t);
 int lora_contract_ksplit(int K);
+void launch_dropout_mask(void*, long, unsigned long long, float,
+                         hipStream_t);
 void launch_lora_contract(const void*, const void*, const void*, float*,
-                          float*, long, int, int, hipStream_t);
+                          float*, long, int, int, unsigned long long,
+                          float, hipStream_t);
 void launch_lora_expand_add(void*, const float*, const void*, const void*,
-                            long, int, int, float, hipStream_t);
+                            long, int, int, float, unsigned long long,
+                            float, hipStream_t);
 int lora_wgrad_splitm(int K, int r, long M);
 void launch_lora_wgrad(const float*, const void*, const void*, float*,
-                       float*, long, int, int, float, hipStream_t);
+                       float*, long, int, int, float, unsigned long long,
+                       float, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, int, float, int, hipStream_t);
 void launch_transpose_sd(const void*, void*, int, int, int, int,
@@ -183,7 +188,8 @@ static const void* opt_mask(const c10::optional<torch::Tensor>& m,
 }
 
 torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w,
-                            c10::optional<torch::Tensor> mask) {
+                            c10::optional<torch::Tensor> mask,
+                            int64_t seed, double keep) {
   check_bf16_contig(x, "x");
   check_bf16_contig(w, "w");
   const int K = (int)x.size(-1);
@@ -202,12 +208,15 @@ torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w,
   }
   launch_lora_contract(x.data_ptr(), w.data_ptr(),
                        opt_mask(mask, x.numel(), "mask"), part_ptr,
-                       t.data_ptr<float>(), M, K, r, cur_stream());
+                       t.data_ptr<float>(), M, K, r,
+                       (unsigned long long)seed, (float)keep,
+                       cur_stream());
   return t;
 }
 
 void lora_expand_add(torch::Tensor y, torch::Tensor t, torch::Tensor w,
-                     double scale, c10::optional<torch::Tensor> mask) {
+                     double scale, c10::optional<torch::Tensor> mask,
+                     int64_t seed, double keep) {
   check_bf16_contig(y, "y");
   check_bf16_contig(w, "w");
   const int N = (int)y.size(-1);
@@ -216,13 +225,18 @@ void lora_expand_add(torch::Tensor y, torch::Tensor t, torch::Tensor w,
   TORCH_CHECK(w.size(0) == N, "w [N,r] mismatch");
   TORCH_CHECK(t.scalar_type() == torch::kFloat && t.is_contiguous());
   auto wt = w.t().contiguous();          // [r,N] for coalesced rows
+  TORCH_CHECK(keep >= 1.0 || (r <= 16 && N % 8 == 0 && M >= 64),
+              "RNG dropout needs the r<=16 fast path; materialize the "
+              "mask for this shape");
   launch_lora_expand_add(y.data_ptr(), t.data_ptr<float>(), wt.data_ptr(),
                          opt_mask(mask, y.numel(), "mask"), M, N, r,
-                         (float)scale, cur_stream());
+                         (float)scale, (unsigned long long)seed,
+                         (float)keep, cur_stream());
 }
 
 torch::Tensor lora_wgrad(torch::Tensor t, torch::Tensor x, double scale,
-                         c10::optional<torch::Tensor> mask) {
+                         c10::optional<torch::Tensor> mask,
+                         int64_t seed, double keep) {
   check_bf16_contig(x, "x");
   TORCH_CHECK(t.scalar_type() == torch::kFloat && t.is_contiguous());
   const int K = (int)x.size(-1);
@@ -234,8 +248,20 @@ torch::Tensor lora_wgrad(torch::Tensor t, torch::Tensor x, double scale,
   launch_lora_wgrad(t.data_ptr<float>(), x.data_ptr(),
                     opt_mask(mask, x.numel(), "mask"),
                     part.data_ptr<float>(), out.data_ptr<float>(), M, K, r,
-                    (float)scale, cur_stream());
+                    (float)scale, (unsigned long long)seed, (float)keep,
+                    cur_stream());
   return out;
+}
+
+// Materialize the RNG dropout mask (bit-identical to what the fused
+// MODE==2 kernels consume) — used by tests and the r>16 fallback.
+torch::Tensor dropout_mask(int64_t M, int64_t K, int64_t seed, double keep,
+                           torch::Tensor like) {
+  TORCH_CHECK((M * K) % 8 == 0, "M*K % 8 == 0");
+  auto m = torch::empty({M, K}, like.options().dtype(torch::kBFloat16));
+  launch_dropout_mask(m.data_ptr(), M * K, (unsigned long long)seed,
+                      (float)keep, cur_stream());
+  return m;
 }
 
 // ---------------------------------------------------------------- AdamW
@@ -405,11 +431,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("xent_fwd", &xent_fwd);
   m.def("xent_bwd", &xent_bwd);
   m.def("lora_contract", &lora_contract, py::arg("x"), py::arg("w"),
-        py::arg("mask") = py::none());
+        py::arg("mask") = py::none(), py::arg("seed") = 0,
+        py::arg("keep") = 1.0);
   m.def("lora_expand_add", &lora_expand_add, py::arg("y"), py::arg("t"),
-        py::arg("w"), py::arg("scale"), py::arg("mask") = py::none());
+        py::arg("w"), py::arg("scale"), py::arg("mask") = py::none(),
+        py::arg("seed") = 0, py::arg("keep") = 1.0);
   m.def("lora_wgrad", &lora_wgrad, py::arg("t"), py::arg("x"),
-        py::arg("scale"), py::arg("mask") = py::none());
+        py::arg("scale"), py::arg("mask") = py::none(),
+        py::arg("seed") = 0, py::arg("keep") = 1.0);
+  m.def("dropout_mask", &dropout_mask, py::arg("m"), py::arg("k"),
+        py::arg("seed"), py::arg("keep"), py::arg("like"));
   m.def("adamw", &adamw);
   m.def("l2_norm", &l2_norm);
   m.def("attn_fwd", &attn_fwd);

@@ -56,6 +56,33 @@ __device__ __forceinline__ unsigned dtx_cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
+// Counter-based dropout RNG (splitmix64 on the element-group index):
+// the SAME (seed, linear-offset) pair yields the same mask bits in
+// every kernel that consumes it (contract / wgrad / expand), so the
+// dropout mask is never materialized in HBM. One hash covers 4
+// elements (16 random bits each; keep-prob quantized to 1/65536).
+__device__ __forceinline__ void dtx_dropout4(unsigned long long seed,
+                                             unsigned long long group,
+                                             unsigned thr16, float inv_keep,
+                                             float mv[4]) {
+  unsigned long long z = seed + group * 0x9E3779B97F4A7C15ull;
+  z ^= z >> 30; z *= 0xBF58476D1CE4E5B9ull;
+  z ^= z >> 27; z *= 0x94D049BB133111EBull;
+  z ^= z >> 31;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    mv[i] = ((unsigned)(z >> (16 * i)) & 0xFFFFu) < thr16 ? inv_keep : 0.f;
+}
+
+// 8-element convenience over an 8-aligned linear offset
+__device__ __forceinline__ void dtx_dropout8(unsigned long long seed,
+                                             long off, unsigned thr16,
+                                             float inv_keep, float mv[8]) {
+  dtx_dropout4(seed, (unsigned long long)(off >> 2), thr16, inv_keep, mv);
+  dtx_dropout4(seed, (unsigned long long)(off >> 2) + 1, thr16, inv_keep,
+               mv + 4);
+}
+
 // ---- wave/block reductions (64-wide wave) ----
 __device__ __forceinline__ float wave_reduce_sum(float x) {
 #pragma unroll

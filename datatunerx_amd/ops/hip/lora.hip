@@ -24,13 +24,16 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 // (each row read once; 8 kc loads/lane = 128 contiguous bytes).
 // B-frag: lane reads the LDS copy of W at row l31 (zero rows >= r).
 // Split-K across blockIdx.y; fp32 partials reduced by reduce_partials.
-template <bool MASKED>
+// MODE: 0 = no mask, 1 = bf16 mask tensor, 2 = in-kernel RNG dropout
+template <int MODE>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_contract_kernel(const unsigned short* __restrict__ X,
                           const unsigned short* __restrict__ W,
                           const unsigned short* __restrict__ Mk,
                           float* __restrict__ part,
-                          long M, int K, int r, int kspan) {
+                          long M, int K, int r, int kspan,
+                          unsigned long long seed, unsigned thr16,
+                          float inv_keep) {
   __shared__ unsigned short wlds[32][1024 + 8];
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -64,12 +67,12 @@ void lora_contract_kernel(const unsigned short* __restrict__ X,
         const long rbase = xrow * K + kc0;
         const unsigned short* xp = xrow < M ? X + rbase : X;
         const unsigned short* mp =
-            (MASKED && xrow < M) ? Mk + rbase : nullptr;
+            (MODE == 1 && xrow < M) ? Mk + rbase : nullptr;
         for (int kc = 0; kc < kc_n; kc += 16) {
           short8v xf = xrow < M
               ? *reinterpret_cast<const short8v*>(xp + kc + hi * 8)
               : short8v{0, 0, 0, 0, 0, 0, 0, 0};
-          if (MASKED && mp) {
+          if (MODE == 1 && mp) {
             short8v mf = *reinterpret_cast<const short8v*>(
                 mp + kc + hi * 8);
             unsigned* xu = reinterpret_cast<unsigned*>(&xf);
@@ -81,6 +84,15 @@ void lora_contract_kernel(const unsigned short* __restrict__ X,
                          bf2f((unsigned short)mf[e + 1]);
               xu[e / 2] = dtx_cvt_pk_bf16(p0, p1);
             }
+          } else if (MODE == 2 && xrow < M) {
+            float mv[8];
+            dtx_dropout8(seed, rbase + kc + hi * 8, thr16, inv_keep, mv);
+            unsigned* xu = reinterpret_cast<unsigned*>(&xf);
+#pragma unroll
+            for (int e = 0; e < 8; e += 2)
+              xu[e / 2] = dtx_cvt_pk_bf16(
+                  bf2f((unsigned short)xf[e]) * mv[e],
+                  bf2f((unsigned short)xf[e + 1]) * mv[e + 1]);
           }
           short8v wf = *reinterpret_cast<const short8v*>(
               &wlds[l31][kc + hi * 8]);
@@ -177,13 +189,15 @@ void lora_expand_add_kernel(unsigned short* __restrict__ Y,
 // part[ms][j][K] = s * sum_{m in split ms} T[m,j] * X[m,k]
 // (k-span 2048/block keeps X re-reads at ceil(K/2048); the partials are
 // streamed back by the vectorized reduce_partials kernel)
-template <int RCH, bool MASKED>
+template <int RCH, int MODE>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_wgrad_kernel(const float* __restrict__ T,
                        const unsigned short* __restrict__ X,
                        const unsigned short* __restrict__ Mk,
                        float* __restrict__ part,
-                       long M, int K, int r, int j0, int splitm, float s) {
+                       long M, int K, int r, int j0, int splitm, float s,
+                       unsigned long long seed, unsigned thr16,
+                       float inv_keep) {
   const int col = blockIdx.x * 2048 + threadIdx.x * 8;
   const int ms = blockIdx.y;
   if (col >= K) return;
@@ -197,9 +211,10 @@ void lora_wgrad_kernel(const float* __restrict__ T,
   for (long m = m_begin; m < m_end; ++m) {
     float xv[8];
     load_bf16x8(X + m * K + col, xv);
-    if (MASKED) {
+    if (MODE >= 1) {
       float mv[8];
-      load_bf16x8(Mk + m * K + col, mv);
+      if (MODE == 1) load_bf16x8(Mk + m * K + col, mv);
+      else dtx_dropout8(seed, m * K + col, thr16, inv_keep, mv);
 #pragma unroll
       for (int i = 0; i < 8; i += 2) {       // match bf16 x*mask numerics
         unsigned pk = dtx_cvt_pk_bf16(xv[i] * mv[i], xv[i + 1] * mv[i + 1]);
@@ -234,13 +249,15 @@ void lora_wgrad_kernel(const float* __restrict__ T,
 // traffic per row is T (lane-uniform 4B*r), Y in/out and the mask --
 // the v1 kernel re-read WT from L2 for every row (64 KB/row/wave) and
 // sat at ~2.8 TB/s; this one runs at the streaming roofline.
-template <int RCH, bool MASKED>
+template <int RCH, int MODE>
 __global__ __launch_bounds__(DTX_BLOCK)
 void lora_expand_add_kernel2(unsigned short* __restrict__ Y,
                              const float* __restrict__ T,
                              const unsigned short* __restrict__ WT,
                              const unsigned short* __restrict__ Mk,
-                             long M, int N, int r, float s, int rowsplit) {
+                             long M, int N, int r, float s, int rowsplit,
+                             unsigned long long seed, unsigned thr16,
+                             float inv_keep) {
   const int col = blockIdx.x * 2048 + threadIdx.x * 8;
   if (col >= N) return;                      // N % 8 == 0 (checked host)
   float w[RCH][8];
@@ -271,9 +288,10 @@ void lora_expand_add_kernel2(unsigned short* __restrict__ Y,
     unsigned short* yp = Y + m * N + col;
     float y[8];
     load_bf16x8(yp, y);
-    if (MASKED) {
+    if (MODE >= 1) {
       float mv[8];
-      load_bf16x8(Mk + m * N + col, mv);
+      if (MODE == 1) load_bf16x8(Mk + m * N + col, mv);
+      else dtx_dropout8(seed, m * N + col, thr16, inv_keep, mv);
 #pragma unroll
       for (int i = 0; i < 8; ++i) y[i] += mv[i] * acc[i];
     } else {
@@ -284,9 +302,46 @@ void lora_expand_add_kernel2(unsigned short* __restrict__ Y,
   }
 }
 
+// Standalone mask materializer (tests + the r>16 fallback path): the
+// EXACT bits the fused MODE==2 kernels consume, as a bf16 tensor.
+__global__ __launch_bounds__(DTX_BLOCK)
+void dropout_mask_kernel(unsigned short* __restrict__ Mk, long n8,
+                         unsigned long long seed, unsigned thr16,
+                         float inv_keep) {
+  long idx = (long)blockIdx.x * DTX_BLOCK + threadIdx.x;
+  long stride = (long)gridDim.x * DTX_BLOCK;
+  for (; idx < n8; idx += stride) {
+    float mv[8];
+    dtx_dropout8(seed, idx * 8, thr16, inv_keep, mv);
+    store_bf16x8(Mk + idx * 8, mv);
+  }
+}
+
+static void dropout_spec(float keep, unsigned* thr16, float* inv_keep) {
+  double t = (double)keep * 65536.0 + 0.5;
+  *thr16 = t >= 65536.0 ? 65536u : (unsigned)t;
+  // the mask VALUE is the bf16-rounded 1/keep (identical numerics to a
+  // materialized bf16 mask tensor)
+  union { float f; unsigned u; } x;
+  x.f = 1.0f / keep;
+  unsigned r = 0x7fffu + ((x.u >> 16) & 1u);
+  x.u = ((x.u + r) >> 16) << 16;
+  *inv_keep = x.f;
+}
+
 // ------------------------------------------------------------- launchers
 void launch_reduce_partials(const float* part, float* out, int P, long L,
                             hipStream_t s);
+
+void launch_dropout_mask(void* Mk, long n, unsigned long long seed,
+                         float keep, hipStream_t s) {
+  unsigned thr16; float ik;
+  dropout_spec(keep, &thr16, &ik);
+  long g = DTX_CDIV(n / 8, DTX_BLOCK);
+  int grid = (int)(g < 2048 ? (g < 1 ? 1 : g) : 2048);
+  dropout_mask_kernel<<<grid, DTX_BLOCK, 0, s>>>(
+      (unsigned short*)Mk, n / 8, seed, thr16, ik);
+}
 
 // split plan: spans are MULTIPLES OF 1024 elements so every staged
 // 16-byte load stays aligned (K=5120 used to produce a 1707-element
@@ -303,51 +358,60 @@ int lora_contract_ksplit(int K) {
 
 void launch_lora_contract(const void* X, const void* W, const void* Mk,
                           float* part, float* out, long M, int K, int r,
+                          unsigned long long seed, float keep,
                           hipStream_t s) {
   const int nsplit = lora_contract_ksplit(K);
   const int kspan = lora_contract_kspan(K);
   long gw = DTX_CDIV(M, 128);
   dim3 grid((int)(gw < 128 ? (gw < 1 ? 1 : gw) : 128), nsplit);
   float* dst = nsplit > 1 ? part : out;
-  if (Mk) {
-    lora_contract_kernel<true><<<grid, DTX_BLOCK, 0, s>>>(
-        (const unsigned short*)X, (const unsigned short*)W,
-        (const unsigned short*)Mk, dst, M, K, r, kspan);
-  } else {
-    lora_contract_kernel<false><<<grid, DTX_BLOCK, 0, s>>>(
-        (const unsigned short*)X, (const unsigned short*)W, nullptr, dst,
-        M, K, r, kspan);
-  }
+  unsigned thr16 = 0; float ik = 1.f;
+  if (!Mk && keep < 1.f) dropout_spec(keep, &thr16, &ik);
+#define CONTRACT(MODE)                                                    \
+  lora_contract_kernel<MODE><<<grid, DTX_BLOCK, 0, s>>>(                  \
+      (const unsigned short*)X, (const unsigned short*)W,                 \
+      (const unsigned short*)Mk, dst, M, K, r, kspan, seed, thr16, ik)
+  if (Mk) CONTRACT(1);
+  else if (keep < 1.f) CONTRACT(2);
+  else CONTRACT(0);
+#undef CONTRACT
   if (nsplit > 1)
     launch_reduce_partials(part, out, nsplit, M * r, s);
 }
 
 void launch_lora_expand_add(void* Y, const float* T, const void* WT,
                             const void* Mk, long M, int N, int r,
-                            float scale, hipStream_t s) {
+                            float scale, unsigned long long seed,
+                            float keep, hipStream_t s) {
+  const bool rng = !Mk && keep < 1.f;
+  unsigned thr16 = 0; float ik = 1.f;
+  if (rng) dropout_spec(keep, &thr16, &ik);
   if (r <= 16 && N % 8 == 0 && M >= 64) {
     const int gx = DTX_CDIV(N, 2048);
     int rowsplit = 1024 / gx;
     if (rowsplit > M) rowsplit = (int)M;
     if (rowsplit < 1) rowsplit = 1;
     dim3 grid(gx, rowsplit);
-#define EX2(RC)                                                           \
+#define EX2(RC, MODE)                                                     \
+    lora_expand_add_kernel2<RC, MODE><<<grid, DTX_BLOCK, 0, s>>>(         \
+        (unsigned short*)Y, T, (const unsigned short*)WT,                 \
+        (const unsigned short*)Mk, M, N, r, scale, rowsplit, seed,        \
+        thr16, ik)
+#define EX2M(RC)                                                          \
     do {                                                                  \
-      if (Mk)                                                             \
-        lora_expand_add_kernel2<RC, true><<<grid, DTX_BLOCK, 0, s>>>(     \
-            (unsigned short*)Y, T, (const unsigned short*)WT,             \
-            (const unsigned short*)Mk, M, N, r, scale, rowsplit);         \
-      else                                                                \
-        lora_expand_add_kernel2<RC, false><<<grid, DTX_BLOCK, 0, s>>>(    \
-            (unsigned short*)Y, T, (const unsigned short*)WT, nullptr,    \
-            M, N, r, scale, rowsplit);                                    \
+      if (Mk) EX2(RC, 1);                                                 \
+      else if (rng) EX2(RC, 2);                                           \
+      else EX2(RC, 0);                                                    \
     } while (0)
-    if (r <= 4) EX2(4);
-    else if (r <= 8) EX2(8);
-    else EX2(16);
+    if (r <= 4) EX2M(4);
+    else if (r <= 8) EX2M(8);
+    else EX2M(16);
+#undef EX2M
 #undef EX2
     return;
   }
+  // v1 fallback (r > 16 or tiny M) has no RNG mode: callers materialize
+  // the mask for those shapes (ops dispatch enforces this)
   long gw = DTX_CDIV(M, 4);
   int grid = (int)(gw < 2048 ? (gw < 1 ? 1 : gw) : 2048);
   if (Mk) {
@@ -375,26 +439,30 @@ int lora_wgrad_splitm(int K, int r, long M) {
 
 void launch_lora_wgrad(const float* T, const void* X, const void* Mk,
                        float* part, float* out, long M, int K, int r,
-                       float s, hipStream_t st) {
+                       float s, unsigned long long seed, float keep,
+                       hipStream_t st) {
   const int splitm = lora_wgrad_splitm(K, r, M);
   dim3 grid(DTX_CDIV(K, 2048), splitm);
+  const bool rng = !Mk && keep < 1.f;
+  unsigned thr16 = 0; float ik = 1.f;
+  if (rng) dropout_spec(keep, &thr16, &ik);
   for (int j0 = 0; j0 < r; j0 += 16) {
     int rch = r - j0;
+#define WG(RC, MODE)                                                      \
+    lora_wgrad_kernel<RC, MODE><<<grid, DTX_BLOCK, 0, st>>>(              \
+        T, (const unsigned short*)X, (const unsigned short*)Mk, part,     \
+        M, K, r, j0, splitm, s, seed, thr16, ik)
 #define CASE(RC)                                                          \
     do {                                                                  \
-      if (Mk)                                                             \
-        lora_wgrad_kernel<RC, true><<<grid, DTX_BLOCK, 0, st>>>(          \
-            T, (const unsigned short*)X, (const unsigned short*)Mk,       \
-            part, M, K, r, j0, splitm, s);                                \
-      else                                                                \
-        lora_wgrad_kernel<RC, false><<<grid, DTX_BLOCK, 0, st>>>(         \
-            T, (const unsigned short*)X, nullptr, part, M, K, r, j0,      \
-            splitm, s);                                                   \
+      if (Mk) WG(RC, 1);                                                  \
+      else if (rng) WG(RC, 2);                                            \
+      else WG(RC, 0);                                                     \
     } while (0)
     if (rch <= 4) CASE(4);
     else if (rch <= 8) CASE(8);
     else CASE(16);
 #undef CASE
+#undef WG
   }
   launch_reduce_partials(part, out, splitm, (long)r * K, st);
 }

@@ -205,27 +205,72 @@ def _masked(x, mask):
     return (x * mask).to(x.dtype)
 
 
-def lora_contract(x: torch.Tensor, w: torch.Tensor, mask=None):
+def dropout_mask(M: int, K: int, seed: int, keep: float,
+                 device=None, dtype=torch.bfloat16):
+    """Bit-exact CPU twin of the GPU counter-based dropout RNG
+    (splitmix64 over 4-element groups, 16 random bits/element,
+    threshold keep*65536, value = bf16-rounded 1/keep). Lets CPU
+    fallbacks and tests reproduce the fused kernels' mask exactly."""
+    assert (M * K) % 4 == 0
+    n_groups = (M * K) // 4
+
+    def _u(c):                       # unsigned 64-bit const -> torch int64
+        return torch.tensor(c - (1 << 64) if c >= (1 << 63) else c,
+                            dtype=torch.int64, device=device)
+
+    def _lshr(z, k):                 # logical shift right on int64
+        return (z >> k) & ((1 << (64 - k)) - 1)
+
+    g = torch.arange(n_groups, dtype=torch.int64, device=device)
+    z = _u(seed & ((1 << 64) - 1)) + g * _u(0x9E3779B97F4A7C15)
+    z = z ^ _lshr(z, 30)
+    z = z * _u(0xBF58476D1CE4E5B9)
+    z = z ^ _lshr(z, 27)
+    z = z * _u(0x94D049BB133111EB)
+    z = z ^ _lshr(z, 31)
+    thr16 = min(65536, int(keep * 65536.0 + 0.5))
+    bits = torch.stack([(z >> (16 * i)) & 0xFFFF for i in range(4)], dim=1)
+    keepers = bits < thr16
+    inv_keep = torch.tensor(1.0 / keep).to(torch.bfloat16).item()
+    return (keepers.reshape(M, K).to(dtype) * inv_keep).to(dtype)
+
+
+def _resolve_mask(mask, seed, keep, shape, device, dtype):
+    if mask is not None or keep >= 1.0:
+        return mask
+    return dropout_mask(shape[0], shape[1], seed, keep, device=device,
+                        dtype=dtype)
+
+
+def lora_contract(x: torch.Tensor, w: torch.Tensor, mask=None,
+                  seed: int = 0, keep: float = 1.0):
     """t[M,r] = (x o mask)[M,K] @ w[r,K]^T  (fp32 out). w is HF lora_A
     layout [r,K], or for the dgrad pass w = lora_B transposed."""
-    return _masked(x, mask).float() @ w.float().t()
+    x2 = x.reshape(-1, x.shape[-1])
+    mask = _resolve_mask(mask, seed, keep, x2.shape, x.device, x.dtype)
+    return _masked(x2, mask).float() @ w.float().t()
 
 
 def lora_expand_add(y: torch.Tensor, t: torch.Tensor, w: torch.Tensor,
-                    scale: float, mask=None):
+                    scale: float, mask=None, seed: int = 0,
+                    keep: float = 1.0):
     """y[M,N] += (mask o) scale * t[M,r] @ w[N,r]^T  (in-place on y).
     HF lora_B layout is [N,r]."""
+    y2 = y.reshape(-1, y.shape[-1])
+    mask = _resolve_mask(mask, seed, keep, y2.shape, y.device, y.dtype)
     d = scale * (t.float() @ w.float().t())
     if mask is not None:
         d = d * mask.float()
-    y.add_(d.to(y.dtype))
+    y2.add_(d.to(y.dtype))
     return y
 
 
 def lora_wgrad(t: torch.Tensor, x: torch.Tensor, scale: float = 1.0,
-               mask=None):
+               mask=None, seed: int = 0, keep: float = 1.0):
     """dW[r,K] = scale * t[M,r]^T @ (x o mask)[M,K]  (fp32)."""
-    return scale * (t.float().t() @ _masked(x, mask).float())
+    x2 = x.reshape(-1, x.shape[-1])
+    mask = _resolve_mask(mask, seed, keep, x2.shape, x.device, x.dtype)
+    return scale * (t.float().t() @ _masked(x2, mask).float())
 
 
 # ------------------------------------------------------------ fused AdamW

@@ -101,6 +101,39 @@ def test_lora_contract(M, K, r):
     assert_close(t.cpu(), t_ref, name="lora contract")
 
 
+def test_lora_rng_dropout():
+    """seed-mode (in-kernel counter-based RNG) must produce EXACTLY the
+    bits of the materialized dropout_mask tensor, in all three kernels,
+    and the GPU mask must match the CPU splitmix64 twin bit-for-bit."""
+    M, K, N, r = 512, 2048, 2048, 8
+    seed, keep = 12345, 0.9
+    x, a = mk(M, K, scale=0.3), mk(r, K, scale=0.3)
+    mask = ops.dropout_mask(M, K, seed, keep, x)
+    # GPU mask == CPU splitmix64 twin (bit-exact)
+    mask_cpu = ref.dropout_mask(M, K, seed, keep)
+    assert torch.equal(mask.cpu(), mask_cpu), "GPU/CPU RNG mismatch"
+    rate = (mask > 0).float().mean().item()
+    assert abs(rate - keep) < 0.02
+
+    t_rng = ops.lora_contract(x, a, seed=seed, keep=keep)
+    t_mask = ops.lora_contract(x, a, mask)
+    assert torch.equal(t_rng, t_mask), "contract rng != mask"
+
+    dt = torch.randn(M, r, device=DEV)
+    dw_rng = ops.lora_wgrad(dt, x, 0.5, seed=seed, keep=keep)
+    dw_mask = ops.lora_wgrad(dt, x, 0.5, mask)
+    assert torch.equal(dw_rng, dw_mask), "wgrad rng != mask"
+
+    y = mk(M, N)
+    y2 = y.clone()
+    b = mk(N, r, scale=0.3)
+    t2 = torch.randn(M, r, device=DEV)
+    ops.lora_expand_add(y, t2, b, 0.5, seed=seed, keep=keep)
+    mask_y = ops.dropout_mask(M, N, seed, keep, y2)
+    ops.lora_expand_add(y2, t2, b, 0.5, mask_y)
+    assert torch.equal(y, y2), "expand rng != mask"
+
+
 def _mk_mask(M, K, keep=0.9):
     m = torch.empty(M, K, device=DEV, dtype=torch.bfloat16)
     m.bernoulli_(keep).mul_(1.0 / keep)

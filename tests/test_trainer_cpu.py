@@ -220,3 +220,22 @@ def test_generation_metrics():
     assert bleu([9, 9, 9, 9], [1, 2, 3, 4]) < 0.01
     m = generation_metrics([[1, 2, 3]], [[1, 2, 4]])
     assert 0 < m["predict_rouge-l"] < 1
+
+
+def test_rng_dropout_cpu_consistency():
+    """The CPU splitmix64 dropout twin: deterministic per seed, correct
+    keep-rate, and the seed-mode reference ops equal the explicit-mask
+    reference ops (the invariant the fused GPU kernels also satisfy)."""
+    import torch
+
+    from datatunerx_amd.ops import reference as ref
+    M, K, r, seed, keep = 128, 256, 8, 77, 0.8
+    m1 = ref.dropout_mask(M, K, seed, keep)
+    m2 = ref.dropout_mask(M, K, seed, keep)
+    assert torch.equal(m1, m2)
+    assert abs((m1 > 0).float().mean().item() - keep) < 0.05
+    x = torch.randn(M, K, dtype=torch.bfloat16)
+    a = torch.randn(r, K, dtype=torch.bfloat16)
+    t_seed = ref.lora_contract(x, a, seed=seed, keep=keep)
+    t_mask = ref.lora_contract(x, a, mask=m1.to(x.dtype))
+    assert torch.equal(t_seed, t_mask)
