@@ -192,39 +192,45 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
         s1v = MFMA32(k1, qfrag[kc], s1v);
       }
 
-      // ---- scale (+ mask on boundary tiles)
+      // ---- mask on boundary tiles (raw scores; kscale is folded into
+      // the exp2 later: p = exp2(fma(s, kscale, -m)), saving the
+      // separate 32-mult scale pass on every tile — guide "exp2+fma
+      // fold")
       const bool interior =
           (kv0 + KVB <= Skv) &&
           (!causal || (kv0 + KVB - 1 <= qw + diag));
       const int qg = qw + l31;
-      if (interior) {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          s0v[r] *= kscale;
-          s1v[r] *= kscale;
-        }
-      } else {
+      if (!interior) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int kv_a = kv0 + crow(r, hi);
           const int kv_b = kv_a + 32;
           bool dead_a = (kv_a >= Skv) | (causal && (kv_a > qg + diag));
           bool dead_b = (kv_b >= Skv) | (causal && (kv_b > qg + diag));
-          s0v[r] = dead_a ? NEG_INF : s0v[r] * kscale;
-          s1v[r] = dead_b ? NEG_INF : s1v[r] * kscale;
+          if (dead_a) s0v[r] = NEG_INF;
+          if (dead_b) s1v[r] = NEG_INF;
         }
       }
 
       // ---- online softmax, lane-local over 32 regs + one cross-half
-      float mx = s0v[0];
+      // (max computed on RAW scores; kscale > 0 commutes with max)
+      float mxr = s0v[0];
 #pragma unroll
-      for (int r = 1; r < 16; ++r) mx = fmaxf(mx, s0v[r]);
+      for (int r = 1; r < 16; ++r) mxr = fmaxf(mxr, s0v[r]);
 #pragma unroll
-      for (int r = 0; r < 16; ++r) mx = fmaxf(mx, s1v[r]);
-      mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
-      const float m_new = fmaxf(m_run, mx);
+      for (int r = 0; r < 16; ++r) mxr = fmaxf(mxr, s1v[r]);
+      float mx = fmaxf(mxr, __shfl_xor(mxr, 32, 64));
+      mx = mx <= NEG_INF ? NEG_INF : mx * kscale;
+      // defer-max (guide ladder): if no lane's tile max exceeds the
+      // running max by more than 8 (exp2 domain -> P bounded by 2^8),
+      // keep m_run and SKIP the O-rescale (16 bpermutes + 64 mults).
+      // First live tile has m_run = NEG_INF so defer is never taken
+      // there; NaN (dead vs dead) compares false -> normal path.
+      const bool defer = __all(mx - m_run <= 8.f);
+      const float m_new = defer ? m_run : fmaxf(m_run, mx);
       const float alpha =
-          (m_new == NEG_INF) ? 1.f : __builtin_exp2f(m_run - m_new);
+          (defer || m_new == NEG_INF) ? 1.f
+                                      : __builtin_exp2f(m_run - m_new);
       float rowsum = 0.f;
       if (m_new == NEG_INF) {
 #pragma unroll
@@ -232,8 +238,8 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          s0v[r] = __builtin_exp2f(s0v[r] - m_new);
-          s1v[r] = __builtin_exp2f(s1v[r] - m_new);
+          s0v[r] = __builtin_exp2f(fmaf(s0v[r], kscale, -m_new));
+          s1v[r] = __builtin_exp2f(fmaf(s1v[r], kscale, -m_new));
           rowsum += s0v[r] + s1v[r];
         }
       }
@@ -267,7 +273,7 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       }
 
       // ---- rescale O rows by alpha[q-row] (one bpermute per row)
-      if (st2 > 0 || half > 0) {
+      if (!defer && (st2 > 0 || half > 0)) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const float a_r = __shfl(alpha, crow(r, hi), 64);
