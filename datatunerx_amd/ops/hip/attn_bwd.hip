@@ -39,6 +39,32 @@ __device__ __forceinline__ int bw_crow(int r, int hi) {
   return (r & 3) + 8 * (r >> 2) + 4 * hi;
 }
 
+// MFMA B-fragment gathered from a ROW-major [k][n] LDS image with
+// ds_read_b64_tr_b16 (gfx950 hardware 4x4 transpose read) — replaces
+// the separately-staged transposed image. Verified mapping
+// (tools/probe_tr.cpp): with 16-lane-group addresses
+// R_i = &img[k0 + (i>>2)][n0 + 4*(i&3)], lane l = 4a+c receives
+// component j = img[k0 + j][n0 + (l&15)] — exactly B[k][n=l&31] when
+// n0 = 32-col base + 16*((l>>4)&1). Two reads cover the lane's 8 ks.
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+#define DTX_AS3 __attribute__((address_space(3)))
+// img must be an address_space(3) pointer so all address math stays in
+// 32-bit LDS offsets (a generic pointer here costs 64-bit address
+// arithmetic per read and ~+120 VGPRs in the dkdv kernel).
+template <int STRIDE>
+__device__ __forceinline__ short8v tr_bfrag(
+    const DTX_AS3 unsigned short* img, int k0, int n0, int lane) {
+  const int row = k0 + ((lane >> 2) & 3);
+  const int col = n0 + ((lane >> 4) & 1) * 16 + (lane & 3) * 4;
+  const DTX_AS3 unsigned short* p = img + row * STRIDE + col;
+  union { bf16x4 v[2]; short8v s; } u;
+  u.v[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (DTX_AS3 bf16x4*)p);
+  u.v[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (DTX_AS3 bf16x4*)(p + 4 * STRIDE));
+  return u.s;
+}
+
 // C-layout f32x16 (rows R on regs, cols on lanes) -> two bf16 A/B
 // fragments with k = R (frag0: R 0..15, frag1: R 16..31).
 __device__ __forceinline__ void conv_c_to_frag(const f32x16& p,
@@ -112,18 +138,17 @@ template <int D>
 struct DkdvLds {
   unsigned short Qr[64][D + 8];     // staged q rows (two 32-row tiles)
   unsigned short dOr[64][D + 8];
-  unsigned short QT[D][72];         // staged q columns (from QT_g)
-  unsigned short dOT[D][72];
+  // no transposed images: dV/dK B-fragments come from these row-major
+  // tiles via ds_read_b64_tr_b16 (tr_bfrag) — halves the LDS footprint
+  // (4 blocks/CU co-resident) and the staged global traffic
 };
 
 template <int D>
 __global__ __launch_bounds__(256, 1)
 void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
-                           const unsigned short* __restrict__ QTg,
                            const unsigned short* __restrict__ Kp,
                            const unsigned short* __restrict__ Vp,
                            const unsigned short* __restrict__ dO,
-                           const unsigned short* __restrict__ dOTg,
                            const float* __restrict__ lse_in,
                            const float* __restrict__ delta_in,
                            unsigned short* __restrict__ dK,
@@ -179,7 +204,6 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
   for (int gi = 0; gi < rep; ++gi) {
     const int hq = hkv * rep + gi;
     const long qbase = (long)b * S * qrowstr + (long)hq * D;
-    const long qtbase = ((long)b * Hq + hq) * (long)D * S;
     const long lbase = ((long)b * Hq + hq) * S;
 
     // q rows that can see this block's kv rows: q + diag >= kv0b
@@ -188,8 +212,7 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
     // T14 async-stage split: stage qs+1's global loads are issued while
     // stage qs computes; the LDS write happens after the barrier.
     constexpr int RIT = (64 * (D / 8)) / 256;   // row-staging iters
-    constexpr int TIT = (D * 4) / 256;          // transposed-staging iters
-    short8v stg[RIT * 2 + TIT * 4];
+    short8v stg[RIT * 2];
     auto issue_stage = [&](int qs) {
 #pragma unroll
       for (int it = 0; it < RIT; ++it) {
@@ -206,36 +229,6 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         stg[it * 2] = q8;
         stg[it * 2 + 1] = d8;
       }
-#pragma unroll
-      for (int it = 0; it < TIT; ++it) {
-        const int idx = threadIdx.x + it * 256;
-        const int dd = idx / 4, g = idx % 4;     // 4 groups of 16 q-cols
-        const int s0 = qs * 64 + g * 16;
-        short8v q8a = {0, 0, 0, 0, 0, 0, 0, 0};
-        short8v d8a = {0, 0, 0, 0, 0, 0, 0, 0};
-        short8v q8b = {0, 0, 0, 0, 0, 0, 0, 0};
-        short8v d8b = {0, 0, 0, 0, 0, 0, 0, 0};
-        const long tb = qtbase + (long)dd * S;
-        if (s0 + 16 <= S) {
-          q8a = *reinterpret_cast<const short8v*>(QTg + tb + s0);
-          d8a = *reinterpret_cast<const short8v*>(dOTg + tb + s0);
-          q8b = *reinterpret_cast<const short8v*>(QTg + tb + s0 + 8);
-          d8b = *reinterpret_cast<const short8v*>(dOTg + tb + s0 + 8);
-        } else if (s0 < S) {
-#pragma unroll
-          for (int j = 0; j < 16; ++j)
-            if (s0 + j < S) {
-              short qv = (short)QTg[tb + s0 + j];
-              short dv = (short)dOTg[tb + s0 + j];
-              if (j < 8) { q8a[j] = qv; d8a[j] = dv; }
-              else { q8b[j - 8] = qv; d8b[j - 8] = dv; }
-            }
-        }
-        stg[RIT * 2 + it * 4] = q8a;
-        stg[RIT * 2 + it * 4 + 1] = q8b;
-        stg[RIT * 2 + it * 4 + 2] = d8a;
-        stg[RIT * 2 + it * 4 + 3] = d8b;
-      }
     };
     auto write_stage = [&]() {
 #pragma unroll
@@ -245,19 +238,6 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = stg[it * 2];
         *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) =
             stg[it * 2 + 1];
-      }
-#pragma unroll
-      for (int it = 0; it < TIT; ++it) {
-        const int idx = threadIdx.x + it * 256;
-        const int dd = idx / 4, g = idx % 4;
-        *reinterpret_cast<short8v*>(&lds.QT[dd][g * 16]) =
-            stg[RIT * 2 + it * 4];
-        *reinterpret_cast<short8v*>(&lds.QT[dd][g * 16 + 8]) =
-            stg[RIT * 2 + it * 4 + 1];
-        *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 16]) =
-            stg[RIT * 2 + it * 4 + 2];
-        *reinterpret_cast<short8v*>(&lds.dOT[dd][g * 16 + 8]) =
-            stg[RIT * 2 + it * 4 + 3];
       }
     };
     if (qs_lo < qs_hi) issue_stage(qs_lo);
@@ -325,18 +305,23 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
       conv_c_to_frag(dpv, df0, df1);
 #pragma unroll
       for (int c = 0; c < ND32; ++c) {
-        short8v dot0 = *reinterpret_cast<const short8v*>(
-            &lds.dOT[c * 32 + l31][qoff + hi * 8]);
-        short8v dot1 = *reinterpret_cast<const short8v*>(
-            &lds.dOT[c * 32 + l31][qoff + 16 + hi * 8]);
+        // B[k=q][n=d] straight from the row-major tiles (tr-read)
+        const DTX_AS3 unsigned short* dor3 =
+            (const DTX_AS3 unsigned short*)&lds.dOr[0][0];
+        const DTX_AS3 unsigned short* qr3 =
+            (const DTX_AS3 unsigned short*)&lds.Qr[0][0];
+        short8v dot0 = tr_bfrag<D + 8>(dor3, qoff + hi * 8, c * 32, lane);
+        short8v dot1 = tr_bfrag<D + 8>(dor3, qoff + 16 + hi * 8, c * 32,
+                                       lane);
         dv_acc[c] = MFMA32(pf0, dot0, dv_acc[c]);
         dv_acc[c] = MFMA32(pf1, dot1, dv_acc[c]);
-        short8v qt0 = *reinterpret_cast<const short8v*>(
-            &lds.QT[c * 32 + l31][qoff + hi * 8]);
-        short8v qt1 = *reinterpret_cast<const short8v*>(
-            &lds.QT[c * 32 + l31][qoff + 16 + hi * 8]);
+        __builtin_amdgcn_sched_barrier(0);    // keep frag live ranges
+        short8v qt0 = tr_bfrag<D + 8>(qr3, qoff + hi * 8, c * 32, lane);
+        short8v qt1 = tr_bfrag<D + 8>(qr3, qoff + 16 + hi * 8, c * 32,
+                                      lane);
         dk_acc[c] = MFMA32(df0, qt0, dk_acc[c]);
         dk_acc[c] = MFMA32(df1, qt1, dk_acc[c]);
+        __builtin_amdgcn_sched_barrier(0);    // short (no cross-c hoist)
       }
       }  // live_tile
       }  // qh
@@ -582,24 +567,22 @@ void launch_attn_delta(const void* dO, const void* O, float* delta,
   }
 }
 
-void launch_attn_bwd_dkdv(const void* q, const void* qt, const void* k,
-                          const void* v, const void* dO, const void* dOT,
-                          const float* lse, const float* delta, void* dk,
-                          void* dv, int B, int Hq, int Hkv, int S, int Skv,
-                          int D, float scale, int causal, hipStream_t st) {
+void launch_attn_bwd_dkdv(const void* q, const void* k, const void* v,
+                          const void* dO, const float* lse,
+                          const float* delta, void* dk, void* dv, int B,
+                          int Hq, int Hkv, int S, int Skv, int D,
+                          float scale, int causal, hipStream_t st) {
   dim3 grid(DTX_CDIV(Skv, 128), B * Hkv);
   if (D == 128) {
     attn_bwd_dkdv2_kernel<128><<<grid, 256, 0, st>>>(
-        (const unsigned short*)q, (const unsigned short*)qt,
-        (const unsigned short*)k, (const unsigned short*)v,
-        (const unsigned short*)dO, (const unsigned short*)dOT,
+        (const unsigned short*)q, (const unsigned short*)k,
+        (const unsigned short*)v, (const unsigned short*)dO,
         lse, delta, (unsigned short*)dk, (unsigned short*)dv,
         B, Hq, Hkv, S, Skv, scale, causal);
   } else if (D == 64) {
     attn_bwd_dkdv2_kernel<64><<<grid, 256, 0, st>>>(
-        (const unsigned short*)q, (const unsigned short*)qt,
-        (const unsigned short*)k, (const unsigned short*)v,
-        (const unsigned short*)dO, (const unsigned short*)dOT,
+        (const unsigned short*)q, (const unsigned short*)k,
+        (const unsigned short*)v, (const unsigned short*)dO,
         lse, delta, (unsigned short*)dk, (unsigned short*)dv,
         B, Hq, Hkv, S, Skv, scale, causal);
   }

@@ -48,9 +48,9 @@ void launch_attn_decode(const void*, const void*, const void*,
 void launch_attn_delta(const void*, const void*, float*, long, int, int,
                        int, hipStream_t);
 void launch_attn_bwd_dkdv(const void*, const void*, const void*,
-                          const void*, const void*, const void*,
-                          const float*, const float*, void*, void*, int,
-                          int, int, int, int, int, float, int, hipStream_t);
+                          const void*, const float*, const float*,
+                          void*, void*, int, int, int, int, int, int,
+                          float, int, hipStream_t);
 void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
                         const void*, const float*, const float*, void*, int,
                         int, int, int, int, int, float, int, hipStream_t);
@@ -394,22 +394,19 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto delta = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat));
   launch_attn_delta(dO_c.data_ptr(), o.data_ptr(), delta.data_ptr<float>(),
                     (long)B * S * Hq, Hq, S, D, cur_stream());
-  auto qt = torch::empty({B, Hq, D, S}, q.options());
-  auto dot = torch::empty({B, Hq, D, S}, q.options());
+  // dkdv gathers its dV/dK B-fragments from the row-major tiles with
+  // ds_read_b64_tr_b16 — no pre-transposed Q/dO copies needed (dq
+  // still stages K columns, so K keeps its transpose for now)
   auto kt = torch::empty({B, Hkv, D, Skv}, q.options());
-  launch_transpose_sd(q.data_ptr(), qt.data_ptr(), B, S, Hq, D,
-                      cur_stream());
-  launch_transpose_sd(dO_c.data_ptr(), dot.data_ptr(), B, S, Hq, D,
-                      cur_stream());
   launch_transpose_sd(k.data_ptr(), kt.data_ptr(), B, Skv, Hkv, D,
                       cur_stream());
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  launch_attn_bwd_dkdv(q.data_ptr(), qt.data_ptr(), k.data_ptr(),
-                       v.data_ptr(), dO_c.data_ptr(), dot.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       dk.data_ptr(), dv.data_ptr(), B, Hq, Hkv, S, Skv, D,
+  launch_attn_bwd_dkdv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                       dO_c.data_ptr(), lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), dk.data_ptr(),
+                       dv.data_ptr(), B, Hq, Hkv, S, Skv, D,
                        (float)scale, causal ? 1 : 0, cur_stream());
   launch_attn_bwd_dq(q.data_ptr(), k.data_ptr(), kt.data_ptr(),
                      v.data_ptr(), dO_c.data_ptr(), lse.data_ptr<float>(),
