@@ -349,14 +349,14 @@ template <int D>
 struct DqLds {
   unsigned short K[128][D + 8];
   unsigned short V[128][D + 8];
-  unsigned short KT[D][128 + 8];
+  // dQ's K B-fragments come from the row-major K tile via
+  // ds_read_b64_tr_b16 (tr_bfrag) — no transposed KT image
 };
 
 template <int D>
 __global__ __launch_bounds__(512, 1)
 void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
                          const unsigned short* __restrict__ Kp,
-                         const unsigned short* __restrict__ KTg,
                          const unsigned short* __restrict__ Vp,
                          const unsigned short* __restrict__ dO,
                          const float* __restrict__ lse_in,
@@ -383,7 +383,6 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
   const int qrowstr = Hq * D, krowstr = Hkv * D;
   const long qbase = (long)b * S * qrowstr + (long)hq * D;
   const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
-  const long ktbase = ((long)b * Hkv + hkv) * (long)D * Skv;
   const long lbase = ((long)b * Hq + hq) * S;
   const int diag = Skv - S;
   const float kscale = scale * LOG2E;
@@ -423,8 +422,7 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
 
   // T14 async-stage split (as in fwd/dkdv)
   constexpr int KIT = (128 * (D / 8)) / 512;
-  constexpr int TIT = (D * (128 / 8)) / 512;
-  short8v stg[KIT * 2 + TIT];
+  short8v stg[KIT * 2];
   auto issue_stage = [&](int st2) {
     const int kvs = st2 * 128;
 #pragma unroll
@@ -441,23 +439,6 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
       stg[it * 2] = k8;
       stg[it * 2 + 1] = v8;
     }
-#pragma unroll
-    for (int it = 0; it < TIT; ++it) {
-      const int idx = threadIdx.x + it * 512;
-      const int dd = idx / (128 / 8), g = idx % (128 / 8);
-      short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      const int s0 = kvs + g * 8;
-      if (s0 + 8 <= Skv) {
-        k8 = *reinterpret_cast<const short8v*>(KTg + ktbase +
-                                               (long)dd * Skv + s0);
-      } else if (s0 < Skv) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (s0 + j < Skv)
-            k8[j] = (short)KTg[ktbase + (long)dd * Skv + s0 + j];
-      }
-      stg[KIT * 2 + it] = k8;
-    }
   };
   auto write_stage = [&]() {
 #pragma unroll
@@ -466,12 +447,6 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
       const int row = idx / (D / 8), g = idx % (D / 8);
       *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it * 2];
       *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = stg[it * 2 + 1];
-    }
-#pragma unroll
-    for (int it = 0; it < TIT; ++it) {
-      const int idx = threadIdx.x + it * 512;
-      const int dd = idx / (128 / 8), g = idx % (128 / 8);
-      *reinterpret_cast<short8v*>(&lds.KT[dd][g * 8]) = stg[KIT * 2 + it];
     }
   };
   issue_stage(0);
@@ -523,14 +498,17 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
         }
         short8v f0, f1;
         conv_c_to_frag(dst, f0, f1);
+        const DTX_AS3 unsigned short* k3 =
+            (const DTX_AS3 unsigned short*)&lds.K[0][0];
 #pragma unroll
         for (int c = 0; c < ND32; ++c) {
-          short8v kt0 = *reinterpret_cast<const short8v*>(
-              &lds.KT[c * 32 + l31][koff + ss * 32 + hi * 8]);
-          short8v kt1 = *reinterpret_cast<const short8v*>(
-              &lds.KT[c * 32 + l31][koff + ss * 32 + 16 + hi * 8]);
+          short8v kt0 = tr_bfrag<D + 8>(k3, koff + ss * 32 + hi * 8,
+                                        c * 32, lane);
+          short8v kt1 = tr_bfrag<D + 8>(k3, koff + ss * 32 + 16 + hi * 8,
+                                        c * 32, lane);
           dq_acc[c] = MFMA32(f0, kt0, dq_acc[c]);
           dq_acc[c] = MFMA32(f1, kt1, dq_acc[c]);
+          __builtin_amdgcn_sched_barrier(0);  // no cross-c frag hoist
         }
       }
     }
@@ -588,8 +566,8 @@ void launch_attn_bwd_dkdv(const void* q, const void* k, const void* v,
   }
 }
 
-void launch_attn_bwd_dq(const void* q, const void* k, const void* kt,
-                        const void* v, const void* dO, const float* lse,
+void launch_attn_bwd_dq(const void* q, const void* k, const void* v,
+                        const void* dO, const float* lse,
                         const float* delta, void* dq, int B, int Hq,
                         int Hkv, int S, int Skv, int D, float scale,
                         int causal, hipStream_t st) {
@@ -597,13 +575,13 @@ void launch_attn_bwd_dq(const void* q, const void* k, const void* kt,
   if (D == 128) {
     attn_bwd_dq2_kernel<128><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)kt, (const unsigned short*)v,
+        (const unsigned short*)v,
         (const unsigned short*)dO, lse, delta, (unsigned short*)dq,
         B, Hq, Hkv, S, Skv, scale, causal);
   } else if (D == 64) {
     attn_bwd_dq2_kernel<64><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)kt, (const unsigned short*)v,
+        (const unsigned short*)v,
         (const unsigned short*)dO, lse, delta, (unsigned short*)dq,
         B, Hq, Hkv, S, Skv, scale, causal);
   }

@@ -47,17 +47,36 @@ __device__ __forceinline__ int crow(int r, int hi) {
   return (r & 3) + 8 * (r >> 2) + 4 * hi;
 }
 
+// MFMA B-fragment from a ROW-major [k][n] LDS image via the gfx950
+// hardware transpose read (mapping verified by tools/probe_tr.cpp;
+// see attn_bwd.hip tr_bfrag for the derivation).
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4f;
+#define DTX_AS3F __attribute__((address_space(3)))
+template <int STRIDE>
+__device__ __forceinline__ short8v fw_tr_bfrag(
+    const DTX_AS3F unsigned short* img, int k0, int n0, int lane) {
+  const int row = k0 + ((lane >> 2) & 3);
+  const int col = n0 + ((lane >> 4) & 1) * 16 + (lane & 3) * 4;
+  const DTX_AS3F unsigned short* p = img + row * STRIDE + col;
+  union { bf16x4f v[2]; short8v s; } u;
+  u.v[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (DTX_AS3F bf16x4f*)p);
+  u.v[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (DTX_AS3F bf16x4f*)(p + 4 * STRIDE));
+  return u.s;
+}
+
 template <int D>
 struct AttnFwdLds {
   unsigned short K[128][D + 8];
-  unsigned short VT[D][128 + 8];
+  unsigned short V[128][D + 8];   // row-major; PV B-frags via tr-read
 };
 
 template <int D>
 __global__ __launch_bounds__(512, 1)
 void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
                       const unsigned short* __restrict__ Kp,
-                      const unsigned short* __restrict__ VTp,
+                      const unsigned short* __restrict__ Vp,
                       unsigned short* __restrict__ O,
                       float* __restrict__ lse_out,
                       int B, int Hq, int Hkv, int S, int Skv,
@@ -82,7 +101,6 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
   const int krowstr = Hkv * D;
   const long qbase = (long)b * S * qrowstr + (long)hq * D;
   const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
-  const long vtbase = ((long)b * Hkv + hkv) * (long)D * Skv;
   const long lbase = ((long)b * Hq + hq) * S;
   const int diag = Skv - S;                // causal diagonal offset
 
@@ -118,8 +136,7 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
   // issued while the current stage computes; LDS writes after the
   // barrier.  KIT/VIT iterations cover the 128-row K and VT tiles.
   constexpr int KIT = (128 * D / 8) / 512;
-  constexpr int VIT = (D * 128 / 8) / 512;
-  short8v stg[KIT + VIT];
+  short8v stg[KIT * 2];
   auto issue_stage = [&](int st2) {
     const int kvs = st2 * 128;
 #pragma unroll
@@ -127,26 +144,14 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       const int idx = threadIdx.x + it * 512;
       const int row = idx / (D / 8), g = idx % (D / 8);
       short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (kvs + row < Skv)
-        k8 = *reinterpret_cast<const short8v*>(
-            Kp + kbase + (long)(kvs + row) * krowstr + g * 8);
-      stg[it] = k8;
-    }
-#pragma unroll
-    for (int it = 0; it < VIT; ++it) {
-      const int idx = threadIdx.x + it * 512;
-      const int dd = idx / (128 / 8), g = idx % (128 / 8);
       short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      const int s0 = kvs + g * 8;
-      const long vrow = vtbase + (long)dd * Skv;
-      if (s0 + 8 <= Skv) {
-        v8 = *reinterpret_cast<const short8v*>(VTp + vrow + s0);
-      } else if (s0 < Skv) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (s0 + j < Skv) v8[j] = (short)VTp[vrow + s0 + j];
+      if (kvs + row < Skv) {
+        const long off = kbase + (long)(kvs + row) * krowstr + g * 8;
+        k8 = *reinterpret_cast<const short8v*>(Kp + off);
+        v8 = *reinterpret_cast<const short8v*>(Vp + off);
       }
-      stg[KIT + it] = v8;
+      stg[it * 2] = k8;
+      stg[it * 2 + 1] = v8;
     }
   };
   auto write_stage = [&]() {
@@ -154,13 +159,8 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
     for (int it = 0; it < KIT; ++it) {
       const int idx = threadIdx.x + it * 512;
       const int row = idx / (D / 8), g = idx % (D / 8);
-      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it];
-    }
-#pragma unroll
-    for (int it = 0; it < VIT; ++it) {
-      const int idx = threadIdx.x + it * 512;
-      const int dd = idx / (128 / 8), g = idx % (128 / 8);
-      *reinterpret_cast<short8v*>(&lds.VT[dd][g * 8]) = stg[KIT + it];
+      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it * 2];
+      *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = stg[it * 2 + 1];
     }
   };
   issue_stage(0);
@@ -282,15 +282,18 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
         }
       }
 
-      // ---- PV: O[q][d] += P[q][kv] x V^T-frag
+      // ---- PV: O[q][d] += P[q][kv] x V-frag (tr-read, row-major V)
+      const DTX_AS3F unsigned short* v3 =
+          (const DTX_AS3F unsigned short*)&lds.V[0][0];
 #pragma unroll
       for (int c = 0; c < ND32; ++c) {
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
-          short8v vf = *reinterpret_cast<const short8v*>(
-              &lds.VT[c * 32 + l31][koff + ks * 16 + hi * 8]);
+          short8v vf = fw_tr_bfrag<D + 8>(v3, koff + ks * 16 + hi * 8,
+                                          c * 32, lane);
           o_acc[c] = MFMA32(pa[ks], vf, o_acc[c]);
         }
+        __builtin_amdgcn_sched_barrier(0);  // no cross-c frag hoist
       }
     }
     }  // half

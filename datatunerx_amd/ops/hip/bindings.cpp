@@ -51,9 +51,10 @@ void launch_attn_bwd_dkdv(const void*, const void*, const void*,
                           const void*, const float*, const float*,
                           void*, void*, int, int, int, int, int, int,
                           float, int, hipStream_t);
-void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
-                        const void*, const float*, const float*, void*, int,
-                        int, int, int, int, int, float, int, hipStream_t);
+void launch_attn_bwd_dq(const void*, const void*, const void*,
+                        const void*, const float*, const float*, void*,
+                        int, int, int, int, int, int, float, int,
+                        hipStream_t);
 
 namespace {
 
@@ -394,12 +395,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto delta = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat));
   launch_attn_delta(dO_c.data_ptr(), o.data_ptr(), delta.data_ptr<float>(),
                     (long)B * S * Hq, Hq, S, D, cur_stream());
-  // dkdv gathers its dV/dK B-fragments from the row-major tiles with
-  // ds_read_b64_tr_b16 — no pre-transposed Q/dO copies needed (dq
-  // still stages K columns, so K keeps its transpose for now)
-  auto kt = torch::empty({B, Hkv, D, Skv}, q.options());
-  launch_transpose_sd(k.data_ptr(), kt.data_ptr(), B, Skv, Hkv, D,
-                      cur_stream());
+  // dkdv and dq gather their B-fragments from the ROW-major LDS tiles
+  // with ds_read_b64_tr_b16 (tr_bfrag) — no pre-transposed Q/K/dO
+  // copies, no transpose_sd pre-passes in the backward at all
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
@@ -408,8 +406,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        delta.data_ptr<float>(), dk.data_ptr(),
                        dv.data_ptr(), B, Hq, Hkv, S, Skv, D,
                        (float)scale, causal ? 1 : 0, cur_stream());
-  launch_attn_bwd_dq(q.data_ptr(), k.data_ptr(), kt.data_ptr(),
-                     v.data_ptr(), dO_c.data_ptr(), lse.data_ptr<float>(),
+  launch_attn_bwd_dq(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                     dO_c.data_ptr(), lse.data_ptr<float>(),
                      delta.data_ptr<float>(), dq.data_ptr(), B, Hq, Hkv, S,
                      Skv, D, (float)scale, causal ? 1 : 0, cur_stream());
   return {dq, dk, dv};
