@@ -118,8 +118,9 @@ def attn_fwd(q, k, v, causal: bool = True, scale: float | None = None,
             # every cached position is visible, causal or not).
             # len_dev: device-resident cache length (hipGraph decode).
             return _EXT.attn_decode(q, k, v, scale, len_dev)
-        vt = _EXT.transpose_sd(v)        # [B,Hkv,D,Skv] for the PV tiles
-        return _EXT.attn_fwd(q, k, vt, causal, scale)
+        # V consumed row-major (PV B-frags via ds_read_b64_tr_b16):
+        # no V pre-transpose
+        return _EXT.attn_fwd(q, k, v.contiguous(), causal, scale)
     if len_dev is not None:
         k = k[:, :int(len_dev)]
         v = v[:, :int(len_dev)]

@@ -372,19 +372,19 @@ void launch_transpose_sd(const void* x, void* xt, int B, int S, int H,
       (const unsigned short*)x, (unsigned short*)xt, B, S, H, D);
 }
 
-void launch_attn_fwd(const void* q, const void* k, const void* vt, void* o,
+void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
                      float* lse, int B, int Hq, int Hkv, int S, int Skv,
                      int D, float scale, int causal, hipStream_t st) {
   dim3 grid(DTX_CDIV(S, 256), B * Hq);
   if (D == 128) {
     attn_fwd2_kernel<128><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)vt, (unsigned short*)o, lse,
+        (const unsigned short*)v, (unsigned short*)o, lse,
         B, Hq, Hkv, S, Skv, scale, causal);
   } else if (D == 64) {
     attn_fwd2_kernel<64><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
-        (const unsigned short*)vt, (unsigned short*)o, lse,
+        (const unsigned short*)v, (unsigned short*)o, lse,
         B, Hq, Hkv, S, Skv, scale, causal);
   }
 }

@@ -304,21 +304,21 @@ torch::Tensor transpose_sd(torch::Tensor x) {
 
 // q,k: [B,S,H,D] BSHD; vt: [B,Hkv,D,Skv] (pre-transposed V).
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor vt, bool causal,
+                                    torch::Tensor v, bool causal,
                                     double scale) {
   check_bf16_contig(q, "q");
   check_bf16_contig(k, "k");
-  check_bf16_contig(vt, "vt");
+  check_bf16_contig(v, "v");
   const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2),
             D = (int)q.size(3);
   const int Skv = (int)k.size(1), Hkv = (int)k.size(2);
-  TORCH_CHECK(vt.size(1) == Hkv && vt.size(2) == D && vt.size(3) == Skv,
-              "vt must be [B,Hkv,D,Skv]");
+  TORCH_CHECK(v.size(1) == Skv && v.size(2) == Hkv && v.size(3) == D,
+              "v must be BSHD like k");
   TORCH_CHECK(D == 64 || D == 128, "D must be 64 or 128");
   TORCH_CHECK(Hq % Hkv == 0, "GQA group");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat));
-  launch_attn_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(),
+  launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                   lse.data_ptr<float>(), B, Hq, Hkv, S, Skv, D,
                   (float)scale, causal ? 1 : 0, cur_stream());
   return {o, lse};
