@@ -313,13 +313,13 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         short8v dot0 = tr_bfrag<D + 8>(dor3, qoff + hi * 8, c * 32, lane);
         short8v dot1 = tr_bfrag<D + 8>(dor3, qoff + 16 + hi * 8, c * 32,
                                        lane);
-        dv_acc[c] = MFMA32(pf0, dot0, dv_acc[c]);
-        dv_acc[c] = MFMA32(pf1, dot1, dv_acc[c]);
-        __builtin_amdgcn_sched_barrier(0);    // keep frag live ranges
         short8v qt0 = tr_bfrag<D + 8>(qr3, qoff + hi * 8, c * 32, lane);
         short8v qt1 = tr_bfrag<D + 8>(qr3, qoff + 16 + hi * 8, c * 32,
                                       lane);
+        // interleave dv/dk accumulators: no back-to-back MFMA RAW
+        dv_acc[c] = MFMA32(pf0, dot0, dv_acc[c]);
         dk_acc[c] = MFMA32(df0, qt0, dk_acc[c]);
+        dv_acc[c] = MFMA32(pf1, dot1, dv_acc[c]);
         dk_acc[c] = MFMA32(df1, qt1, dk_acc[c]);
         __builtin_amdgcn_sched_barrier(0);    // short (no cross-c hoist)
       }
@@ -500,16 +500,22 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
         conv_c_to_frag(dst, f0, f1);
         const DTX_AS3 unsigned short* k3 =
             (const DTX_AS3 unsigned short*)&lds.K[0][0];
+        // two passes over c so consecutive MFMAs hit different
+        // accumulators (dq_acc[0..3]) instead of pairing on one
 #pragma unroll
         for (int c = 0; c < ND32; ++c) {
           short8v kt0 = tr_bfrag<D + 8>(k3, koff + ss * 32 + hi * 8,
                                         c * 32, lane);
+          dq_acc[c] = MFMA32(f0, kt0, dq_acc[c]);
+        }
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int c = 0; c < ND32; ++c) {
           short8v kt1 = tr_bfrag<D + 8>(k3, koff + ss * 32 + 16 + hi * 8,
                                         c * 32, lane);
-          dq_acc[c] = MFMA32(f0, kt0, dq_acc[c]);
           dq_acc[c] = MFMA32(f1, kt1, dq_acc[c]);
-          __builtin_amdgcn_sched_barrier(0);  // no cross-c frag hoist
         }
+        __builtin_amdgcn_sched_barrier(0);  // no cross-stage frag hoist
       }
     }
     }  // kh

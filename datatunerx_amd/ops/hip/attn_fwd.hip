@@ -282,18 +282,21 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
         }
       }
 
-      // ---- PV: O[q][d] += P[q][kv] x V-frag (tr-read, row-major V)
+      // ---- PV: O[q][d] += P[q][kv] x V-frag (tr-read, row-major V).
+      // ks OUTER so consecutive MFMAs hit DIFFERENT accumulators
+      // (o_acc[0..3]) — back-to-back MFMAs on one accumulator pay the
+      // full RAW latency (guide: SQ_WAIT_INST_ANY).
       const DTX_AS3F unsigned short* v3 =
           (const DTX_AS3F unsigned short*)&lds.V[0][0];
 #pragma unroll
-      for (int c = 0; c < ND32; ++c) {
+      for (int ks = 0; ks < 4; ++ks) {
 #pragma unroll
-        for (int ks = 0; ks < 4; ++ks) {
+        for (int c = 0; c < ND32; ++c) {
           short8v vf = fw_tr_bfrag<D + 8>(v3, koff + ks * 16 + hi * 8,
                                           c * 32, lane);
           o_acc[c] = MFMA32(pa[ks], vf, o_acc[c]);
         }
-        __builtin_amdgcn_sched_barrier(0);  // no cross-c frag hoist
+        __builtin_amdgcn_sched_barrier(0);  // no cross-ks frag hoist
       }
     }
     }  // half
