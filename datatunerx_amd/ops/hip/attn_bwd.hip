@@ -132,8 +132,13 @@ void attn_delta2_kernel(const unsigned short* __restrict__ dO,
 }
 
 // ------------------------------------------------------------- dk/dv
-// 4 waves (256 threads), one wave per SIMD: the 512-register budget
-// holds K, V, dK, dV and the softmax tiles with zero spill.
+// 8 waves (512 threads), TWO waves per SIMD: waves 0-3 accumulate dV
+// and waves 4-7 accumulate dK for the same 4x32 kv rows. Splitting the
+// roles keeps each wave's accumulator set at 64 VGPRs (one acc array,
+// ~225 total) instead of the fused kernel's 128 (344 total -> one wave
+// per SIMD, 50% of wave time parked with nothing co-resident to hide
+// it). The price is S recomputed by both roles (+25% MFMA issue on a
+// pipe that idles ~85% of the time).
 template <int D>
 struct DkdvLds {
   unsigned short Qr[64][D + 8];     // staged q rows (two 32-row tiles)
@@ -144,7 +149,7 @@ struct DkdvLds {
 };
 
 template <int D>
-__global__ __launch_bounds__(256, 1)
+__global__ __launch_bounds__(512, 1)
 void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
                            const unsigned short* __restrict__ Kp,
                            const unsigned short* __restrict__ Vp,
@@ -167,39 +172,41 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
   const int l31 = lane & 31;
   const int hi = lane >> 5;
   const int rep = Hq / Hkv;
+  const bool role_dv = wid < 4;            // waves 0-3 dV, 4-7 dK
 
   const int bh = blockIdx.y;
   const int b = bh / Hkv, hkv = bh % Hkv;
   const int kv0b = blockIdx.x * 128;
-  const int kw = kv0b + wid * 32;          // wave's first kv row
+  const int kw = kv0b + (wid & 3) * 32;    // wave's first kv row
 
   const int qrowstr = Hq * D, krowstr = Hkv * D;
   const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
   const int diag = Skv - S;
   const float kscale = scale * LOG2E;
 
-  // wave-resident K and V fragments: lane holds X[kw+l31][kc*16+hi*8+j]
+  // wave-resident K fragments (+ V for the dK waves, which need
+  // dP = dO V^T): lane holds X[kw+l31][kc*16+hi*8+j]
   short8v kfrag[DC16], vfrag[DC16];
   {
     const int krow = kw + l31;
 #pragma unroll
     for (int kc = 0; kc < DC16; ++kc) {
+      kfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
+      vfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
       if (krow < Skv) {
         const long off = kbase + (long)krow * krowstr + kc * 16 + hi * 8;
         kfrag[kc] = *reinterpret_cast<const short8v*>(Kp + off);
-        vfrag[kc] = *reinterpret_cast<const short8v*>(Vp + off);
-      } else {
-        kfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
-        vfrag[kc] = short8v{0, 0, 0, 0, 0, 0, 0, 0};
+        if (!role_dv)
+          vfrag[kc] = *reinterpret_cast<const short8v*>(Vp + off);
       }
     }
   }
 
-  f32x16 dk_acc[ND32], dv_acc[ND32];
+  f32x16 acc[ND32];                        // dV or dK by role
 #pragma unroll
   for (int c = 0; c < ND32; ++c)
 #pragma unroll
-    for (int r = 0; r < 16; ++r) { dk_acc[c][r] = 0.f; dv_acc[c][r] = 0.f; }
+    for (int r = 0; r < 16; ++r) acc[c][r] = 0.f;
 
   for (int gi = 0; gi < rep; ++gi) {
     const int hq = hkv * rep + gi;
@@ -211,12 +218,12 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
     const int qs_hi = (S + 63) / 64;
     // T14 async-stage split: stage qs+1's global loads are issued while
     // stage qs computes; the LDS write happens after the barrier.
-    constexpr int RIT = (64 * (D / 8)) / 256;   // row-staging iters
+    constexpr int RIT = (64 * (D / 8)) / 512;   // row-staging iters
     short8v stg[RIT * 2];
     auto issue_stage = [&](int qs) {
 #pragma unroll
       for (int it = 0; it < RIT; ++it) {
-        const int idx = threadIdx.x + it * 256;
+        const int idx = threadIdx.x + it * 512;
         const int row = idx / (D / 8), g = idx % (D / 8);
         const int q0s = qs * 64;
         short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -233,7 +240,7 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
     auto write_stage = [&]() {
 #pragma unroll
       for (int it = 0; it < RIT; ++it) {
-        const int idx = threadIdx.x + it * 256;
+        const int idx = threadIdx.x + it * 512;
         const int row = idx / (D / 8), g = idx % (D / 8);
         *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = stg[it * 2];
         *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) =
@@ -262,19 +269,28 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
           !(q0 >= S || (causal && (q0 + 31 + diag < kw)));
       if (live_tile) {
 
-      // ---- S[q][kv] and dP[q][kv] (C-layout: q rows on regs, kv on
-      // lanes = the wave's kv rows kw + l31)
+      // ---- S[q][kv] (both roles) and dP[q][kv] (dK waves only)
+      // C-layout: q rows on regs, kv on lanes = the wave's kw + l31
       f32x16 sv, dpv;
 #pragma unroll
       for (int r = 0; r < 16; ++r) { sv[r] = 0.f; dpv[r] = 0.f; }
+      if (role_dv) {
 #pragma unroll
-      for (int kc = 0; kc < DC16; ++kc) {
-        short8v qa = *reinterpret_cast<const short8v*>(
-            &lds.Qr[qoff + l31][kc * 16 + hi * 8]);
-        short8v da = *reinterpret_cast<const short8v*>(
-            &lds.dOr[qoff + l31][kc * 16 + hi * 8]);
-        sv = MFMA32(qa, kfrag[kc], sv);
-        dpv = MFMA32(da, vfrag[kc], dpv);
+        for (int kc = 0; kc < DC16; ++kc) {
+          short8v qa = *reinterpret_cast<const short8v*>(
+              &lds.Qr[qoff + l31][kc * 16 + hi * 8]);
+          sv = MFMA32(qa, kfrag[kc], sv);
+        }
+      } else {
+#pragma unroll
+        for (int kc = 0; kc < DC16; ++kc) {
+          short8v qa = *reinterpret_cast<const short8v*>(
+              &lds.Qr[qoff + l31][kc * 16 + hi * 8]);
+          short8v da = *reinterpret_cast<const short8v*>(
+              &lds.dOr[qoff + l31][kc * 16 + hi * 8]);
+          sv = MFMA32(qa, kfrag[kc], sv);
+          dpv = MFMA32(da, vfrag[kc], dpv);
+        }
       }
 
       // ---- P = exp2(s*kscale - lse2); dS = P o (dP - delta) * scale
@@ -299,28 +315,21 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         dpv[r] = p * (dpv[r] - dlt) * scale;
       }
 
-      // ---- fragments (k = q) and accumulate dV, dK
-      short8v pf0, pf1, df0, df1;
-      conv_c_to_frag(sv, pf0, pf1);
-      conv_c_to_frag(dpv, df0, df1);
+      // ---- fragments (k = q) and accumulate this role's output
+      short8v f0, f1;
+      if (role_dv) conv_c_to_frag(sv, f0, f1);
+      else conv_c_to_frag(dpv, f0, f1);
+      const DTX_AS3 unsigned short* img3 = role_dv
+          ? (const DTX_AS3 unsigned short*)&lds.dOr[0][0]
+          : (const DTX_AS3 unsigned short*)&lds.Qr[0][0];
 #pragma unroll
       for (int c = 0; c < ND32; ++c) {
-        // B[k=q][n=d] straight from the row-major tiles (tr-read)
-        const DTX_AS3 unsigned short* dor3 =
-            (const DTX_AS3 unsigned short*)&lds.dOr[0][0];
-        const DTX_AS3 unsigned short* qr3 =
-            (const DTX_AS3 unsigned short*)&lds.Qr[0][0];
-        short8v dot0 = tr_bfrag<D + 8>(dor3, qoff + hi * 8, c * 32, lane);
-        short8v dot1 = tr_bfrag<D + 8>(dor3, qoff + 16 + hi * 8, c * 32,
-                                       lane);
-        short8v qt0 = tr_bfrag<D + 8>(qr3, qoff + hi * 8, c * 32, lane);
-        short8v qt1 = tr_bfrag<D + 8>(qr3, qoff + 16 + hi * 8, c * 32,
-                                      lane);
-        // interleave dv/dk accumulators: no back-to-back MFMA RAW
-        dv_acc[c] = MFMA32(pf0, dot0, dv_acc[c]);
-        dk_acc[c] = MFMA32(df0, qt0, dk_acc[c]);
-        dv_acc[c] = MFMA32(pf1, dot1, dv_acc[c]);
-        dk_acc[c] = MFMA32(df1, qt1, dk_acc[c]);
+        // B[k=q][n=d] straight from the row-major tile (tr-read)
+        short8v b0 = tr_bfrag<D + 8>(img3, qoff + hi * 8, c * 32, lane);
+        short8v b1 = tr_bfrag<D + 8>(img3, qoff + 16 + hi * 8, c * 32,
+                                     lane);
+        acc[c] = MFMA32(f0, b0, acc[c]);
+        acc[c] = MFMA32(f1, b1, acc[c]);
         __builtin_amdgcn_sched_barrier(0);    // short (no cross-c hoist)
       }
       }  // live_tile
@@ -328,18 +337,16 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
     }
   }
 
-  // ---- epilogue: dK/dV C-layout [kv regs][d lanes] -> BSHD stores
+  // ---- epilogue: this role's C-layout [kv regs][d lanes] -> BSHD
+  unsigned short* out = role_dv ? dV : dK;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int kvg = kw + bw_crow(r, hi);
     if (kvg < Skv) {
-      unsigned short* krow = dK + kbase + (long)kvg * krowstr;
-      unsigned short* vrow = dV + kbase + (long)kvg * krowstr;
+      unsigned short* orow = out + kbase + (long)kvg * krowstr;
 #pragma unroll
-      for (int c = 0; c < ND32; ++c) {
-        krow[c * 32 + l31] = f2bf(dk_acc[c][r]);
-        vrow[c * 32 + l31] = f2bf(dv_acc[c][r]);
-      }
+      for (int c = 0; c < ND32; ++c)
+        orow[c * 32 + l31] = f2bf(acc[c][r]);
     }
   }
 }
@@ -558,13 +565,13 @@ void launch_attn_bwd_dkdv(const void* q, const void* k, const void* v,
                           float scale, int causal, hipStream_t st) {
   dim3 grid(DTX_CDIV(Skv, 128), B * Hkv);
   if (D == 128) {
-    attn_bwd_dkdv2_kernel<128><<<grid, 256, 0, st>>>(
+    attn_bwd_dkdv2_kernel<128><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
         (const unsigned short*)v, (const unsigned short*)dO,
         lse, delta, (unsigned short*)dk, (unsigned short*)dv,
         B, Hq, Hkv, S, Skv, scale, causal);
   } else if (D == 64) {
-    attn_bwd_dkdv2_kernel<64><<<grid, 256, 0, st>>>(
+    attn_bwd_dkdv2_kernel<64><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
         (const unsigned short*)v, (const unsigned short*)dO,
         lse, delta, (unsigned short*)dk, (unsigned short*)dv,
