@@ -100,7 +100,7 @@ class LlamaAttention(nn.Module):
         self.o_proj = _proj(cfg, "o_proj", H * D, cfg.hidden_size, lora, dtype)
 
     def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None,
-                pos_dev=None):
+                pos_dev=None, residual=None):
         B, S, _ = x.shape
         cfg = self.cfg
         H, Hkv, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
@@ -113,7 +113,11 @@ class LlamaAttention(nn.Module):
         if kv_cache is not None:
             k, v, len_dev = kv_cache.update(k, v)   # serving path (BSHD)
         o = attention(q, k, v, causal=True, len_dev=len_dev)
-        return self.o_proj(o.reshape(B, S, H * D))
+        o = o.reshape(B, S, H * D)
+        if residual is not None and isinstance(self.o_proj, FrozenLinear):
+            return self.o_proj(o, residual)   # residual in GEMM epilogue
+        y = self.o_proj(o)
+        return y if residual is None else residual + y
 
 
 class LlamaMLP(nn.Module):
@@ -126,8 +130,12 @@ class LlamaMLP(nn.Module):
         self.down_proj = _proj(cfg, "down_proj", cfg.intermediate_size,
                                cfg.hidden_size, lora, dtype)
 
-    def forward(self, x):
-        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+    def forward(self, x, residual=None):
+        h = swiglu(self.gate_proj(x), self.up_proj(x))
+        if residual is not None and isinstance(self.down_proj, FrozenLinear):
+            return self.down_proj(h, residual)    # residual in epilogue
+        y = self.down_proj(h)
+        return y if residual is None else residual + y
 
 
 class LlamaDecoderLayer(nn.Module):
@@ -144,9 +152,10 @@ class LlamaDecoderLayer(nn.Module):
     def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None,
                 pos_dev=None):
         h = rmsnorm(x, self.input_layernorm, self.cfg.rms_norm_eps)
-        x = x + self.self_attn(h, cos, sin, pos0, kv_cache, pos_dev)
+        x = self.self_attn(h, cos, sin, pos0, kv_cache, pos_dev,
+                           residual=x)
         h = rmsnorm(x, self.post_attention_layernorm, self.cfg.rms_norm_eps)
-        return x + self.mlp(h)
+        return self.mlp(h, residual=x)
 
 
 class LlamaForCausalLM(nn.Module):

@@ -34,10 +34,19 @@ class FrozenLinear(nn.Module):
             torch.empty(out_features, in_features, dtype=dtype),
             requires_grad=False)
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         if x.is_cuda and x.numel() == x.shape[-1]:
             from ..ops import gemv
-            return gemv(x, self.weight)      # decode: streaming GEMV
+            y = gemv(x, self.weight)         # decode: streaming GEMV
+            return y if residual is None else y + residual
+        if residual is not None:
+            # fold the residual add into the GEMM epilogue (addmm,
+            # beta=1): one kernel instead of linear + elementwise add
+            xs = x.shape
+            x2 = x.reshape(-1, xs[-1])
+            r2 = residual.reshape(-1, self.out_features)
+            return torch.addmm(r2, x2, self.weight.t()).reshape(
+                *xs[:-1], self.out_features)
         # F.linear routes to addmm/hipBLASLt: +18% over x @ w.t() on the
         # [16K,4096]x[4096,4096] shape (measured, tools/bench_gemm.py)
         return F.linear(x, self.weight)
