@@ -235,23 +235,13 @@ void swiglu_bwd_kernel(const unsigned short* __restrict__ dout,
                        const unsigned short* __restrict__ up,
                        unsigned short* __restrict__ dgate,
                        unsigned short* __restrict__ dup, long n8) {
-  // 2 independent 16B groups per iteration: more loads in flight per
-  // wave (the 5-stream pattern is latency-, not issue-, limited)
   long idx = (long)blockIdx.x * DTX_BLOCK + threadIdx.x;
-  const long stride = (long)gridDim.x * DTX_BLOCK;
-  for (; idx < n8; idx += stride * 2) {
-    const long i2 = idx + stride;
-    float d[8], g[8], u[8], d2[8], g2[8], u2[8];
+  long stride = (long)gridDim.x * DTX_BLOCK;
+  for (; idx < n8; idx += stride) {
+    float d[8], g[8], u[8], dg[8], du[8];
     load_bf16x8(dout + idx * 8, d);
     load_bf16x8(gate + idx * 8, g);
     load_bf16x8(up + idx * 8, u);
-    const bool has2 = i2 < n8;
-    if (has2) {
-      load_bf16x8(dout + i2 * 8, d2);
-      load_bf16x8(gate + i2 * 8, g2);
-      load_bf16x8(up + i2 * 8, u2);
-    }
-    float dg[8], du[8];
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       float sig = __builtin_amdgcn_rcpf(1.f + __expf(-g[i]));
@@ -261,17 +251,6 @@ void swiglu_bwd_kernel(const unsigned short* __restrict__ dout,
     }
     store_bf16x8(dgate + idx * 8, dg);
     store_bf16x8(dup + idx * 8, du);
-    if (has2) {
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        float sig = __builtin_amdgcn_rcpf(1.f + __expf(-g2[i]));
-        float silu = g2[i] * sig;
-        dg[i] = d2[i] * u2[i] * (sig + silu * (1.f - sig));
-        du[i] = d2[i] * silu;
-      }
-      store_bf16x8(dgate + i2 * 8, dg);
-      store_bf16x8(dup + i2 * 8, du);
-    }
   }
 }
 
