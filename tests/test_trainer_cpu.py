@@ -239,3 +239,27 @@ def test_rng_dropout_cpu_consistency():
     t_seed = ref.lora_contract(x, a, seed=seed, keep=keep)
     t_mask = ref.lora_contract(x, a, mask=m1.to(x.dtype))
     assert torch.equal(t_seed, t_mask)
+
+
+def test_residual_epilogue_fold_matches_separate_add():
+    """o_proj/down_proj fold the residual into the GEMM epilogue
+    (addmm); the result must equal the separate linear + add (same op
+    order in fp32 on CPU)."""
+    import torch
+
+    from datatunerx_amd.models.lora import FrozenLinear
+    torch.manual_seed(0)
+    lin = FrozenLinear(32, 48, dtype=torch.float32)
+    with torch.no_grad():
+        lin.weight.copy_(torch.randn(48, 32))
+    x = torch.randn(2, 5, 32)
+    res = torch.randn(2, 5, 48)
+    fused = lin(x, residual=res)
+    sep = res + lin(x)
+    assert torch.allclose(fused, sep, atol=1e-5)
+    # gradient flows through both x and residual
+    x2 = x.clone().requires_grad_(True)
+    r2 = res.clone().requires_grad_(True)
+    lin(x2, residual=r2).sum().backward()
+    assert torch.allclose(r2.grad, torch.ones_like(r2))
+    assert x2.grad is not None
