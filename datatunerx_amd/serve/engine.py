@@ -200,12 +200,46 @@ class InferenceEngine:
         ids = self.generate(src, max_tokens, temperature, top_p)
         return self.tok.decode(ids)
 
+    def chat_stream(self, messages: List[dict], max_tokens: int = 64,
+                    temperature: float = 0.0, top_p: float = 1.0):
+        """Like chat() but yields text DELTAS as tokens decode (the
+        serving endpoint streams these as SSE chunks). Token-identical
+        to chat(): same generate loop, so TP followers running chat()
+        stay in collective lockstep with a streaming front rank."""
+        system, history, pending_user = "", [], None
+        for m in messages:
+            if m["role"] == "system":
+                system = m["content"]
+            elif m["role"] == "user":
+                pending_user = m["content"]
+            elif m["role"] == "assistant" and pending_user is not None:
+                history.append((pending_user, m["content"]))
+                pending_user = None
+        t = get_template(self.template)
+        src, _ = t.encode_oneturn(self.tok, pending_user or "", "",
+                                  history, system)
+        ids, sent = [], ""
+        for tok_id in self.generate_stream(src, max_tokens, temperature,
+                                           top_p):
+            ids.append(tok_id)
+            full = self.tok.decode(ids)
+            if len(full) > len(sent):
+                delta, sent = full[len(sent):], full
+                yield delta
+
     @torch.no_grad()
     def generate(self, prompt_ids: List[int], max_new_tokens: int = 64,
                  temperature: float = 0.0, top_p: float = 1.0) -> List[int]:
+        return list(self.generate_stream(prompt_ids, max_new_tokens,
+                                         temperature, top_p))
+
+    @torch.no_grad()
+    def generate_stream(self, prompt_ids: List[int],
+                        max_new_tokens: int = 64, temperature: float = 0.0,
+                        top_p: float = 1.0):
         ids = torch.tensor([prompt_ids], dtype=torch.long,
                            device=self.device)
-        out = []
+        n_out = 0
         gd = self._get_graphed() if self.is_llama else None
         if gd is not None and len(prompt_ids) + max_new_tokens + 1 < \
                 gd.max_s:
@@ -218,12 +252,13 @@ class InferenceEngine:
             for _ in range(max_new_tokens):
                 if nxt == self.tok.eos_token_id:
                     break
-                out.append(nxt)
-                if len(prompt_ids) + len(out) + 1 >= gd.max_s:
+                yield nxt
+                n_out += 1
+                if len(prompt_ids) + n_out + 1 >= gd.max_s:
                     break
                 logits = gd.step(nxt)
                 nxt = self._sample(logits[0, -1], temperature, top_p)
-            return out
+            return
         if self.is_llama:
             cfg = self.model.cfg
             max_s = min(cfg.max_position_embeddings,
@@ -242,7 +277,7 @@ class InferenceEngine:
                 nxt = self._sample(logits[0, -1], temperature, top_p)
                 if nxt == self.tok.eos_token_id:
                     break
-                out.append(nxt)
+                yield nxt
                 cur = torch.tensor([[nxt]], dtype=torch.long,
                                    device=self.device)
         else:
@@ -254,9 +289,8 @@ class InferenceEngine:
                 nxt = self._sample(logits[0, -1], temperature, top_p)
                 if nxt == self.tok.eos_token_id:
                     break
-                out.append(nxt)
+                yield nxt
                 seq.append(nxt)
-        return out
 
     def _sample(self, logits: torch.Tensor, temperature: float,
                 top_p: float) -> int:

@@ -50,12 +50,34 @@ def build_handler(engine):
                 return
             try:
                 if self.path == "/chat/completions":
-                    with lock:
-                        text = engine.chat(
-                            body.get("messages", []),
+                    args = (body.get("messages", []),
                             int(body.get("max_tokens", 64)),
                             float(body.get("temperature", 0.0)),
                             float(body.get("top_p", 1.0)))
+                    if body.get("stream"):
+                        # SSE token streaming (OpenAI chunk format)
+                        self.send_response(200)
+                        self.send_header("Content-Type",
+                                         "text/event-stream")
+                        self.send_header("Cache-Control", "no-cache")
+                        self.end_headers()
+                        rid = f"chatcmpl-{int(time.time()*1000)}"
+                        with lock:
+                            for delta in engine.chat_stream(*args):
+                                chunk = json.dumps({
+                                    "id": rid,
+                                    "object": "chat.completion.chunk",
+                                    "choices": [{
+                                        "index": 0,
+                                        "delta": {"content": delta},
+                                        "finish_reason": None}]})
+                                self.wfile.write(
+                                    f"data: {chunk}\n\n".encode())
+                                self.wfile.flush()
+                        self.wfile.write(b"data: [DONE]\n\n")
+                        return
+                    with lock:
+                        text = engine.chat(*args)
                     self._send(200, {
                         "id": f"chatcmpl-{int(time.time()*1000)}",
                         "object": "chat.completion",
@@ -100,6 +122,15 @@ class TPFrontEngine:
                      "max_tokens": max_tokens, "temperature": temperature,
                      "top_p": top_p})
         return self.engine.chat(messages, max_tokens, temperature, top_p)
+
+    def chat_stream(self, messages, max_tokens, temperature, top_p):
+        # followers run the non-streaming chat(): token-identical loop,
+        # so the TP collectives stay in lockstep with the streaming front
+        self._bcast({"op": "chat", "messages": messages,
+                     "max_tokens": max_tokens, "temperature": temperature,
+                     "top_p": top_p})
+        return self.engine.chat_stream(messages, max_tokens, temperature,
+                                       top_p)
 
     def perplexity(self, texts):
         self._bcast({"op": "ppl", "texts": texts})

@@ -213,3 +213,62 @@ dist.destroy_process_group()
         assert p.wait(timeout=300) == 0
     out = json.load(open(tmp_path / "ad_out.json"))
     assert out["err"] < 1e-4 * max(1.0, out["ref"])
+
+
+def test_streaming_matches_nonstreaming(tmp_path):
+    """SSE streaming (stream=true) yields the same completion text as
+    the non-streaming endpoint, chunk by chunk."""
+    import json as _json
+    import subprocess
+    import time
+    import urllib.request
+
+    port = 18991
+    env = dict(os.environ)
+    env.update({"PYTHONPATH": ROOT})
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "datatunerx_amd.serve.server",
+         "--model", "llama-tiny", "--port", str(port),
+         "--template", "vanilla"], env=env)
+    try:
+        body = _json.dumps({"messages": [{"role": "user", "content": "hi"}],
+                            "max_tokens": 8}).encode()
+        deadline = time.time() + 120
+        resp = None
+        while time.time() < deadline:
+            try:
+                req = urllib.request.Request(
+                    f"http://127.0.0.1:{port}/chat/completions", body,
+                    {"Content-Type": "application/json"})
+                resp = _json.load(urllib.request.urlopen(req, timeout=10))
+                break
+            except Exception:
+                if proc.poll() is not None:
+                    raise AssertionError("server died early")
+                time.sleep(1.0)
+        assert resp is not None
+        full = resp["choices"][0]["message"]["content"]
+
+        sbody = _json.dumps({"messages": [{"role": "user", "content": "hi"}],
+                             "max_tokens": 8, "stream": True}).encode()
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}/chat/completions", sbody,
+            {"Content-Type": "application/json"})
+        streamed = ""
+        n_chunks = 0
+        with urllib.request.urlopen(req, timeout=60) as r:
+            assert r.headers["Content-Type"].startswith("text/event-stream")
+            for line in r:
+                line = line.decode().strip()
+                if not line.startswith("data: "):
+                    continue
+                payload = line[len("data: "):]
+                if payload == "[DONE]":
+                    break
+                streamed += _json.loads(payload)[
+                    "choices"][0]["delta"]["content"]
+                n_chunks += 1
+        assert streamed == full
+        assert n_chunks >= 2, "should stream multiple chunks"
+    finally:
+        proc.kill()
