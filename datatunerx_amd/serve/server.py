@@ -158,6 +158,8 @@ def _llama_config(name: str):
         return LlamaConfig.llama2_7b()
     if name in ("llama2-13b", "llama-2-13b"):
         return LlamaConfig.llama2_13b()
+    if name in ("llama3-8b", "llama-3-8b"):
+        return LlamaConfig.llama3_8b()
     if name == "llama-tiny":
         return LlamaConfig.tiny()
     raise ValueError(f"TP serving supports llama models, not {name!r}")

@@ -315,6 +315,10 @@ def test_serve_engine_decode_gpu():
     assert out1 == out3
     ppl = eng.perplexity(["hello world", "the quick brown fox"])
     assert 0 < ppl < float("inf")
+    # streaming yields the same tokens as batch generation (graph mode)
+    streamed = list(eng.generate_stream(list(range(4, 20)),
+                                        max_new_tokens=8))
+    assert streamed == out1
 
 
 @pytest.mark.parametrize("B,Hq,Hkv,Skv,D", [
