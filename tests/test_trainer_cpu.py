@@ -263,3 +263,21 @@ def test_residual_epilogue_fold_matches_separate_add():
     lin(x2, residual=r2).sum().backward()
     assert torch.allclose(r2.grad, torch.ones_like(r2))
     assert x2.grad is not None
+
+
+def test_engine_stream_matches_chat_cpu():
+    """chat_stream deltas concatenate to exactly chat()'s text (eager
+    CPU path; the GPU graph path is covered in test_ops_gpu)."""
+    import torch
+
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    model = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                             dtype=torch.float32).init_random(seed=0)
+    model.eval()
+    eng = InferenceEngine(model, template="vanilla",
+                          device=torch.device("cpu"))
+    msgs = [{"role": "user", "content": "hello"}]
+    full = eng.chat(msgs, 8)
+    streamed = "".join(eng.chat_stream(msgs, 8))
+    assert streamed == full
