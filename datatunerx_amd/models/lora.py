@@ -19,8 +19,28 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import lora_contract, lora_expand_add
+from ..ops import gemm_nt_supported, lora_contract, lora_expand_add
+from ..ops.autograd import FrozenGemm
 from ..ops.autograd import LoRALinear as _FusedLoRAFn
+
+
+def _weight_t(mod: nn.Module):
+    """Cached W^T for a frozen base weight: both the forward (x @ W^T)
+    and the dgrad (dy @ W) then run the hand-written NT MFMA GEMM with
+    the contraction contiguous — 288 GB HBM3E per MI355X makes the
+    persistent transposed copy free. Built lazily on first GPU forward;
+    None when the shape is outside the kernel's support."""
+    wt = getattr(mod, "_wt", False)
+    if wt is False:
+        w = mod.weight
+        n, k = w.shape
+        if (w.is_cuda and w.dtype == torch.bfloat16
+                and gemm_nt_supported(n, k) and gemm_nt_supported(k, n)):
+            wt = w.detach().t().contiguous()
+        else:
+            wt = None
+        mod._wt = wt
+    return wt
 
 
 class FrozenLinear(nn.Module):
@@ -39,6 +59,12 @@ class FrozenLinear(nn.Module):
             from ..ops import gemv
             y = gemv(x, self.weight)         # decode: streaming GEMV
             return y if residual is None else y + residual
+        if x.is_cuda:
+            wt = _weight_t(self)
+            if wt is not None:
+                # hand-written MFMA GEMM; residual add fused into the
+                # kernel epilogue (no separate elementwise pass)
+                return FrozenGemm.apply(x, self.weight, wt, residual)
         if residual is not None:
             # fold the residual add into the GEMM epilogue (addmm,
             # beta=1): one kernel instead of linear + elementwise add
@@ -47,8 +73,7 @@ class FrozenLinear(nn.Module):
             r2 = residual.reshape(-1, self.out_features)
             return torch.addmm(r2, x2, self.weight.t()).reshape(
                 *xs[:-1], self.out_features)
-        # F.linear routes to addmm/hipBLASLt: +18% over x @ w.t() on the
-        # [16K,4096]x[4096,4096] shape (measured, tools/bench_gemm.py)
+        # hipBLASLt fallback for unsupported shapes / CPU
         return F.linear(x, self.weight)
 
 
@@ -62,31 +87,33 @@ class LoRAFunctionWithDropout(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x, w, a, b, scale, mask, seed, keep):
+    def forward(ctx, x, w, a, b, scale, mask, seed, keep, wt=None):
+        from ..ops.autograd import _base_gemm
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
-        y = F.linear(x2, w)
+        y = _base_gemm(x2, w, wt)
         t = lora_contract(x2, a, mask, seed, keep)   # (x o mask) @ A^T
         lora_expand_add(y, t, b, scale)
-        ctx.save_for_backward(x2, w, a, b, t, mask)
+        ctx.save_for_backward(x2, w, a, b, t, mask, wt)
         ctx.scale, ctx.xshape = scale, xs
         ctx.seed, ctx.keep = seed, keep
         return y.reshape(*xs[:-1], w.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        from ..ops import lora_wgrad
-        x2, w, a, b, t, mask = ctx.saved_tensors
+        from ..ops import gemm_nt, lora_wgrad
+        x2, w, a, b, t, mask, wt = ctx.saved_tensors
         s, seed, keep = ctx.scale, ctx.seed, ctx.keep
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
-        dx = dy2 @ w
+        dx = gemm_nt(dy2, wt) if (wt is not None and dy2.is_cuda) \
+            else dy2 @ w
         dt = lora_contract(dy2, b.t().contiguous())       # [M,r] = dy @ B
         da = lora_wgrad(dt, x2, s, mask, seed, keep)  # dt^T @ (x o mask)
         db = lora_wgrad(t, dy2, s).t().contiguous()
         # dx += mask o (s * dt @ A)
         lora_expand_add(dx, dt, a.t().contiguous(), s, mask, seed, keep)
         return (dx.reshape(ctx.xshape), None, da.to(a.dtype),
-                db.to(b.dtype), None, None, None, None)
+                db.to(b.dtype), None, None, None, None, None)
 
 
 class LoRALinearModule(nn.Module):
@@ -109,9 +136,10 @@ class LoRALinearModule(nn.Module):
         nn.init.kaiming_uniform_(self.lora_A, a=math.sqrt(5))
 
     def forward(self, x):
+        wt = _weight_t(self) if x.is_cuda else None
         if not (self.training and self.dropout > 0.0):
             return _FusedLoRAFn.apply(x, self.weight, self.lora_A,
-                                      self.lora_B, self.scale)
+                                      self.lora_B, self.scale, wt)
         keep = 1.0 - self.dropout
         # draw the per-call RNG seed from torch's generator so dropout
         # stays reproducible under torch.manual_seed
@@ -120,13 +148,13 @@ class LoRALinearModule(nn.Module):
             # fused counter-based RNG: mask never materialized
             return LoRAFunctionWithDropout.apply(
                 x, self.weight, self.lora_A, self.lora_B, self.scale,
-                None, seed, keep)
+                None, seed, keep, wt)
         from ..ops import dropout_mask
         mask = dropout_mask(x.numel() // x.shape[-1], x.shape[-1], seed,
                             keep, x).to(x.dtype)
         return LoRAFunctionWithDropout.apply(x, self.weight, self.lora_A,
                                              self.lora_B, self.scale,
-                                             mask, 0, 1.0)
+                                             mask, 0, 1.0, wt)
 
     def merged_weight(self):
         """W + s·B@A — used by the serving engine (no adapter overhead)."""

@@ -143,6 +143,23 @@ def gemv(x, w):
     return torch.nn.functional.linear(x, w)
 
 
+def gemm_nt_supported(n: int, k: int) -> bool:
+    """Shapes the hand-written MFMA GEMM handles (all Llama projection /
+    lm_head shapes); anything else falls back to hipBLASLt."""
+    return n % 256 == 0 and k % 64 == 0 and k >= 128
+
+
+def gemm_nt(a, b, src=None):
+    """C[...,N] = a[...,K] @ b[N,K]^T (+ src) via the hand-written
+    256x256 MFMA kernel (gemm.hip). bf16 in/out, fp32 accumulate."""
+    if _gpu(a):
+        return _EXT.gemm_nt(a, b, src)
+    out = (a.float() @ b.float().t()).to(a.dtype)
+    if src is not None:
+        out = out + src.reshape(out.shape)
+    return out
+
+
 # -------------------------------------------------------------- LoRA
 # Dropout enters the kernels one of two ways:
 #   - `mask`: a materialized bf16 mask tensor (same shape as x / y)

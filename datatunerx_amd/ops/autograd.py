@@ -121,44 +121,87 @@ def cross_entropy(logits, targets, ignore_index: int = -100):
     return CrossEntropy.apply(logits, targets, ignore_index)
 
 
+def _base_gemm(x2, w, wt):
+    """Base-projection GEMM: the hand-written MFMA kernel (gemm.hip) when
+    a cached W^T is provided (both fwd and dgrad then run the fast NT
+    layout), hipBLASLt otherwise."""
+    from . import gemm_nt
+    if wt is not None and x2.is_cuda and x2.shape[0] > 0:
+        return gemm_nt(x2, w)
+    return torch.nn.functional.linear(x2, w)
+
+
 class LoRALinear(torch.autograd.Function):
     """y = x @ W^T + scale * (x @ A^T) @ B^T  with frozen W.
 
-    W: [N,K] (frozen base, no wgrad) — the base GEMM goes through
-    torch.matmul (hipBLASLt on ROCm); the low-rank path is the fused HIP
-    contract/expand pair. A: [r,K], B: [N,r] (HF PEFT adapter layout, the
-    checkpoint-compat contract — SURVEY.md §5 Checkpoint/resume).
+    W: [N,K] (frozen base, no wgrad) — the base GEMM is the hand-written
+    gfx950 MFMA kernel (ops/hip/gemm.hip) with `wt` = W^T cached so the
+    dgrad dx = dy @ W is the same contraction-contiguous NT kernel; the
+    low-rank path is the fused HIP contract/expand pair. A: [r,K],
+    B: [N,r] (HF PEFT adapter layout, the checkpoint-compat contract —
+    SURVEY.md §5 Checkpoint/resume).
     """
 
     @staticmethod
-    def forward(ctx, x, w, a, b, scale):
+    def forward(ctx, x, w, a, b, scale, wt=None):
         xs = x.shape
         x2 = x.reshape(-1, xs[-1])
         if x2.is_cuda and x2.shape[0] == 1:
             from . import gemv
             y = gemv(x2, w)                      # decode: streaming GEMV
         else:
-            y = torch.nn.functional.linear(x2, w)
+            y = _base_gemm(x2, w, wt)
         t = lora_contract(x2, a)                 # [M,r] f32
         lora_expand_add(y, t, b, scale)          # y += s * t @ B^T
-        ctx.save_for_backward(x2, w, a, b, t)
+        ctx.save_for_backward(x2, w, a, b, t, wt)
         ctx.scale = scale
         ctx.xshape = xs
         return y.reshape(*xs[:-1], w.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        x2, w, a, b, t = ctx.saved_tensors
+        x2, w, a, b, t, wt = ctx.saved_tensors
         s = ctx.scale
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
-        dx = dy2 @ w                                  # base dgrad (hipBLASLt)
+        if wt is not None and dy2.is_cuda:
+            from . import gemm_nt
+            dx = gemm_nt(dy2, wt)                # base dgrad (NT MFMA)
+        else:
+            dx = dy2 @ w
         dt = lora_contract(dy2, b.t().contiguous())   # [M,r] = dy @ B
         db = lora_wgrad(t, dy2, s)                    # [r,N] -> B grad is [N,r]
         da = lora_wgrad(dt, x2, s)                    # [r,K]
         lora_expand_add(dx, dt, a.t().contiguous(), s)  # dx += s * dt @ A
         return (dx.reshape(ctx.xshape), None, da.to(a.dtype),
-                db.t().contiguous().to(b.dtype), None)
+                db.t().contiguous().to(b.dtype), None, None)
 
 
-def lora_linear(x, w, a, b, scale: float):
-    return LoRALinear.apply(x, w, a, b, scale)
+def lora_linear(x, w, a, b, scale: float, wt=None):
+    return LoRALinear.apply(x, w, a, b, scale, wt)
+
+
+class FrozenGemm(torch.autograd.Function):
+    """y = x @ W^T (+ residual) with frozen W via the MFMA NT kernel; the
+    residual add runs in the GEMM epilogue (one kernel, no extra HBM
+    round-trip). dgrad reuses the kernel with the cached W^T."""
+
+    @staticmethod
+    def forward(ctx, x, w, wt, residual):
+        from . import gemm_nt
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1])
+        r2 = residual.reshape(-1, w.shape[0]).contiguous() \
+            if residual is not None else None
+        y = gemm_nt(x2, w, r2)
+        ctx.save_for_backward(wt)
+        ctx.xshape, ctx.has_res = xs, residual is not None
+        return y.reshape(*xs[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        from . import gemm_nt
+        (wt,) = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = gemm_nt(dy2, wt)
+        return (dx.reshape(ctx.xshape), None, None,
+                dy if ctx.has_res else None)

@@ -42,6 +42,8 @@ void launch_transpose_sd(const void*, void*, int, int, int, int,
 int attn_decode_nsplit(int Skv);
 void launch_gemv_bf16(const void*, const void*, void*, int, int,
                       hipStream_t);
+void launch_gemm_nt(const void*, const void*, const void*, void*, long,
+                    int, int, hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*,
                         const int*, float*, void*, float*, int, int, int,
                         int, int, float, hipStream_t);
@@ -324,6 +326,32 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   return {o, lse};
 }
 
+// C[M,N] = A[M,K] @ B[N,K]^T (+ optional bf16 src) — the hand-written
+// 256x256-tile MFMA GEMM (gemm.hip). Requires N % 256 == 0, K % 64 == 0,
+// K >= 128; any M (SRSRC-clamped tail).
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> src) {
+  check_bf16_contig(a, "a");
+  check_bf16_contig(b, "b");
+  const int K = (int)b.size(1), N = (int)b.size(0);
+  const long M = a.numel() / K;
+  TORCH_CHECK(a.size(-1) == K, "gemm_nt: inner dims");
+  TORCH_CHECK(N % 256 == 0 && K % 64 == 0 && K >= 128,
+              "gemm_nt: unsupported shape N=", N, " K=", K);
+  auto sizes = a.sizes().vec();
+  sizes[sizes.size() - 1] = N;
+  auto c = torch::empty(sizes, a.options());
+  const void* sp = nullptr;
+  if (src.has_value()) {
+    check_bf16_contig(*src, "src");
+    TORCH_CHECK(src->numel() == M * N, "gemm_nt: src shape");
+    sp = src->data_ptr();
+  }
+  launch_gemm_nt(a.data_ptr(), b.data_ptr(), sp, c.data_ptr(), M, N, K,
+                 cur_stream());
+  return c;
+}
+
 // y[1,N] = x[1,K] @ W[N,K]^T (decode GEMV; fp32 accumulate)
 torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
   check_bf16_contig(x, "x");
@@ -442,6 +470,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("transpose_sd", &transpose_sd);
   m.def("gemv", &gemv);
+  m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"),
+        py::arg("src") = py::none());
   m.def("attn_decode", &attn_decode, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("scale"), py::arg("len_dev") = py::none());
 }

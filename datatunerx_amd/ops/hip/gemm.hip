@@ -1,0 +1,309 @@
+// Hand-written CDNA4 (gfx950) bf16 GEMM for the frozen-base projections —
+// the kernel SURVEY.md §2.4 row 1 names first ("base GEMMs ... HIP MFMA
+// tiled GEMM"; reference exercise site: HF LlamaForCausalLM fwd/bwd,
+// /root/reference/cmd/tuning/train.py:236-242).
+//
+//   C[M,N] = A[M,K] @ B[N,K]^T (+ optional bf16 source add)
+//
+// Everything is row-major bf16 with the CONTRACTION contiguous in both
+// operands ("NT"). The dgrad pass dX = dY @ W reuses this same kernel
+// with a cached W^T copy: base weights are frozen under LoRA, and 288 GB
+// of HBM3E per MI355X makes a persistent transposed copy free — so both
+// hot GEMMs of the training step run the one fast layout instead of a
+// strided-B variant.
+//
+// Structure (guide: cdna_hip_programming.md §5 "256^2 8-phase template"):
+//   - 256x256 output tile, K-step 64 as two 32-deep k-halves.
+//   - 512 threads = 8 waves in a 2(M) x 4(N) grid; per-wave output
+//     128x64 = 8x4 fragments of mfma_f32_16x16x32_bf16 (f32 acc).
+//   - LDS ring: 4 A half-slots + 4 B half-slots of 16 KiB ([256 rows] x
+//     [32 bf16]), 128 KiB total — compute tile kt while tile kt+1 lands.
+//   - Staging by buffer_load_dwordx4 ... lds (LDS-DMA, inline asm: hipcc
+//     never auto-emits it and must not count it), one half-tile per
+//     phase, counted s_waitcnt vmcnt(4) at odd-phase ends ONLY (loads
+//     span barriers — T3+T4), raw s_barrier (never __syncthreads: with
+//     LDS-DMA in flight its fence drains vmcnt to 0).
+//   - Per-phase s_setprio(1) around the 16-MFMA cluster (T5).
+//   - XCD-aware bijective block remap, nb-major so each XCD's contiguous
+//     chunk re-reads the same 2 MB B-panel through its private L2 (T1).
+//
+// LDS swizzle (bank-conflict-free, derived for this [256][32] half
+// layout): a half-row is 64 B = 4 16-B sub-slots; data sub-slot s of row
+// r is stored at position s ^ ((-(r>>2))&3). Each ds_read_b128 lane
+// group then touches 16 distinct (row%4, slot) pairs = all 64 banks
+// exactly once (verified for all four hardware lane groups). The
+// LDS-DMA destination is lane-linear, so the swizzle is applied to the
+// per-lane GLOBAL source address (guide §5.4 rule 21); it permutes
+// 16-B chunks only WITHIN one row's 64-B k-half, so global requests
+// still cover whole 64-B lines (no FETCH cost).
+#include "dtx_common.h"
+
+typedef __attribute__((ext_vector_type(4))) unsigned uint4v;
+typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
+
+#define MFMA16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
+
+// Buffer resource descriptor (T8/T20): built from readfirstlane'd scalars
+// so hipcc proves uniformity (no waterfall loops), num_records = bytes so
+// out-of-bounds rows of the M-tail clamp to zero loads / dropped stores.
+__device__ __forceinline__ uint4v dtx_srd(const void* base,
+                                          unsigned long long bytes) {
+  unsigned long long b = (unsigned long long)base;
+  uint4v r;
+  r.x = __builtin_amdgcn_readfirstlane((unsigned)b);
+  r.y = __builtin_amdgcn_readfirstlane((unsigned)(b >> 32));
+  r.z = __builtin_amdgcn_readfirstlane((unsigned)bytes);
+  r.w = 0x00020000u;  // dfmt/nfmt raw buffer config
+  return r;
+}
+
+// LDS-DMA: stage 64 lanes x 16 B at global (srd + voff) into LDS at
+// wave-uniform byte base `lds_dst` + lane*16. M0 written in the same
+// statement (s_nop 0 = the required wait state after the M0 write).
+__device__ __forceinline__ void glds16(uint4v srd, unsigned voff,
+                                       unsigned lds_dst) {
+  asm volatile(
+      "s_mov_b32 m0, %1\n\t"
+      "s_nop 0\n\t"
+      "buffer_load_dwordx4 %0, %2, 0 offen lds"
+      :: "v"(voff), "s"(lds_dst), "s"(srd) : "memory");
+}
+
+__device__ __forceinline__ void dtx_vmcnt4() {
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+}
+__device__ __forceinline__ void dtx_vmcnt0() {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+}
+__device__ __forceinline__ void dtx_bar() {
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+}
+
+#define GEMM_BM 256
+#define GEMM_BN 256
+#define GEMM_BK 64
+#define HALF_BYTES (256 * 64)          // one [256][32] bf16 half-slot
+#define A_RING 0                       // 4 A slots
+#define B_RING (4 * HALF_BYTES)        // 4 B slots
+
+// f((r>>2)&3) of the swizzle: 0->0, 1->3, 2->2, 3->1  == (-x)&3
+__device__ __forceinline__ unsigned swz_f(unsigned x) { return (0u - x) & 3u; }
+
+template <bool HAS_SRC>
+__global__ __launch_bounds__(512, 2)
+void gemm_nt_kernel(const unsigned short* __restrict__ A,
+                    const unsigned short* __restrict__ B,
+                    const unsigned short* __restrict__ SRC,
+                    unsigned short* __restrict__ C,
+                    long M, int N, int K, int mb_n) {
+  __shared__ __attribute__((aligned(16))) unsigned char lds[8 * HALF_BYTES];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid_div = tid >> 6;                       // divergent to hipcc
+  const int wid = __builtin_amdgcn_readfirstlane(wid_div);
+  const int wm = wid >> 2, wn = wid & 3;
+
+  // ---- XCD-aware bijective remap; nb-major chunks (B-panel L2 reuse)
+  const int nwg = gridDim.x;
+  const int q = nwg >> 3, r8 = nwg & 7;
+  const int xcd = blockIdx.x & 7, pos = blockIdx.x >> 3;
+  const int wgid = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q)
+                   + pos;
+  const int mb = wgid % mb_n, nb = wgid / mb_n;
+  const long m0 = (long)mb * GEMM_BM;
+  const int n0 = nb * GEMM_BN;
+
+  // SRDs are rebased per block (A at row m0, B at row n0, C/SRC at row
+  // m0) so every voffset fits 32 bits even for multi-GB logits tensors;
+  // num_records clamps the M-tail (OOB loads read 0, OOB stores drop).
+  const long ldab = (long)K * 2;                       // A/B row bytes
+  const uint4v srdA = dtx_srd(A + m0 * K, (unsigned long long)(M - m0) * ldab);
+  const uint4v srdB = dtx_srd(B + (long)n0 * K,
+                              (unsigned long long)(N - n0) * ldab);
+
+  // ---- per-lane staging constants (swizzle on the SOURCE address)
+  // stage: lane l covers (row = base + l>>2, sub-slot pos q = l&3);
+  // the data sub-slot is s = q ^ f((l>>2)&3) (row bits 2..3 == (l>>2)&3
+  // because every stage base row is a multiple of 16).
+  const unsigned st_row = lane >> 2;                   // 0..15
+  const unsigned st_s = (lane & 3) ^ swz_f((lane >> 2) & 3);
+  // fragment read: lane l reads (row = frag_base + (l&15),
+  // slot s = l>>4) at position s ^ f(((l&15)>>2)&3).
+  const unsigned fr_off = (lane & 15) * 64
+      + ((unsigned)(lane >> 4) ^ swz_f(((lane & 15) >> 2) & 3)) * 16;
+
+  // A stage: wave wid covers rows wid*32 + piece*16 + st_row of the
+  // 256-row tile. Global voffset (bytes):
+  //   (m0 + rows)*ldab + kbyte + st_s*16
+  const long a_row0 = wid * 32 + st_row;               // rows local to SRD
+  const long b_row0 = wid * 32 + st_row;
+  // lds dst base for this wave's slice of a half-slot
+  const unsigned st_lds = wid * 2048;
+
+  const int KT = K / GEMM_BK;
+
+  float4v acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = float4v{0.f, 0.f, 0.f, 0.f};
+
+  // ---- staging helper: stage half (a_or_b, kt, kh): 2 glds per wave
+#define STAGE_A(kt, kh)                                                     \
+  {                                                                         \
+    const unsigned slot = ((unsigned)(2 * (kt) + (kh)) & 3) * HALF_BYTES;   \
+    const long kb = (long)(kt) * 128 + (kh) * 64 + st_s * 16;               \
+    glds16(srdA, (unsigned)(a_row0 * ldab + kb), A_RING + slot + st_lds);   \
+    glds16(srdA, (unsigned)((a_row0 + 16) * ldab + kb),                     \
+           A_RING + slot + st_lds + 1024);                                  \
+  }
+#define STAGE_B(kt, kh)                                                     \
+  {                                                                         \
+    const unsigned slot = ((unsigned)(2 * (kt) + (kh)) & 3) * HALF_BYTES;   \
+    const long kb = (long)(kt) * 128 + (kh) * 64 + st_s * 16;               \
+    glds16(srdB, (unsigned)(b_row0 * ldab + kb), B_RING + slot + st_lds);   \
+    glds16(srdB, (unsigned)((b_row0 + 16) * ldab + kb),                     \
+           B_RING + slot + st_lds + 1024);                                  \
+  }
+
+  // ---- prologue: stage tile 0 (kh0 pair first), wait for the kh0 pair
+  STAGE_B(0, 0);
+  STAGE_A(0, 0);
+  STAGE_B(0, 1);
+  STAGE_A(0, 1);
+  dtx_vmcnt4();
+  dtx_bar();
+
+  // ---- K loop: 4 phases per tile; phase p: kh = p>>1, m-half = p&1.
+  // B fragments are read at even phases and reused at the odd phase.
+  // Stage schedule (tile kt stages kt+1): p0: B-kh0, p1: A-kh0,
+  // p2: B-kh1, p3: A-kh1. vmcnt(4) at the ends of p1 and p3 retires
+  // exactly the halves the next two phases read (2 half-tiles stay in
+  // flight across each barrier).
+  short8v bfr[4];
+  for (int kt = 0; kt < KT; ++kt) {
+    const unsigned aslot = A_RING + ((unsigned)(2 * kt) & 3) * HALF_BYTES;
+    const unsigned bslot = B_RING + ((unsigned)(2 * kt) & 3) * HALF_BYTES;
+    const bool pre = kt + 1 < KT;
+
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const unsigned kh = p >> 1, mh = p & 1;
+      const unsigned as = aslot + kh * HALF_BYTES;
+      const unsigned bs = bslot + kh * HALF_BYTES;
+      // ds_read register subtile for this phase
+      short8v afr[4];
+      if (mh == 0) {
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          bfr[fn] = *reinterpret_cast<const short8v*>(
+              &lds[bs + wn * 4096 + fn * 1024 + fr_off]);
+      }
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+        afr[fm] = *reinterpret_cast<const short8v*>(
+            &lds[as + wm * 8192 + (mh * 4 + fm) * 1024 + fr_off]);
+      // issue next tile's stage for this phase
+      if (pre) {
+        if (p == 0) STAGE_B(kt + 1, 0)
+        else if (p == 1) STAGE_A(kt + 1, 0)
+        else if (p == 2) STAGE_B(kt + 1, 1)
+        else STAGE_A(kt + 1, 1)
+      }
+      dtx_bar();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[mh * 4 + fm][fn] = MFMA16(afr[fm], bfr[fn], acc[mh * 4 + fm][fn]);
+      __builtin_amdgcn_s_setprio(0);
+      // counted waits only (loads span barriers); the LAST tile issues
+      // no new stages, so its in-flight count is already <= 4 at p1 and
+      // vmcnt(4) would not retire the kh1 pair its p2/p3 read — drain
+      // fully there (once per kernel, the loop is over anyway).
+      if (p & 1) { if (pre) dtx_vmcnt4(); else dtx_vmcnt0(); }
+      dtx_bar();
+    }
+  }
+  dtx_vmcnt0();
+  dtx_bar();
+
+  // ---- epilogue: stage f32 through LDS ([256][128] f32 = 128 KiB per
+  // round, 2 rounds of 128 columns), re-read coalesced, cvt_pk to bf16,
+  // (optional source add), buffer stores (OOB rows clamp on num_records).
+  const long ldc = (long)N * 2;
+  const unsigned long long c_bytes = (unsigned long long)(M - m0) * ldc;
+  const __amdgpu_buffer_rsrc_t rsC = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)(C + m0 * N), (short)0, (int)(c_bytes > 0xffffffffull
+                                           ? 0xffffffffu : c_bytes),
+      0x00020000);
+  const __amdgpu_buffer_rsrc_t rsS = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)(SRC ? SRC + m0 * N : C), (short)0,
+      (int)(HAS_SRC ? (c_bytes > 0xffffffffull ? 0xffffffffu : c_bytes) : 0),
+      0x00020000);
+  float* fl = reinterpret_cast<float*>(lds);
+
+#pragma unroll
+  for (int rnd = 0; rnd < 2; ++rnd) {
+    if ((wn >> 1) == rnd) {
+      const int colb = (wn & 1) * 64 + (lane & 15);
+#pragma unroll
+      for (int fm = 0; fm < 8; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          const int row = wm * 128 + fm * 16 + (lane >> 4) * 4;
+          const int col = colb + fn * 16;
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            fl[(row + j) * 128 + col] = acc[fm][fn][j];
+        }
+    }
+    dtx_bar();
+    // read-out: 512 threads, 16 passes; thread t handles row t>>5 (+16
+    // per pass), cols (t&31)*4 .. +3 of this round's 128-col block.
+#pragma unroll
+    for (int pass = 0; pass < 16; ++pass) {
+      const int row = pass * 16 + (tid >> 5);
+      const int col = (tid & 31) * 4;
+      const float4v v = *reinterpret_cast<const float4v*>(
+          &fl[row * 128 + col]);
+      const unsigned voff =
+          (unsigned)(row * ldc + (n0 + rnd * 128 + col) * 2);
+      unsigned lo, hi;
+      if (HAS_SRC) {
+        uint2v sv = __builtin_amdgcn_raw_buffer_load_b64(rsS, voff, 0, 0);
+        lo = dtx_cvt_pk_bf16(v[0] + bf2f((unsigned short)(sv.x & 0xffff)),
+                             v[1] + bf2f((unsigned short)(sv.x >> 16)));
+        hi = dtx_cvt_pk_bf16(v[2] + bf2f((unsigned short)(sv.y & 0xffff)),
+                             v[3] + bf2f((unsigned short)(sv.y >> 16)));
+      } else {
+        lo = dtx_cvt_pk_bf16(v[0], v[1]);
+        hi = dtx_cvt_pk_bf16(v[2], v[3]);
+      }
+      uint2v out; out.x = lo; out.y = hi;
+      __builtin_amdgcn_raw_buffer_store_b64(out, rsC, voff, 0, 0);
+    }
+    dtx_bar();
+  }
+#undef STAGE_A
+#undef STAGE_B
+}
+
+void launch_gemm_nt(const void* A, const void* B, const void* SRC, void* C,
+                    long M, int N, int K, hipStream_t stream) {
+  const int mb_n = (int)((M + GEMM_BM - 1) / GEMM_BM);
+  const int nb_n = N / GEMM_BN;
+  dim3 grid(mb_n * nb_n), block(512);
+  if (SRC)
+    hipLaunchKernelGGL(gemm_nt_kernel<true>, grid, block, 0, stream,
+                       (const unsigned short*)A, (const unsigned short*)B,
+                       (const unsigned short*)SRC, (unsigned short*)C,
+                       M, N, K, mb_n);
+  else
+    hipLaunchKernelGGL(gemm_nt_kernel<false>, grid, block, 0, stream,
+                       (const unsigned short*)A, (const unsigned short*)B,
+                       nullptr, (unsigned short*)C, M, N, K, mb_n);
+}

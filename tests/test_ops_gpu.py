@@ -347,3 +347,46 @@ def test_gemv(N, K):
     y = _dtx_hip.gemv(x, w)
     want = torch.nn.functional.linear(x.float(), w.float())
     assert_close(y.cpu(), want.cpu(), name="gemv")
+
+
+# ------------------------------------------------------- base MFMA GEMM
+@pytest.mark.parametrize("M,N,K", [
+    (512, 512, 128),          # minimal tile coverage
+    (1024, 1024, 4096),       # square projection-ish
+    (777, 1024, 4096),        # M tail (SRSRC clamp + store predicate)
+    (2048, 11008, 4096),      # gate/up shape family
+    (1024, 4096, 11008),      # down-proj (big K)
+    (384, 1280, 192),         # odd-but-supported dims
+])
+def test_gemm_nt(M, N, K):
+    a, b = mk(M, K), mk(N, K)
+    c = ops.gemm_nt(a, b)
+    want = a.float() @ b.float().t()
+    assert_close(c, want, name=f"gemm_nt {M}x{N}x{K}")
+
+
+def test_gemm_nt_src_fused():
+    M, N, K = 777, 1024, 512
+    a, b, src = mk(M, K), mk(N, K), mk(M, N)
+    c = ops.gemm_nt(a, b, src)
+    want = a.float() @ b.float().t() + src.float()
+    assert_close(c, want, name="gemm_nt+src")
+
+
+def test_frozen_gemm_autograd():
+    from datatunerx_amd.ops.autograd import FrozenGemm
+    M, N, K = 512, 1024, 512
+    x = mk(M, K)
+    x.requires_grad_(True)
+    w = mk(N, K)
+    wt = w.t().contiguous()
+    res = mk(M, N)
+    res.requires_grad_(True)
+    y = FrozenGemm.apply(x, w, wt, res)
+    dy = mk(M, N)
+    y.backward(dy)
+    want_y = x.detach().float() @ w.float().t() + res.detach().float()
+    want_dx = dy.float() @ w.float()
+    assert_close(y, want_y, name="frozen_gemm y")
+    assert_close(x.grad, want_dx, name="frozen_gemm dx")
+    assert_close(res.grad, dy, name="frozen_gemm dres")
