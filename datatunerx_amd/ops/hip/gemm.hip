@@ -90,7 +90,12 @@ __device__ __forceinline__ void dtx_bar() {
 // f((r>>2)&3) of the swizzle: 0->0, 1->3, 2->2, 3->1  == (-x)&3
 __device__ __forceinline__ unsigned swz_f(unsigned x) { return (0u - x) & 3u; }
 
-template <bool HAS_SRC>
+// GRID: 0 = nb-major linear chunks per XCD; 1 = 8mb x 4nb clusters
+//       inside each XCD chunk (A-panel reuse in L2/L3 as well as B);
+//       requires mb_n % 8 == 0 and nb_n % 4 == 0 (launcher checks).
+// VS:   0 = vmcnt(4) at both odd phases; 1 = vmcnt(4) at phase 3 only,
+//       vmcnt(6) at phase 1 (deeper in-flight window).
+template <bool HAS_SRC, int GRID = 0, int VS = 0>
 __global__ __launch_bounds__(512, 2)
 void gemm_nt_kernel(const unsigned short* __restrict__ A,
                     const unsigned short* __restrict__ B,
@@ -111,7 +116,17 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
   const int xcd = blockIdx.x & 7, pos = blockIdx.x >> 3;
   const int wgid = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q)
                    + pos;
-  const int mb = wgid % mb_n, nb = wgid / mb_n;
+  int mb, nb;
+  if (GRID == 1) {
+    // 32-block clusters of 8(mb) x 4(nb); clusters enumerated nb-major
+    const int mbc = mb_n >> 3;
+    const int cl = wgid >> 5, ci = wgid & 31;
+    mb = (cl % mbc) * 8 + (ci & 7);
+    nb = (cl / mbc) * 4 + (ci >> 3);
+  } else {
+    mb = wgid % mb_n;
+    nb = wgid / mb_n;
+  }
   const long m0 = (long)mb * GEMM_BM;
   const int n0 = nb * GEMM_BN;
 
@@ -125,10 +140,10 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
 
   // ---- per-lane staging constants (swizzle on the SOURCE address)
   // stage: lane l covers (row = base + l>>2, sub-slot pos q = l&3);
-  // the data sub-slot is s = q ^ f((l>>2)&3) (row bits 2..3 == (l>>2)&3
-  // because every stage base row is a multiple of 16).
+  // the data sub-slot is s = q ^ f(row bits 2..3). row = l>>2 and every
+  // stage base row is a multiple of 16, so row bits 2..3 == (l>>4)&3.
   const unsigned st_row = lane >> 2;                   // 0..15
-  const unsigned st_s = (lane & 3) ^ swz_f((lane >> 2) & 3);
+  const unsigned st_s = (lane & 3) ^ swz_f((lane >> 4) & 3);
   // fragment read: lane l reads (row = frag_base + (l&15),
   // slot s = l>>4) at position s ^ f(((l&15)>>2)&3).
   const unsigned fr_off = (lane & 15) * 64
@@ -205,12 +220,19 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       for (int fm = 0; fm < 4; ++fm)
         afr[fm] = *reinterpret_cast<const short8v*>(
             &lds[as + wm * 8192 + (mh * 4 + fm) * 1024 + fr_off]);
-      // issue next tile's stage for this phase
+      // issue next tile's stage for this phase. VS=0: one half per
+      // phase; VS=1: both kh0 halves at p0, both kh1 at p1 (earlier
+      // issue, deeper in-flight window: 8 ops instead of 4).
       if (pre) {
-        if (p == 0) STAGE_B(kt + 1, 0)
-        else if (p == 1) STAGE_A(kt + 1, 0)
-        else if (p == 2) STAGE_B(kt + 1, 1)
-        else STAGE_A(kt + 1, 1)
+        if (VS == 0) {
+          if (p == 0) STAGE_B(kt + 1, 0)
+          else if (p == 1) STAGE_A(kt + 1, 0)
+          else if (p == 2) STAGE_B(kt + 1, 1)
+          else STAGE_A(kt + 1, 1)
+        } else {
+          if (p == 0) { STAGE_B(kt + 1, 0) STAGE_A(kt + 1, 0) }
+          else if (p == 1) { STAGE_B(kt + 1, 1) STAGE_A(kt + 1, 1) }
+        }
       }
       dtx_bar();
       __builtin_amdgcn_s_setprio(1);
@@ -224,7 +246,11 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       // no new stages, so its in-flight count is already <= 4 at p1 and
       // vmcnt(4) would not retire the kh1 pair its p2/p3 read — drain
       // fully there (once per kernel, the loop is over anyway).
-      if (p & 1) { if (pre) dtx_vmcnt4(); else dtx_vmcnt0(); }
+      if (p & 1) {
+        if (!pre) dtx_vmcnt0();
+        else if (p == 3 || VS == 0) dtx_vmcnt4();
+        else asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      }
       dtx_bar();
     }
   }

@@ -5,11 +5,14 @@ interleaved rounds (guide §5.4 rule 24); random [-1,1) data (rule 25).
 
   gpurun -- 'python tools/bench_mygemm.py'
 """
+import os
+import sys
 import time
 
 import torch
 
-from datatunerx_amd import ops
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+from datatunerx_amd import ops  # noqa: E402
 
 assert torch.cuda.is_available() and ops.have_ext()
 dev = torch.device("cuda:0")

@@ -1,0 +1,105 @@
+// Standalone A/B probe for the hand-written NT GEMM (no torch): variants
+// run interleaved in ONE process (guide §5.4 rule 24), random data
+// (rule 25), bitwise cross-check between variants (same math order =>
+// identical bits).
+//   hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/gemm_probe.cpp -o tools/gemm_probe.bin
+//   ./tools/gemm_probe.bin bench 24576 4096 4096 [rounds]
+//   ./tools/gemm_probe.bin one <grid> <vs> 24576 4096 4096 <iters>   # for rocprofv3
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+#include <chrono>
+
+#include "../datatunerx_amd/ops/hip/gemm.hip"
+
+#define CK(x) do { hipError_t e = (x); if (e != hipSuccess) { \
+  fprintf(stderr, "HIP error %s at %s:%d\n", hipGetErrorString(e), \
+          __FILE__, __LINE__); exit(1); } } while (0)
+
+__global__ void fill_rand(unsigned short* p, long n, unsigned seed) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  unsigned long long z = seed + (unsigned long long)i * 0x9E3779B97F4A7C15ull;
+  z ^= z >> 30; z *= 0xBF58476D1CE4E5B9ull; z ^= z >> 27;
+  float f = ((z >> 40) & 0xFFFFFF) / 8388608.0f * 2.f - 1.f;  // [-1,1)
+  union { float f; unsigned u; } u; u.f = f;
+  p[i] = (unsigned short)(u.u >> 16);
+}
+
+template <int GRID, int VS>
+void run(const unsigned short* A, const unsigned short* B,
+         unsigned short* C, long M, int N, int K) {
+  const int mb_n = (int)((M + 255) / 256), nb_n = N / 256;
+  hipLaunchKernelGGL((gemm_nt_kernel<false, GRID, VS>), dim3(mb_n * nb_n),
+                     dim3(512), 0, 0, A, B, nullptr, C, M, N, K, mb_n);
+}
+
+typedef void (*runfn)(const unsigned short*, const unsigned short*,
+                      unsigned short*, long, int, int);
+static runfn FNS[4] = {run<0, 0>, run<1, 0>, run<0, 1>, run<1, 1>};
+static const char* NAMES[4] = {"g0v0", "g1v0", "g0v1", "g1v1"};
+
+int main(int argc, char** argv) {
+  if (argc < 5) { fprintf(stderr, "usage: see header\n"); return 1; }
+  const bool one = !strcmp(argv[1], "one");
+  int ai = one ? 4 : 2;
+  long M = atol(argv[ai]); int N = atoi(argv[ai + 1]), K = atoi(argv[ai + 2]);
+  int rounds = argc > ai + 3 ? atoi(argv[ai + 3]) : 6;
+
+  unsigned short *A, *B, *C;
+  CK(hipMalloc(&A, M * (long)K * 2));
+  CK(hipMalloc(&B, (long)N * K * 2));
+  CK(hipMalloc(&C, M * (long)N * 2));
+  hipLaunchKernelGGL(fill_rand, dim3((M * K + 255) / 256), dim3(256), 0, 0,
+                     A, M * (long)K, 1u);
+  hipLaunchKernelGGL(fill_rand, dim3(((long)N * K + 255) / 256), dim3(256),
+                     0, 0, B, (long)N * K, 2u);
+  CK(hipDeviceSynchronize());
+  const double fl = 2.0 * M * N * K;
+
+  if (one) {
+    int g = atoi(argv[2]), v = atoi(argv[3]);
+    runfn f = FNS[(v ? 2 : 0) + (g ? 1 : 0)];
+    for (int i = 0; i < rounds; ++i) f(A, B, C, M, N, K);
+    CK(hipDeviceSynchronize());
+    printf("done %s\n", NAMES[(v ? 2 : 0) + (g ? 1 : 0)]);
+    return 0;
+  }
+
+  const bool cluster_ok = (M / 256) % 8 == 0 && (N / 256) % 4 == 0
+                          && M % 256 == 0;
+  // bitwise cross-check of variants against variant 0
+  std::vector<unsigned short> ref(4096), got(4096);
+  FNS[0](A, B, C, M, N, K);
+  CK(hipDeviceSynchronize());
+  CK(hipMemcpy(ref.data(), C + M * (long)N / 2, 8192, hipMemcpyDeviceToHost));
+  for (int v = 1; v < 4; ++v) {
+    if ((v & 1) && !cluster_ok) continue;
+    CK(hipMemset(C + M * (long)N / 2, 0, 8192));
+    FNS[v](A, B, C, M, N, K);
+    CK(hipDeviceSynchronize());
+    CK(hipMemcpy(got.data(), C + M * (long)N / 2, 8192,
+                 hipMemcpyDeviceToHost));
+    if (memcmp(ref.data(), got.data(), 8192))
+      printf("MISMATCH variant %s vs g0v0\n", NAMES[v]);
+  }
+
+  double best[4] = {1e30, 1e30, 1e30, 1e30};
+  for (int r = 0; r < rounds; ++r)
+    for (int v = 0; v < 4; ++v) {
+      if ((v & 1) && !cluster_ok) continue;
+      CK(hipDeviceSynchronize());
+      auto t0 = std::chrono::steady_clock::now();
+      for (int i = 0; i < 3; ++i) FNS[v](A, B, C, M, N, K);
+      CK(hipDeviceSynchronize());
+      double dt = std::chrono::duration<double>(
+                      std::chrono::steady_clock::now() - t0).count() / 3;
+      if (dt < best[v]) best[v] = dt;
+    }
+  for (int v = 0; v < 4; ++v)
+    if (best[v] < 1e29)
+      printf("%s  M=%ld N=%d K=%d  %7.3f ms  %7.1f TF/s\n", NAMES[v], M, N,
+             K, best[v] * 1e3, fl / best[v] / 1e12);
+  return 0;
+}
