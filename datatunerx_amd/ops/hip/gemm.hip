@@ -84,25 +84,35 @@ __device__ __forceinline__ void dtx_bar() {
 #define GEMM_BN 256
 #define GEMM_BK 64
 #define HALF_BYTES (256 * 64)          // one [256][32] bf16 half-slot
-#define A_RING 0                       // 4 A slots
-#define B_RING (4 * HALF_BYTES)        // 4 B slots
 
 // f((r>>2)&3) of the swizzle: 0->0, 1->3, 2->2, 3->1  == (-x)&3
 __device__ __forceinline__ unsigned swz_f(unsigned x) { return (0u - x) & 3u; }
 
 // GRID: 0 = nb-major linear chunks per XCD; 1 = 8mb x 4nb clusters
-//       inside each XCD chunk (A-panel reuse in L2/L3 as well as B);
-//       requires mb_n % 8 == 0 and nb_n % 4 == 0 (launcher checks).
-// VS:   0 = vmcnt(4) at both odd phases; 1 = vmcnt(4) at phase 3 only,
-//       vmcnt(6) at phase 1 (deeper in-flight window).
-template <bool HAS_SRC, int GRID = 0, int VS = 0>
+//       inside each XCD chunk (A-panel reuse in L2/L3 as well as B;
+//       needs mb_n % 8 == 0 and nb_n % 4 == 0); 2 = 8-row mb-groups
+//       sweeping all nb (general shapes, L3-bounded window).
+// PIPE: 0 = 4-slot ring, 2 barriers/phase, vmcnt(4) at p1/p3 (lockstep
+//       phases); 1 = 4-slot ring, barriers ONLY at the two publication
+//       points (p1/p3 end) — waves de-lockstep so one wave's loads
+//       overlap its SIMD partner's MFMAs; 2 = 5-slot ring (all 160 KiB
+//       LDS), stage TWO tiles ahead, ONE vmcnt(8) + two barriers per
+//       tile (max latency slack); 3 = ring5 + sparse barriers +
+//       vmcnt(4) at both odd phases (publication one phase earlier) +
+//       fragment PRELOAD: phase p+1's ds_reads issue before phase p's
+//       MFMA cluster (incl. across the tile seam), so LDS latency and
+//       read issue hide inside the MFMA stream.
+template <bool HAS_SRC, int GRID = 0, int PIPE = 0>
 __global__ __launch_bounds__(512, 2)
 void gemm_nt_kernel(const unsigned short* __restrict__ A,
                     const unsigned short* __restrict__ B,
                     const unsigned short* __restrict__ SRC,
                     unsigned short* __restrict__ C,
                     long M, int N, int K, int mb_n) {
-  __shared__ __attribute__((aligned(16))) unsigned char lds[8 * HALF_BYTES];
+  constexpr int RING = (PIPE >= 2) ? 5 : 4;
+  constexpr unsigned B_RING = RING * HALF_BYTES;
+  __shared__ __attribute__((aligned(16))) unsigned char
+      lds[2 * RING * HALF_BYTES];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -111,6 +121,7 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
   const int wm = wid >> 2, wn = wid & 3;
 
   // ---- XCD-aware bijective remap; nb-major chunks (B-panel L2 reuse)
+  const int nbn = N >> 8;
   const int nwg = gridDim.x;
   const int q = nwg >> 3, r8 = nwg & 7;
   const int xcd = blockIdx.x & 7, pos = blockIdx.x >> 3;
@@ -123,6 +134,17 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
     const int cl = wgid >> 5, ci = wgid & 31;
     mb = (cl % mbc) * 8 + (ci & 7);
     nb = (cl / mbc) * 4 + (ci >> 3);
+  } else if (GRID == 2) {
+    // 8-row mb-groups sweeping ALL nb columns, dispatch order (no XCD
+    // remap): the ~256 concurrently-resident blocks then span <= 2
+    // groups, so the A-panel (16 rows x K) and B re-reads stay
+    // L3-resident chip-wide instead of streaming A from HBM once per
+    // nb column (PMC: FETCH dropped ~4x on the mb24 shapes).
+    const int g = blockIdx.x / (nbn * 8);
+    const int rem = blockIdx.x % (nbn * 8);
+    const int rows = min(8, mb_n - g * 8);
+    nb = rem / rows;
+    mb = g * 8 + rem % rows;
   } else {
     mb = wgid % mb_n;
     nb = wgid / mb_n;
@@ -165,49 +187,146 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = float4v{0.f, 0.f, 0.f, 0.f};
 
-  // ---- staging helper: stage half (a_or_b, kt, kh): 2 glds per wave
-#define STAGE_A(kt, kh)                                                     \
+  // ---- staging helpers: one half = 2 glds per wave (its 2 KiB slice).
+  // `slot` is the ring slot byte offset, `kt`/`kh` pick the source k.
+#define STAGE_A(slot, kt, kh)                                               \
   {                                                                         \
-    const unsigned slot = ((unsigned)(2 * (kt) + (kh)) & 3) * HALF_BYTES;   \
     const long kb = (long)(kt) * 128 + (kh) * 64 + st_s * 16;               \
-    glds16(srdA, (unsigned)(a_row0 * ldab + kb), A_RING + slot + st_lds);   \
+    glds16(srdA, (unsigned)(a_row0 * ldab + kb), (slot) + st_lds);          \
     glds16(srdA, (unsigned)((a_row0 + 16) * ldab + kb),                     \
-           A_RING + slot + st_lds + 1024);                                  \
+           (slot) + st_lds + 1024);                                         \
   }
-#define STAGE_B(kt, kh)                                                     \
+#define STAGE_B(slot, kt, kh)                                               \
   {                                                                         \
-    const unsigned slot = ((unsigned)(2 * (kt) + (kh)) & 3) * HALF_BYTES;   \
     const long kb = (long)(kt) * 128 + (kh) * 64 + st_s * 16;               \
-    glds16(srdB, (unsigned)(b_row0 * ldab + kb), B_RING + slot + st_lds);   \
+    glds16(srdB, (unsigned)(b_row0 * ldab + kb), B_RING + (slot) + st_lds); \
     glds16(srdB, (unsigned)((b_row0 + 16) * ldab + kb),                     \
-           B_RING + slot + st_lds + 1024);                                  \
+           B_RING + (slot) + st_lds + 1024);                                \
   }
+#define SLOT4(h) ((unsigned)((h) & 3) * HALF_BYTES)
+#define SLOT5(h) ((unsigned)((h) % 5) * HALF_BYTES)
 
-  // ---- prologue: stage tile 0 (kh0 pair first), wait for the kh0 pair
-  STAGE_B(0, 0);
-  STAGE_A(0, 0);
-  STAGE_B(0, 1);
-  STAGE_A(0, 1);
-  dtx_vmcnt4();
+  // ---- prologue
+  if (PIPE >= 2) {
+    // stage tiles 0 and 1 (slots h%5), wait for tile 0's 8 ops
+    STAGE_B(SLOT5(0), 0, 0); STAGE_A(SLOT5(0), 0, 0);
+    STAGE_B(SLOT5(1), 0, 1); STAGE_A(SLOT5(1), 0, 1);
+    STAGE_B(SLOT5(2), 1, 0); STAGE_A(SLOT5(2), 1, 0);
+    STAGE_B(SLOT5(3), 1, 1); STAGE_A(SLOT5(3), 1, 1);
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  } else {
+    // stage tile 0 (kh0 pair first), wait for the kh0 pair
+    STAGE_B(SLOT4(0), 0, 0);
+    STAGE_A(SLOT4(0), 0, 0);
+    STAGE_B(SLOT4(1), 0, 1);
+    STAGE_A(SLOT4(1), 0, 1);
+    dtx_vmcnt4();
+  }
   dtx_bar();
 
   // ---- K loop: 4 phases per tile; phase p: kh = p>>1, m-half = p&1.
   // B fragments are read at even phases and reused at the odd phase.
-  // Stage schedule (tile kt stages kt+1): p0: B-kh0, p1: A-kh0,
-  // p2: B-kh1, p3: A-kh1. vmcnt(4) at the ends of p1 and p3 retires
-  // exactly the halves the next two phases read (2 half-tiles stay in
-  // flight across each barrier).
+  // PIPE 0/1 (4-ring): tile kt stages kt+1, one half per phase
+  //   (B-kh0@p0, A-kh0@p1, B-kh1@p2, A-kh1@p3); vmcnt(4) at p1/p3 ends
+  //   retires exactly the halves the next two phases read.
+  // PIPE 2 (5-ring): tile kt stages kt+2; ONE vmcnt(8) at p3 end
+  //   retires ALL four halves of tile kt+1 (staged during kt-1);
+  //   barriers at p1/p3 ends gate slot reuse (WAR) and publication.
   short8v bfr[4];
+  short8v afr2[2][4], bfr2[2][4];      // PIPE3 double-buffered fragments
+  // ring-5 slot indices maintained incrementally ((2kt+i) % 5)
+  int h5 = 0;                          // (2*kt) % 5
+
+#define RD_A(dst, as, mh)                                                   \
+  _Pragma("unroll") for (int fm = 0; fm < 4; ++fm)                          \
+      dst[fm] = *reinterpret_cast<const short8v*>(                          \
+          &lds[(as) + wm * 8192 + ((mh) * 4 + fm) * 1024 + fr_off]);
+#define RD_B(dst, bs)                                                       \
+  _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                          \
+      dst[fn] = *reinterpret_cast<const short8v*>(                          \
+          &lds[(bs) + wn * 4096 + fn * 1024 + fr_off]);
+#define MFMA_PHASE(af, bf, mh)                                              \
+  _Pragma("unroll") for (int fm = 0; fm < 4; ++fm)                          \
+      _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                      \
+          acc[(mh) * 4 + fm][fn] =                                          \
+              MFMA16(af[fm], bf[fn], acc[(mh) * 4 + fm][fn]);
+
+  if (PIPE == 3) {
+    // preload tile 0 phase 0 fragments
+    RD_A(afr2[0], 0u, 0);
+    RD_B(bfr2[0], B_RING + 0u);
+    for (int kt = 0; kt < KT; ++kt) {
+      const unsigned as0 = (unsigned)h5 * HALF_BYTES;
+      const int h5b = h5 + 1 - (h5 + 1 >= 5 ? 5 : 0);
+      const unsigned as1 = (unsigned)h5b * HALF_BYTES;
+      const int h5s0 = h5 + 4 - (h5 + 4 >= 5 ? 5 : 0);
+      const int h5s1 = h5s0 + 1 - (h5s0 + 1 >= 5 ? 5 : 0);
+      const unsigned ss0 = (unsigned)h5s0 * HALF_BYTES;
+      const unsigned ss1 = (unsigned)h5s1 * HALF_BYTES;
+      const bool pre = kt + 2 < KT;
+      const bool last = kt + 1 == KT;
+      h5 = h5 + 2 - (h5 + 2 >= 5 ? 5 : 0);
+      const unsigned as0n = (unsigned)h5 * HALF_BYTES;  // next tile kh0
+#pragma unroll
+      for (int p = 0; p < 4; ++p) {
+        const unsigned mh = p & 1;
+        if (pre) {
+          if (p == 0) STAGE_B(ss0, kt + 2, 0)
+          else if (p == 1) STAGE_A(ss0, kt + 2, 0)
+          else if (p == 2) STAGE_B(ss1, kt + 2, 1)
+          else STAGE_A(ss1, kt + 2, 1)
+        }
+        // preload next phase's fragments (phase 3 preloads the NEXT
+        // tile's phase 0: its halves were published at this tile's
+        // p1-end vmcnt, and its slot's next writer stages at kt+1.p2 —
+        // after the read).
+        if (p == 0) {
+          RD_A(afr2[1], as0, 1);
+        } else if (p == 1) {
+          RD_A(afr2[0], as1, 0);
+          RD_B(bfr2[1], as1 + B_RING);
+        } else if (p == 2) {
+          RD_A(afr2[1], as1, 1);
+        } else if (!last) {
+          RD_A(afr2[0], as0n, 0);
+          RD_B(bfr2[0], as0n + B_RING);
+        }
+        __builtin_amdgcn_s_setprio(1);
+        if (p == 0) { MFMA_PHASE(afr2[0], bfr2[0], 0); }
+        else if (p == 1) { MFMA_PHASE(afr2[1], bfr2[0], 1); }
+        else if (p == 2) { MFMA_PHASE(afr2[0], bfr2[1], 0); }
+        else { MFMA_PHASE(afr2[1], bfr2[1], 1); }
+        __builtin_amdgcn_s_setprio(0);
+        if (p & 1) {
+          if (pre) dtx_vmcnt4(); else dtx_vmcnt0();
+          dtx_bar();
+        }
+      }
+    }
+  } else
   for (int kt = 0; kt < KT; ++kt) {
-    const unsigned aslot = A_RING + ((unsigned)(2 * kt) & 3) * HALF_BYTES;
-    const unsigned bslot = B_RING + ((unsigned)(2 * kt) & 3) * HALF_BYTES;
-    const bool pre = kt + 1 < KT;
+
+    const unsigned as0 = (PIPE == 2) ? (unsigned)h5 * HALF_BYTES
+                                     : SLOT4(2 * kt);
+    const int h5b = h5 + 1 - (h5 + 1 >= 5 ? 5 : 0);
+    const unsigned as1 = (PIPE == 2) ? (unsigned)h5b * HALF_BYTES
+                                     : SLOT4(2 * kt + 1);
+    // stage destination slots (kt+1 for 4-ring, kt+2 for 5-ring)
+    const int h5s0 = h5 + 4 - (h5 + 4 >= 5 ? 5 : 0);   // (2kt+4) % 5
+    const int h5s1 = h5s0 + 1 - (h5s0 + 1 >= 5 ? 5 : 0);
+    const unsigned ss0 = (PIPE == 2) ? (unsigned)h5s0 * HALF_BYTES
+                                     : SLOT4(2 * kt + 2);
+    const unsigned ss1 = (PIPE == 2) ? (unsigned)h5s1 * HALF_BYTES
+                                     : SLOT4(2 * kt + 3);
+    const int kt_s = (PIPE == 2) ? kt + 2 : kt + 1;
+    const bool pre = kt_s < KT;
+    h5 = h5 + 2 - (h5 + 2 >= 5 ? 5 : 0);
 
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      const unsigned kh = p >> 1, mh = p & 1;
-      const unsigned as = aslot + kh * HALF_BYTES;
-      const unsigned bs = bslot + kh * HALF_BYTES;
+      const unsigned mh = p & 1;
+      const unsigned as = (p < 2) ? as0 : as1;
+      const unsigned bs = ((p < 2) ? as0 : as1) + B_RING;
       // ds_read register subtile for this phase
       short8v afr[4];
       if (mh == 0) {
@@ -220,21 +339,14 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       for (int fm = 0; fm < 4; ++fm)
         afr[fm] = *reinterpret_cast<const short8v*>(
             &lds[as + wm * 8192 + (mh * 4 + fm) * 1024 + fr_off]);
-      // issue next tile's stage for this phase. VS=0: one half per
-      // phase; VS=1: both kh0 halves at p0, both kh1 at p1 (earlier
-      // issue, deeper in-flight window: 8 ops instead of 4).
+      // issue the stage for this phase (one half per phase)
       if (pre) {
-        if (VS == 0) {
-          if (p == 0) STAGE_B(kt + 1, 0)
-          else if (p == 1) STAGE_A(kt + 1, 0)
-          else if (p == 2) STAGE_B(kt + 1, 1)
-          else STAGE_A(kt + 1, 1)
-        } else {
-          if (p == 0) { STAGE_B(kt + 1, 0) STAGE_A(kt + 1, 0) }
-          else if (p == 1) { STAGE_B(kt + 1, 1) STAGE_A(kt + 1, 1) }
-        }
+        if (p == 0) STAGE_B(ss0, kt_s, 0)
+        else if (p == 1) STAGE_A(ss0, kt_s, 0)
+        else if (p == 2) STAGE_B(ss1, kt_s, 1)
+        else STAGE_A(ss1, kt_s, 1)
       }
-      dtx_bar();
+      if (PIPE == 0) dtx_bar();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
@@ -242,16 +354,20 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
         for (int fn = 0; fn < 4; ++fn)
           acc[mh * 4 + fm][fn] = MFMA16(afr[fm], bfr[fn], acc[mh * 4 + fm][fn]);
       __builtin_amdgcn_s_setprio(0);
-      // counted waits only (loads span barriers); the LAST tile issues
-      // no new stages, so its in-flight count is already <= 4 at p1 and
-      // vmcnt(4) would not retire the kh1 pair its p2/p3 read — drain
-      // fully there (once per kernel, the loop is over anyway).
+      // publication waits: counted only (loads span barriers); the last
+      // tile(s) issue no stages, so their in-flight count is too low
+      // for the counted wait to retire what the next phases read —
+      // drain fully there (end of the loop anyway).
       if (p & 1) {
-        if (!pre) dtx_vmcnt0();
-        else if (p == 3 || VS == 0) dtx_vmcnt4();
-        else asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        if (PIPE == 2) {
+          if (p == 3) { if (pre) { asm volatile(
+              "s_waitcnt vmcnt(8)" ::: "memory"); } else dtx_vmcnt0(); }
+          dtx_bar();
+        } else {
+          if (!pre) dtx_vmcnt0(); else dtx_vmcnt4();
+          dtx_bar();
+        }
       }
-      dtx_bar();
     }
   }
   dtx_vmcnt0();
@@ -316,6 +432,8 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
   }
 #undef STAGE_A
 #undef STAGE_B
+#undef SLOT4
+#undef SLOT5
 }
 
 void launch_gemm_nt(const void* A, const void* B, const void* SRC, void* C,

@@ -27,23 +27,26 @@ __global__ void fill_rand(unsigned short* p, long n, unsigned seed) {
   p[i] = (unsigned short)(u.u >> 16);
 }
 
-template <int GRID, int VS>
+template <int GRID, int PIPE>
 void run(const unsigned short* A, const unsigned short* B,
          unsigned short* C, long M, int N, int K) {
   const int mb_n = (int)((M + 255) / 256), nb_n = N / 256;
-  hipLaunchKernelGGL((gemm_nt_kernel<false, GRID, VS>), dim3(mb_n * nb_n),
+  hipLaunchKernelGGL((gemm_nt_kernel<false, GRID, PIPE>), dim3(mb_n * nb_n),
                      dim3(512), 0, 0, A, B, nullptr, C, M, N, K, mb_n);
 }
 
 typedef void (*runfn)(const unsigned short*, const unsigned short*,
                       unsigned short*, long, int, int);
-static runfn FNS[4] = {run<0, 0>, run<1, 0>, run<0, 1>, run<1, 1>};
-static const char* NAMES[4] = {"g0v0", "g1v0", "g0v1", "g1v1"};
+#define NVAR 6
+static runfn FNS[NVAR] = {run<1, 1>, run<1, 2>, run<1, 3>,
+                          run<2, 1>, run<2, 2>, run<2, 3>};
+static const char* NAMES[NVAR] = {"g1p1", "g1p2", "g1p3",
+                                  "g2p1", "g2p2", "g2p3"};
 
 int main(int argc, char** argv) {
   if (argc < 5) { fprintf(stderr, "usage: see header\n"); return 1; }
   const bool one = !strcmp(argv[1], "one");
-  int ai = one ? 4 : 2;
+  int ai = one ? 3 : 2;  // one <grid> M N K [iters]
   long M = atol(argv[ai]); int N = atoi(argv[ai + 1]), K = atoi(argv[ai + 2]);
   int rounds = argc > ai + 3 ? atoi(argv[ai + 3]) : 6;
 
@@ -59,11 +62,10 @@ int main(int argc, char** argv) {
   const double fl = 2.0 * M * N * K;
 
   if (one) {
-    int g = atoi(argv[2]), v = atoi(argv[3]);
-    runfn f = FNS[(v ? 2 : 0) + (g ? 1 : 0)];
-    for (int i = 0; i < rounds; ++i) f(A, B, C, M, N, K);
+    int g = atoi(argv[2]);
+    for (int i = 0; i < rounds; ++i) FNS[g](A, B, C, M, N, K);
     CK(hipDeviceSynchronize());
-    printf("done %s\n", NAMES[(v ? 2 : 0) + (g ? 1 : 0)]);
+    printf("done %s\n", NAMES[g]);
     return 0;
   }
 
@@ -71,11 +73,11 @@ int main(int argc, char** argv) {
                           && M % 256 == 0;
   // bitwise cross-check of variants against variant 0
   std::vector<unsigned short> ref(4096), got(4096);
-  FNS[0](A, B, C, M, N, K);
+  FNS[3](A, B, C, M, N, K);
   CK(hipDeviceSynchronize());
   CK(hipMemcpy(ref.data(), C + M * (long)N / 2, 8192, hipMemcpyDeviceToHost));
-  for (int v = 1; v < 4; ++v) {
-    if ((v & 1) && !cluster_ok) continue;
+  for (int v = 0; v < NVAR; ++v) {
+    if (v == 3 || (v < 3 && !cluster_ok)) continue;
     CK(hipMemset(C + M * (long)N / 2, 0, 8192));
     FNS[v](A, B, C, M, N, K);
     CK(hipDeviceSynchronize());
@@ -85,10 +87,10 @@ int main(int argc, char** argv) {
       printf("MISMATCH variant %s vs g0v0\n", NAMES[v]);
   }
 
-  double best[4] = {1e30, 1e30, 1e30, 1e30};
+  double best[NVAR] = {1e30, 1e30, 1e30, 1e30, 1e30, 1e30};
   for (int r = 0; r < rounds; ++r)
-    for (int v = 0; v < 4; ++v) {
-      if ((v & 1) && !cluster_ok) continue;
+    for (int v = 0; v < NVAR; ++v) {
+      if (v < 3 && !cluster_ok) continue;
       CK(hipDeviceSynchronize());
       auto t0 = std::chrono::steady_clock::now();
       for (int i = 0; i < 3; ++i) FNS[v](A, B, C, M, N, K);
@@ -97,7 +99,7 @@ int main(int argc, char** argv) {
                       std::chrono::steady_clock::now() - t0).count() / 3;
       if (dt < best[v]) best[v] = dt;
     }
-  for (int v = 0; v < 4; ++v)
+  for (int v = 0; v < NVAR; ++v)
     if (best[v] < 1e29)
       printf("%s  M=%ld N=%d K=%d  %7.3f ms  %7.1f TF/s\n", NAMES[v], M, N,
              K, best[v] * 1e3, fl / best[v] / 1e12);
