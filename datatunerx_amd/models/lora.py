@@ -29,7 +29,15 @@ def _weight_t(mod: nn.Module):
     and the dgrad (dy @ W) then run the hand-written NT MFMA GEMM with
     the contraction contiguous — 288 GB HBM3E per MI355X makes the
     persistent transposed copy free. Built lazily on first GPU forward;
-    None when the shape is outside the kernel's support."""
+    None when the shape is outside the kernel's support.
+
+    Gated by DTX_CUSTOM_GEMM=1: the hand-written kernel currently
+    measures 1.25-1.33 PF/s on the mb24 training shapes vs the tuned
+    hipBLASLt picks' 1.47-1.67 (tools/bench_mygemm.py, within-probe
+    A/B), so routing the 60%-of-step base GEMMs through it would
+    regress tokens/s; it stays opt-in until it wins."""
+    if os.environ.get("DTX_CUSTOM_GEMM", "0") != "1":
+        return None
     wt = getattr(mod, "_wt", False)
     if wt is False:
         w = mod.weight

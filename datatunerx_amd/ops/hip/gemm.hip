@@ -42,6 +42,8 @@ typedef __attribute__((ext_vector_type(4))) unsigned uint4v;
 typedef __attribute__((ext_vector_type(2))) unsigned uint2v;
 
 #define MFMA16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
+#define MFMA32(a, b, c) __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
+typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 // Buffer resource descriptor (T8/T20): built from readfirstlane'd scalars
 // so hipcc proves uniformity (no waterfall loops), num_records = bytes so
@@ -102,7 +104,10 @@ __device__ __forceinline__ unsigned swz_f(unsigned x) { return (0u - x) & 3u; }
 //       fragment PRELOAD: phase p+1's ds_reads issue before phase p's
 //       MFMA cluster (incl. across the tile seam), so LDS latency and
 //       read issue hide inside the MFMA stream.
-template <bool HAS_SRC, int GRID = 0, int PIPE = 0>
+// MF:   0 = mfma_f32_16x16x32_bf16; 1 = mfma_f32_32x32x16_bf16 (the
+//       higher-ceiling intrinsic: 2382 vs 2075 TF ubench — and half the
+//       MFMA + fragment-read instructions per phase).
+template <bool HAS_SRC, int GRID = 0, int PIPE = 0, int MF = 0>
 __global__ __launch_bounds__(512, 2)
 void gemm_nt_kernel(const unsigned short* __restrict__ A,
                     const unsigned short* __restrict__ B,
@@ -181,11 +186,27 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
 
   const int KT = K / GEMM_BK;
 
-  float4v acc[8][4];
+  float4v acc[8][4];          // MF0 accumulators (8x4 16x16 frags)
+  f32x16 acc32[4][2];         // MF1 accumulators (4x2 32x32 frags)
+  if (MF == 0) {
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
+    for (int i = 0; i < 8; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = float4v{0.f, 0.f, 0.f, 0.f};
+      for (int j = 0; j < 4; ++j) acc[i][j] = float4v{0.f, 0.f, 0.f, 0.f};
+  } else {
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int e = 0; e < 16; ++e) acc32[i][j][e] = 0.f;
+  }
+  // MF1 fragment addressing: lane covers row (l&31) of its 32-row frag
+  // at 16-B sub-slot (ks*2 + (l>>5)) ^ f(((l&31)>>2)&3) of the 64-B row.
+  const unsigned f31 = lane & 31;
+  const unsigned mf1_posx = swz_f((f31 >> 2) & 3);
+  const unsigned mf1_row = f31 * 64;
+  const unsigned mf1_hi = lane >> 5;
 
   // ---- staging helpers: one half = 2 glds per wave (its 2 KiB slice).
   // `slot` is the ring slot byte offset, `kt`/`kh` pick the source k.
@@ -329,16 +350,30 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       const unsigned bs = ((p < 2) ? as0 : as1) + B_RING;
       // ds_read register subtile for this phase
       short8v afr[4];
-      if (mh == 0) {
+      short8v bfr32[2];
+      if (MF == 0) {
+        if (mh == 0) {
 #pragma unroll
-        for (int fn = 0; fn < 4; ++fn)
-          bfr[fn] = *reinterpret_cast<const short8v*>(
-              &lds[bs + wn * 4096 + fn * 1024 + fr_off]);
+          for (int fn = 0; fn < 4; ++fn)
+            bfr[fn] = *reinterpret_cast<const short8v*>(
+                &lds[bs + wn * 4096 + fn * 1024 + fr_off]);
+        }
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+          afr[fm] = *reinterpret_cast<const short8v*>(
+              &lds[as + wm * 8192 + (mh * 4 + fm) * 1024 + fr_off]);
+      } else {
+        // MF1: phase p = kstep; A 4 frags (32 rows), B 2 frags
+        const unsigned pos = ((mh * 2 + mf1_hi) ^ mf1_posx) * 16;
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+          afr[fm] = *reinterpret_cast<const short8v*>(
+              &lds[as + wm * 8192 + fm * 2048 + mf1_row + pos]);
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+          bfr32[fn] = *reinterpret_cast<const short8v*>(
+              &lds[bs + wn * 4096 + fn * 2048 + mf1_row + pos]);
       }
-#pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
-        afr[fm] = *reinterpret_cast<const short8v*>(
-            &lds[as + wm * 8192 + (mh * 4 + fm) * 1024 + fr_off]);
       // issue the stage for this phase (one half per phase)
       if (pre) {
         if (p == 0) STAGE_B(ss0, kt_s, 0)
@@ -348,11 +383,20 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       }
       if (PIPE == 0) dtx_bar();
       __builtin_amdgcn_s_setprio(1);
+      if (MF == 0) {
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
+        for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-        for (int fn = 0; fn < 4; ++fn)
-          acc[mh * 4 + fm][fn] = MFMA16(afr[fm], bfr[fn], acc[mh * 4 + fm][fn]);
+          for (int fn = 0; fn < 4; ++fn)
+            acc[mh * 4 + fm][fn] =
+                MFMA16(afr[fm], bfr[fn], acc[mh * 4 + fm][fn]);
+      } else {
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+            acc32[fm][fn] = MFMA32(afr[fm], bfr32[fn], acc32[fm][fn]);
+      }
       __builtin_amdgcn_s_setprio(0);
       // publication waits: counted only (loads span barriers); the last
       // tile(s) issue no stages, so their in-flight count is too low
@@ -391,17 +435,33 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
 #pragma unroll
   for (int rnd = 0; rnd < 2; ++rnd) {
     if ((wn >> 1) == rnd) {
-      const int colb = (wn & 1) * 64 + (lane & 15);
+      if (MF == 0) {
+        const int colb = (wn & 1) * 64 + (lane & 15);
 #pragma unroll
-      for (int fm = 0; fm < 8; ++fm)
+        for (int fm = 0; fm < 8; ++fm)
 #pragma unroll
-        for (int fn = 0; fn < 4; ++fn) {
-          const int row = wm * 128 + fm * 16 + (lane >> 4) * 4;
-          const int col = colb + fn * 16;
+          for (int fn = 0; fn < 4; ++fn) {
+            const int row = wm * 128 + fm * 16 + (lane >> 4) * 4;
+            const int col = colb + fn * 16;
 #pragma unroll
-          for (int j = 0; j < 4; ++j)
-            fl[(row + j) * 128 + col] = acc[fm][fn][j];
-        }
+            for (int j = 0; j < 4; ++j)
+              fl[(row + j) * 128 + col] = acc[fm][fn][j];
+          }
+      } else {
+        // 32x32 C layout: col = lane&31, row = (reg&3)+8*(reg>>2)+4*hi
+        const int colb = (wn & 1) * 64 + (lane & 31);
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn) {
+            const int rowb = wm * 128 + fm * 32 + (lane >> 5) * 4;
+            const int col = colb + fn * 32;
+#pragma unroll
+            for (int r = 0; r < 16; ++r)
+              fl[(rowb + (r & 3) + 8 * (r >> 2)) * 128 + col]
+                  = acc32[fm][fn][r];
+          }
+      }
     }
     dtx_bar();
     // read-out: 512 threads, 16 passes; thread t handles row t>>5 (+16
@@ -441,13 +501,16 @@ void launch_gemm_nt(const void* A, const void* B, const void* SRC, void* C,
   const int mb_n = (int)((M + GEMM_BM - 1) / GEMM_BM);
   const int nb_n = N / GEMM_BN;
   dim3 grid(mb_n * nb_n), block(512);
-  if (SRC)
-    hipLaunchKernelGGL(gemm_nt_kernel<true>, grid, block, 0, stream,
-                       (const unsigned short*)A, (const unsigned short*)B,
-                       (const unsigned short*)SRC, (unsigned short*)C,
-                       M, N, K, mb_n);
-  else
-    hipLaunchKernelGGL(gemm_nt_kernel<false>, grid, block, 0, stream,
-                       (const unsigned short*)A, (const unsigned short*)B,
-                       nullptr, (unsigned short*)C, M, N, K, mb_n);
+  const bool cluster = (mb_n % 8 == 0) && (nb_n % 4 == 0) && (M % 256 == 0);
+  // best measured config (tools/gemm_probe.bin A/B): PIPE=2 (5-slot
+  // ring + sparse barriers), GRID=1 clusters when divisible else
+  // GRID=2 snake groups.
+#define LAUNCH(HS, G)                                                     \
+  hipLaunchKernelGGL((gemm_nt_kernel<HS, G, 2, 0>), grid, block, 0,       \
+                     stream, (const unsigned short*)A,                    \
+                     (const unsigned short*)B, (const unsigned short*)SRC,\
+                     (unsigned short*)C, M, N, K, mb_n)
+  if (SRC) { if (cluster) LAUNCH(true, 1); else LAUNCH(true, 2); }
+  else     { if (cluster) LAUNCH(false, 1); else LAUNCH(false, 2); }
+#undef LAUNCH
 }

@@ -27,21 +27,22 @@ __global__ void fill_rand(unsigned short* p, long n, unsigned seed) {
   p[i] = (unsigned short)(u.u >> 16);
 }
 
-template <int GRID, int PIPE>
+template <int GRID, int PIPE, int MF = 0>
 void run(const unsigned short* A, const unsigned short* B,
          unsigned short* C, long M, int N, int K) {
   const int mb_n = (int)((M + 255) / 256), nb_n = N / 256;
-  hipLaunchKernelGGL((gemm_nt_kernel<false, GRID, PIPE>), dim3(mb_n * nb_n),
+  hipLaunchKernelGGL((gemm_nt_kernel<false, GRID, PIPE, MF>),
+                     dim3(mb_n * nb_n),
                      dim3(512), 0, 0, A, B, nullptr, C, M, N, K, mb_n);
 }
 
 typedef void (*runfn)(const unsigned short*, const unsigned short*,
                       unsigned short*, long, int, int);
 #define NVAR 6
-static runfn FNS[NVAR] = {run<1, 1>, run<1, 2>, run<1, 3>,
-                          run<2, 1>, run<2, 2>, run<2, 3>};
-static const char* NAMES[NVAR] = {"g1p1", "g1p2", "g1p3",
-                                  "g2p1", "g2p2", "g2p3"};
+static runfn FNS[NVAR] = {run<1, 2, 0>, run<1, 2, 1>, run<1, 0, 1>,
+                          run<2, 2, 0>, run<2, 2, 1>, run<2, 1, 1>};
+static const char* NAMES[NVAR] = {"g1p2mf0", "g1p2mf1", "g1p0mf1",
+                                  "g2p2mf0", "g2p2mf1", "g2p1mf1"};
 
 int main(int argc, char** argv) {
   if (argc < 5) { fprintf(stderr, "usage: see header\n"); return 1; }
@@ -83,8 +84,18 @@ int main(int argc, char** argv) {
     CK(hipDeviceSynchronize());
     CK(hipMemcpy(got.data(), C + M * (long)N / 2, 8192,
                  hipMemcpyDeviceToHost));
-    if (memcmp(ref.data(), got.data(), 8192))
-      printf("MISMATCH variant %s vs g0v0\n", NAMES[v]);
+    // MF1 variants have a different accumulation split (32x32x16 vs
+    // 16x16x32) so compare with tolerance, not bitwise
+    float mx = 0, sc = 0;
+    for (int i = 0; i < 4096; ++i) {
+      union { unsigned u; float f; } a, b;
+      a.u = (unsigned)ref[i] << 16; b.u = (unsigned)got[i] << 16;
+      float d = a.f - b.f; if (d < 0) d = -d; if (d > mx) mx = d;
+      float m = a.f < 0 ? -a.f : a.f; if (m > sc) sc = m;
+    }
+    if (mx > 0.02f * (sc > 1 ? sc : 1))
+      printf("MISMATCH variant %s vs base: maxdiff %f scale %f\n",
+             NAMES[v], mx, sc);
   }
 
   double best[NVAR] = {1e30, 1e30, 1e30, 1e30, 1e30, 1e30};
