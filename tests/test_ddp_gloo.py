@@ -131,3 +131,102 @@ def test_allreduce_flat_mean():
         assert p.exitcode == 0
     for _, v in vals:
         assert v == pytest.approx(1.5)
+
+
+RESUME_WORKER = r"""
+import json, os, sys
+import torch
+sys.path.insert(0, os.environ["DTX_ROOT"])
+from datatunerx_amd.data.dataset import SFTDataset
+from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+from datatunerx_amd.parallel.ddp import init_distributed
+from datatunerx_amd.train.trainer import SFTTrainer, TrainerConfig
+
+rank, world, local, device = init_distributed(backend="gloo")
+mode = os.environ["DTX_OPT_MODE"]
+phase = os.environ["DTX_PHASE"]      # "full" | "save" | "resume"
+out = os.environ["DTX_OUT"]
+
+
+def build(max_steps, save_steps=0):
+    torch.manual_seed(7)
+    cfg = LlamaConfig.tiny()
+    model = LlamaForCausalLM(cfg, lora=True,
+                             dtype=torch.float32).init_random()
+    ds = SFTDataset.synthetic(32, 32, 512, seed=0)
+    tr = SFTTrainer(model, ds,
+                    TrainerConfig(output_dir=out + f"/{phase}_r{rank}",
+                                  max_steps=max_steps,
+                                  save_steps=save_steps,
+                                  micro_batch_size=2,
+                                  optimizer_mode=mode,
+                                  comm_bucket_bytes=1 << 12,
+                                  logging_steps=0, learning_rate=1e-3,
+                                  lora_dropout=0.0),
+                    device=device, rank=rank, world_size=world)
+    return tr
+
+
+if phase == "full":
+    tr = build(4)
+    tr.train()
+elif phase == "save":
+    # save at step 2 into a SHARED dir, stop at 2
+    tr = build(2, save_steps=2)
+    tr.cfg.output_dir = out + "/shared"
+    tr.train()
+    sys.exit(0)
+else:
+    tr = build(4)
+    tr.load_checkpoint(out + "/shared/checkpoint-2")
+    tr.train()
+
+res = {"params": tr.opt.param_flat.tolist()[:128],
+       "master": tr.opt.master.tolist()[:64],
+       "t": tr.opt.t}
+with open(out + f"/{phase}_rank{rank}.json", "w") as f:
+    json.dump(res, f)
+import torch.distributed as dist
+if dist.is_initialized():
+    dist.destroy_process_group()
+"""
+
+
+def _run_resume(nproc, out_dir, port, mode, phase):
+    script = os.path.join(out_dir, "resume_worker.py")
+    with open(script, "w") as f:
+        f.write(RESUME_WORKER)
+    procs = []
+    for rank in range(nproc):
+        env = dict(os.environ)
+        env.update({"RANK": str(rank), "WORLD_SIZE": str(nproc),
+                    "LOCAL_RANK": str(rank),
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+                    "DTX_ROOT": ROOT, "DTX_OUT": out_dir,
+                    "DTX_OPT_MODE": mode, "DTX_PHASE": phase})
+        procs.append(subprocess.Popen([sys.executable, script], env=env))
+    for p in procs:
+        assert p.wait(timeout=300) == 0
+
+
+@pytest.mark.parametrize("mode,port", [("flat", 29722), ("zero1", 29726)])
+def test_save_restart_resume_matches_uninterrupted(tmp_path, mode, port):
+    """4 straight steps == 2 steps -> checkpoint -> restart -> 2 more, for
+    both the replicated and the ZeRO-1-sharded optimizer (VERDICT r1
+    weak #2: every rank saves its shard; resume all-gathers the restored
+    params before the first forward)."""
+    out = str(tmp_path)
+    _run_resume(2, out, port, mode, "full")
+    _run_resume(2, out, port + 1, mode, "save")
+    assert os.path.exists(os.path.join(out, "shared/checkpoint-2",
+                                       "trainer_state.pt"))
+    if mode == "zero1":
+        assert os.path.exists(os.path.join(
+            out, "shared/checkpoint-2", "trainer_state_rank1.pt"))
+    _run_resume(2, out, port + 2, mode, "resume")
+    for rank in range(2):
+        full = json.load(open(os.path.join(out, f"full_rank{rank}.json")))
+        res = json.load(open(os.path.join(out, f"resume_rank{rank}.json")))
+        assert res["t"] == full["t"]
+        assert res["params"] == pytest.approx(full["params"], abs=2e-6)
+        assert res["master"] == pytest.approx(full["master"], abs=2e-6)
