@@ -60,6 +60,11 @@ def validate_(obj: ApiObject) -> None:
     if isinstance(obj, FinetuneJob):
         ft = (obj.spec.get("fineTune") or {}).get("finetuneSpec") or {}
         _validate_finetune_spec(ft, "spec.fineTune.finetuneSpec", errs)
+        sc = obj.spec.get("serveConfig") or {}
+        tp = sc.get("tensorParallel", sc.get("gpus", 1))
+        _require(isinstance(tp, int) and 1 <= tp <= 8,
+                 "spec.serveConfig.tensorParallel must be an int in "
+                 "[1, 8] (one 8-GPU MI355X node)", errs)
     elif isinstance(obj, Finetune):
         _validate_finetune_spec(obj.spec, "spec", errs)
     elif isinstance(obj, FinetuneExperiment):

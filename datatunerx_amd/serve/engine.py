@@ -223,9 +223,19 @@ class InferenceEngine:
                                            top_p):
             ids.append(tok_id)
             full = self.tok.decode(ids)
-            if len(full) > len(sent):
-                delta, sent = full[len(sent):], full
+            # hold back trailing U+FFFD: a multi-byte UTF-8 char split
+            # across tokens decodes as a replacement char until the
+            # rest arrives; emitting it early would make the
+            # concatenated stream differ from chat().
+            stable = full
+            while stable.endswith("�"):
+                stable = stable[:-1]
+            if len(stable) > len(sent):
+                delta, sent = stable[len(sent):], stable
                 yield delta
+        full = self.tok.decode(ids)
+        if len(full) > len(sent):
+            yield full[len(sent):]      # flush any held-back tail
 
     @torch.no_grad()
     def generate(self, prompt_ids: List[int], max_new_tokens: int = 64,
