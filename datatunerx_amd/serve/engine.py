@@ -74,8 +74,14 @@ def build_model(name: str, device, adapter_dir: Optional[str] = None,
                        "lora_alpha": float(ac.get("lora_alpha", 32)),
                        "lora_targets": tuple(ac.get("target_modules") or
                                              ("q_proj", "v_proj"))}
+    from ..models.hf_io import (is_hf_model_dir, load_hf_config,
+                                load_hf_weights)
+    hf_dir = is_hf_model_dir(name)
     with torch.device(device):
-        if name in ("llama2-7b", "llama-2-7b"):
+        if hf_dir:
+            model = LlamaForCausalLM(load_hf_config(name, **lora_kw),
+                                     lora=bool(adapter_dir), dtype=dtype)
+        elif name in ("llama2-7b", "llama-2-7b"):
             model = LlamaForCausalLM(LlamaConfig.llama2_7b(**lora_kw),
                                      dtype=dtype)
         elif name in ("llama2-13b", "llama-2-13b"):
@@ -90,7 +96,10 @@ def build_model(name: str, device, adapter_dir: Optional[str] = None,
             model = GPT2ForCausalLM(GPT2Config.tiny(), dtype=dtype)
         else:
             raise ValueError(f"unknown model {name!r}")
-    model.init_random()          # no network: base weights random-init
+    if hf_dir:
+        load_hf_weights(model, name)   # real local weights
+    else:
+        model.init_random()      # builtin names: random-init (no network)
     if adapter_dir:
         load_adapter(model, adapter_dir)
     model.eval()

@@ -223,16 +223,20 @@ def main(argv=None):
         if args.adapter:
             load_adapter_tp(model, args.adapter, cfg, rank, world)
         model.eval()
-        engine = InferenceEngine(model, template=args.template,
+        from ..models.hf_io import load_tokenizer
+        engine = InferenceEngine(model, tokenizer=load_tokenizer(args.model),
+                                 template=args.template,
                                  device=device)
         if rank == 0:
             serve_forever(TPFrontEngine(engine), args.host, args.port)
         else:
             tp_follower_loop(engine)
         return
+    from ..models.hf_io import load_tokenizer
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     model = build_model(args.model, device, adapter_dir=args.adapter)
-    engine = InferenceEngine(model, template=args.template, device=device)
+    engine = InferenceEngine(model, tokenizer=load_tokenizer(args.model),
+                             template=args.template, device=device)
     serve_forever(engine, args.host, args.port)
 
 

@@ -50,8 +50,18 @@ def main(argv=None):
         lora_kw = dict(lora_r=fargs.lora_rank, lora_alpha=fargs.lora_alpha,
                        lora_dropout=fargs.lora_dropout,
                        lora_targets=fargs.lora_targets)
+        from ..models.hf_io import (is_hf_model_dir, load_hf_config,
+                                    load_hf_weights, load_tokenizer)
+        hf_dir = is_hf_model_dir(name)
         with torch.device(device):
-            if name in ("llama2-7b", "llama-2-7b"):
+            if hf_dir:
+                # real HF-format checkpoint from a local dir (the
+                # reference's from_pretrained contract, train.py:236-242)
+                model = LlamaForCausalLM(load_hf_config(
+                    name,
+                    gradient_checkpointing=fargs.gradient_checkpointing,
+                    **lora_kw), lora=lora, dtype=dtype, train_base=full)
+            elif name in ("llama2-7b", "llama-2-7b"):
                 model = LlamaForCausalLM(LlamaConfig.llama2_7b(
                     gradient_checkpointing=fargs.gradient_checkpointing,
                     **lora_kw), lora=lora, dtype=dtype, train_base=full)
@@ -76,7 +86,11 @@ def main(argv=None):
             else:
                 raise SystemExit(f"unknown model {name!r}")
         # no network: checkpoints load from local dirs, else random init
-        if margs.checkpoint_dir and os.path.isdir(margs.checkpoint_dir):
+        if hf_dir:
+            n_t = load_hf_weights(model, name)
+            if rank == 0:
+                print(f"loaded {n_t} weight tensors from {name}")
+        elif margs.checkpoint_dir and os.path.isdir(margs.checkpoint_dir):
             from safetensors.torch import load_file
             sd = load_file(os.path.join(margs.checkpoint_dir,
                                         "model.safetensors"))
@@ -92,7 +106,7 @@ def main(argv=None):
                 print(f"quantized {n_q} frozen base layers to "
                       f"{margs.quantization}")
 
-        tok = ByteTokenizer()
+        tok = load_tokenizer(name)
         vocab = getattr(model.cfg, "vocab_size")
         if dargs.dataset_path and os.path.exists(dargs.dataset_path):
             ds = SFTDataset.from_csv(

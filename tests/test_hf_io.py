@@ -1,0 +1,134 @@
+"""Real-checkpoint contract: load an HF-format Llama
+checkpoint + tokenizer from a LOCAL directory and match HF transformers'
+logits. transformers is a TEST-ONLY dependency (creates the fixture and
+the reference logits); the loader itself uses safetensors/tokenizers/
+sentencepiece only."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from datatunerx_amd.models import LlamaForCausalLM
+from datatunerx_amd.models.hf_io import (HFTokenizer, is_hf_model_dir,
+                                         load_hf_config, load_hf_weights,
+                                         load_tokenizer)
+
+transformers = pytest.importorskip("transformers")
+
+
+@pytest.fixture(scope="module")
+def hf_dir(tmp_path_factory):
+    d = str(tmp_path_factory.mktemp("hfmodel"))
+    cfg = transformers.LlamaConfig(
+        vocab_size=320, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+        attn_implementation="eager")
+    torch.manual_seed(3)
+    m = transformers.LlamaForCausalLM(cfg)
+    m.save_pretrained(d, safe_serialization=True)
+
+    # build a real BPE tokenizer.json locally (no network)
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+    tok = Tokenizer(models.BPE(unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    trainer = trainers.BpeTrainer(
+        vocab_size=320, special_tokens=["<unk>", "<s>", "</s>"])
+    tok.train_from_iterator(
+        ["the quick brown fox jumps over the lazy dog",
+         "llama models fine tune with low rank adapters",
+         "hello world example text for byte pair merges"] * 30, trainer)
+    tok.save(os.path.join(d, "tokenizer.json"))
+    return d
+
+
+def test_hf_config_parse(hf_dir):
+    assert is_hf_model_dir(hf_dir)
+    cfg = load_hf_config(hf_dir)
+    assert cfg.vocab_size == 320 and cfg.hidden_size == 64
+    assert cfg.num_key_value_heads == 2 and cfg.num_hidden_layers == 2
+
+
+def test_hf_weights_match_transformers_logits(hf_dir):
+    cfg = load_hf_config(hf_dir)
+    ours = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32)
+    n = load_hf_weights(ours, hf_dir)
+    assert n >= 2 * 9 + 3            # per-layer tensors + embed/norm/head
+    ours.eval()
+
+    ref = transformers.LlamaForCausalLM.from_pretrained(
+        hf_dir, torch_dtype=torch.float32, attn_implementation="eager")
+    ref.eval()
+
+    torch.manual_seed(0)
+    ids = torch.randint(0, 320, (2, 17))
+    with torch.no_grad():
+        got = ours(ids)
+        want = ref(ids).logits
+    assert torch.allclose(got, want, atol=2e-4, rtol=1e-3), \
+        (got - want).abs().max()
+
+
+def test_hf_weights_sharded_index(hf_dir, tmp_path):
+    """Sharded checkpoints (model-0000x-of-0000y.safetensors + index)."""
+    import shutil
+
+    from safetensors.torch import load_file, save_file
+    d = str(tmp_path / "sharded")
+    shutil.copytree(hf_dir, d)
+    sd = load_file(os.path.join(d, "model.safetensors"))
+    os.remove(os.path.join(d, "model.safetensors"))
+    keys = sorted(sd)
+    half = len(keys) // 2
+    save_file({k: sd[k] for k in keys[:half]},
+              os.path.join(d, "model-00001-of-00002.safetensors"))
+    save_file({k: sd[k] for k in keys[half:]},
+              os.path.join(d, "model-00002-of-00002.safetensors"))
+    wm = {k: ("model-00001-of-00002.safetensors" if i < half else
+              "model-00002-of-00002.safetensors")
+          for i, k in enumerate(keys)}
+    with open(os.path.join(d, "model.safetensors.index.json"), "w") as f:
+        json.dump({"weight_map": wm}, f)
+
+    cfg = load_hf_config(d)
+    ours = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32)
+    load_hf_weights(ours, d)
+    ref = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32)
+    load_hf_weights(ref, hf_dir)
+    for (n1, p1), (_, p2) in zip(ours.named_parameters(),
+                                 ref.named_parameters()):
+        assert torch.equal(p1, p2), n1
+
+
+def test_tokenizer_roundtrip_and_specials(hf_dir):
+    tok = load_tokenizer(hf_dir)
+    assert isinstance(tok, HFTokenizer)
+    ids = tok.encode("the quick brown fox", add_special_tokens=True)
+    assert ids[0] == tok.bos_token_id
+    text = tok.decode(ids)
+    assert "quick" in text and "fox" in text
+    assert tok.vocab_size > 0
+
+
+def test_train_on_hf_dir_end_to_end(hf_dir, tmp_path):
+    """The trainer accepts model_name_or_path = a local HF dir: loads
+    real weights + tokenizer, trains LoRA, saves an adapter the engine
+    can serve on those weights."""
+    from datatunerx_amd.serve.engine import InferenceEngine, build_model
+    from datatunerx_amd.train.run import main as train_main
+    out = str(tmp_path / "out")
+    train_main([
+        "--model_name_or_path", hf_dir, "--output_dir", out,
+        "--max_steps", "2", "--per_device_train_batch_size", "2",
+        "--synthetic_examples", "8", "--block_size", "32",
+        "--lora_rank", "4", "--lora_target", "q_proj,v_proj",
+        "--logging_steps", "0"])
+    ckpt = os.path.join(out, "checkpoint")
+    assert os.path.exists(os.path.join(ckpt, "adapter_model.safetensors"))
+    model = build_model(hf_dir, torch.device("cpu"), adapter_dir=ckpt)
+    eng = InferenceEngine(model, tokenizer=load_tokenizer(hf_dir),
+                          template="vanilla", device=torch.device("cpu"))
+    txt = eng.chat([{"role": "user", "content": "hello"}], max_tokens=4)
+    assert isinstance(txt, str)
