@@ -10,6 +10,7 @@ for the loss. No HF dependency in the training hot path.
 
 from __future__ import annotations
 
+import os
 from dataclasses import dataclass
 
 import torch
@@ -223,8 +224,15 @@ class LlamaForCausalLM(nn.Module):
             return self.lm_head(h)
         # shift: predict token t+1 from position t
         h = h[:, :-1, :].reshape(-1, self.cfg.hidden_size)
-        logits = self.lm_head(h)
         targets = labels[:, 1:].reshape(-1)
+        if os.environ.get("DTX_NO_FUSED_CE") != "1" and \
+                self.cfg.vocab_size % 8 == 0:
+            # chunked-vocab fused lm_head+CE: no [M,V] logits tensor
+            from ..ops.autograd import fused_linear_cross_entropy
+            return fused_linear_cross_entropy(
+                h.contiguous(), self.lm_head.weight, targets,
+                ignore_index=-100)
+        logits = self.lm_head(h)
         return cross_entropy(logits, targets, ignore_index=-100)
 
     def trainable_parameters(self):

@@ -106,6 +106,23 @@ def softmax_xent_bwd(logits, targets, lse, dloss, ignore_index: int = -100):
     return ref.softmax_xent_bwd(logits, targets, lse, dloss, ignore_index)
 
 
+def xent_lse_merge(logits_c, targets, m_run, l_run, tgt, v0: int,
+                   ignore_index: int = -100):
+    if _gpu(logits_c):
+        _EXT.xent_lse_merge(logits_c, targets, m_run, l_run, tgt, v0,
+                            ignore_index)
+        return
+    ref.xent_lse_merge(logits_c, targets, m_run, l_run, tgt, v0,
+                       ignore_index)
+
+
+def xent_dlogits(logits_c, targets, lse, v0: int,
+                 ignore_index: int = -100):
+    if _gpu(logits_c):
+        return _EXT.xent_dlogits(logits_c, targets, lse, v0, ignore_index)
+    return ref.xent_dlogits(logits_c, targets, lse, v0, ignore_index)
+
+
 # --------------------------------------------------------- attention
 def attn_fwd(q, k, v, causal: bool = True, scale: float | None = None,
              len_dev=None):

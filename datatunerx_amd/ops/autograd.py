@@ -6,6 +6,8 @@ and the torch reference on CPU via datatunerx_amd.ops dispatch.
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import (attn_bwd, attn_fwd, lora_contract, lora_expand_add,
@@ -205,3 +207,69 @@ class FrozenGemm(torch.autograd.Function):
         dx = gemm_nt(dy2, wt)
         return (dx.reshape(ctx.xshape), None, None,
                 dy if ctx.has_res else None)
+
+
+class FusedLinearCrossEntropy(torch.autograd.Function):
+    """lm_head GEMM fused with cross-entropy by vocab chunking: the
+    [M, V] logits tensor is NEVER materialized (SURVEY.md §2.4 row 1:
+    "lm_head fused with cross-entropy to skip the logit
+    materialization"). Each chunk's logits [M, Vc] stay cache-resident
+    between the chunk GEMM and the online-logsumexp / dlogits kernels.
+
+    The mean-CE backward scale (grad_out / n_valid) is a per-call
+    SCALAR, so dX (and dW when the head is trainable) are accumulated
+    unscaled during a second forward sweep and backward is a single
+    multiply — total GEMM work equals the unfused fwd+dgrad pair while
+    logits never round-trip HBM.
+    """
+
+    CHUNK = int(os.environ.get("DTX_CE_CHUNK", "4096"))
+
+    @staticmethod
+    def forward(ctx, h2, w, targets, ignore_index, need_grads):
+        from . import xent_dlogits, xent_lse_merge
+        M, V = h2.shape[0], w.shape[0]
+        dev = h2.device
+        m = torch.full((M,), -3.4e38, device=dev, dtype=torch.float32)
+        l = torch.zeros(M, device=dev, dtype=torch.float32)
+        tgt = torch.zeros(M, device=dev, dtype=torch.float32)
+        ch = FusedLinearCrossEntropy.CHUNK
+        for v0 in range(0, V, ch):
+            wc = w[v0:v0 + ch]
+            lc = torch.nn.functional.linear(h2, wc)
+            xent_lse_merge(lc, targets, m, l, tgt, v0, ignore_index)
+        lse = m + l.log()
+        valid = targets != ignore_index
+        n_valid = valid.sum().clamp(min=1)
+        loss = torch.where(valid, lse - tgt,
+                           torch.zeros((), device=dev)).sum() \
+            / n_valid.float()
+        dx = dw = None
+        if need_grads:
+            dx = torch.zeros_like(h2)
+            if w.requires_grad:
+                dw = torch.zeros_like(w)
+            for v0 in range(0, V, ch):
+                wc = w[v0:v0 + ch]
+                lc = torch.nn.functional.linear(h2, wc)
+                dl = xent_dlogits(lc, targets, lse, v0, ignore_index)
+                dx.addmm_(dl, wc)
+                if dw is not None:
+                    dw[v0:v0 + ch].addmm_(dl.t(), h2)
+        ctx.save_for_backward(dx, dw, n_valid)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gout):
+        dx, dw, n_valid = ctx.saved_tensors
+        s = (gout.float() / n_valid.float()).to(dx.dtype) \
+            if dx is not None else None
+        return (dx * s if dx is not None else None, None if dw is None
+                else dw * s, None, None, None)
+
+
+def fused_linear_cross_entropy(h2, w, targets, ignore_index: int = -100):
+    need = torch.is_grad_enabled() and (h2.requires_grad or
+                                        w.requires_grad)
+    return FusedLinearCrossEntropy.apply(h2, w, targets, ignore_index,
+                                         need)

@@ -19,6 +19,10 @@ void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
                      long, hipStream_t);
 void launch_xent_bwd(const void*, const long*, const float*, const float*,
                      void*, long, int, long, hipStream_t);
+void launch_xent_lse_merge(const void*, const long*, float*, float*,
+                           float*, long, int, long, long, hipStream_t);
+void launch_xent_dlogits(const void*, const long*, const float*, void*,
+                         long, int, long, long, hipStream_t);
 void launch_adamw(void*, float*, const float*, float*, float*, long, float,
                   float, float, float, float, int, hipStream_t);
 void launch_l2_norm(const float*, float*, float*, long, hipStream_t);
@@ -352,6 +356,34 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+// chunked-vocab CE: online LSE merge over one logits chunk (in-place
+// m/l/tgt update) and the per-chunk dlogits for the dX sweep.
+void xent_lse_merge(torch::Tensor logits, torch::Tensor targets,
+                    torch::Tensor m_run, torch::Tensor l_run,
+                    torch::Tensor tgt, long v0, long ignore_index) {
+  check_bf16_contig(logits, "logits");
+  const int Vc = (int)logits.size(-1);
+  const long N = logits.numel() / Vc;
+  TORCH_CHECK(Vc % 8 == 0, "Vc % 8");
+  launch_xent_lse_merge(logits.data_ptr(), targets.data_ptr<long>(),
+                        m_run.data_ptr<float>(), l_run.data_ptr<float>(),
+                        tgt.data_ptr<float>(), N, Vc, v0, ignore_index,
+                        cur_stream());
+}
+
+torch::Tensor xent_dlogits(torch::Tensor logits, torch::Tensor targets,
+                           torch::Tensor lse, long v0, long ignore_index) {
+  check_bf16_contig(logits, "logits");
+  const int Vc = (int)logits.size(-1);
+  const long N = logits.numel() / Vc;
+  TORCH_CHECK(Vc % 8 == 0, "Vc % 8");
+  auto dl = torch::empty_like(logits);
+  launch_xent_dlogits(logits.data_ptr(), targets.data_ptr<long>(),
+                      lse.data_ptr<float>(), dl.data_ptr(), N, Vc, v0,
+                      ignore_index, cur_stream());
+  return dl;
+}
+
 // y[1,N] = x[1,K] @ W[N,K]^T (decode GEMV; fp32 accumulate)
 torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
   check_bf16_contig(x, "x");
@@ -453,6 +485,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("xent_fwd", &xent_fwd);
   m.def("xent_bwd", &xent_bwd);
+  m.def("xent_lse_merge", &xent_lse_merge);
+  m.def("xent_dlogits", &xent_dlogits);
   m.def("lora_contract", &lora_contract, py::arg("x"), py::arg("w"),
         py::arg("mask") = py::none(), py::arg("seed") = 0,
         py::arg("keep") = 1.0);

@@ -293,3 +293,34 @@ def adamw_step(p_bf16: torch.Tensor, master: torch.Tensor,
 
 def l2_norm(flat: torch.Tensor) -> torch.Tensor:
     return flat.float().pow(2).sum().sqrt()
+
+
+def xent_lse_merge(logits_c, targets, m_run, l_run, tgt, v0: int,
+                   ignore_index: int = -100):
+    """CPU fp32 reference of the chunked online-LSE merge (in-place)."""
+    x = logits_c.float()
+    cmax = x.max(dim=-1).values
+    csum = (x - cmax[:, None]).exp().sum(dim=-1)
+    m_new = torch.maximum(m_run, cmax)
+    l_run.mul_((m_run - m_new).exp()).add_(csum * (cmax - m_new).exp())
+    m_run.copy_(m_new)
+    vc = x.shape[-1]
+    inside = (targets != ignore_index) & (targets >= v0) & \
+        (targets < v0 + vc)
+    idx = inside.nonzero(as_tuple=True)[0]
+    if idx.numel():
+        tgt[idx] = x[idx, (targets[idx] - v0)]
+
+
+def xent_dlogits(logits_c, targets, lse, v0: int,
+                 ignore_index: int = -100):
+    x = logits_c.float()
+    dl = (x - lse[:, None]).exp()
+    vc = x.shape[-1]
+    inside = (targets != ignore_index) & (targets >= v0) & \
+        (targets < v0 + vc)
+    idx = inside.nonzero(as_tuple=True)[0]
+    if idx.numel():
+        dl[idx, (targets[idx] - v0)] -= 1.0
+    dl[targets == ignore_index] = 0.0
+    return dl.to(logits_c.dtype)

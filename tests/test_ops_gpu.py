@@ -390,3 +390,54 @@ def test_frozen_gemm_autograd():
     assert_close(y, want_y, name="frozen_gemm y")
     assert_close(x.grad, want_dx, name="frozen_gemm dx")
     assert_close(res.grad, dy, name="frozen_gemm dres")
+
+
+# ------------------------------------------------- chunked fused CE
+def test_xent_chunk_kernels():
+    M, V = 513, 1024
+    logits = mk(M, V)
+    targets = torch.randint(0, 2 * V, (M,), device=DEV)  # half outside
+    targets[::7] = -100
+    m = torch.full((M,), -3.4e38, device=DEV)
+    l = torch.zeros(M, device=DEV)
+    tgt = torch.zeros(M, device=DEV)
+    v0 = 512
+    ops.xent_lse_merge(logits, targets, m, l, tgt, v0)
+    m2 = torch.full((M,), -3.4e38, dtype=torch.float32)
+    l2 = torch.zeros(M)
+    t2 = torch.zeros(M)
+    ref.xent_lse_merge(logits.cpu(), targets.cpu(), m2, l2, t2, v0)
+    assert_close(m.cpu(), m2, name="chunk m")
+    assert_close(l.cpu(), l2, name="chunk l")
+    assert_close(tgt.cpu(), t2, name="chunk tgt")
+
+    lse = m + l.log()
+    dl = ops.xent_dlogits(logits, targets, lse, v0)
+    dl_ref = ref.xent_dlogits(logits.cpu(), targets.cpu(), lse.cpu(), v0)
+    assert_close(dl.cpu(), dl_ref, name="chunk dlogits")
+
+
+def test_fused_linear_ce_gpu_matches_unfused():
+    from datatunerx_amd.ops.autograd import (FusedLinearCrossEntropy,
+                                             cross_entropy,
+                                             fused_linear_cross_entropy)
+    M, V, E = 777, 2048, 256
+    h = mk(M, E)
+    h.requires_grad_(True)
+    w = mk(V, E)
+    t = torch.randint(0, V, (M,), device=DEV)
+    t[::9] = -100
+    old = FusedLinearCrossEntropy.CHUNK
+    FusedLinearCrossEntropy.CHUNK = 512
+    try:
+        loss = fused_linear_cross_entropy(h, w, t)
+        loss.backward()
+        g1 = h.grad.clone()
+        h.grad = None
+        loss2 = cross_entropy(torch.nn.functional.linear(h, w), t)
+        loss2.backward()
+        assert abs(float(loss) - float(loss2)) < 2e-3 * max(
+            1.0, abs(float(loss2)))
+        assert_close(g1, h.grad, rtol=3e-2, name="fused-ce dx")
+    finally:
+        FusedLinearCrossEntropy.CHUNK = old

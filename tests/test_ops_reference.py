@@ -150,3 +150,33 @@ def test_adamw_matches_torch_adamw():
         ref.adamw_step(p_bf, mine, g, m, v, 1e-3, 0.9, 0.999, 1e-8,
                        0.01, step)
     assert torch.allclose(mine, p_ref.detach(), atol=1e-6)
+
+
+def test_fused_linear_ce_matches_unfused_cpu():
+    """Chunked fused lm_head+CE == materialized logits + CE (loss, dX,
+    dW), including ignore_index rows and a non-dividing final chunk."""
+    import torch
+
+    from datatunerx_amd.ops.autograd import (FusedLinearCrossEntropy,
+                                             cross_entropy,
+                                             fused_linear_cross_entropy)
+    torch.manual_seed(0)
+    M, V, E = 37, 104, 24
+    h = torch.randn(M, E, requires_grad=True)
+    w = torch.randn(V, E, requires_grad=True)
+    t = torch.randint(0, V, (M,))
+    t[::5] = -100
+    old = FusedLinearCrossEntropy.CHUNK
+    FusedLinearCrossEntropy.CHUNK = 40          # 40+40+24 chunks
+    try:
+        loss = fused_linear_cross_entropy(h, w, t)
+        loss.backward()
+        g_h, g_w = h.grad.clone(), w.grad.clone()
+        h.grad = w.grad = None
+        loss2 = cross_entropy(torch.nn.functional.linear(h, w), t)
+        loss2.backward()
+        assert torch.allclose(loss, loss2, atol=1e-5)
+        assert torch.allclose(g_h, h.grad, atol=1e-5)
+        assert torch.allclose(g_w, w.grad, atol=1e-5)
+    finally:
+        FusedLinearCrossEntropy.CHUNK = old
