@@ -17,7 +17,8 @@ import torch
 import torch.nn as nn
 
 from ..ops import rope_tables
-from ..ops.autograd import attention, cross_entropy, rmsnorm, rope, swiglu
+from ..ops.autograd import (attention, cross_entropy, rmsnorm,
+                            rmsnorm_tap, rope, swiglu)
 from .lora import FrozenLinear, LoRALinearModule
 
 
@@ -152,11 +153,13 @@ class LlamaDecoderLayer(nn.Module):
 
     def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None,
                 pos_dev=None):
-        h = rmsnorm(x, self.input_layernorm, self.cfg.rms_norm_eps)
+        h, res = rmsnorm_tap(x, self.input_layernorm,
+                             self.cfg.rms_norm_eps)
         x = self.self_attn(h, cos, sin, pos0, kv_cache, pos_dev,
-                           residual=x)
-        h = rmsnorm(x, self.post_attention_layernorm, self.cfg.rms_norm_eps)
-        return self.mlp(h, residual=x)
+                           residual=res)
+        h, res = rmsnorm_tap(x, self.post_attention_layernorm,
+                             self.cfg.rms_norm_eps)
+        return self.mlp(h, residual=res)
 
 
 class LlamaForCausalLM(nn.Module):

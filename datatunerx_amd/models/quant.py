@@ -98,6 +98,11 @@ class QuantFrozenLinear(nn.Module):
 
     @property
     def weight(self):
+        if self.qweight.is_cuda and self.dtype == torch.bfloat16:
+            from ..ops import dequant_int4, dequant_int8
+            if self.bits == 8:
+                return dequant_int8(self.qweight, self.scale)
+            return dequant_int4(self.qweight, self.scale, self.group)
         if self.bits == 8:
             return dequantize_int8(self.qweight, self.scale, self.dtype)
         return dequantize_int4(self.qweight, self.scale, self.dtype,

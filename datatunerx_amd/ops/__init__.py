@@ -56,10 +56,16 @@ def rmsnorm_fwd(x, w, eps: float = 1e-5):
     return ref.rmsnorm_fwd(x, w, eps)
 
 
-def rmsnorm_bwd(dy, x, w, inv):
+def rmsnorm_bwd(dy, x, w, inv, dres=None):
+    """dres: optional residual-branch gradient accumulated into dx
+    IN-KERNEL (the fused residual+rmsnorm backward — kills the autograd
+    grad-accumulation adds on every decoder-layer residual)."""
     if _gpu(x):
-        return _EXT.rmsnorm_bwd(dy, x, w, inv)
-    return ref.rmsnorm_bwd(dy, x, w, inv)
+        return _EXT.rmsnorm_bwd(dy, x, w, inv, dres)
+    dx, dw = ref.rmsnorm_bwd(dy, x, w, inv)
+    if dres is not None:
+        dx = dx + dres
+    return dx, dw
 
 
 # ------------------------------------------------------------------ RoPE
@@ -212,6 +218,20 @@ def dropout_mask(M: int, K: int, seed: int, keep: float, like):
         return _EXT.dropout_mask(M, K, seed, keep, like)
     return ref.dropout_mask(M, K, seed, keep, device=like.device,
                             dtype=torch.bfloat16)
+
+
+# -------------------------------------------------- weight dequant
+def dequant_int8(q, scale):
+    if _gpu(q):
+        return _EXT.dequant_int8(q, scale)
+    return (q.float() * scale[:, None]).to(torch.bfloat16)
+
+
+def dequant_int4(packed, scale, group: int = 64):
+    if _gpu(packed):
+        return _EXT.dequant_int4(packed, scale, group)
+    from ..models.quant import dequantize_int4
+    return dequantize_int4(packed, scale, torch.bfloat16, group)
 
 
 # ------------------------------------------------------------- AdamW

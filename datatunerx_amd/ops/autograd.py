@@ -33,6 +33,32 @@ def rmsnorm(x, w, eps: float = 1e-5):
     return RMSNorm.apply(x, w, eps)
 
 
+class ResidualTapRMSNorm(torch.autograd.Function):
+    """(norm(x), x) in one node: the layer uses output 1 as the residual
+    branch, so x has ONE autograd consumer and the two branch gradients
+    merge inside the rmsnorm_bwd kernel instead of a separate
+    grad-accumulation add per residual (round-1 STATUS note: ~128 such
+    adds/step)."""
+
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        y, inv = rmsnorm_fwd(x, w, eps)
+        ctx.save_for_backward(x, w, inv)
+        return y, x
+
+    @staticmethod
+    def backward(ctx, dy, dres):
+        x, w, inv = ctx.saved_tensors
+        from . import rmsnorm_bwd as _bwd
+        dx, dw = _bwd(dy.contiguous(), x, w, inv,
+                      dres.contiguous() if dres is not None else None)
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm_tap(x, w, eps: float = 1e-5):
+    return ResidualTapRMSNorm.apply(x, w, eps)
+
+
 class Rope(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, cos, sin, pos0, pos_dev):

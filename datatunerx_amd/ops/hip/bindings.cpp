@@ -9,7 +9,8 @@ void launch_rmsnorm_fwd(const void*, const void*, void*, float*, int, int,
                         float, hipStream_t);
 int rmsnorm_bwd_nblocks(int M);
 void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
-                        void*, float*, float*, int, int, hipStream_t);
+                        const void*, void*, float*, float*, int, int,
+                        hipStream_t);
 void launch_rope(const void*, const float*, const float*, void*, long, int,
                  int, int, int, int, const int*, hipStream_t);
 void launch_swiglu_fwd(const void*, const void*, void*, long, hipStream_t);
@@ -19,6 +20,10 @@ void launch_xent_fwd(const void*, const long*, float*, float*, long, int,
                      long, hipStream_t);
 void launch_xent_bwd(const void*, const long*, const float*, const float*,
                      void*, long, int, long, hipStream_t);
+void launch_dequant_int8(const void*, const float*, void*, long, int,
+                         hipStream_t);
+void launch_dequant_int4(const void*, const float*, void*, long, int, int,
+                         hipStream_t);
 void launch_xent_lse_merge(const void*, const long*, float*, float*,
                            float*, long, int, long, long, hipStream_t);
 void launch_xent_dlogits(const void*, const long*, const float*, void*,
@@ -91,7 +96,8 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
 }
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
-                                       torch::Tensor w, torch::Tensor inv) {
+                                       torch::Tensor w, torch::Tensor inv,
+                                       c10::optional<torch::Tensor> dres) {
   check_bf16_contig(dy, "dy");
   check_bf16_contig(x, "x");
   const int H = (int)x.size(-1);
@@ -100,8 +106,14 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dw = torch::empty({H}, x.options().dtype(torch::kFloat));
   const int nb = rmsnorm_bwd_nblocks((int)M);
   auto part = torch::empty({nb, H}, x.options().dtype(torch::kFloat));
+  const void* drp = nullptr;
+  if (dres.has_value()) {
+    check_bf16_contig(*dres, "dres");
+    TORCH_CHECK(dres->numel() == x.numel(), "dres shape");
+    drp = dres->data_ptr();
+  }
   launch_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
-                     inv.data_ptr<float>(), dx.data_ptr(),
+                     inv.data_ptr<float>(), drp, dx.data_ptr(),
                      part.data_ptr<float>(), dw.data_ptr<float>(), (int)M, H,
                      cur_stream());
   return {dx, dw};
@@ -356,6 +368,34 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+// weight-only dequant (int8 per-row / int4 group-64) -> bf16
+torch::Tensor dequant_int8(torch::Tensor q, torch::Tensor scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kChar &&
+              q.is_contiguous());
+  const int K = (int)q.size(1);
+  const long N = q.size(0);
+  TORCH_CHECK(K % 8 == 0, "K % 8");
+  auto out = torch::empty({N, K},
+                          q.options().dtype(torch::kBFloat16));
+  launch_dequant_int8(q.data_ptr(), scale.data_ptr<float>(),
+                      out.data_ptr(), N, K, cur_stream());
+  return out;
+}
+
+torch::Tensor dequant_int4(torch::Tensor packed, torch::Tensor scale,
+                           long group) {
+  TORCH_CHECK(packed.is_cuda() && packed.scalar_type() == torch::kByte &&
+              packed.is_contiguous());
+  const int K = (int)packed.size(1) * 2;
+  const long N = packed.size(0);
+  TORCH_CHECK(K % 8 == 0 && group % 8 == 0, "K/group alignment");
+  auto out = torch::empty({N, K},
+                          packed.options().dtype(torch::kBFloat16));
+  launch_dequant_int4(packed.data_ptr(), scale.data_ptr<float>(),
+                      out.data_ptr(), N, K, (int)group, cur_stream());
+  return out;
+}
+
 // chunked-vocab CE: online LSE merge over one logits chunk (in-place
 // m/l/tgt update) and the per-chunk dlogits for the dX sweep.
 void xent_lse_merge(torch::Tensor logits, torch::Tensor targets,
@@ -477,7 +517,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
-  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
+        py::arg("w"), py::arg("inv"), py::arg("dres") = py::none());
   m.def("rope", &rope, py::arg("x"), py::arg("cosb"), py::arg("sinb"),
         py::arg("pos0"), py::arg("backward"),
         py::arg("pos_dev") = py::none());
@@ -486,6 +527,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("xent_fwd", &xent_fwd);
   m.def("xent_bwd", &xent_bwd);
   m.def("xent_lse_merge", &xent_lse_merge);
+  m.def("dequant_int8", &dequant_int8);
+  m.def("dequant_int4", &dequant_int4);
   m.def("xent_dlogits", &xent_dlogits);
   m.def("lora_contract", &lora_contract, py::arg("x"), py::arg("w"),
         py::arg("mask") = py::none(), py::arg("seed") = 0,

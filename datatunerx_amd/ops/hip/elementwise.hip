@@ -60,6 +60,7 @@ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
                         const unsigned short* __restrict__ x,
                         const unsigned short* __restrict__ w,
                         const float* __restrict__ inv_in,
+                        const unsigned short* __restrict__ dres,
                         unsigned short* __restrict__ dx,
                         float* __restrict__ dw_part,
                         int M, int H) {
@@ -93,15 +94,20 @@ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
     float tot = block_reduce_sum(dot, scratch);
     const float c = inv * inv * inv / (float)H * tot;
     unsigned short* dxr = dx + (long)row * H;
+    const unsigned short* drr =
+        dres ? dres + (long)row * H : nullptr;
 #pragma unroll
     for (int n = 0; n < NG; ++n) {
       const int g = tid + n * DTX_BLOCK;
       if (g < groups) {
         float out[8];
+        float dr[8];
+        if (drr) load_bf16x8(drr + g * 8, dr);
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
           float gi = dys[n][i] * wv[n][i];
           out[i] = inv * gi - c * xs[n][i];
+          if (drr) out[i] += dr[i];   // fused residual-branch gradient
           dwacc[n][i] += dys[n][i] * xs[n][i] * inv;
         }
         store_bf16x8(dxr + g * 8, out);
@@ -282,13 +288,15 @@ void launch_rmsnorm_fwd(const void* x, const void* w, void* y, float* inv,
 int rmsnorm_bwd_nblocks(int M) { return M < 1024 ? (M < 1 ? 1 : M) : 1024; }
 
 void launch_rmsnorm_bwd(const void* dy, const void* x, const void* w,
-                        const float* inv, void* dx, float* dw_part,
-                        float* dw, int M, int H, hipStream_t s) {
+                        const float* inv, const void* dres, void* dx,
+                        float* dw_part, float* dw, int M, int H,
+                        hipStream_t s) {
   int grid = rmsnorm_bwd_nblocks(M);
   const int ng = DTX_CDIV(H / 8, DTX_BLOCK);
 #define CASE(N) rmsnorm_bwd_kernel<N><<<grid, DTX_BLOCK, 0, s>>>( \
       (const unsigned short*)dy, (const unsigned short*)x, \
-      (const unsigned short*)w, inv, (unsigned short*)dx, dw_part, M, H)
+      (const unsigned short*)w, inv, (const unsigned short*)dres, \
+      (unsigned short*)dx, dw_part, M, H)
   switch (ng) {
     case 1: CASE(1); break;
     case 2: CASE(2); break;

@@ -441,3 +441,43 @@ def test_fused_linear_ce_gpu_matches_unfused():
         assert_close(g1, h.grad, rtol=3e-2, name="fused-ce dx")
     finally:
         FusedLinearCrossEntropy.CHUNK = old
+
+
+# ----------------------------------------------------- weight dequant
+def test_dequant_kernels_match_reference():
+    from datatunerx_amd.models.quant import (dequantize_int4,
+                                             dequantize_int8,
+                                             quantize_int4, quantize_int8)
+    w = mk(512, 1024)
+    q8, s8 = quantize_int8(w.cpu())
+    got8 = ops.dequant_int8(q8.to(DEV), s8.to(DEV))
+    assert_close(got8.cpu(), dequantize_int8(q8, s8, torch.float32),
+                 name="int8 dequant")
+    q4, s4 = quantize_int4(w.cpu())
+    got4 = ops.dequant_int4(q4.to(DEV), s4.to(DEV))
+    assert_close(got4.cpu(), dequantize_int4(q4, s4, torch.float32),
+                 name="int4 dequant")
+
+
+@pytest.mark.parametrize("bits", [8, 4])
+def test_quantized_train_step_gpu(bits):
+    """One quantized LoRA train step on GPU: the base dequant runs the
+    HIP kernels and the step produces finite loss + adapter grads
+    (VERDICT r1 item 8)."""
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.models.quant import quantize_model_
+    torch.manual_seed(0)
+    cfg = LlamaConfig.tiny()
+    model = LlamaForCausalLM(cfg, lora=True,
+                             dtype=torch.bfloat16).init_random()
+    model = model.to(DEV)
+    n_q = quantize_model_(model, bits=bits)
+    assert n_q > 0
+    ids = torch.randint(3, cfg.vocab_size, (2, 64), device=DEV)
+    labels = ids.clone()
+    loss = model(ids, labels=labels)
+    loss.backward()
+    assert torch.isfinite(loss)
+    for n, p in model.trainable_parameters():
+        if "lora_A" in n:
+            assert p.grad is not None and torch.isfinite(p.grad).all(), n
