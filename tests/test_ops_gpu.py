@@ -467,7 +467,11 @@ def test_quantized_train_step_gpu(bits):
     from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
     from datatunerx_amd.models.quant import quantize_model_
     torch.manual_seed(0)
-    cfg = LlamaConfig.tiny()
+    # head_dim 64 (the attention kernels support D in {64, 128})
+    cfg = LlamaConfig(vocab_size=512, hidden_size=256,
+                      intermediate_size=512, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=4,
+                      max_position_embeddings=256, lora_r=8)
     model = LlamaForCausalLM(cfg, lora=True,
                              dtype=torch.bfloat16).init_random()
     model = model.to(DEV)
