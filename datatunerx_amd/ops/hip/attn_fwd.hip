@@ -27,6 +27,7 @@
 
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+typedef __attribute__((ext_vector_type(2))) unsigned int uint2_t;
 
 #define MFMA32(a, b, c) __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
 #define NEG_INF (-3.0e38f)
@@ -72,8 +73,13 @@ struct AttnFwdLds {
   unsigned short V[128][D + 8];   // row-major; PV B-frags via tr-read
 };
 
-template <int D>
-__global__ __launch_bounds__(512, 1)
+// WAVES: q-rows per workgroup = 32*WAVES. 8 (256 rows) shares each
+// staged K/V tile across more q-rows; 4 (128 rows) halves the causal
+// wave-skew inside the workgroup (utilization 82.5% -> 91.7% at
+// S=1024) and lets TWO workgroups co-reside per CU. Measured A/B picks
+// the launcher default.
+template <int D, int WAVES = 8>
+__global__ __launch_bounds__(64 * WAVES, 1)
 void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
                       const unsigned short* __restrict__ Kp,
                       const unsigned short* __restrict__ Vp,
@@ -94,7 +100,9 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
   const int bh = blockIdx.y;
   const int b = bh / Hq, hq = bh % Hq;
   const int hkv = hq / (Hq / Hkv);
-  const int q0 = blockIdx.x * 256;
+  constexpr int TPB = 64 * WAVES;
+  constexpr int QROWS = 32 * WAVES;
+  const int q0 = blockIdx.x * QROWS;
   const int qw = q0 + wid * 32;            // wave's first q row
 
   const int qrowstr = Hq * D;              // BSHD row stride
@@ -128,20 +136,20 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
   float m_run = NEG_INF, l_run = 0.f;      // per-lane: q-row qw + l31
   const float kscale = scale * LOG2E;      // softmax in exp2 domain
 
-  const int q_hi_blk = min(q0 + 255, S - 1);
+  const int q_hi_blk = min(q0 + QROWS - 1, S - 1);
   const int kv_hi = causal ? min(Skv - 1, q_hi_blk + diag) : (Skv - 1);
   const int nstages = kv_hi / 128 + 1;     // 128 kv rows per stage
 
   // T14 async-stage split: the next stage's K/VT global loads are
   // issued while the current stage computes; LDS writes after the
   // barrier.  KIT/VIT iterations cover the 128-row K and VT tiles.
-  constexpr int KIT = (128 * D / 8) / 512;
+  constexpr int KIT = (128 * D / 8) / TPB;
   short8v stg[KIT * 2];
   auto issue_stage = [&](int st2) {
     const int kvs = st2 * 128;
 #pragma unroll
     for (int it = 0; it < KIT; ++it) {
-      const int idx = threadIdx.x + it * 512;
+      const int idx = threadIdx.x + it * TPB;
       const int row = idx / (D / 8), g = idx % (D / 8);
       short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
       short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -157,7 +165,7 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
   auto write_stage = [&]() {
 #pragma unroll
     for (int it = 0; it < KIT; ++it) {
-      const int idx = threadIdx.x + it * 512;
+      const int idx = threadIdx.x + it * TPB;
       const int row = idx / (D / 8), g = idx % (D / 8);
       *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it * 2];
       *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = stg[it * 2 + 1];
@@ -302,17 +310,26 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
     }  // half
   }
 
-  // ---- epilogue: normalize rows, store O (bf16, coalesced over d) + lse
+  // ---- epilogue: normalize rows, store O per wave IMMEDIATELY (an
+  // LDS-restaged barrier epilogue was measured 20% SLOWER end-to-end:
+  // the block-wide sync serializes the causal wave skew that per-wave
+  // stores overlap). cvt_pk packs row-pairs so the convert is one VALU
+  // op per two elements instead of the branchy scalar f2bf.
   const float rcp = l_run > 0.f ? 1.f / l_run : 0.f;
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int qg = qw + crow(r, hi);
-    const float n_r = __shfl(rcp, crow(r, hi), 64);
-    if (qg < S) {
-      unsigned short* orow = O + qbase + (long)qg * qrowstr;
+  for (int r = 0; r < 16; r += 2) {
+    const int qa = qw + crow(r, hi);
+    const int qb = qw + crow(r + 1, hi);
+    const float n_a = __shfl(rcp, crow(r, hi), 64);
+    const float n_b = __shfl(rcp, crow(r + 1, hi), 64);
+    unsigned short* orow_a = O + qbase + (long)qa * qrowstr;
+    unsigned short* orow_b = O + qbase + (long)qb * qrowstr;
 #pragma unroll
-      for (int c = 0; c < ND32; ++c)
-        orow[c * 32 + l31] = f2bf(o_acc[c][r] * n_r);
+    for (int c = 0; c < ND32; ++c) {
+      const unsigned pk = dtx_cvt_pk_bf16(o_acc[c][r] * n_a,
+                                          o_acc[c][r + 1] * n_b);
+      if (qa < S) orow_a[c * 32 + l31] = (unsigned short)(pk & 0xffff);
+      if (qb < S) orow_b[c * 32 + l31] = (unsigned short)(pk >> 16);
     }
   }
   if (hi == 0) {
@@ -378,14 +395,33 @@ void launch_transpose_sd(const void* x, void* xt, int B, int S, int H,
 void launch_attn_fwd(const void* q, const void* k, const void* v, void* o,
                      float* lse, int B, int Hq, int Hkv, int S, int Skv,
                      int D, float scale, int causal, hipStream_t st) {
+  // 4-wave workgroups measured 45% SLOWER than 8-wave on the training
+  // shape (a second 256-thread WG does not co-reside at 256 VGPRs, so
+  // half the CU idles); keep 8-wave and leave the variant for probes.
+  const bool w4 = getenv("DTX_ATTN_W4") != nullptr;
+  if (w4) {
+    dim3 grid(DTX_CDIV(S, 128), B * Hq);
+    if (D == 128) {
+      attn_fwd2_kernel<128, 4><<<grid, 256, 0, st>>>(
+          (const unsigned short*)q, (const unsigned short*)k,
+          (const unsigned short*)v, (unsigned short*)o, lse,
+          B, Hq, Hkv, S, Skv, scale, causal);
+    } else if (D == 64) {
+      attn_fwd2_kernel<64, 4><<<grid, 256, 0, st>>>(
+          (const unsigned short*)q, (const unsigned short*)k,
+          (const unsigned short*)v, (unsigned short*)o, lse,
+          B, Hq, Hkv, S, Skv, scale, causal);
+    }
+    return;
+  }
   dim3 grid(DTX_CDIV(S, 256), B * Hq);
   if (D == 128) {
-    attn_fwd2_kernel<128><<<grid, 512, 0, st>>>(
+    attn_fwd2_kernel<128, 8><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
         (const unsigned short*)v, (unsigned short*)o, lse,
         B, Hq, Hkv, S, Skv, scale, causal);
   } else if (D == 64) {
-    attn_fwd2_kernel<64><<<grid, 512, 0, st>>>(
+    attn_fwd2_kernel<64, 8><<<grid, 512, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
         (const unsigned short*)v, (unsigned short*)o, lse,
         B, Hq, Hkv, S, Skv, scale, causal);

@@ -103,7 +103,11 @@ __device__ __forceinline__ unsigned swz_f(unsigned x) { return (0u - x) & 3u; }
 //       vmcnt(4) at both odd phases (publication one phase earlier) +
 //       fragment PRELOAD: phase p+1's ds_reads issue before phase p's
 //       MFMA cluster (incl. across the tile seam), so LDS latency and
-//       read issue hide inside the MFMA stream.
+//       read issue hide inside the MFMA stream; 4 = PIPE3 +
+//       sched_group_barrier [MFMA,MFMA,ds_read] interleave (the reads
+//       are EMITTED inside the MFMA cluster so the matrix pipe is never
+//       starved during the read window) + static young-half s_setprio
+//       instead of per-phase flips (T5 static form).
 // MF:   0 = mfma_f32_16x16x32_bf16; 1 = mfma_f32_32x32x16_bf16 (the
 //       higher-ceiling intrinsic: 2382 vs 2075 TF ubench — and half the
 //       MFMA + fragment-read instructions per phase).
@@ -115,6 +119,7 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
                     unsigned short* __restrict__ C,
                     long M, int N, int K, int mb_n) {
   constexpr int RING = (PIPE >= 2) ? 5 : 4;
+  constexpr bool PRELOAD = (PIPE == 3) || (PIPE == 4);
   constexpr unsigned B_RING = RING * HALF_BYTES;
   __shared__ __attribute__((aligned(16))) unsigned char
       lds[2 * RING * HALF_BYTES];
@@ -254,7 +259,13 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
   //   retires ALL four halves of tile kt+1 (staged during kt-1);
   //   barriers at p1/p3 ends gate slot reuse (WAR) and publication.
   short8v bfr[4];
-  short8v afr2[2][4], bfr2[2][4];      // PIPE3 double-buffered fragments
+  short8v afr2[2][4], bfr2[2][4];      // PIPE3/4 double-buffered fragments
+  if (PIPE == 4) {
+    // T5 static form: the second-dispatched half loses VALU arbitration
+    // on every segment; one wave-uniform setprio before the loop.
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+      __builtin_amdgcn_s_setprio(1);
+  }
   // ring-5 slot indices maintained incrementally ((2kt+i) % 5)
   int h5 = 0;                          // (2*kt) % 5
 
@@ -272,7 +283,7 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
           acc[(mh) * 4 + fm][fn] =                                          \
               MFMA16(af[fm], bf[fn], acc[(mh) * 4 + fm][fn]);
 
-  if (PIPE == 3) {
+  if (PRELOAD) {
     // preload tile 0 phase 0 fragments
     RD_A(afr2[0], 0u, 0);
     RD_B(bfr2[0], B_RING + 0u);
@@ -290,8 +301,7 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       const unsigned as0n = (unsigned)h5 * HALF_BYTES;  // next tile kh0
 #pragma unroll
       for (int p = 0; p < 4; ++p) {
-        const unsigned mh = p & 1;
-        if (pre) {
+        if (PIPE == 3 && pre) {
           if (p == 0) STAGE_B(ss0, kt + 2, 0)
           else if (p == 1) STAGE_A(ss0, kt + 2, 0)
           else if (p == 2) STAGE_B(ss1, kt + 2, 1)
@@ -301,23 +311,52 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
         // tile's phase 0: its halves were published at this tile's
         // p1-end vmcnt, and its slot's next writer stages at kt+1.p2 —
         // after the read).
+        int nreads = 0;
         if (p == 0) {
           RD_A(afr2[1], as0, 1);
+          nreads = 4;
         } else if (p == 1) {
           RD_A(afr2[0], as1, 0);
           RD_B(bfr2[1], as1 + B_RING);
+          nreads = 8;
         } else if (p == 2) {
           RD_A(afr2[1], as1, 1);
+          nreads = 4;
         } else if (!last) {
           RD_A(afr2[0], as0n, 0);
           RD_B(bfr2[0], as0n + B_RING);
+          nreads = 8;
         }
-        __builtin_amdgcn_s_setprio(1);
+        if (PIPE == 3) __builtin_amdgcn_s_setprio(1);
         if (p == 0) { MFMA_PHASE(afr2[0], bfr2[0], 0); }
         else if (p == 1) { MFMA_PHASE(afr2[1], bfr2[0], 1); }
         else if (p == 2) { MFMA_PHASE(afr2[0], bfr2[1], 0); }
         else { MFMA_PHASE(afr2[1], bfr2[1], 1); }
-        __builtin_amdgcn_s_setprio(0);
+        if (PIPE == 3) __builtin_amdgcn_s_setprio(0);
+        if (PIPE == 4) {
+          // emit the region as [2 MFMA, 1 ds_read] x nreads, remaining
+          // MFMAs after (guide T19 masks: MFMA=0x8, DS_READ=0x100)
+          if (nreads == 8) {
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+              __builtin_amdgcn_sched_group_barrier(0x8, 2, 0);
+              __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);
+            }
+          } else if (nreads == 4) {
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+              __builtin_amdgcn_sched_group_barrier(0x8, 3, 0);
+              __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);
+            }
+            __builtin_amdgcn_sched_group_barrier(0x8, 4, 0);
+          }
+        }
+        if (PIPE == 4 && pre) {
+          if (p == 0) STAGE_B(ss0, kt + 2, 0)
+          else if (p == 1) STAGE_A(ss0, kt + 2, 0)
+          else if (p == 2) STAGE_B(ss1, kt + 2, 1)
+          else STAGE_A(ss1, kt + 2, 1)
+        }
         if (p & 1) {
           if (pre) dtx_vmcnt4(); else dtx_vmcnt0();
           dtx_bar();

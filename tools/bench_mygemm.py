@@ -68,3 +68,29 @@ for name, m, n, k in shapes:
           f"ratio {tl/tm:5.2f}x")
     del a, b, src
     torch.cuda.empty_cache()
+
+
+# ---- dgrad comparison: lib dy@W (NN layout) vs mine dy@(W^T)^T (NT via
+# the cached transpose). The W^T trick means our dgrad runs at NT speed.
+print("\n==== dgrad ====")
+for name, m, n, k in shapes:
+    dy = (torch.rand((m, n), device=dev) * 2 - 1).to(torch.bfloat16)
+    w = (torch.rand((n, k), device=dev) * 2 - 1).to(torch.bfloat16)
+    wt = w.t().contiguous()
+    ref = dy[:256].float() @ w.float()
+    got = ops.gemm_nt(dy, wt)
+    err = (got[:256].float() - ref).abs().max() / ref.abs().max()
+    fl = 2.0 * m * n * k
+    for _ in range(3):
+        ops.gemm_nt(dy, wt)
+        dy @ w
+    mine, lib = [], []
+    for _ in range(5):
+        mine.append(timeit(lambda: ops.gemm_nt(dy, wt), 5))
+        lib.append(timeit(lambda: dy @ w, 5))
+    tm, tl = min(mine), min(lib)
+    print(f"{name} dgrad err {err:.2e}  mine {tm*1e3:7.3f} ms "
+          f"({fl/tm/1e12:7.1f} TF/s)  lib {tl*1e3:7.3f} ms "
+          f"({fl/tl/1e12:7.1f} TF/s)  ratio {tl/tm:5.2f}x")
+    del dy, w, wt
+    torch.cuda.empty_cache()
