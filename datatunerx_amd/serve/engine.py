@@ -162,8 +162,16 @@ class _GraphedDecoder:
 
 
 class InferenceEngine:
+    """One generation context over a (shared, read-only) model. For
+    concurrent serving, build several engines over the SAME model
+    (serve/server.py EnginePool): each holds its own KV caches, hip
+    graph and HIP stream, so batch-1 decodes overlap on the GPU instead
+    of queueing on one engine lock (decode is latency-bound, not
+    occupancy-bound)."""
+
     def __init__(self, model, tokenizer=None, template: str = "llama2",
-                 device=None, graph_decode: bool = True):
+                 device=None, graph_decode: bool = True,
+                 own_stream: bool = False):
         self.model = model
         self.tok = tokenizer or ByteTokenizer()
         self.template = template
@@ -171,10 +179,19 @@ class InferenceEngine:
         self.is_llama = isinstance(model, LlamaForCausalLM)
         import torch.distributed as dist
         self._graphed = None
+        self._stream = (torch.cuda.Stream(device=self.device)
+                        if own_stream and self.device.type == "cuda"
+                        else None)
         self._graph_ok = (graph_decode and self.is_llama and
                           self.device.type == "cuda" and
                           not dist.is_initialized() and
                           os.environ.get("DTX_NO_GRAPH") != "1")
+
+    def _stream_ctx(self):
+        import contextlib
+        if self._stream is None:
+            return contextlib.nullcontext()
+        return torch.cuda.stream(self._stream)
 
     def _get_graphed(self):
         if self._graphed is None and self._graph_ok:
@@ -191,6 +208,11 @@ class InferenceEngine:
     def chat(self, messages: List[dict], max_tokens: int = 64,
              temperature: float = 0.0, top_p: float = 1.0) -> str:
         """messages: [{role, content}] -> completion text."""
+        with self._stream_ctx():
+            return self._chat(messages, max_tokens, temperature, top_p)
+
+    def _chat(self, messages: List[dict], max_tokens: int = 64,
+              temperature: float = 0.0, top_p: float = 1.0) -> str:
         system = ""
         history = []
         query = ""
@@ -211,6 +233,12 @@ class InferenceEngine:
 
     def chat_stream(self, messages: List[dict], max_tokens: int = 64,
                     temperature: float = 0.0, top_p: float = 1.0):
+        for delta in self._chat_stream_inner(messages, max_tokens,
+                                             temperature, top_p):
+            yield delta
+
+    def _chat_stream_inner(self, messages, max_tokens, temperature,
+                           top_p):
         """Like chat() but yields text DELTAS as tokens decode (the
         serving endpoint streams these as SSE chunks). Token-identical
         to chat(): same generate loop, so TP followers running chat()
@@ -256,6 +284,14 @@ class InferenceEngine:
     def generate_stream(self, prompt_ids: List[int],
                         max_new_tokens: int = 64, temperature: float = 0.0,
                         top_p: float = 1.0):
+        with self._stream_ctx():
+            yield from self._generate_stream(prompt_ids, max_new_tokens,
+                                             temperature, top_p)
+
+    @torch.no_grad()
+    def _generate_stream(self, prompt_ids: List[int],
+                         max_new_tokens: int = 64,
+                         temperature: float = 0.0, top_p: float = 1.0):
         ids = torch.tensor([prompt_ids], dtype=torch.long,
                            device=self.device)
         n_out = 0
@@ -343,6 +379,10 @@ class InferenceEngine:
     # ----------------------------------------------------------- score
     @torch.no_grad()
     def perplexity(self, texts: List[str]) -> float:
+        with self._stream_ctx():
+            return self._perplexity(texts)
+
+    def _perplexity(self, texts: List[str]) -> float:
         """Mean perplexity over texts (built-in Scoring metric)."""
         losses = []
         for t in texts:

@@ -2,17 +2,20 @@
 reference's Scoring hits — finetunejob_controller.go:433), /v1/score
 (built-in scoring metric), /health.
 
-Std-lib ThreadingHTTPServer (no web-framework dependency): the service
-is an ephemeral per-job evaluation endpoint (torn down after scoring,
-SURVEY.md §3.4), so a single-model, low-QPS server is the right shape.
-Generation requests are serialized through an engine lock (one GPU, one
-model instance).
+Std-lib ThreadingHTTPServer (no web-framework dependency). Concurrency:
+a pool of generation engines shares the (read-only) model weights —
+each engine holds its own KV caches, hip graph and HIP stream, so
+batch-1 decodes from concurrent requests overlap on the GPU
+(DTX_SERVE_CONCURRENCY, default 2). TP mode stays
+single-engine + lock: follower ranks step collectives in lockstep with
+exactly one request at a time.
 """
 
 from __future__ import annotations
 
 import argparse
 import json
+import os
 import threading
 import time
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
@@ -20,8 +23,35 @@ from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 import torch
 
 
-def build_handler(engine):
-    lock = threading.Lock()
+class EnginePool:
+    """Blocking pool of engines over one shared model."""
+
+    def __init__(self, engines):
+        import queue
+        self._q = queue.Queue()
+        for e in engines:
+            self._q.put(e)
+
+    def acquire(self):
+        return self._q.get()
+
+    def release(self, e):
+        self._q.put(e)
+
+
+def build_handler(pool):
+    if not isinstance(pool, EnginePool):
+        pool = EnginePool([pool])
+
+    class _Lease:
+        def __enter__(self):
+            self.e = pool.acquire()
+            return self.e
+
+        def __exit__(self, *a):
+            pool.release(self.e)
+
+    lock = _Lease()
 
     class Handler(BaseHTTPRequestHandler):
         def log_message(self, *a):       # quiet
@@ -68,7 +98,7 @@ def build_handler(engine):
                         # would leave follower ranks blocked in
                         # dist.broadcast — desyncing every later
                         # request's collectives.
-                        with lock:
+                        with lock as engine:
                             gen = engine.chat_stream(*args)
                             client_gone = False
                             for delta in gen:
@@ -95,7 +125,7 @@ def build_handler(engine):
                                     OSError):
                                 pass
                         return
-                    with lock:
+                    with lock as engine:
                         text = engine.chat(*args)
                     self._send(200, {
                         "id": f"chatcmpl-{int(time.time()*1000)}",
@@ -109,7 +139,7 @@ def build_handler(engine):
                         }],
                     })
                 elif self.path == "/v1/score":
-                    with lock:
+                    with lock as engine:
                         ppl = engine.perplexity(body.get("texts", []))
                     self._send(200, {"perplexity": ppl})
                 else:
@@ -235,9 +265,13 @@ def main(argv=None):
     from ..models.hf_io import load_tokenizer
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     model = build_model(args.model, device, adapter_dir=args.adapter)
-    engine = InferenceEngine(model, tokenizer=load_tokenizer(args.model),
-                             template=args.template, device=device)
-    serve_forever(engine, args.host, args.port)
+    tok = load_tokenizer(args.model)
+    n_slots = max(1, int(os.environ.get("DTX_SERVE_CONCURRENCY", "2")))
+    pool = EnginePool([
+        InferenceEngine(model, tokenizer=tok, template=args.template,
+                        device=device, own_stream=(n_slots > 1))
+        for _ in range(n_slots)])
+    serve_forever(pool, args.host, args.port)
 
 
 if __name__ == "__main__":
