@@ -225,16 +225,21 @@ class LlamaForCausalLM(nn.Module):
         h = self.hidden_states(input_ids, pos0, kv_caches, pos_dev)
         if labels is None:
             return self.lm_head(h)
+        if os.environ.get("DTX_NO_FUSED_CE") != "1" and \
+                self.cfg.vocab_size % 8 == 0:
+            # chunked-vocab fused lm_head+CE: no [M,V] logits tensor.
+            # Keep ALL positions (h stays the contiguous buffer — no
+            # [M,E] copy) and mark each row's LAST position ignored
+            # instead of slicing the shift.
+            from ..ops.autograd import fused_linear_cross_entropy
+            t = torch.full_like(labels, -100)
+            t[:, :-1] = labels[:, 1:]
+            return fused_linear_cross_entropy(
+                h.reshape(-1, self.cfg.hidden_size),
+                self.lm_head.weight, t.reshape(-1), ignore_index=-100)
         # shift: predict token t+1 from position t
         h = h[:, :-1, :].reshape(-1, self.cfg.hidden_size)
         targets = labels[:, 1:].reshape(-1)
-        if os.environ.get("DTX_NO_FUSED_CE") != "1" and \
-                self.cfg.vocab_size % 8 == 0:
-            # chunked-vocab fused lm_head+CE: no [M,V] logits tensor
-            from ..ops.autograd import fused_linear_cross_entropy
-            return fused_linear_cross_entropy(
-                h.contiguous(), self.lm_head.weight, targets,
-                ignore_index=-100)
         logits = self.lm_head(h)
         return cross_entropy(logits, targets, ignore_index=-100)
 
