@@ -250,6 +250,7 @@ class FusedLinearCrossEntropy(torch.autograd.Function):
     """
 
     CHUNK = int(os.environ.get("DTX_CE_CHUNK", "4096"))
+    MBLOCK = int(os.environ.get("DTX_CE_MBLOCK", "8192"))
 
     @staticmethod
     def forward(ctx, h2, w, targets, ignore_index, need_grads):
@@ -260,10 +261,18 @@ class FusedLinearCrossEntropy(torch.autograd.Function):
         l = torch.zeros(M, device=dev, dtype=torch.float32)
         tgt = torch.zeros(M, device=dev, dtype=torch.float32)
         ch = FusedLinearCrossEntropy.CHUNK
-        for v0 in range(0, V, ch):
-            wc = w[v0:v0 + ch]
-            lc = torch.nn.functional.linear(h2, wc)
-            xent_lse_merge(lc, targets, m, l, tgt, v0, ignore_index)
+        mb = FusedLinearCrossEntropy.MBLOCK
+        # block over rows TOO: at mb=8192, ch=4096 the chunk logits
+        # (64 MB), the dx row-block and the W chunk stay L3-resident
+        # across the vocab sweep; without the row block a chunk is
+        # ~200 MB and every re-read/accumulate round-trips HBM.
+        for r0 in range(0, M, mb):
+            hr = h2[r0:r0 + mb]
+            tr = targets[r0:r0 + mb]
+            for v0 in range(0, V, ch):
+                lc = torch.nn.functional.linear(hr, w[v0:v0 + ch])
+                xent_lse_merge(lc, tr, m[r0:r0 + mb], l[r0:r0 + mb],
+                               tgt[r0:r0 + mb], v0, ignore_index)
         lse = m + l.log()
         valid = targets != ignore_index
         n_valid = valid.sum().clamp(min=1)
@@ -275,13 +284,18 @@ class FusedLinearCrossEntropy(torch.autograd.Function):
             dx = torch.zeros_like(h2)
             if w.requires_grad:
                 dw = torch.zeros_like(w)
-            for v0 in range(0, V, ch):
-                wc = w[v0:v0 + ch]
-                lc = torch.nn.functional.linear(h2, wc)
-                dl = xent_dlogits(lc, targets, lse, v0, ignore_index)
-                dx.addmm_(dl, wc)
-                if dw is not None:
-                    dw[v0:v0 + ch].addmm_(dl.t(), h2)
+            for r0 in range(0, M, mb):
+                hr = h2[r0:r0 + mb]
+                tr = targets[r0:r0 + mb]
+                lr = lse[r0:r0 + mb]
+                dxr = dx[r0:r0 + mb]
+                for v0 in range(0, V, ch):
+                    wc = w[v0:v0 + ch]
+                    lc = torch.nn.functional.linear(hr, wc)
+                    dl = xent_dlogits(lc, tr, lr, v0, ignore_index)
+                    dxr.addmm_(dl, wc)
+                    if dw is not None:
+                        dw[v0:v0 + ch].addmm_(dl.t(), hr)
         ctx.save_for_backward(dx, dw, n_valid)
         return loss
 
