@@ -281,3 +281,19 @@ def test_engine_stream_matches_chat_cpu():
     full = eng.chat(msgs, 8)
     streamed = "".join(eng.chat_stream(msgs, 8))
     assert streamed == full
+
+
+def test_model_loss_fused_ce_env(monkeypatch):
+    """DTX_FUSED_CE=1 routes the model loss through the chunked fused
+    path and matches the materialized default."""
+    import torch
+
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=True,
+                         dtype=torch.float32).init_random()
+    ids = torch.randint(3, 512, (2, 32))
+    base = float(m(ids, labels=ids.clone()))
+    monkeypatch.setenv("DTX_FUSED_CE", "1")
+    fused = float(m(ids, labels=ids.clone()))
+    assert abs(base - fused) < 1e-4
