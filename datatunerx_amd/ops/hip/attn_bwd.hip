@@ -141,8 +141,10 @@ void attn_delta2_kernel(const unsigned short* __restrict__ dO,
 // pipe that idles ~85% of the time).
 template <int D>
 struct DkdvLds {
-  unsigned short Qr[64][D + 8];     // staged q rows (two 32-row tiles)
-  unsigned short dOr[64][D + 8];
+  // double-buffered staged q rows (two 32-row tiles per slot): one
+  // barrier per stage instead of the [sync; write; sync] full stop
+  unsigned short Qr[2][64][D + 8];
+  unsigned short dOr[2][64][D + 8];
   // no transposed images: dV/dK B-fragments come from these row-major
   // tiles via ds_read_b64_tr_b16 (tr_bfrag) — halves the LDS footprint
   // (4 blocks/CU co-resident) and the staged global traffic
@@ -165,7 +167,7 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
   __shared__ DkdvLds<D> lds;
   // separate object: float arrays INSIDE DkdvLds made hipcc scalarize
   // every b128 fragment read of the struct (see profiles/README.md)
-  __shared__ float lsed[64 + 64];
+  __shared__ float lsed[2][64 + 64];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -237,27 +239,35 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
         stg[it * 2 + 1] = d8;
       }
     };
-    auto write_stage = [&]() {
+    auto write_stage = [&](int slot) {
 #pragma unroll
       for (int it = 0; it < RIT; ++it) {
         const int idx = threadIdx.x + it * 512;
         const int row = idx / (D / 8), g = idx % (D / 8);
-        *reinterpret_cast<short8v*>(&lds.Qr[row][g * 8]) = stg[it * 2];
-        *reinterpret_cast<short8v*>(&lds.dOr[row][g * 8]) =
+        *reinterpret_cast<short8v*>(&lds.Qr[slot][row][g * 8]) =
+            stg[it * 2];
+        *reinterpret_cast<short8v*>(&lds.dOr[slot][row][g * 8]) =
             stg[it * 2 + 1];
       }
     };
-    if (qs_lo < qs_hi) issue_stage(qs_lo);
+    auto write_lsed = [&](int slot, int qs) {
+      if (threadIdx.x < 64) {
+        const int qg = qs * 64 + threadIdx.x;
+        lsed[slot][threadIdx.x] = qg < S ? lse_in[lbase + qg] : 0.f;
+        lsed[slot][64 + threadIdx.x] =
+            qg < S ? delta_in[lbase + qg] : 0.f;
+      }
+    };
+    int cur = 0;
+    if (qs_lo < qs_hi) {
+      // prologue: first stage lands in slot 0 before the loop
+      issue_stage(qs_lo);
+      write_stage(0);
+      write_lsed(0, qs_lo);
+      __syncthreads();
+    }
     for (int qs = qs_lo; qs < qs_hi; ++qs) {
       const int q0s = qs * 64;
-      __syncthreads();
-      write_stage();
-      if (threadIdx.x < 64) {
-        const int qg = q0s + threadIdx.x;
-        lsed[threadIdx.x] = qg < S ? lse_in[lbase + qg] : 0.f;
-        lsed[64 + threadIdx.x] = qg < S ? delta_in[lbase + qg] : 0.f;
-      }
-      __syncthreads();
       if (qs + 1 < qs_hi) issue_stage(qs + 1);
 
 #pragma unroll
@@ -278,16 +288,16 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
         for (int kc = 0; kc < DC16; ++kc) {
           short8v qa = *reinterpret_cast<const short8v*>(
-              &lds.Qr[qoff + l31][kc * 16 + hi * 8]);
+              &lds.Qr[cur][qoff + l31][kc * 16 + hi * 8]);
           sv = MFMA32(qa, kfrag[kc], sv);
         }
       } else {
 #pragma unroll
         for (int kc = 0; kc < DC16; ++kc) {
           short8v qa = *reinterpret_cast<const short8v*>(
-              &lds.Qr[qoff + l31][kc * 16 + hi * 8]);
+              &lds.Qr[cur][qoff + l31][kc * 16 + hi * 8]);
           short8v da = *reinterpret_cast<const short8v*>(
-              &lds.dOr[qoff + l31][kc * 16 + hi * 8]);
+              &lds.dOr[cur][qoff + l31][kc * 16 + hi * 8]);
           sv = MFMA32(qa, kfrag[kc], sv);
           dpv = MFMA32(da, vfrag[kc], dpv);
         }
@@ -301,8 +311,8 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qg = q0 + bw_crow(r, hi);
-        const float lse2 = lsed[qoff + bw_crow(r, hi)] * LOG2E;
-        const float dlt = lsed[64 + qoff + bw_crow(r, hi)];
+        const float lse2 = lsed[cur][qoff + bw_crow(r, hi)] * LOG2E;
+        const float dlt = lsed[cur][64 + qoff + bw_crow(r, hi)];
         float p;
         if (interior) {
           p = __builtin_exp2f(sv[r] * kscale - lse2);
@@ -320,8 +330,8 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
       if (role_dv) conv_c_to_frag(sv, f0, f1);
       else conv_c_to_frag(dpv, f0, f1);
       const DTX_AS3 unsigned short* img3 = role_dv
-          ? (const DTX_AS3 unsigned short*)&lds.dOr[0][0]
-          : (const DTX_AS3 unsigned short*)&lds.Qr[0][0];
+          ? (const DTX_AS3 unsigned short*)&lds.dOr[cur][0][0]
+          : (const DTX_AS3 unsigned short*)&lds.Qr[cur][0][0];
 #pragma unroll
       for (int c = 0; c < ND32; ++c) {
         // B[k=q][n=d] straight from the row-major tile (tr-read)
@@ -334,6 +344,14 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
       }
       }  // live_tile
       }  // qh
+      if (qs + 1 < qs_hi) {
+        write_stage(cur ^ 1);
+        write_lsed(cur ^ 1, qs + 1);
+      }
+      // one barrier per stage: publishes slot cur^1 AND guards the
+      // next iteration's write into the slot every wave just read
+      __syncthreads();
+      cur ^= 1;
     }
   }
 
@@ -354,8 +372,9 @@ void attn_bwd_dkdv2_kernel(const unsigned short* __restrict__ Q,
 // ---------------------------------------------------------------- dq
 template <int D>
 struct DqLds {
-  unsigned short K[128][D + 8];
-  unsigned short V[128][D + 8];
+  // double-buffered (one barrier per 128-row stage; see DkdvLds)
+  unsigned short K[2][128][D + 8];
+  unsigned short V[2][128][D + 8];
   // dQ's K B-fragments come from the row-major K tile via
   // ds_read_b64_tr_b16 (tr_bfrag) — no transposed KT image
 };
@@ -447,20 +466,22 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
       stg[it * 2 + 1] = v8;
     }
   };
-  auto write_stage = [&]() {
+  auto write_stage = [&](int slot) {
 #pragma unroll
     for (int it = 0; it < KIT; ++it) {
       const int idx = threadIdx.x + it * 512;
       const int row = idx / (D / 8), g = idx % (D / 8);
-      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it * 2];
-      *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = stg[it * 2 + 1];
+      *reinterpret_cast<short8v*>(&lds.K[slot][row][g * 8]) =
+          stg[it * 2];
+      *reinterpret_cast<short8v*>(&lds.V[slot][row][g * 8]) =
+          stg[it * 2 + 1];
     }
   };
   issue_stage(0);
+  write_stage(0);
+  __syncthreads();
+  int cur = 0;
   for (int st2 = 0; st2 < nstages; ++st2) {
-    __syncthreads();
-    write_stage();
-    __syncthreads();
     if (st2 + 1 < nstages) issue_stage(st2 + 1);
     const int kvs = st2 * 128;
 
@@ -483,9 +504,9 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
         for (int kc = 0; kc < DC16; ++kc) {
           short8v ka = *reinterpret_cast<const short8v*>(
-              &lds.K[koff + ss * 32 + l31][kc * 16 + hi * 8]);
+              &lds.K[cur][koff + ss * 32 + l31][kc * 16 + hi * 8]);
           short8v va = *reinterpret_cast<const short8v*>(
-              &lds.V[koff + ss * 32 + l31][kc * 16 + hi * 8]);
+              &lds.V[cur][koff + ss * 32 + l31][kc * 16 + hi * 8]);
           st = MFMA32(ka, qfrag[kc], st);
           dpt = MFMA32(va, dofrag[kc], dpt);
         }
@@ -506,7 +527,7 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
         short8v f0, f1;
         conv_c_to_frag(dst, f0, f1);
         const DTX_AS3 unsigned short* k3 =
-            (const DTX_AS3 unsigned short*)&lds.K[0][0];
+            (const DTX_AS3 unsigned short*)&lds.K[cur][0][0];
         // two passes over c so consecutive MFMAs hit different
         // accumulators (dq_acc[0..3]) instead of pairing on one
 #pragma unroll
@@ -526,6 +547,9 @@ void attn_bwd_dq2_kernel(const unsigned short* __restrict__ Q,
       }
     }
     }  // kh
+    if (st2 + 1 < nstages) write_stage(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: dQ C-layout [q regs][d lanes] -> BSHD stores

@@ -69,8 +69,10 @@ __device__ __forceinline__ short8v fw_tr_bfrag(
 
 template <int D>
 struct AttnFwdLds {
-  unsigned short K[128][D + 8];
-  unsigned short V[128][D + 8];   // row-major; PV B-frags via tr-read
+  // double-buffered stages: ONE barrier per 128 kv rows instead of the
+  // [sync; write; sync] full stop (same restructure as the bwd kernels)
+  unsigned short K[2][128][D + 8];
+  unsigned short V[2][128][D + 8];  // row-major; PV B-frags via tr-read
 };
 
 // WAVES: q-rows per workgroup = 32*WAVES. 8 (256 rows) shares each
@@ -162,20 +164,22 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       stg[it * 2 + 1] = v8;
     }
   };
-  auto write_stage = [&]() {
+  auto write_stage = [&](int slot) {
 #pragma unroll
     for (int it = 0; it < KIT; ++it) {
       const int idx = threadIdx.x + it * TPB;
       const int row = idx / (D / 8), g = idx % (D / 8);
-      *reinterpret_cast<short8v*>(&lds.K[row][g * 8]) = stg[it * 2];
-      *reinterpret_cast<short8v*>(&lds.V[row][g * 8]) = stg[it * 2 + 1];
+      *reinterpret_cast<short8v*>(&lds.K[slot][row][g * 8]) =
+          stg[it * 2];
+      *reinterpret_cast<short8v*>(&lds.V[slot][row][g * 8]) =
+          stg[it * 2 + 1];
     }
   };
   issue_stage(0);
+  write_stage(0);
+  __syncthreads();
+  int cur = 0;
   for (int st2 = 0; st2 < nstages; ++st2) {
-    __syncthreads();
-    write_stage();
-    __syncthreads();
     if (st2 + 1 < nstages) issue_stage(st2 + 1);
     const int kvs = st2 * 128;
 
@@ -193,9 +197,9 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
       for (int kc = 0; kc < DC16; ++kc) {
         short8v k0 = *reinterpret_cast<const short8v*>(
-            &lds.K[koff + l31][kc * 16 + hi * 8]);
+            &lds.K[cur][koff + l31][kc * 16 + hi * 8]);
         short8v k1 = *reinterpret_cast<const short8v*>(
-            &lds.K[koff + 32 + l31][kc * 16 + hi * 8]);
+            &lds.K[cur][koff + 32 + l31][kc * 16 + hi * 8]);
         s0v = MFMA32(k0, qfrag[kc], s0v);
         s1v = MFMA32(k1, qfrag[kc], s1v);
       }
@@ -295,7 +299,7 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       // (o_acc[0..3]) — back-to-back MFMAs on one accumulator pay the
       // full RAW latency (guide: SQ_WAIT_INST_ANY).
       const DTX_AS3F unsigned short* v3 =
-          (const DTX_AS3F unsigned short*)&lds.V[0][0];
+          (const DTX_AS3F unsigned short*)&lds.V[cur][0][0];
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
 #pragma unroll
@@ -308,6 +312,9 @@ void attn_fwd2_kernel(const unsigned short* __restrict__ Q,
       }
     }
     }  // half
+    if (st2 + 1 < nstages) write_stage(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: normalize rows, store O per wave IMMEDIATELY (an
