@@ -207,3 +207,47 @@ def load_tokenizer(name_or_dir: str):
         return HFTokenizer.from_dir(name_or_dir)
     from ..data.dataset import ByteTokenizer
     return ByteTokenizer()
+
+
+@torch.no_grad()
+def save_hf_model(model, out_dir: str):
+    """Export this repo's LlamaForCausalLM in HF format (config.json +
+    model.safetensors with HF names) so transformers can load the
+    trained full-param model — the inverse of load_hf_weights."""
+    from safetensors.torch import save_file
+    os.makedirs(out_dir, exist_ok=True)
+    cfg = model.cfg
+    head_dim = getattr(cfg, "head_dim", None) or (
+        cfg.hidden_size // cfg.num_attention_heads)
+    with open(os.path.join(out_dir, "config.json"), "w") as f:
+        json.dump({
+            "architectures": ["LlamaForCausalLM"],
+            "model_type": "llama",
+            "vocab_size": cfg.vocab_size,
+            "hidden_size": cfg.hidden_size,
+            "intermediate_size": cfg.intermediate_size,
+            "num_hidden_layers": cfg.num_hidden_layers,
+            "num_attention_heads": cfg.num_attention_heads,
+            "num_key_value_heads": cfg.num_key_value_heads,
+            "max_position_embeddings": cfg.max_position_embeddings,
+            "rms_norm_eps": cfg.rms_norm_eps,
+            "rope_theta": cfg.rope_theta,
+            "head_dim": head_dim,
+            "torch_dtype": "bfloat16",
+            "tie_word_embeddings": False,
+        }, f, indent=2)
+    sd = {}
+    for name, p in model.named_parameters():
+        if "lora_" in name:
+            continue                    # adapters export separately
+        if name == "norm":
+            hf = "model.norm.weight"
+        elif name == "lm_head.weight":
+            hf = "lm_head.weight"
+        elif name.endswith(("input_layernorm", "post_attention_layernorm")):
+            hf = f"model.{name}.weight"
+        else:
+            hf = f"model.{name}"
+        sd[hf] = p.detach().cpu().contiguous()
+    save_file(sd, os.path.join(out_dir, "model.safetensors"))
+    return out_dir

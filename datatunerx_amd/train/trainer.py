@@ -535,10 +535,18 @@ class SFTTrainer:
                          target_modules=list(self.cfg.lora_targets),
                          base_model_name_or_path=self.cfg.base_model)
         else:
-            from safetensors.torch import save_file
-            sd = {n: p.detach().cpu().contiguous()
-                  for n, p in self.model.trainable_parameters()}
-            save_file(sd, os.path.join(out_dir, "model.safetensors"))
+            from ..models import LlamaForCausalLM
+            if isinstance(self.model, LlamaForCausalLM):
+                # full-param Llama: HF-format export (config.json +
+                # HF-named safetensors) so transformers / the hf_io
+                # loader can consume the checkpoint directly
+                from ..models.hf_io import save_hf_model
+                save_hf_model(self.model, out_dir)
+            else:
+                from safetensors.torch import save_file
+                sd = {n: p.detach().cpu().contiguous()
+                      for n, p in self.model.trainable_parameters()}
+                save_file(sd, os.path.join(out_dir, "model.safetensors"))
         torch.save({"optimizer": self.opt.state_dict(),
                     "global_step": self.global_step},
                    os.path.join(out_dir, self._opt_state_name(self.rank)))

@@ -132,3 +132,22 @@ def test_train_on_hf_dir_end_to_end(hf_dir, tmp_path):
                           template="vanilla", device=torch.device("cpu"))
     txt = eng.chat([{"role": "user", "content": "hello"}], max_tokens=4)
     assert isinstance(txt, str)
+
+
+def test_hf_export_roundtrip(hf_dir, tmp_path):
+    """save_hf_model output loads in transformers with matching logits
+    (full-param checkpoint interchange)."""
+    from datatunerx_amd.models.hf_io import save_hf_model
+    cfg = load_hf_config(hf_dir)
+    ours = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32)
+    load_hf_weights(ours, hf_dir)
+    out = str(tmp_path / "export")
+    save_hf_model(ours, out)
+    ref = transformers.LlamaForCausalLM.from_pretrained(
+        out, torch_dtype=torch.float32, attn_implementation="eager")
+    ref.eval()
+    ours.eval()
+    ids = torch.randint(0, 320, (1, 13))
+    with torch.no_grad():
+        assert torch.allclose(ours(ids), ref(ids).logits, atol=2e-4,
+                              rtol=1e-3)
