@@ -103,7 +103,10 @@ __device__ __forceinline__ unsigned swz_f(unsigned x) { return (0u - x) & 3u; }
 //       vmcnt(4) at both odd phases (publication one phase earlier) +
 //       fragment PRELOAD: phase p+1's ds_reads issue before phase p's
 //       MFMA cluster (incl. across the tile seam), so LDS latency and
-//       read issue hide inside the MFMA stream; 4 = PIPE3 +
+//       read issue hide inside the MFMA stream; 5 = B on the 4-slot
+//       ring, A on the 5-slot ring with the A stages shifted one phase
+//       pair earlier — every slot reuse then crosses the p3 barrier, so
+//       ONE barrier + ONE vmcnt(2) per tile; 4 = PIPE3 +
 //       sched_group_barrier [MFMA,MFMA,ds_read] interleave (the reads
 //       are EMITTED inside the MFMA cluster so the matrix pipe is never
 //       starved during the read window) + static young-half s_setprio
@@ -118,7 +121,7 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
                     const unsigned short* __restrict__ SRC,
                     unsigned short* __restrict__ C,
                     long M, int N, int K, int mb_n) {
-  constexpr int RING = (PIPE >= 2) ? 5 : 4;
+  constexpr int RING = (PIPE >= 2) ? 5 : 4;   // PIPE5: A-ring only
   constexpr bool PRELOAD = (PIPE == 3) || (PIPE == 4);
   constexpr unsigned B_RING = RING * HALF_BYTES;
   __shared__ __attribute__((aligned(16))) unsigned char
@@ -233,7 +236,19 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
 #define SLOT5(h) ((unsigned)((h) % 5) * HALF_BYTES)
 
   // ---- prologue
-  if (PIPE >= 2) {
+  if (PIPE == 5) {
+    // tile0's four halves + tile1's A-kh0 (the one half the in-loop
+    // schedule stages at kt-1.p2); vmcnt(2) leaves exactly that A-kh0
+    // in flight = the steady-state invariant at tile entry
+    STAGE_B(SLOT4(0), 0, 0); STAGE_B(SLOT4(1), 0, 1);
+    STAGE_A(SLOT5(0), 0, 0); STAGE_A(SLOT5(1), 0, 1);
+    if (KT > 1) {
+      STAGE_A(SLOT5(2), 1, 0);
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    } else {
+      dtx_vmcnt0();
+    }
+  } else if (PIPE >= 2) {
     // stage tiles 0 and 1 (slots h%5), wait for tile 0's 8 ops
     STAGE_B(SLOT5(0), 0, 0); STAGE_A(SLOT5(0), 0, 0);
     STAGE_B(SLOT5(1), 0, 1); STAGE_A(SLOT5(1), 0, 1);
@@ -260,6 +275,7 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
   //   barriers at p1/p3 ends gate slot reuse (WAR) and publication.
   short8v bfr[4];
   short8v afr2[2][4], bfr2[2][4];      // PIPE3/4 double-buffered fragments
+  // PIPE5 slot bookkeeping: B slots (2kt+kh)&3, A slots (2kt+kh)%5
   if (PIPE == 4) {
     // T5 static form: the second-dispatched half loses VALU arbitration
     // on every segment; one wave-uniform setprio before the loop.
@@ -366,12 +382,17 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
   } else
   for (int kt = 0; kt < KT; ++kt) {
 
-    const unsigned as0 = (PIPE == 2) ? (unsigned)h5 * HALF_BYTES
-                                     : SLOT4(2 * kt);
+    const unsigned as0 = (PIPE == 2 || PIPE == 5)
+                             ? (unsigned)h5 * HALF_BYTES
+                             : SLOT4(2 * kt);
     const int h5b = h5 + 1 - (h5 + 1 >= 5 ? 5 : 0);
-    const unsigned as1 = (PIPE == 2) ? (unsigned)h5b * HALF_BYTES
-                                     : SLOT4(2 * kt + 1);
+    const unsigned as1 = (PIPE == 2 || PIPE == 5)
+                             ? (unsigned)h5b * HALF_BYTES
+                             : SLOT4(2 * kt + 1);
+    // PIPE5: B lives on its own 4-slot ring
+    const unsigned bs0 = SLOT4(2 * kt), bs1 = SLOT4(2 * kt + 1);
     // stage destination slots (kt+1 for 4-ring, kt+2 for 5-ring)
+    const int h5b2 = h5b + 2 - (h5b + 2 >= 5 ? 5 : 0); // (2kt+3) % 5
     const int h5s0 = h5 + 4 - (h5 + 4 >= 5 ? 5 : 0);   // (2kt+4) % 5
     const int h5s1 = h5s0 + 1 - (h5s0 + 1 >= 5 ? 5 : 0);
     const unsigned ss0 = (PIPE == 2) ? (unsigned)h5s0 * HALF_BYTES
@@ -386,7 +407,8 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
     for (int p = 0; p < 4; ++p) {
       const unsigned mh = p & 1;
       const unsigned as = (p < 2) ? as0 : as1;
-      const unsigned bs = ((p < 2) ? as0 : as1) + B_RING;
+      const unsigned bs = (PIPE == 5 ? ((p < 2) ? bs0 : bs1)
+                                     : ((p < 2) ? as0 : as1)) + B_RING;
       // ds_read register subtile for this phase
       short8v afr[4];
       short8v bfr32[2];
@@ -414,7 +436,19 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
               &lds[bs + wn * 4096 + fn * 2048 + mf1_row + pos]);
       }
       // issue the stage for this phase (one half per phase)
-      if (pre) {
+      if (PIPE == 5) {
+        // p0: B-kh0(kt+1) + A-kh1(kt+1); p1: B-kh1(kt+1);
+        // p2: A-kh0(kt+2). Every overwritten slot's last read is
+        // separated from the stage by the kt-1.p3 barrier.
+        if (p == 0 && kt + 1 < KT) {
+          STAGE_B(SLOT4(2 * kt + 2), kt + 1, 0)
+          STAGE_A((unsigned)h5b2 * HALF_BYTES, kt + 1, 1)
+        } else if (p == 1 && kt + 1 < KT) {
+          STAGE_B(SLOT4(2 * kt + 3), kt + 1, 1)
+        } else if (p == 2 && kt + 2 < KT) {
+          STAGE_A((unsigned)h5s0 * HALF_BYTES, kt + 2, 0)
+        }
+      } else if (pre) {
         if (p == 0) STAGE_B(ss0, kt_s, 0)
         else if (p == 1) STAGE_A(ss0, kt_s, 0)
         else if (p == 2) STAGE_B(ss1, kt_s, 1)
@@ -441,7 +475,16 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       // tile(s) issue no stages, so their in-flight count is too low
       // for the counted wait to retire what the next phases read —
       // drain fully there (end of the loop anyway).
-      if (p & 1) {
+      if (PIPE == 5) {
+        if (p == 3) {
+          if (kt + 2 < KT) {
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+          } else {
+            dtx_vmcnt0();
+          }
+          dtx_bar();
+        }
+      } else if (p & 1) {
         if (PIPE == 2) {
           if (p == 3) { if (pre) { asm volatile(
               "s_waitcnt vmcnt(8)" ::: "memory"); } else dtx_vmcnt0(); }

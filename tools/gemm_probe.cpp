@@ -39,10 +39,10 @@ void run(const unsigned short* A, const unsigned short* B,
 typedef void (*runfn)(const unsigned short*, const unsigned short*,
                       unsigned short*, long, int, int);
 #define NVAR 6
-static runfn FNS[NVAR] = {run<1, 2, 0>, run<1, 3, 0>, run<1, 4, 0>,
-                          run<2, 2, 0>, run<2, 4, 0>, run<2, 4, 0>};
-static const char* NAMES[NVAR] = {"g1p2", "g1p3", "g1p4",
-                                  "g2p2", "g2p4", "g2p4b"};
+static runfn FNS[NVAR] = {run<1, 2, 0>, run<1, 5, 0>, run<1, 5, 0>,
+                          run<2, 2, 0>, run<2, 5, 0>, run<2, 5, 0>};
+static const char* NAMES[NVAR] = {"g1p2", "g1p5", "g1p5b",
+                                  "g2p2", "g2p5", "g2p5b"};
 
 int main(int argc, char** argv) {
   if (argc < 5) { fprintf(stderr, "usage: see header\n"); return 1; }
