@@ -193,16 +193,33 @@ class InferenceEngine:
             return contextlib.nullcontext()
         return torch.cuda.stream(self._stream)
 
+    # hip graph CAPTURE must never run concurrently with another
+    # engine's capture or in-flight kernels (capture mode is
+    # context-global); all captures serialize on this lock. Replays of
+    # distinct graphs on distinct streams are concurrency-safe.
+    _capture_lock = __import__("threading").Lock()
+
     def _get_graphed(self):
         if self._graphed is None and self._graph_ok:
             try:
-                cfg = self.model.cfg
-                self._graphed = _GraphedDecoder(
-                    self.model, cfg, self.device,
-                    cfg.max_position_embeddings)
+                with InferenceEngine._capture_lock, self._stream_ctx():
+                    cfg = self.model.cfg
+                    self._graphed = _GraphedDecoder(
+                        self.model, cfg, self.device,
+                        cfg.max_position_embeddings)
             except Exception:
                 self._graph_ok = False
         return self._graphed
+
+    def warmup(self):
+        """Build the graphed decoder + run one tiny generation NOW (a
+        pool builds engines sequentially before serving concurrent
+        requests, so no capture happens mid-traffic)."""
+        with self._stream_ctx():
+            self._get_graphed()
+            list(self.generate([1], max_new_tokens=2))
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
 
     # ------------------------------------------------------------ chat
     def chat(self, messages: List[dict], max_tokens: int = 64,

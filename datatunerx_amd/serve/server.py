@@ -24,12 +24,18 @@ import torch
 
 
 class EnginePool:
-    """Blocking pool of engines over one shared model."""
+    """Blocking pool of engines over one shared model. Engines warm up
+    (hip graph capture + first kernels) SEQUENTIALLY here: capture is
+    context-global and must not overlap another engine's traffic."""
 
     def __init__(self, engines):
         import queue
         self._q = queue.Queue()
         for e in engines:
+            try:
+                e.warmup()
+            except Exception:
+                pass
             self._q.put(e)
 
     def acquire(self):
