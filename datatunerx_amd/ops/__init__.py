@@ -75,6 +75,11 @@ rope_tables = ref.rope_tables
 def rope_fwd(x, cos, sin, pos0: int = 0, pos_dev=None):
     if _gpu(x):
         return _EXT.rope(x, cos, sin, pos0, False, pos_dev)
+    if pos_dev is not None and pos_dev.numel() > 1:
+        # batched ragged decode: per-row positions
+        return torch.cat([ref.rope_fwd(x[i:i + 1], cos, sin,
+                                       pos0 + int(p))
+                          for i, p in enumerate(pos_dev)], dim=0)
     if pos_dev is not None:
         pos0 = pos0 + int(pos_dev)
     return ref.rope_fwd(x, cos, sin, pos0)
@@ -144,6 +149,16 @@ def attn_fwd(q, k, v, causal: bool = True, scale: float | None = None,
         # V consumed row-major (PV B-frags via ds_read_b64_tr_b16):
         # no V pre-transpose
         return _EXT.attn_fwd(q, k, v.contiguous(), causal, scale)
+    if len_dev is not None and len_dev.numel() > 1:
+        # batched ragged decode on CPU: per-row cache lengths
+        outs, lses = [], []
+        for i in range(q.shape[0]):
+            n = int(len_dev[i])
+            o, ls = ref.attn_fwd(q[i:i + 1], k[i:i + 1, :n],
+                                 v[i:i + 1, :n], causal, scale)
+            outs.append(o)
+            lses.append(ls)
+        return torch.cat(outs, 0), torch.cat(lses, 0)
     if len_dev is not None:
         k = k[:, :int(len_dev)]
         v = v[:, :int(len_dev)]

@@ -454,7 +454,7 @@ void attn_decode_kernel(const unsigned short* __restrict__ Q,
                         const int* __restrict__ len_dev,
                         float* __restrict__ part,   // [B*Hq, NS, D+2]
                         int B, int Hq, int Hkv, int Skv, int chunk,
-                        float scale) {
+                        float scale, int len_stride) {
   constexpr int LPR = D / 8;                   // lanes per row
   constexpr int SLOTS = 64 / LPR;              // row slots per wave
   const int lane = threadIdx.x & 63;
@@ -467,7 +467,7 @@ void attn_decode_kernel(const unsigned short* __restrict__ Q,
   const int hkv = hq / (Hq / Hkv);
   const int krowstr = Hkv * D;
   const long kbase = (long)b * Skv * krowstr + (long)hkv * D;
-  const int len = len_dev ? *len_dev : Skv;
+  const int len = len_dev ? len_dev[b * len_stride] : Skv;
   const int kv0 = blockIdx.x * chunk;
   const int kv_end = min(len, kv0 + chunk);
 
@@ -591,7 +591,8 @@ int attn_decode_nsplit(int Skv) {
 void launch_attn_decode(const void* q, const void* k, const void* v,
                         const int* len_dev, float* part, void* o,
                         float* lse, int B, int Hq, int Hkv, int Skv,
-                        int D, float scale, hipStream_t st) {
+                        int D, float scale, int len_stride,
+                        hipStream_t st) {
   const int ns = attn_decode_nsplit(Skv);
   const int chunk = DTX_CDIV(Skv, ns);
   dim3 grid(ns, B * Hq);
@@ -599,14 +600,14 @@ void launch_attn_decode(const void* q, const void* k, const void* v,
     attn_decode_kernel<128><<<grid, DTX_BLOCK, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
         (const unsigned short*)v, len_dev, part, B, Hq, Hkv, Skv, chunk,
-        scale);
+        scale, len_stride);
     attn_decode_combine_kernel<128><<<B * Hq, 64, 0, st>>>(
         part, (unsigned short*)o, lse, B, Hq, ns);
   } else if (D == 64) {
     attn_decode_kernel<64><<<grid, DTX_BLOCK, 0, st>>>(
         (const unsigned short*)q, (const unsigned short*)k,
         (const unsigned short*)v, len_dev, part, B, Hq, Hkv, Skv, chunk,
-        scale);
+        scale, len_stride);
     attn_decode_combine_kernel<64><<<B * Hq, 64, 0, st>>>(
         part, (unsigned short*)o, lse, B, Hq, ns);
   }

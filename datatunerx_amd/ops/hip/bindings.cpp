@@ -12,7 +12,7 @@ void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
                         const void*, void*, float*, float*, int, int,
                         hipStream_t);
 void launch_rope(const void*, const float*, const float*, void*, long, int,
-                 int, int, int, int, const int*, hipStream_t);
+                 int, int, int, int, const int*, int, hipStream_t);
 void launch_swiglu_fwd(const void*, const void*, void*, long, hipStream_t);
 void launch_swiglu_bwd(const void*, const void*, const void*, void*, void*,
                        long, hipStream_t);
@@ -55,7 +55,7 @@ void launch_gemm_nt(const void*, const void*, const void*, void*, long,
                     int, int, hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*,
                         const int*, float*, void*, float*, int, int, int,
-                        int, int, float, hipStream_t);
+                        int, int, float, int, hipStream_t);
 void launch_attn_delta(const void*, const void*, float*, long, int, int,
                        int, hipStream_t);
 void launch_attn_bwd_dkdv(const void*, const void*, const void*,
@@ -130,10 +130,14 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cosb, torch::Tensor sinb,
             D = (int)x.size(3);
   TORCH_CHECK(D % 8 == 0, "D % 8");
   const int* pd = nullptr;
+  int pos_per_b = 0;
   if (pos_dev.has_value()) {
     TORCH_CHECK(pos_dev->scalar_type() == torch::kInt &&
                 pos_dev->is_cuda(), "pos_dev must be int32 on GPU");
     pd = pos_dev->data_ptr<int>();
+    pos_per_b = pos_dev->numel() > 1;
+    if (pos_per_b)
+      TORCH_CHECK(pos_dev->numel() == B, "pos_dev must be [1] or [B]");
     TORCH_CHECK(cosb.size(0) >= S, "rope table too short");
   } else {
     TORCH_CHECK(cosb.size(0) >= pos0 + S, "rope table too short");
@@ -141,7 +145,7 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cosb, torch::Tensor sinb,
   auto y = torch::empty_like(x);
   launch_rope(x.data_ptr(), cosb.data_ptr<float>(), sinb.data_ptr<float>(),
               y.data_ptr(), B, S, H, D, (int)pos0, backward ? 1 : 0, pd,
-              cur_stream());
+              pos_per_b, cur_stream());
   return y;
 }
 
@@ -468,15 +472,19 @@ std::vector<torch::Tensor> attn_decode(torch::Tensor q, torch::Tensor k,
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, Hq, 1}, q.options().dtype(torch::kFloat));
   const int* ld = nullptr;
+  int len_stride = 0;
   if (len_dev.has_value()) {
     TORCH_CHECK(len_dev->scalar_type() == torch::kInt &&
                 len_dev->is_cuda(), "len_dev must be int32 on GPU");
     ld = len_dev->data_ptr<int>();
+    len_stride = len_dev->numel() > 1;
+    if (len_stride)
+      TORCH_CHECK(len_dev->numel() == B, "len_dev must be [1] or [B]");
   }
   launch_attn_decode(q.data_ptr(), k.data_ptr(), v.data_ptr(), ld,
                      part.data_ptr<float>(), o.data_ptr(),
                      lse.data_ptr<float>(), B, Hq, Hkv, Skv, D,
-                     (float)scale, cur_stream());
+                     (float)scale, len_stride, cur_stream());
   return {o, lse};
 }
 

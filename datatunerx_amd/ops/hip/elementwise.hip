@@ -184,8 +184,10 @@ void rope_kernel(const unsigned short* __restrict__ x,
                  unsigned short* __restrict__ y,
                  const int* __restrict__ pos_dev,  // graph-mode position
                  long total_quads,  // B*S*H*(D/2/4)
-                 int S, int H, int D, int pos0, int backward) {
-  if (pos_dev) pos0 += *pos_dev;
+                 int S, int H, int D, int pos0, int backward,
+                 int pos_per_b) {
+  // pos_per_b: pos_dev is a PER-ROW [B] vector (batched ragged decode)
+  if (pos_dev && !pos_per_b) pos0 += *pos_dev;
   const int qpr = D / 8;                      // 4-pair groups per head-row
   long idx = (long)blockIdx.x * DTX_BLOCK + threadIdx.x;
   long stride = (long)gridDim.x * DTX_BLOCK;
@@ -193,6 +195,8 @@ void rope_kernel(const unsigned short* __restrict__ x,
     long row = idx / qpr;                     // (b*S + s)*H + h
     int q = (int)(idx % qpr);
     int s_pos = (int)((row / H) % S);
+    if (pos_dev && pos_per_b)
+      s_pos += pos_dev[row / ((long)S * H)];
     const long base = row * D;
     short4v x1 = *reinterpret_cast<const short4v*>(x + base + q * 4);
     short4v x2 = *reinterpret_cast<const short4v*>(x + base + D / 2 + q * 4);
@@ -318,11 +322,12 @@ void launch_reduce_partials(const float* part, float* out, int P, long L,
 
 void launch_rope(const void* x, const float* cosb, const float* sinb,
                  void* y, long B, int S, int H, int D, int pos0,
-                 int backward, const int* pos_dev, hipStream_t s) {
+                 int backward, const int* pos_dev, int pos_per_b,
+                 hipStream_t s) {
   long quads = B * S * H * (D / 8);
   rope_kernel<<<ew_grid(quads), DTX_BLOCK, 0, s>>>(
       (const unsigned short*)x, cosb, sinb, (unsigned short*)y, pos_dev, quads,
-      S, H, D, pos0, backward);
+      S, H, D, pos0, backward, pos_per_b);
 }
 
 void launch_swiglu_fwd(const void* g, const void* u, void* o, long n,

@@ -58,6 +58,45 @@ class KVCache:
         return kk, vv, None
 
 
+class BatchedKVCache:
+    """Ragged batched decode cache: rows prefill individually (row_view
+    writes through a dim-0 slice), then decode appends ONE token per row
+    at per-row positions (pos64 [B] int64) and attends per-row lengths
+    (len32 [B] int32 — the decode kernel / CPU ref read them per b).
+    Batched decode shares one weight stream across the batch — the
+    single-stream decode is WEIGHT-bandwidth-bound, so this is the
+    scaling axis for serving throughput."""
+
+    def __init__(self, B, Hkv, max_s, D, device, dtype):
+        self.k = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
+        self.v = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
+        self.pos64 = torch.zeros(B, dtype=torch.long, device=device)
+        self.len32 = torch.zeros(B, dtype=torch.int32, device=device)
+        self._rows = torch.arange(B, device=device)
+
+    class _RowView:
+        def __init__(self, parent, i):
+            self.parent, self.i = parent, i
+            self.cur = 0
+
+        def update(self, k, v):
+            S = k.shape[1]
+            self.parent.k[self.i, self.cur:self.cur + S] = k[0]
+            self.parent.v[self.i, self.cur:self.cur + S] = v[0]
+            self.cur += S
+            return (self.parent.k[self.i:self.i + 1, :self.cur],
+                    self.parent.v[self.i:self.i + 1, :self.cur], None)
+
+    def row_view(self, i):
+        return BatchedKVCache._RowView(self, i)
+
+    def update(self, k, v):
+        # batched decode append: k/v [B, 1, Hkv, D] at per-row positions
+        self.k[self._rows, self.pos64] = k[:, 0]
+        self.v[self._rows, self.pos64] = v[:, 0]
+        return self.k, self.v, self.len32
+
+
 def build_model(name: str, device, adapter_dir: Optional[str] = None,
                 lora_kw: Optional[dict] = None):
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
@@ -392,6 +431,105 @@ class InferenceEngine:
             dist.broadcast(torch.tensor([nxt], dtype=torch.long,
                                         device=comm_dev), src=0)
         return nxt
+
+    # ------------------------------------------------- batched decode
+    @torch.no_grad()
+    def generate_batch(self, prompts: List[List[int]],
+                       max_new_tokens: int = 64,
+                       temperature: float = 0.0,
+                       top_p: float = 1.0) -> List[List[int]]:
+        """Ragged batched generation: per-row prefill, then ONE decode
+        step per token for the whole batch (the weight stream is read
+        once per step instead of once per request)."""
+        with self._stream_ctx():
+            return self._generate_batch(prompts, max_new_tokens,
+                                        temperature, top_p)
+
+    @torch.no_grad()
+    def _generate_batch(self, prompts, max_new_tokens, temperature,
+                        top_p):
+        assert self.is_llama, "batched decode is the Llama path"
+        cfg = self.model.cfg
+        B = len(prompts)
+        lens = [len(p) for p in prompts]
+        max_s = min(cfg.max_position_embeddings,
+                    max(lens) + max_new_tokens + 8)
+        dtype = next(self.model.parameters()).dtype
+        caches = [BatchedKVCache(B, cfg.num_key_value_heads, max_s,
+                                 cfg.head_dim, self.device, dtype)
+                  for _ in range(cfg.num_hidden_layers)]
+        nxt = [0] * B
+        for i, ids in enumerate(prompts):
+            row = [c.row_view(i) for c in caches]
+            t = torch.tensor([ids], dtype=torch.long, device=self.device)
+            logits = self.model(t, pos0=0, kv_caches=row)
+            nxt[i] = self._sample(logits[0, -1], temperature, top_p)
+        pos = torch.tensor(lens, dtype=torch.long, device=self.device)
+        for c in caches:
+            c.pos64.copy_(pos)
+            c.len32.copy_((pos + 1).to(torch.int32))
+        pos32 = pos.to(torch.int32)
+        outs = [[] for _ in range(B)]
+        done = [False] * B
+        eos = self.tok.eos_token_id
+        for i in range(B):
+            if nxt[i] == eos:
+                done[i] = True
+            else:
+                outs[i].append(nxt[i])
+        cur = torch.tensor(nxt, dtype=torch.long,
+                           device=self.device).view(B, 1)
+        budget = max_s - max(lens) - 2
+        for _ in range(min(max_new_tokens - 1, budget)):
+            if all(done):
+                break
+            logits = self.model(cur, kv_caches=caches, pos_dev=pos32)
+            for c in caches:
+                c.pos64 += 1
+                c.len32 += 1
+            pos32 += 1
+            if temperature and temperature > 0:
+                toks = [self._sample(logits[i, -1], temperature, top_p)
+                        for i in range(B)]
+            else:
+                toks = logits[:, -1].argmax(-1).tolist()
+            for i, t in enumerate(toks):
+                if done[i]:
+                    continue
+                if t == eos:
+                    done[i] = True
+                else:
+                    outs[i].append(t)
+            cur = torch.tensor(toks, dtype=torch.long,
+                               device=self.device).view(B, 1)
+        return outs
+
+    def chat_batch(self, requests: List[dict]) -> List[str]:
+        """requests: [{messages, max_tokens, temperature, top_p}] ->
+        completion texts (one batched generation)."""
+        t = get_template(self.template)
+        prompts = []
+        for r in requests:
+            system, history, pending = "", [], None
+            for m in r.get("messages", []):
+                if m["role"] == "system":
+                    system = m["content"]
+                elif m["role"] == "user":
+                    pending = m["content"]
+                elif m["role"] == "assistant" and pending is not None:
+                    history.append((pending, m["content"]))
+                    pending = None
+            src, _ = t.encode_oneturn(self.tok, pending or "", "",
+                                      history, system)
+            prompts.append(src)
+        max_new = max(int(r.get("max_tokens", 64)) for r in requests)
+        temp = float(requests[0].get("temperature", 0.0))
+        top_p = float(requests[0].get("top_p", 1.0))
+        outs = self.generate_batch(prompts, max_new, temp, top_p)
+        # honor each request's own max_tokens
+        return [self.tok.decode(
+                    o[:int(r.get("max_tokens", 64))])
+                for o, r in zip(outs, requests)]
 
     # ----------------------------------------------------------- score
     @torch.no_grad()

@@ -45,7 +45,60 @@ class EnginePool:
         self._q.put(e)
 
 
-def build_handler(pool):
+class BatchingFront:
+    """Coalesces concurrent non-streaming /chat/completions into ONE
+    batched generation (engine.chat_batch): single-stream decode is
+    weight-bandwidth-bound, so a batch of B shares one weight read per
+    token instead of B. A worker thread collects requests for up to
+    `linger` seconds (or `max_batch`), runs the batch, and resolves
+    each waiter."""
+
+    def __init__(self, engine, max_batch: int = 8,
+                 linger: float = 0.003):
+        import queue
+        self.engine = engine
+        self.max_batch = max_batch
+        self.linger = linger
+        self._q = queue.Queue()
+        t = threading.Thread(target=self._worker, daemon=True)
+        t.start()
+
+    def chat(self, messages, max_tokens, temperature, top_p):
+        ev = threading.Event()
+        slot = {"req": {"messages": messages, "max_tokens": max_tokens,
+                        "temperature": temperature, "top_p": top_p},
+                "ev": ev}
+        self._q.put(slot)
+        ev.wait()
+        if "err" in slot:
+            raise slot["err"]
+        return slot["out"]
+
+    def _worker(self):
+        import queue
+        while True:
+            batch = [self._q.get()]
+            deadline = time.time() + self.linger
+            while len(batch) < self.max_batch:
+                left = deadline - time.time()
+                if left <= 0:
+                    break
+                try:
+                    batch.append(self._q.get(timeout=left))
+                except queue.Empty:
+                    break
+            try:
+                outs = self.engine.chat_batch([b["req"] for b in batch])
+                for b, o in zip(batch, outs):
+                    b["out"] = o
+            except Exception as e:      # pragma: no cover
+                for b in batch:
+                    b["err"] = e
+            for b in batch:
+                b["ev"].set()
+
+
+def build_handler(pool, batcher=None):
     if not isinstance(pool, EnginePool):
         pool = EnginePool([pool])
 
@@ -131,8 +184,12 @@ def build_handler(pool):
                                     OSError):
                                 pass
                         return
-                    with lock as engine:
-                        text = engine.chat(*args)
+                    if batcher is not None and len(body.get(
+                            "messages", [])) >= 0:
+                        text = batcher.chat(*args)
+                    else:
+                        with lock as engine:
+                            text = engine.chat(*args)
                     self._send(200, {
                         "id": f"chatcmpl-{int(time.time()*1000)}",
                         "object": "chat.completion",
@@ -156,8 +213,9 @@ def build_handler(pool):
     return Handler
 
 
-def serve_forever(engine, host: str, port: int):
-    httpd = ThreadingHTTPServer((host, port), build_handler(engine))
+def serve_forever(engine, host: str, port: int, batcher=None):
+    httpd = ThreadingHTTPServer((host, port),
+                                build_handler(engine, batcher))
     httpd.serve_forever()
 
 
@@ -268,6 +326,7 @@ def main(argv=None):
         else:
             tp_follower_loop(engine)
         return
+    from ..models import LlamaForCausalLM
     from ..models.hf_io import load_tokenizer
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     model = build_model(args.model, device, adapter_dir=args.adapter)
@@ -277,7 +336,17 @@ def main(argv=None):
         InferenceEngine(model, tokenizer=tok, template=args.template,
                         device=device, own_stream=(n_slots > 1))
         for _ in range(n_slots)])
-    serve_forever(pool, args.host, args.port)
+    batcher = None
+    if os.environ.get("DTX_SERVE_BATCH", "1") != "0" and \
+            isinstance(model, LlamaForCausalLM):
+        batcher = BatchingFront(
+            InferenceEngine(model, tokenizer=tok, template=args.template,
+                            device=device, graph_decode=False,
+                            own_stream=True),
+            max_batch=int(os.environ.get("DTX_SERVE_MAX_BATCH", "8")))
+    httpd = ThreadingHTTPServer((args.host, args.port),
+                                build_handler(pool, batcher))
+    httpd.serve_forever()
 
 
 if __name__ == "__main__":
