@@ -491,8 +491,17 @@ def test_quantized_train_step_gpu(bits):
 def test_batched_decode_matches_sequential_gpu():
     """GPU batched ragged decode (per-row rope positions + per-row
     cache lengths through the HIP kernels) == per-request generation."""
-    from datatunerx_amd.serve.engine import InferenceEngine, build_model
-    model = build_model("llama-tiny", DEV)
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=512, hidden_size=256,
+                      intermediate_size=512, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2,
+                      max_position_embeddings=256)
+    with torch.device(DEV):
+        model = LlamaForCausalLM(cfg, lora=False, dtype=torch.bfloat16)
+    model.init_random()
+    model.eval()
     eng = InferenceEngine(model, template="vanilla", device=DEV,
                           graph_decode=False)
     prompts = [eng.tok.encode("hello world", add_special_tokens=True),
