@@ -58,6 +58,60 @@ class KVCache:
         return kk, vv, None
 
 
+class _BatchedGraphedDecoder:
+    """hipGraph-captured BATCHED decode step (fixed batch MAXB, padded
+    rows): one replay per token for the whole batch, greedy argmax
+    in-graph, per-row position/length counters device-resident. Built
+    once per engine (under the capture lock) and reused; requests pad
+    to MAXB with dummy rows."""
+
+    def __init__(self, model, cfg, device, B, max_s):
+        self.B, self.max_s = B, max_s
+        dtype = next(model.parameters()).dtype
+        self.pos64 = torch.zeros(B, dtype=torch.long, device=device)
+        self.pos32 = torch.zeros(B, dtype=torch.int32, device=device)
+        self.len32 = torch.zeros(B, dtype=torch.int32, device=device)
+        self.caches = [
+            BatchedKVCache(B, cfg.num_key_value_heads, max_s,
+                           cfg.head_dim, device, dtype,
+                           pos64=self.pos64, len32=self.len32)
+            for _ in range(cfg.num_hidden_layers)]
+        self.input = torch.ones(B, 1, dtype=torch.long, device=device)
+        self.model = model
+        with torch.no_grad():
+            for _ in range(3):
+                lg = model(self.input, pos_dev=self.pos32,
+                           kv_caches=self.caches)
+                lg[:, -1].argmax(-1)
+            torch.cuda.synchronize()
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                lg = model(self.input, pos_dev=self.pos32,
+                           kv_caches=self.caches)
+                self.out = lg[:, -1].argmax(-1)
+
+    def reset(self):
+        self.pos64.zero_()
+        self.pos32.zero_()
+        self.len32.zero_()
+
+    def set_state(self, lens):
+        pos = torch.tensor(lens, dtype=torch.long,
+                           device=self.pos64.device)
+        self.pos64.copy_(pos)
+        self.pos32.copy_(pos.to(torch.int32))
+        self.len32.copy_((pos + 1).to(torch.int32))
+
+    def step(self, cur):
+        """cur: [B,1] int64 device. Returns next tokens [B] (device)."""
+        self.input.copy_(cur)
+        self.graph.replay()
+        self.pos64 += 1
+        self.pos32 += 1
+        self.len32 += 1
+        return self.out
+
+
 class BatchedKVCache:
     """Ragged batched decode cache: rows prefill individually (row_view
     writes through a dim-0 slice), then decode appends ONE token per row
@@ -67,11 +121,16 @@ class BatchedKVCache:
     single-stream decode is WEIGHT-bandwidth-bound, so this is the
     scaling axis for serving throughput."""
 
-    def __init__(self, B, Hkv, max_s, D, device, dtype):
+    def __init__(self, B, Hkv, max_s, D, device, dtype,
+                 pos64=None, len32=None):
         self.k = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
         self.v = torch.zeros(B, max_s, Hkv, D, device=device, dtype=dtype)
-        self.pos64 = torch.zeros(B, dtype=torch.long, device=device)
-        self.len32 = torch.zeros(B, dtype=torch.int32, device=device)
+        # pos/len may be SHARED across layers (the graphed decoder
+        # advances one set of counters for all layer caches)
+        self.pos64 = (pos64 if pos64 is not None else
+                      torch.zeros(B, dtype=torch.long, device=device))
+        self.len32 = (len32 if len32 is not None else
+                      torch.zeros(B, dtype=torch.int32, device=device))
         self._rows = torch.arange(B, device=device)
 
     class _RowView:
@@ -201,6 +260,8 @@ class _GraphedDecoder:
 
 
 class InferenceEngine:
+    MAX_BATCH = int(os.environ.get("DTX_SERVE_MAX_BATCH", "8"))
+
     """One generation context over a (shared, read-only) model. For
     concurrent serving, build several engines over the SAME model
     (serve/server.py EnginePool): each holds its own KV caches, hip
@@ -218,6 +279,7 @@ class InferenceEngine:
         self.is_llama = isinstance(model, LlamaForCausalLM)
         import torch.distributed as dist
         self._graphed = None
+        self._bgraphed = None
         self._stream = (torch.cuda.Stream(device=self.device)
                         if own_stream and self.device.type == "cuda"
                         else None)
@@ -445,19 +507,50 @@ class InferenceEngine:
             return self._generate_batch(prompts, max_new_tokens,
                                         temperature, top_p)
 
+    def _get_batched_graphed(self, max_s):
+        if self._bgraphed is None and self._graph_ok:
+            try:
+                with InferenceEngine._capture_lock, self._stream_ctx():
+                    self._bgraphed = _BatchedGraphedDecoder(
+                        self.model, self.model.cfg, self.device,
+                        InferenceEngine.MAX_BATCH, max_s)
+            except Exception:
+                self._bgraphed = False
+        return self._bgraphed or None
+
     @torch.no_grad()
     def _generate_batch(self, prompts, max_new_tokens, temperature,
                         top_p):
         assert self.is_llama, "batched decode is the Llama path"
         cfg = self.model.cfg
+        greedy = not (temperature and temperature > 0)
+        lens0 = [len(p) for p in prompts]
+        gd = None
+        if greedy and self.device.type == "cuda" and \
+                len(prompts) <= InferenceEngine.MAX_BATCH:
+            cap = min(cfg.max_position_embeddings,
+                      int(os.environ.get("DTX_SERVE_MAXS", "2048")))
+            if max(lens0) + max_new_tokens + 8 < cap:
+                gd = self._get_batched_graphed(cap)
+        if gd is not None:
+            # pad the batch to the graph's fixed MAXB with dummy rows
+            prompts = list(prompts) + \
+                [[self.tok.bos_token_id]] * (gd.B - len(prompts))
         B = len(prompts)
         lens = [len(p) for p in prompts]
-        max_s = min(cfg.max_position_embeddings,
-                    max(lens) + max_new_tokens + 8)
+        max_s = gd.max_s if gd is not None else min(
+            cfg.max_position_embeddings,
+            max(lens) + max_new_tokens + 8)
         dtype = next(self.model.parameters()).dtype
-        caches = [BatchedKVCache(B, cfg.num_key_value_heads, max_s,
-                                 cfg.head_dim, self.device, dtype)
-                  for _ in range(cfg.num_hidden_layers)]
+        if gd is not None:
+            # no cache zeroing needed: prefill overwrites rows up to
+            # len and attention never reads past len32
+            gd.reset()
+            caches = gd.caches
+        else:
+            caches = [BatchedKVCache(B, cfg.num_key_value_heads, max_s,
+                                     cfg.head_dim, self.device, dtype)
+                      for _ in range(cfg.num_hidden_layers)]
         nxt = [0] * B
         for i, ids in enumerate(prompts):
             row = [c.row_view(i) for c in caches]
@@ -465,10 +558,14 @@ class InferenceEngine:
             logits = self.model(t, pos0=0, kv_caches=row)
             nxt[i] = self._sample(logits[0, -1], temperature, top_p)
         pos = torch.tensor(lens, dtype=torch.long, device=self.device)
-        for c in caches:
-            c.pos64.copy_(pos)
-            c.len32.copy_((pos + 1).to(torch.int32))
-        pos32 = pos.to(torch.int32)
+        if gd is not None:
+            gd.set_state(lens)
+            pos32 = gd.pos32
+        else:
+            for c in caches:
+                c.pos64.copy_(pos)
+                c.len32.copy_((pos + 1).to(torch.int32))
+            pos32 = pos.to(torch.int32)
         outs = [[] for _ in range(B)]
         done = [False] * B
         eos = self.tok.eos_token_id
@@ -477,22 +574,31 @@ class InferenceEngine:
                 done[i] = True
             else:
                 outs[i].append(nxt[i])
+        n_live = len(lens0)
         cur = torch.tensor(nxt, dtype=torch.long,
                            device=self.device).view(B, 1)
         budget = max_s - max(lens) - 2
         for _ in range(min(max_new_tokens - 1, budget)):
-            if all(done):
+            if all(done[:n_live]):
                 break
-            logits = self.model(cur, kv_caches=caches, pos_dev=pos32)
-            for c in caches:
-                c.pos64 += 1
-                c.len32 += 1
-            pos32 += 1
-            if temperature and temperature > 0:
-                toks = [self._sample(logits[i, -1], temperature, top_p)
-                        for i in range(B)]
+            if gd is not None:
+                out = gd.step(cur)
+                toks = out.tolist()
+                cur = out.view(B, 1)
             else:
-                toks = logits[:, -1].argmax(-1).tolist()
+                logits = self.model(cur, kv_caches=caches,
+                                    pos_dev=pos32)
+                for c in caches:
+                    c.pos64 += 1
+                    c.len32 += 1
+                pos32 += 1
+                if temperature and temperature > 0:
+                    toks = [self._sample(logits[i, -1], temperature,
+                                         top_p) for i in range(B)]
+                else:
+                    toks = logits[:, -1].argmax(-1).tolist()
+                cur = torch.tensor(toks, dtype=torch.long,
+                                   device=self.device).view(B, 1)
             for i, t in enumerate(toks):
                 if done[i]:
                     continue
@@ -500,9 +606,7 @@ class InferenceEngine:
                     done[i] = True
                 else:
                     outs[i].append(t)
-            cur = torch.tensor(toks, dtype=torch.long,
-                               device=self.device).view(B, 1)
-        return outs
+        return outs[:len(lens0)]
 
     def chat_batch(self, requests: List[dict]) -> List[str]:
         """requests: [{messages, max_tokens, temperature, top_p}] ->

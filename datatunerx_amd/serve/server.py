@@ -341,8 +341,7 @@ def main(argv=None):
             isinstance(model, LlamaForCausalLM):
         batcher = BatchingFront(
             InferenceEngine(model, tokenizer=tok, template=args.template,
-                            device=device, graph_decode=False,
-                            own_stream=True),
+                            device=device, own_stream=True),
             max_batch=int(os.environ.get("DTX_SERVE_MAX_BATCH", "8")))
     httpd = ThreadingHTTPServer((args.host, args.port),
                                 build_handler(pool, batcher))

@@ -43,10 +43,11 @@ from datatunerx_amd.serve.server import BatchingFront
 for mb in (4, 8):
     pool = EnginePool([InferenceEngine(model, template="llama2",
                                        device=torch.device("cuda:0"))])
+    import datatunerx_amd.serve.engine as _e
+    _e.InferenceEngine.MAX_BATCH = mb
     batcher = BatchingFront(
         InferenceEngine(model, template="llama2",
-                        device=torch.device("cuda:0"),
-                        graph_decode=False, own_stream=False),
+                        device=torch.device("cuda:0"), own_stream=False),
         max_batch=mb, linger=0.02)
     httpd = ThreadingHTTPServer(("127.0.0.1", 0),
                                 build_handler(pool, batcher))
