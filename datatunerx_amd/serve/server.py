@@ -249,6 +249,12 @@ class TPFrontEngine:
         self._bcast({"op": "ppl", "texts": texts})
         return self.engine.perplexity(texts)
 
+    def chat_batch(self, requests):
+        # followers run the identical batched generation so the TP
+        # collectives stay in lockstep token for token
+        self._bcast({"op": "chat_batch", "requests": requests})
+        return self.engine.chat_batch(requests)
+
 
 def tp_follower_loop(engine):
     import torch.distributed as dist
@@ -261,6 +267,8 @@ def tp_follower_loop(engine):
                         req["temperature"], req["top_p"])
         elif req["op"] == "ppl":
             engine.perplexity(req["texts"])
+        elif req["op"] == "chat_batch":
+            engine.chat_batch(req["requests"])
         elif req["op"] == "stop":
             return
 
@@ -322,7 +330,16 @@ def main(argv=None):
                                  template=args.template,
                                  device=device)
         if rank == 0:
-            serve_forever(TPFrontEngine(engine), args.host, args.port)
+            front = TPFrontEngine(engine)
+            batcher = None
+            if os.environ.get("DTX_SERVE_BATCH", "1") != "0":
+                # ONE batcher and NO pool concurrency in TP mode: every
+                # request (batched or not) broadcasts to followers, so
+                # requests must serialize per group — but they still
+                # coalesce into batched generations.
+                batcher = BatchingFront(front, max_batch=int(
+                    os.environ.get("DTX_SERVE_MAX_BATCH", "8")))
+            serve_forever(front, args.host, args.port, batcher)
         else:
             tp_follower_loop(engine)
         return
