@@ -221,39 +221,50 @@ def serve_forever(engine, host: str, port: int, batcher=None):
 
 class TPFrontEngine:
     """Rank-0 wrapper: broadcasts each request to the follower ranks so
-    the TP group steps through the model collectives in lockstep."""
+    the TP group steps through the model collectives in lockstep. EVERY
+    entry point serializes on one internal lock — the batching front and
+    the handler pool both call in, and interleaved broadcasts would
+    desync the followers."""
 
     def __init__(self, engine):
         self.engine = engine
+        self._mx = threading.RLock()
 
     def _bcast(self, req: dict):
         import torch.distributed as dist
         dist.broadcast_object_list([req], src=0)
 
     def chat(self, messages, max_tokens, temperature, top_p):
-        self._bcast({"op": "chat", "messages": messages,
-                     "max_tokens": max_tokens, "temperature": temperature,
-                     "top_p": top_p})
-        return self.engine.chat(messages, max_tokens, temperature, top_p)
+        with self._mx:
+            self._bcast({"op": "chat", "messages": messages,
+                         "max_tokens": max_tokens,
+                         "temperature": temperature, "top_p": top_p})
+            return self.engine.chat(messages, max_tokens, temperature,
+                                    top_p)
 
     def chat_stream(self, messages, max_tokens, temperature, top_p):
         # followers run the non-streaming chat(): token-identical loop,
-        # so the TP collectives stay in lockstep with the streaming front
-        self._bcast({"op": "chat", "messages": messages,
-                     "max_tokens": max_tokens, "temperature": temperature,
-                     "top_p": top_p})
-        return self.engine.chat_stream(messages, max_tokens, temperature,
-                                       top_p)
+        # so the TP collectives stay in lockstep with the streaming
+        # front. The lock spans the DRAIN (this generator's body); the
+        # server always consumes the generator to the end.
+        with self._mx:
+            self._bcast({"op": "chat", "messages": messages,
+                         "max_tokens": max_tokens,
+                         "temperature": temperature, "top_p": top_p})
+            yield from self.engine.chat_stream(messages, max_tokens,
+                                               temperature, top_p)
 
     def perplexity(self, texts):
-        self._bcast({"op": "ppl", "texts": texts})
-        return self.engine.perplexity(texts)
+        with self._mx:
+            self._bcast({"op": "ppl", "texts": texts})
+            return self.engine.perplexity(texts)
 
     def chat_batch(self, requests):
         # followers run the identical batched generation so the TP
         # collectives stay in lockstep token for token
-        self._bcast({"op": "chat_batch", "requests": requests})
-        return self.engine.chat_batch(requests)
+        with self._mx:
+            self._bcast({"op": "chat_batch", "requests": requests})
+            return self.engine.chat_batch(requests)
 
 
 def tp_follower_loop(engine):
