@@ -520,7 +520,20 @@ class InferenceEngine:
 
     @torch.no_grad()
     def _generate_batch(self, prompts, max_new_tokens, temperature,
-                        top_p):
+                        top_p, budgets=None):
+        outs = [[] for _ in prompts]
+        for emitted in self._generate_batch_steps(
+                prompts, max_new_tokens, temperature, top_p, budgets):
+            for i, t in emitted:
+                outs[i].append(t)
+        return outs
+
+    @torch.no_grad()
+    def _generate_batch_steps(self, prompts, max_new_tokens, temperature,
+                              top_p, budgets=None):
+        """Yields, per decode step, the [(row, token)] emissions.
+        budgets: optional per-row max_tokens (rows stop emitting — and
+        count as done — at their own budget)."""
         assert self.is_llama, "batched decode is the Llama path"
         cfg = self.model.cfg
         greedy = not (temperature and temperature > 0)
@@ -566,19 +579,28 @@ class InferenceEngine:
                 c.pos64.copy_(pos)
                 c.len32.copy_((pos + 1).to(torch.int32))
             pos32 = pos.to(torch.int32)
-        outs = [[] for _ in range(B)]
-        done = [False] * B
         eos = self.tok.eos_token_id
+        n_live = len(lens0)
+        if budgets is None:
+            budgets = [max_new_tokens] * n_live
+        count = [0] * n_live
+        done = [False] * B
+        first = []
         for i in range(B):
             if nxt[i] == eos:
                 done[i] = True
-            else:
-                outs[i].append(nxt[i])
-        n_live = len(lens0)
+            elif i < n_live:
+                if count[i] < budgets[i]:
+                    first.append((i, nxt[i]))
+                    count[i] += 1
+                if count[i] >= budgets[i]:
+                    done[i] = True
+        yield first
         cur = torch.tensor(nxt, dtype=torch.long,
                            device=self.device).view(B, 1)
-        budget = max_s - max(lens) - 2
-        for _ in range(min(max_new_tokens - 1, budget)):
+        p32 = pos32
+        cap = max_s - max(lens) - 2
+        for _ in range(min(max_new_tokens - 1, cap)):
             if all(done[:n_live]):
                 break
             if gd is not None:
@@ -586,12 +608,11 @@ class InferenceEngine:
                 toks = out.tolist()
                 cur = out.view(B, 1)
             else:
-                logits = self.model(cur, kv_caches=caches,
-                                    pos_dev=pos32)
+                logits = self.model(cur, kv_caches=caches, pos_dev=p32)
                 for c in caches:
                     c.pos64 += 1
                     c.len32 += 1
-                pos32 += 1
+                p32 += 1
                 if temperature and temperature > 0:
                     toks = [self._sample(logits[i, -1], temperature,
                                          top_p) for i in range(B)]
@@ -599,18 +620,20 @@ class InferenceEngine:
                     toks = logits[:, -1].argmax(-1).tolist()
                 cur = torch.tensor(toks, dtype=torch.long,
                                    device=self.device).view(B, 1)
+            emitted = []
             for i, t in enumerate(toks):
                 if done[i]:
                     continue
                 if t == eos:
                     done[i] = True
-                else:
-                    outs[i].append(t)
-        return outs[:len(lens0)]
+                elif i < n_live:
+                    emitted.append((i, t))
+                    count[i] += 1
+                    if count[i] >= budgets[i]:
+                        done[i] = True
+            yield emitted
 
-    def chat_batch(self, requests: List[dict]) -> List[str]:
-        """requests: [{messages, max_tokens, temperature, top_p}] ->
-        completion texts (one batched generation)."""
+    def _encode_requests(self, requests):
         t = get_template(self.template)
         prompts = []
         for r in requests:
@@ -626,14 +649,52 @@ class InferenceEngine:
             src, _ = t.encode_oneturn(self.tok, pending or "", "",
                                       history, system)
             prompts.append(src)
-        max_new = max(int(r.get("max_tokens", 64)) for r in requests)
+        budgets = [int(r.get("max_tokens", 64)) for r in requests]
         temp = float(requests[0].get("temperature", 0.0))
         top_p = float(requests[0].get("top_p", 1.0))
-        outs = self.generate_batch(prompts, max_new, temp, top_p)
-        # honor each request's own max_tokens
-        return [self.tok.decode(
-                    o[:int(r.get("max_tokens", 64))])
-                for o, r in zip(outs, requests)]
+        return prompts, budgets, temp, top_p
+
+    def chat_batch(self, requests: List[dict]) -> List[str]:
+        """requests: [{messages, max_tokens, temperature, top_p}] ->
+        completion texts (one batched generation)."""
+        prompts, budgets, temp, top_p = self._encode_requests(requests)
+        with self._stream_ctx():
+            outs = self._generate_batch(prompts, max(budgets), temp,
+                                        top_p, budgets=budgets)
+        return [self.tok.decode(o) for o in outs]
+
+    def chat_batch_stream(self, requests: List[dict]):
+        """Batched STREAMING generation: yields (row_index, text_delta)
+        as tokens decode, then (row_index, None) when a row finishes.
+        Deltas per row concatenate to exactly chat_batch()'s text (the
+        same U+FFFD holdback as chat_stream)."""
+        prompts, budgets, temp, top_p = self._encode_requests(requests)
+        n = len(requests)
+        ids = [[] for _ in range(n)]
+        sent = [""] * n
+        live = [True] * n
+        with self._stream_ctx():
+            for emitted in self._generate_batch_steps(
+                    prompts, max(budgets), temp, top_p, budgets=budgets):
+                step_rows = set()
+                for i, t in emitted:
+                    ids[i].append(t)
+                    step_rows.add(i)
+                for i in step_rows:
+                    full = self.tok.decode(ids[i])
+                    stable = full
+                    while stable.endswith("\ufffd"):
+                        stable = stable[:-1]
+                    if len(stable) > len(sent[i]):
+                        delta = stable[len(sent[i]):]
+                        sent[i] = stable
+                        yield i, delta
+        for i in range(n):
+            if live[i]:
+                full = self.tok.decode(ids[i])
+                if len(full) > len(sent[i]):
+                    yield i, full[len(sent[i]):]
+                yield i, None
 
     # ----------------------------------------------------------- score
     @torch.no_grad()

@@ -74,6 +74,23 @@ class BatchingFront:
             raise slot["err"]
         return slot["out"]
 
+    def chat_stream(self, messages, max_tokens, temperature, top_p):
+        """Streamed request joining the batch: yields text deltas from
+        a per-request queue fed by the worker (None = end)."""
+        import queue
+        tokq = queue.Queue()
+        slot = {"req": {"messages": messages, "max_tokens": max_tokens,
+                        "temperature": temperature, "top_p": top_p},
+                "tokq": tokq}
+        self._q.put(slot)
+        while True:
+            d = tokq.get()
+            if d is None:
+                break
+            if isinstance(d, Exception):
+                raise d
+            yield d
+
     def _worker(self):
         import queue
         while True:
@@ -87,15 +104,43 @@ class BatchingFront:
                     batch.append(self._q.get(timeout=left))
                 except queue.Empty:
                     break
+            reqs = [b["req"] for b in batch]
+            streaming = any("tokq" in b for b in batch)
             try:
-                outs = self.engine.chat_batch([b["req"] for b in batch])
-                for b, o in zip(batch, outs):
-                    b["out"] = o
+                if streaming and hasattr(self.engine,
+                                         "chat_batch_stream"):
+                    texts = [""] * len(batch)
+                    for i, d in self.engine.chat_batch_stream(reqs):
+                        b = batch[i]
+                        if d is None:
+                            if "tokq" in b:
+                                b["tokq"].put(None)
+                            else:
+                                b["out"] = texts[i]
+                                b["ev"].set()
+                        elif "tokq" in b:
+                            b["tokq"].put(d)
+                        else:
+                            texts[i] += d
+                else:
+                    outs = self.engine.chat_batch(reqs)
+                    for b, o in zip(batch, outs):
+                        if "tokq" in b:
+                            # engine without chat_batch_stream (TP
+                            # front): deliver the whole text as one
+                            # delta so stream slots still resolve
+                            b["tokq"].put(o)
+                            b["tokq"].put(None)
+                        else:
+                            b["out"] = o
+                            b["ev"].set()
             except Exception as e:      # pragma: no cover
                 for b in batch:
-                    b["err"] = e
-            for b in batch:
-                b["ev"].set()
+                    if "tokq" in b:
+                        b["tokq"].put(e)
+                    else:
+                        b["err"] = e
+                        b["ev"].set()
 
 
 def build_handler(pool, batcher=None):
@@ -157,8 +202,16 @@ def build_handler(pool, batcher=None):
                         # would leave follower ranks blocked in
                         # dist.broadcast — desyncing every later
                         # request's collectives.
-                        with lock as engine:
-                            gen = engine.chat_stream(*args)
+                        if batcher is not None:
+                            import contextlib
+                            lease = contextlib.nullcontext()
+                            gen = batcher.chat_stream(*args)
+                        else:
+                            lease = None
+                        with (lease if lease is not None else lock) \
+                                as engine:
+                            if lease is None:
+                                gen = engine.chat_stream(*args)
                             client_gone = False
                             for delta in gen:
                                 if client_gone:
@@ -184,8 +237,7 @@ def build_handler(pool, batcher=None):
                                     OSError):
                                 pass
                         return
-                    if batcher is not None and len(body.get(
-                            "messages", [])) >= 0:
+                    if batcher is not None:
                         text = batcher.chat(*args)
                     else:
                         with lock as engine:

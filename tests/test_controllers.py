@@ -558,3 +558,68 @@ def test_serve_batched_requests_match_sequential(tmp_path):
         assert results == seq
     finally:
         httpd.shutdown()
+
+
+def test_serve_batched_stream_matches_batch(tmp_path):
+    """Streamed requests join the batch: concurrent SSE streams through
+    the BatchingFront concatenate to the same greedy completions."""
+    import json as _json
+    import threading
+    import urllib.request
+
+    import torch
+
+    from datatunerx_amd.serve.engine import InferenceEngine, build_model
+    from datatunerx_amd.serve.server import (BatchingFront, EnginePool,
+                                             build_handler)
+    from http.server import ThreadingHTTPServer
+
+    model = build_model("llama-tiny", torch.device("cpu"))
+    pool = EnginePool([InferenceEngine(model, template="vanilla",
+                                       device=torch.device("cpu"))])
+    batcher = BatchingFront(
+        InferenceEngine(model, template="vanilla",
+                        device=torch.device("cpu")), max_batch=4,
+        linger=0.05)
+    httpd = ThreadingHTTPServer(("127.0.0.1", 0),
+                                build_handler(pool, batcher))
+    port = httpd.server_address[1]
+    threading.Thread(target=httpd.serve_forever, daemon=True).start()
+    try:
+        def ask(content, stream):
+            data = {"messages": [{"role": "user", "content": content}],
+                    "max_tokens": 8, "temperature": 0.0}
+            if stream:
+                data["stream"] = True
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{port}/chat/completions",
+                data=_json.dumps(data).encode(),
+                headers={"Content-Type": "application/json"})
+            with urllib.request.urlopen(req, timeout=120) as r:
+                if not stream:
+                    return _json.load(r)["choices"][0]["message"][
+                        "content"]
+                text = ""
+                for line in r:
+                    line = line.decode().strip()
+                    if line.startswith("data: ") and \
+                            line != "data: [DONE]":
+                        text += _json.loads(line[6:])["choices"][0][
+                            "delta"]["content"]
+                return text
+
+        want = [ask("alpha", False), ask("beta", False)]
+        got = [None, None]
+
+        def worker(i, c):
+            got[i] = ask(c, True)
+
+        ts = [threading.Thread(target=worker, args=(i, c))
+              for i, c in enumerate(["alpha", "beta"])]
+        for th in ts:
+            th.start()
+        for th in ts:
+            th.join(timeout=120)
+        assert got == want
+    finally:
+        httpd.shutdown()
