@@ -250,3 +250,58 @@ def shard_adapter_state(sd: dict, cfg: LlamaConfig, rank: int, ws: int):
         else:
             out[key] = t
     return out
+
+
+@torch.no_grad()
+def load_hf_weights_tp(model, model_dir: str, cfg: LlamaConfig,
+                       rank: int, ws: int) -> int:
+    """Load a REAL local HF-Llama checkpoint into rank `rank`'s TP
+    shard (the 13B inference-compare service on actual weights):
+    column-parallel q/k/v/gate/up slice output rows, row-parallel
+    o/down slice input columns, lm_head is vocab-parallel, embeddings
+    and norms replicate. Tied lm_head falls back to the embedding."""
+    from ..models.hf_io import _iter_hf_tensors, _map_name
+    params = dict(model.named_parameters())
+    n = 0
+    embed = None
+    got_head = False
+
+    def dst(name):
+        # row-parallel wrappers insert ".inner" in the module path
+        for cand in (name,
+                     name.replace("o_proj.", "o_proj.inner.")
+                     .replace("down_proj.", "down_proj.inner."),
+                     name.replace("lm_head.", "lm_head.inner.")):
+            if cand in params:
+                return params[cand]
+        return None
+
+    for hf_name, t in _iter_hf_tensors(model_dir):
+        name = _map_name(hf_name)
+        if name is None:
+            continue
+        p = dst(name)
+        if p is None:
+            continue
+        if any(k in name for k in ("q_proj", "k_proj", "v_proj",
+                                   "gate_proj", "up_proj")):
+            sh = t.shape[0] // ws
+            p.copy_(t[rank * sh:(rank + 1) * sh].to(p.dtype))
+        elif any(k in name for k in ("o_proj", "down_proj")):
+            sh = t.shape[1] // ws
+            p.copy_(t[:, rank * sh:(rank + 1) * sh].to(p.dtype))
+        elif name == "lm_head.weight":
+            sh = t.shape[0] // ws
+            p.copy_(t[rank * sh:(rank + 1) * sh].to(p.dtype))
+            got_head = True
+        else:                         # embed, norms: replicated
+            p.copy_(t.to(p.dtype))
+            if name == "embed_tokens.weight":
+                embed = t
+        n += 1
+    if not got_head and embed is not None:       # tied weights
+        p = dst("lm_head.weight")
+        sh = embed.shape[0] // ws
+        p.copy_(embed[rank * sh:(rank + 1) * sh].to(p.dtype))
+        n += 1
+    return n

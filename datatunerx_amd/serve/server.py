@@ -369,7 +369,10 @@ def main(argv=None):
         from ..parallel.ddp import init_distributed
         from ..parallel.tp import build_tp_llama, load_adapter_tp
         rank, world, local_rank, device = init_distributed()
-        cfg = _llama_config(args.model)
+        from ..models.hf_io import is_hf_model_dir, load_hf_config
+        hf_dir = is_hf_model_dir(args.model)
+        cfg = (load_hf_config(args.model) if hf_dir
+               else _llama_config(args.model))
         if args.adapter:
             # match the checkpoint's LoRA geometry (same contract as the
             # single-GPU build_model path): serve whatever r/alpha/
@@ -385,6 +388,9 @@ def main(argv=None):
         dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
         model = build_tp_llama(cfg, rank, world, lora=bool(args.adapter),
                                dtype=dtype, device=device)
+        if hf_dir:
+            from ..parallel.tp import load_hf_weights_tp
+            load_hf_weights_tp(model, args.model, cfg, rank, world)
         if args.adapter:
             load_adapter_tp(model, args.adapter, cfg, rank, world)
         model.eval()

@@ -272,3 +272,76 @@ def test_streaming_matches_nonstreaming(tmp_path):
         assert n_chunks >= 2, "should stream multiple chunks"
     finally:
         proc.kill()
+
+
+HF_TP_WORKER = r"""
+import os, sys
+import torch
+sys.path.insert(0, os.environ["DTX_ROOT"])
+import torch.distributed as dist
+from datatunerx_amd.models.hf_io import load_hf_config
+from datatunerx_amd.parallel.tp import build_tp_llama, load_hf_weights_tp
+
+rank = int(os.environ["RANK"]); world = int(os.environ["WORLD_SIZE"])
+dist.init_process_group("gloo", rank=rank, world_size=world)
+cfg = load_hf_config(os.environ["DTX_HF_DIR"])
+model = build_tp_llama(cfg, rank, world, lora=False,
+                       dtype=torch.float32,
+                       device=torch.device("cpu"))
+n = load_hf_weights_tp(model, os.environ["DTX_HF_DIR"], cfg, rank, world)
+assert n > 0, n
+model.eval()
+torch.manual_seed(0)
+ids = torch.randint(0, cfg.vocab_size, (1, 9))
+with torch.no_grad():
+    logits = model(ids)
+if rank == 0:
+    torch.save(logits, os.environ["DTX_OUT"] + "/tp_logits.pt")
+dist.destroy_process_group()
+"""
+
+
+def test_tp_real_weights_match_single_process(tmp_path):
+    """2-rank TP sharding of a REAL HF-format checkpoint produces the
+    same logits as the single-process hf_io load (the 13B
+    inference-compare-on-real-weights contract, gloo on CPU)."""
+    import subprocess
+
+    transformers = pytest.importorskip("transformers")
+    d = str(tmp_path / "hf")
+    hc = transformers.LlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, max_position_embeddings=64,
+        attn_implementation="eager")
+    torch.manual_seed(5)
+    transformers.LlamaForCausalLM(hc).save_pretrained(
+        d, safe_serialization=True)
+
+    script = str(tmp_path / "w.py")
+    with open(script, "w") as f:
+        f.write(HF_TP_WORKER)
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({"RANK": str(rank), "WORLD_SIZE": "2",
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29741",
+                    "DTX_ROOT": ROOT, "DTX_OUT": str(tmp_path),
+                    "DTX_HF_DIR": d})
+        procs.append(subprocess.Popen([sys.executable, script], env=env))
+    for p in procs:
+        assert p.wait(timeout=180) == 0
+    tp_logits = torch.load(str(tmp_path / "tp_logits.pt"))
+
+    from datatunerx_amd.models import LlamaForCausalLM
+    from datatunerx_amd.models.hf_io import load_hf_config, load_hf_weights
+    cfg = load_hf_config(d)
+    single = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32)
+    load_hf_weights(single, d)
+    single.eval()
+    torch.manual_seed(0)
+    ids = torch.randint(0, cfg.vocab_size, (1, 9))
+    with torch.no_grad():
+        want = single(ids)
+    assert torch.allclose(tp_logits, want, atol=1e-4), \
+        (tp_logits - want).abs().max()
