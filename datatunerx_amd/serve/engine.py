@@ -150,6 +150,18 @@ class BatchedKVCache:
         return BatchedKVCache._RowView(self, i)
 
     def update(self, k, v):
+        if k.shape[1] > 1:
+            # batched RIGHT-PADDED prefill: write all rows up to the
+            # padded length; per-row len32 (set by the caller) bounds
+            # every later read, and decode overwrites the first pad
+            # slot, so pad K/V beyond a row's real length is never
+            # consumed. Returns len_dev=None: prefill attention runs
+            # plain causal over the padded batch (pads sit AFTER real
+            # tokens, so no real query attends one).
+            S = k.shape[1]
+            self.k[:, :S] = k
+            self.v[:, :S] = v
+            return k, v, None
         # batched decode append: k/v [B, 1, Hkv, D] at per-row positions
         self.k[self._rows, self.pos64] = k[:, 0]
         self.v[self._rows, self.pos64] = v[:, 0]
@@ -565,11 +577,26 @@ class InferenceEngine:
                                      cfg.head_dim, self.device, dtype)
                       for _ in range(cfg.num_hidden_layers)]
         nxt = [0] * B
-        for i, ids in enumerate(prompts):
-            row = [c.row_view(i) for c in caches]
-            t = torch.tensor([ids], dtype=torch.long, device=self.device)
-            logits = self.model(t, pos0=0, kv_caches=row)
-            nxt[i] = self._sample(logits[0, -1], temperature, top_p)
+        lmax = max(lens)
+        if B > 1 and self.device.type == "cuda":
+            # ONE right-padded batched prefill (pads never attended:
+            # causal + pads-after-real; len32 bounds decode reads)
+            padded = torch.zeros(B, lmax, dtype=torch.long,
+                                 device=self.device)
+            for i, ids in enumerate(prompts):
+                padded[i, :len(ids)] = torch.tensor(ids,
+                                                    dtype=torch.long)
+            logits = self.model(padded, pos0=0, kv_caches=caches)
+            for i in range(B):
+                nxt[i] = self._sample(logits[i, lens[i] - 1],
+                                      temperature, top_p)
+        else:
+            for i, ids in enumerate(prompts):
+                row = [c.row_view(i) for c in caches]
+                t = torch.tensor([ids], dtype=torch.long,
+                                 device=self.device)
+                logits = self.model(t, pos0=0, kv_caches=row)
+                nxt[i] = self._sample(logits[0, -1], temperature, top_p)
         pos = torch.tensor(lens, dtype=torch.long, device=self.device)
         if gd is not None:
             gd.set_state(lens)
