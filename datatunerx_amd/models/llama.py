@@ -17,8 +17,9 @@ import torch
 import torch.nn as nn
 
 from ..ops import rope_tables
-from ..ops.autograd import (attention, cross_entropy, rmsnorm,
-                            rmsnorm_tap, rope, swiglu)
+from ..ops.autograd import (PairedFrozenGemm, QKVProj, attention,
+                            cross_entropy, rmsnorm, rmsnorm_tap,
+                            rope, swiglu)
 from .lora import FrozenLinear, LoRALinearModule
 
 
@@ -101,14 +102,43 @@ class LlamaAttention(nn.Module):
         self.v_proj = _proj(cfg, "v_proj", cfg.hidden_size, Hkv * D, lora, dtype)
         self.o_proj = _proj(cfg, "o_proj", H * D, cfg.hidden_size, lora, dtype)
 
+    def _fused_qkv(self) -> bool:
+        """One autograd node for q/k/v (backward dgrads accumulate via
+        addmm_ instead of autograd's per-consumer grad adds): the
+        standard LoRA shape — q/v LoRA (same scale, fused-RNG-capable),
+        k frozen, base weights untrained."""
+        qm, km, vm = self.q_proj, self.k_proj, self.v_proj
+        return (type(km) is FrozenLinear and
+                isinstance(qm, LoRALinearModule) and
+                isinstance(vm, LoRALinearModule) and
+                not km.weight.requires_grad and
+                not qm.weight.requires_grad and
+                qm.scale == vm.scale and qm.r == vm.r and
+                qm.r <= 16 and qm.in_features % 8 == 0 and
+                qm.dropout == vm.dropout)
+
     def forward(self, x, cos, sin, pos0: int = 0, kv_cache=None,
                 pos_dev=None, residual=None):
         B, S, _ = x.shape
         cfg = self.cfg
         H, Hkv, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
-        q = self.q_proj(x).view(B, S, H, D)
-        k = self.k_proj(x).view(B, S, Hkv, D)
-        v = self.v_proj(x).view(B, S, Hkv, D)
+        if self._fused_qkv():
+            qm, km, vm = self.q_proj, self.k_proj, self.v_proj
+            keep, sq, sv = 1.0, 0, 0
+            if self.training and qm.dropout > 0.0:
+                keep = 1.0 - qm.dropout
+                sq = int(torch.randint(0, 2 ** 31 - 1, (1,)).item())
+                sv = int(torch.randint(0, 2 ** 31 - 1, (1,)).item())
+            q, k, v = QKVProj.apply(x, qm.weight, qm.lora_A, qm.lora_B,
+                                    km.weight, vm.weight, vm.lora_A,
+                                    vm.lora_B, qm.scale, sq, sv, keep)
+            q = q.view(B, S, H, D)
+            k = k.view(B, S, Hkv, D)
+            v = v.view(B, S, Hkv, D)
+        else:
+            q = self.q_proj(x).view(B, S, H, D)
+            k = self.k_proj(x).view(B, S, Hkv, D)
+            v = self.v_proj(x).view(B, S, Hkv, D)
         q = rope(q, cos, sin, pos0, pos_dev)
         k = rope(k, cos, sin, pos0, pos_dev)
         len_dev = None
@@ -133,7 +163,16 @@ class LlamaMLP(nn.Module):
                                cfg.hidden_size, lora, dtype)
 
     def forward(self, x, residual=None):
-        h = swiglu(self.gate_proj(x), self.up_proj(x))
+        if type(self.gate_proj) is FrozenLinear and \
+                type(self.up_proj) is FrozenLinear and \
+                not self.gate_proj.weight.requires_grad:
+            # one node for the gate/up pair: backward accumulates both
+            # dgrads with addmm_ instead of a grad-add per layer
+            g, u = PairedFrozenGemm.apply(x, self.gate_proj.weight,
+                                          self.up_proj.weight)
+            h = swiglu(g, u)
+        else:
+            h = swiglu(self.gate_proj(x), self.up_proj(x))
         if residual is not None and isinstance(self.down_proj, FrozenLinear):
             return self.down_proj(h, residual)    # residual in epilogue
         y = self.down_proj(h)

@@ -313,3 +313,83 @@ def fused_linear_cross_entropy(h2, w, targets, ignore_index: int = -100):
                                         w.requires_grad)
     return FusedLinearCrossEntropy.apply(h2, w, targets, ignore_index,
                                          need)
+
+
+class PairedFrozenGemm(torch.autograd.Function):
+    """(x@Wg^T, x@Wu^T) for two frozen weights sharing one input (the
+    MLP gate/up pair): backward ACCUMULATES the two dgrads into one
+    buffer with addmm_ (beta=1 GEMM epilogue), eliminating the autograd
+    grad-add that a shared input otherwise costs per layer."""
+
+    @staticmethod
+    def forward(ctx, x, wg, wu):
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1])
+        g = torch.nn.functional.linear(x2, wg)
+        u = torch.nn.functional.linear(x2, wu)
+        ctx.save_for_backward(wg, wu)
+        ctx.xshape = xs
+        n = wg.shape[0]
+        return (g.reshape(*xs[:-1], n), u.reshape(*xs[:-1], n))
+
+    @staticmethod
+    def backward(ctx, dg, du):
+        wg, wu = ctx.saved_tensors
+        dg2 = dg.reshape(-1, dg.shape[-1])
+        du2 = du.reshape(-1, du.shape[-1])
+        dx = dg2 @ wg
+        dx.addmm_(du2, wu)              # accumulate in the GEMM epilogue
+        return dx.reshape(ctx.xshape), None, None
+
+
+class QKVProj(torch.autograd.Function):
+    """q/k/v projections as ONE autograd node (k frozen; q/v LoRA):
+    forward runs the three base GEMMs + the fused LoRA low-rank pairs;
+    backward accumulates all three dgrads (and the LoRA dx terms, which
+    were already in-place) into ONE dx buffer via addmm_ — removing the
+    two [M,E] grad-adds autograd inserts per decoder layer for the
+    shared attention input."""
+
+    @staticmethod
+    def forward(ctx, x, wq, aq, bq, wk, wv, av, bv, scale,
+                seed_q, seed_v, keep):
+        xs = x.shape
+        x2 = x.reshape(-1, xs[-1])
+        yq = torch.nn.functional.linear(x2, wq)
+        yk = torch.nn.functional.linear(x2, wk)
+        yv = torch.nn.functional.linear(x2, wv)
+        tq = lora_contract(x2, aq, None, seed_q, keep)
+        lora_expand_add(yq, tq, bq, scale)
+        tv = lora_contract(x2, av, None, seed_v, keep)
+        lora_expand_add(yv, tv, bv, scale)
+        ctx.save_for_backward(x2, wq, aq, bq, wk, wv, av, bv, tq, tv)
+        ctx.meta = (scale, seed_q, seed_v, keep, xs)
+        return (yq.reshape(*xs[:-1], wq.shape[0]),
+                yk.reshape(*xs[:-1], wk.shape[0]),
+                yv.reshape(*xs[:-1], wv.shape[0]))
+
+    @staticmethod
+    def backward(ctx, dyq, dyk, dyv):
+        from . import lora_wgrad
+        x2, wq, aq, bq, wk, wv, av, bv, tq, tv = ctx.saved_tensors
+        scale, seed_q, seed_v, keep, xs = ctx.meta
+        dq2 = dyq.reshape(-1, dyq.shape[-1]).contiguous()
+        dk2 = dyk.reshape(-1, dyk.shape[-1]).contiguous()
+        dv2 = dyv.reshape(-1, dyv.shape[-1]).contiguous()
+        dx = dq2 @ wq
+        dx.addmm_(dk2, wk)
+        dx.addmm_(dv2, wv)
+        # LoRA grads + low-rank dx terms (expand_add is in-place on dx)
+        dtq = lora_contract(dq2, bq.t().contiguous())
+        daq = lora_wgrad(dtq, x2, scale, None, seed_q, keep)
+        dbq = lora_wgrad(tq, dq2, scale).t().contiguous()
+        lora_expand_add(dx, dtq, aq.t().contiguous(), scale, None,
+                        seed_q, keep)
+        dtv = lora_contract(dv2, bv.t().contiguous())
+        dav = lora_wgrad(dtv, x2, scale, None, seed_v, keep)
+        dbv = lora_wgrad(tv, dv2, scale).t().contiguous()
+        lora_expand_add(dx, dtv, av.t().contiguous(), scale, None,
+                        seed_v, keep)
+        return (dx.reshape(xs), None, daq.to(aq.dtype),
+                dbq.to(bq.dtype), None, None, dav.to(av.dtype),
+                dbv.to(bv.dtype), None, None, None, None)
