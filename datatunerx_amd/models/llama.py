@@ -81,6 +81,17 @@ class LlamaConfig:
         d.update(kw)
         return cls(**d)
 
+    @classmethod
+    def mini(cls, **kw):
+        """GPU-capable small config (head_dim 64 — the attention
+        kernels support D in {64, 128}); for on-hardware control-plane
+        and smoke tests."""
+        d = dict(vocab_size=512, hidden_size=256, intermediate_size=512,
+                 num_hidden_layers=2, num_attention_heads=4,
+                 num_key_value_heads=4, max_position_embeddings=512)
+        d.update(kw)
+        return cls(**d)
+
 
 def _proj(cfg: LlamaConfig, name: str, in_f: int, out_f: int, lora: bool,
           dtype):

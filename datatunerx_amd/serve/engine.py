@@ -200,6 +200,9 @@ def build_model(name: str, device, adapter_dir: Optional[str] = None,
         elif name == "llama-tiny":
             model = LlamaForCausalLM(LlamaConfig.tiny(**lora_kw),
                                      dtype=dtype)
+        elif name == "llama-mini":
+            model = LlamaForCausalLM(LlamaConfig.mini(**lora_kw),
+                                     dtype=dtype)
         elif name in ("gpt2-small", "gpt2"):
             model = GPT2ForCausalLM(GPT2Config.small(), dtype=dtype)
         elif name == "gpt2-tiny":

@@ -346,6 +346,8 @@ def _llama_config(name: str):
         return LlamaConfig.llama3_8b()
     if name == "llama-tiny":
         return LlamaConfig.tiny()
+    if name == "llama-mini":
+        return LlamaConfig.mini()
     raise ValueError(f"TP serving supports llama models, not {name!r}")
 
 

@@ -77,6 +77,10 @@ def main(argv=None):
                 model = LlamaForCausalLM(LlamaConfig.tiny(**lora_kw),
                                          lora=lora, dtype=dtype,
                                          train_base=full)
+            elif name == "llama-mini":
+                model = LlamaForCausalLM(LlamaConfig.mini(**lora_kw),
+                                         lora=lora, dtype=dtype,
+                                         train_base=full)
             elif name in ("gpt2-small", "gpt2"):
                 model = GPT2ForCausalLM(GPT2Config.small(
                     lora_r=fargs.lora_rank, lora_alpha=fargs.lora_alpha,
