@@ -79,9 +79,11 @@ def main():
         args.model = "llama-tiny"
         args.seq_len = min(args.seq_len, 128)
 
-    # mb 24 measured fastest on MI355X (tools A/B: 16 -> 31.8k, 24 ->
-    # 32.2k, 32 -> 32.2k tok/s); activations fit comfortably in 288 GB
-    mb = args.micro_batch or (24 if device.type == "cuda" else 2)
+    # micro-batch sweep on MI355X (r2, post-fusion): 24 -> 35.4k,
+    # 48 -> 36.2k, 64 -> 36.5k tok/s, 96 OOM; 64 keeps ~100 GB headroom
+    # of the 288 GB HBM3E (large batches amortize kernel tails; the
+    # 288 GB capacity is exactly what makes mb64 seq-1024 feasible)
+    mb = args.micro_batch or (64 if device.type == "cuda" else 2)
     model, cfg = build_model(args.model, device, args.lora_dropout,
                              args.full_param, args.grad_ckpt)
 
