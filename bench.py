@@ -62,7 +62,7 @@ def main():
     ap.add_argument("--model", default="llama2-7b")
     ap.add_argument("--seq-len", type=int, default=1024)
     ap.add_argument("--micro-batch", type=int, default=0,
-                    help="0 = auto (16 on GPU, 2 on CPU)")
+                    help="0 = auto (64 on GPU, 2 on CPU)")
     ap.add_argument("--grad-accum", type=int, default=1)
     ap.add_argument("--lora-dropout", type=float, default=0.1)
     ap.add_argument("--full-param", action="store_true",
