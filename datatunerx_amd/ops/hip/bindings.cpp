@@ -31,7 +31,7 @@ void launch_xent_dlogits(const void*, const long*, const float*, void*,
 void launch_adamw(void*, float*, const float*, float*, float*, long, float,
                   float, float, float, float, int, hipStream_t);
 void launch_l2_norm(const float*, float*, float*, long, hipStream_t);
-int lora_contract_ksplit(int K);
+int lora_contract_ksplit(int K, long M);
 void launch_dropout_mask(void*, long, unsigned long long, float,
                          hipStream_t);
 void launch_lora_contract(const void*, const void*, const void*, float*,
@@ -222,7 +222,7 @@ torch::Tensor lora_contract(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(r <= 64, "r <= 64");
   TORCH_CHECK(K % 8 == 0);
   auto t = torch::empty({M, r}, x.options().dtype(torch::kFloat));
-  const int nsplit = lora_contract_ksplit(K);
+  const int nsplit = lora_contract_ksplit(K, M);
   torch::Tensor part;
   float* part_ptr = nullptr;
   if (nsplit > 1) {

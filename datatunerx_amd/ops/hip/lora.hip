@@ -352,7 +352,13 @@ static int lora_contract_kspan(int K) {
   return DTX_CDIV(units, nsplit) * 1024;
 }
 
-int lora_contract_ksplit(int K) {
+int lora_contract_ksplit(int K, long M) {
+  // K-split exists to FILL the chip when M alone cannot (grid.x tops
+  // out at 128 four-wave blocks). At M >= 32K rows the m-tiles already
+  // saturate every CU, and splitting only adds the fp32 partials +
+  // reduce pass — skip it (mb64 shapes: M = 65536).
+  if (M >= (128L * 256))
+    return 1;
   return DTX_CDIV(K, lora_contract_kspan(K));
 }
 
@@ -360,8 +366,9 @@ void launch_lora_contract(const void* X, const void* W, const void* Mk,
                           float* part, float* out, long M, int K, int r,
                           unsigned long long seed, float keep,
                           hipStream_t s) {
-  const int nsplit = lora_contract_ksplit(K);
-  const int kspan = lora_contract_kspan(K);
+  const int nsplit = lora_contract_ksplit(K, M);
+  const int kspan = nsplit == 1 ? DTX_CDIV(K, 1024) * 1024
+                                : lora_contract_kspan(K);
   long gw = DTX_CDIV(M, 128);
   dim3 grid((int)(gw < 128 ? (gw < 1 ? 1 : gw) : 128), nsplit);
   float* dst = nsplit > 1 ? part : out;
