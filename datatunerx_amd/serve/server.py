@@ -107,6 +107,24 @@ class BatchingFront:
             reqs = [b["req"] for b in batch]
             streaming = any("tokq" in b for b in batch)
             try:
+                if len(batch) == 1:
+                    # single request: the padded batched graph would
+                    # decode MAXB rows for one stream — use the
+                    # single-stream (graphed) path instead
+                    b = batch[0]
+                    r = b["req"]
+                    if "tokq" in b:
+                        for d in self.engine.chat_stream(
+                                r["messages"], r["max_tokens"],
+                                r["temperature"], r["top_p"]):
+                            b["tokq"].put(d)
+                        b["tokq"].put(None)
+                    else:
+                        b["out"] = self.engine.chat(
+                            r["messages"], r["max_tokens"],
+                            r["temperature"], r["top_p"])
+                        b["ev"].set()
+                    continue
                 if streaming and hasattr(self.engine,
                                          "chat_batch_stream"):
                     texts = [""] * len(batch)
