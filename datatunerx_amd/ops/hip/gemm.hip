@@ -475,7 +475,16 @@ void gemm_nt_kernel(const unsigned short* __restrict__ A,
       // tile(s) issue no stages, so their in-flight count is too low
       // for the counted wait to retire what the next phases read —
       // drain fully there (end of the loop anyway).
-      if (PIPE == 5) {
+      if (PIPE == 6 || PIPE == 7) {
+        // TIMING PROBES ONLY (results are racy/wrong): 6 = PIPE2
+        // without the counted vmcnt (isolates the wait cost);
+        // 7 = PIPE2 without the barriers (isolates barrier parking).
+        if (PIPE == 6 && (p & 1)) dtx_bar();
+        if (PIPE == 7 && (p & 1)) {
+          if (p == 3) { if (pre) { asm volatile(
+              "s_waitcnt vmcnt(8)" ::: "memory"); } else dtx_vmcnt0(); }
+        }
+      } else if (PIPE == 5) {
         if (p == 3) {
           if (kt + 2 < KT) {
             asm volatile("s_waitcnt vmcnt(2)" ::: "memory");

@@ -39,10 +39,12 @@ void run(const unsigned short* A, const unsigned short* B,
 typedef void (*runfn)(const unsigned short*, const unsigned short*,
                       unsigned short*, long, int, int);
 #define NVAR 6
-static runfn FNS[NVAR] = {run<1, 2, 0>, run<1, 5, 0>, run<1, 5, 0>,
-                          run<2, 2, 0>, run<2, 5, 0>, run<2, 5, 0>};
-static const char* NAMES[NVAR] = {"g1p2", "g1p5", "g1p5b",
-                                  "g2p2", "g2p5", "g2p5b"};
+static runfn FNS[NVAR] = {run<1, 2, 0>, run<1, 6, 0>, run<1, 7, 0>,
+                          run<2, 2, 0>, run<2, 6, 0>, run<2, 7, 0>};
+static const char* NAMES[NVAR] = {"g1p2", "g1p6-novmcnt-RACY",
+                                  "g1p7-nobar-RACY",
+                                  "g2p2", "g2p6-novmcnt-RACY",
+                                  "g2p7-nobar-RACY"};
 
 int main(int argc, char** argv) {
   if (argc < 5) { fprintf(stderr, "usage: see header\n"); return 1; }
