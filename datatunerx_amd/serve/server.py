@@ -220,6 +220,7 @@ def build_handler(pool, batcher=None):
                         # would leave follower ranks blocked in
                         # dist.broadcast — desyncing every later
                         # request's collectives.
+                        client_gone = False
                         if batcher is not None:
                             import contextlib
                             lease = contextlib.nullcontext()
@@ -230,7 +231,6 @@ def build_handler(pool, batcher=None):
                                 as engine:
                             if lease is None:
                                 gen = engine.chat_stream(*args)
-                            client_gone = False
                             for delta in gen:
                                 if client_gone:
                                     continue  # keep consuming to the end
@@ -255,6 +255,8 @@ def build_handler(pool, batcher=None):
                                     OSError):
                                 pass
                         return
+                    # (non-streaming paths below; a mid-stream failure
+                    # must not fall through to _send: headers are gone)
                     if batcher is not None:
                         text = batcher.chat(*args)
                     else:
@@ -278,7 +280,10 @@ def build_handler(pool, batcher=None):
                 else:
                     self._send(404, {"error": "not found"})
             except Exception as e:      # surface engine errors as 500s
-                self._send(500, {"error": f"{type(e).__name__}: {e}"})
+                try:
+                    self._send(500, {"error": f"{type(e).__name__}: {e}"})
+                except Exception:
+                    pass                # headers already sent (stream)
 
     return Handler
 
