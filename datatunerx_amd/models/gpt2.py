@@ -20,7 +20,12 @@ from .lora import FrozenLinear, LoRALinearModule
 
 @dataclass
 class GPT2Config:
-    vocab_size: int = 50257
+    # padded from GPT-2's 50257 to the next multiple of 8: the
+    # fused CE kernel wants V % 8 == 0, ids stay < 50257, and the
+    # pad logits train to -inf-ish harmlessly (random-init path;
+    # there is no HF-GPT-2 weight loader to stay byte-compatible
+    # with)
+    vocab_size: int = 50264
     hidden_size: int = 768
     num_hidden_layers: int = 12
     num_attention_heads: int = 12
