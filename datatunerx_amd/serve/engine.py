@@ -168,6 +168,27 @@ class BatchedKVCache:
         return self.k, self.v, self.len32
 
 
+def builtin_config(name: str, **lora_kw):
+    """Resolve a builtin model name to ("llama"|"gpt2", config) without
+    allocating weights — the single registry the engine, TP server and
+    tests share. Raises ValueError on unknown names."""
+    if name in ("llama2-7b", "llama-2-7b"):
+        return "llama", LlamaConfig.llama2_7b(**lora_kw)
+    if name in ("llama2-13b", "llama-2-13b"):
+        return "llama", LlamaConfig.llama2_13b(**lora_kw)
+    if name in ("llama3-8b", "llama-3-8b"):
+        return "llama", LlamaConfig.llama3_8b(**lora_kw)
+    if name == "llama-tiny":
+        return "llama", LlamaConfig.tiny(**lora_kw)
+    if name == "llama-mini":
+        return "llama", LlamaConfig.mini(**lora_kw)
+    if name in ("gpt2-small", "gpt2"):
+        return "gpt2", GPT2Config.small()
+    if name == "gpt2-tiny":
+        return "gpt2", GPT2Config.tiny()
+    raise ValueError(f"unknown model {name!r}")
+
+
 def build_model(name: str, device, adapter_dir: Optional[str] = None,
                 lora_kw: Optional[dict] = None):
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
@@ -191,24 +212,12 @@ def build_model(name: str, device, adapter_dir: Optional[str] = None,
         if hf_dir:
             model = LlamaForCausalLM(load_hf_config(name, **lora_kw),
                                      lora=bool(adapter_dir), dtype=dtype)
-        elif name in ("llama2-7b", "llama-2-7b"):
-            model = LlamaForCausalLM(LlamaConfig.llama2_7b(**lora_kw),
-                                     dtype=dtype)
-        elif name in ("llama2-13b", "llama-2-13b"):
-            model = LlamaForCausalLM(LlamaConfig.llama2_13b(**lora_kw),
-                                     dtype=dtype)
-        elif name == "llama-tiny":
-            model = LlamaForCausalLM(LlamaConfig.tiny(**lora_kw),
-                                     dtype=dtype)
-        elif name == "llama-mini":
-            model = LlamaForCausalLM(LlamaConfig.mini(**lora_kw),
-                                     dtype=dtype)
-        elif name in ("gpt2-small", "gpt2"):
-            model = GPT2ForCausalLM(GPT2Config.small(), dtype=dtype)
-        elif name == "gpt2-tiny":
-            model = GPT2ForCausalLM(GPT2Config.tiny(), dtype=dtype)
         else:
-            raise ValueError(f"unknown model {name!r}")
+            family, cfg = builtin_config(name, **lora_kw)
+            if family == "llama":
+                model = LlamaForCausalLM(cfg, dtype=dtype)
+            else:
+                model = GPT2ForCausalLM(cfg, dtype=dtype)
     if hf_dir:
         load_hf_weights(model, name)   # real local weights
     else:

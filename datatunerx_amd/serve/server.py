@@ -360,18 +360,11 @@ def tp_follower_loop(engine):
 
 
 def _llama_config(name: str):
-    from ..models import LlamaConfig
-    if name in ("llama2-7b", "llama-2-7b"):
-        return LlamaConfig.llama2_7b()
-    if name in ("llama2-13b", "llama-2-13b"):
-        return LlamaConfig.llama2_13b()
-    if name in ("llama3-8b", "llama-3-8b"):
-        return LlamaConfig.llama3_8b()
-    if name == "llama-tiny":
-        return LlamaConfig.tiny()
-    if name == "llama-mini":
-        return LlamaConfig.mini()
-    raise ValueError(f"TP serving supports llama models, not {name!r}")
+    from .engine import builtin_config
+    family, cfg = builtin_config(name)
+    if family != "llama":
+        raise ValueError(f"TP serving supports llama models, not {name!r}")
+    return cfg
 
 
 def main(argv=None):

@@ -623,3 +623,27 @@ def test_serve_batched_stream_matches_batch(tmp_path):
         assert got == want
     finally:
         httpd.shutdown()
+
+
+def test_builtin_model_registry_covers_all_serve_names():
+    """Every builtin model name the platform advertises resolves to a
+    config through the ONE registry the engine and the TP server share
+    (a llama3-8b serve request once failed only on GPU because the
+    engine's name list had drifted from the server's)."""
+    from datatunerx_amd.serve.engine import builtin_config
+    from datatunerx_amd.serve.server import _llama_config
+    names = ["llama2-7b", "llama-2-7b", "llama2-13b", "llama-2-13b",
+             "llama3-8b", "llama-3-8b", "llama-tiny", "llama-mini",
+             "gpt2-small", "gpt2", "gpt2-tiny"]
+    for n in names:
+        family, cfg = builtin_config(n)
+        assert cfg.vocab_size > 0
+        if family == "llama":
+            assert _llama_config(n) is not None
+    with pytest.raises(ValueError):
+        builtin_config("no-such-model")
+    with pytest.raises(ValueError):
+        _llama_config("gpt2-small")
+    # GQA geometry reaches the engine path for llama3
+    _, c3 = builtin_config("llama3-8b", lora_r=4)
+    assert c3.num_key_value_heads == 8 and c3.lora_r == 4
