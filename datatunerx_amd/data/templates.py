@@ -168,6 +168,12 @@ register_template("chatglm2", Template(
 register_template("chatglm3", Template(
     prefix=["<|system|>\n{{system}}"],
     prompt=["<|user|>\n{{query}}<|assistant|>\n"]))
+register_template("llama3", Template(
+    prefix=["<|start_header_id|>system<|end_header_id|>\n\n"
+            "{{system}}<|eot_id|>"],
+    prompt=["<|start_header_id|>user<|end_header_id|>\n\n{{query}}<|eot_id|>"
+            "<|start_header_id|>assistant<|end_header_id|>\n\n"],
+    stop_words=["<|eot_id|>"]))
 register_template("openchat", Template(
     prompt=["GPT4 User: {{query}}<|end_of_turn|>GPT4 Assistant: "],
     sep=["<|end_of_turn|>"], stop_words=["<|end_of_turn|>"]))

@@ -144,17 +144,21 @@ class HFTokenizer:
     (SentencePiece)."""
 
     def __init__(self, backend, kind: str, bos: int, eos: int, pad: int,
-                 vocab: int):
+                 vocab: int, stop_token_ids=()):
         self._t = backend
         self.kind = kind
         self.bos_token_id = bos
         self.eos_token_id = eos
         self.pad_token_id = pad
         self.vocab_size = vocab
+        # llama3-style checkpoints declare eos_token_id as a LIST
+        # ([128001, 128008, 128009]); the engine stops on any of these
+        self.stop_token_ids = set(stop_token_ids) | {eos}
 
     @classmethod
     def from_dir(cls, model_dir: str) -> "HFTokenizer":
         bos, eos, pad = 1, 2, None
+        stops = set()
         cfg_path = os.path.join(model_dir, "config.json")
         if os.path.exists(cfg_path):
             with open(cfg_path) as f:
@@ -162,6 +166,9 @@ class HFTokenizer:
             bos = hc.get("bos_token_id", bos) or bos
             eos = hc.get("eos_token_id", eos) or eos
             pad = hc.get("pad_token_id", pad)
+            if isinstance(eos, (list, tuple)):
+                stops = set(int(i) for i in eos)
+                eos = int(eos[0])
         tj = os.path.join(model_dir, "tokenizer.json")
         tm = os.path.join(model_dir, "tokenizer.model")
         if os.path.exists(tj):
@@ -169,7 +176,8 @@ class HFTokenizer:
             t = Tokenizer.from_file(tj)
             vocab = t.get_vocab_size()
             return cls(t, "tokenizers", bos, eos,
-                       pad if pad is not None else eos, vocab)
+                       pad if pad is not None else eos, vocab,
+                       stop_token_ids=stops)
         if os.path.exists(tm):
             import sentencepiece as spm
             sp = spm.SentencePieceProcessor(model_file=tm)
@@ -177,7 +185,7 @@ class HFTokenizer:
                        sp.eos_id() if sp.eos_id() >= 0 else eos,
                        pad if pad is not None else
                        (sp.pad_id() if sp.pad_id() >= 0 else 0),
-                       sp.get_piece_size())
+                       sp.get_piece_size(), stop_token_ids=stops)
         raise FileNotFoundError(
             f"no tokenizer.json or tokenizer.model in {model_dir}")
 
@@ -191,8 +199,8 @@ class HFTokenizer:
         return ids
 
     def decode(self, ids):
-        ids = [int(i) for i in ids
-               if int(i) not in (self.bos_token_id, self.eos_token_id)]
+        drop = self.stop_token_ids | {self.bos_token_id}
+        ids = [int(i) for i in ids if int(i) not in drop]
         if self.kind == "tokenizers":
             return self._t.decode(ids)
         return self._t.decode(ids)

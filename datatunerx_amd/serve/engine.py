@@ -311,6 +311,28 @@ class InferenceEngine:
                           self.device.type == "cuda" and
                           not dist.is_initialized() and
                           os.environ.get("DTX_NO_GRAPH") != "1")
+        # stop set: tokenizer eos (may be a LIST in llama3-style
+        # config.json), plus any template stop_words that the tokenizer
+        # encodes to a single special id (llama3's <|eot_id|> ends a
+        # turn but is distinct from eos)
+        eos = getattr(self.tok, "eos_token_id", None)
+        self._stop_ids = set(eos) if isinstance(eos, (list, tuple)) \
+            else ({eos} if eos is not None else set())
+        extra = getattr(self.tok, "stop_token_ids", None)
+        if extra:
+            self._stop_ids.update(int(i) for i in extra)
+        try:
+            tmpl = get_template(template)
+            for w in tmpl.stop_words:
+                ids = self.tok.encode(w, add_special_tokens=False) \
+                    if hasattr(self.tok, "encode") else []
+                if len(ids) == 1:
+                    self._stop_ids.add(int(ids[0]))
+        except (KeyError, TypeError):
+            pass
+
+    def _is_stop(self, tok_id) -> bool:
+        return int(tok_id) in self._stop_ids
 
     def _stream_ctx(self):
         import contextlib
@@ -447,7 +469,7 @@ class InferenceEngine:
             gd.after_prefill(len(prompt_ids))
             nxt = self._sample(logits[0, -1], temperature, top_p)
             for _ in range(max_new_tokens):
-                if nxt == self.tok.eos_token_id:
+                if self._is_stop(nxt):
                     break
                 yield nxt
                 n_out += 1
@@ -472,7 +494,7 @@ class InferenceEngine:
                 if pos >= max_s:
                     break
                 nxt = self._sample(logits[0, -1], temperature, top_p)
-                if nxt == self.tok.eos_token_id:
+                if self._is_stop(nxt):
                     break
                 yield nxt
                 cur = torch.tensor([[nxt]], dtype=torch.long,
@@ -484,7 +506,7 @@ class InferenceEngine:
                                    device=self.device)
                 logits = self.model(cur)
                 nxt = self._sample(logits[0, -1], temperature, top_p)
-                if nxt == self.tok.eos_token_id:
+                if self._is_stop(nxt):
                     break
                 yield nxt
                 seq.append(nxt)
@@ -618,7 +640,6 @@ class InferenceEngine:
                 c.pos64.copy_(pos)
                 c.len32.copy_((pos + 1).to(torch.int32))
             pos32 = pos.to(torch.int32)
-        eos = self.tok.eos_token_id
         n_live = len(lens0)
         if budgets is None:
             budgets = [max_new_tokens] * n_live
@@ -626,7 +647,7 @@ class InferenceEngine:
         done = [False] * B
         first = []
         for i in range(B):
-            if nxt[i] == eos:
+            if self._is_stop(nxt[i]):
                 done[i] = True
             elif i < n_live:
                 if count[i] < budgets[i]:
@@ -663,7 +684,7 @@ class InferenceEngine:
             for i, t in enumerate(toks):
                 if done[i]:
                     continue
-                if t == eos:
+                if self._is_stop(t):
                     done[i] = True
                 elif i < n_live:
                     emitted.append((i, t))

@@ -151,3 +151,45 @@ def test_hf_export_roundtrip(hf_dir, tmp_path):
     with torch.no_grad():
         assert torch.allclose(ours(ids), ref(ids).logits, atol=2e-4,
                               rtol=1e-3)
+
+
+def test_list_eos_and_engine_stop_set(hf_dir, tmp_path):
+    """llama3-style checkpoints declare eos_token_id as a LIST; the
+    tokenizer keeps all of them as stop ids and the engine stops on any
+    (a single-eos engine would run past <|eot_id|> on real llama3)."""
+    import shutil
+    d = str(tmp_path / "l3style")
+    shutil.copytree(hf_dir, d)
+    cfgp = os.path.join(d, "config.json")
+    with open(cfgp) as f:
+        hc = json.load(f)
+    hc["eos_token_id"] = [2, 7, 9]
+    with open(cfgp, "w") as f:
+        json.dump(hc, f)
+    tok = HFTokenizer.from_dir(d)
+    assert tok.eos_token_id == 2
+    assert tok.stop_token_ids == {2, 7, 9}
+    # decode drops every stop id, not just the primary eos
+    ids = tok.encode("fox", add_special_tokens=False)
+    assert tok.decode(ids + [7, 9, 2]) == tok.decode(ids)
+
+    from datatunerx_amd.models import LlamaConfig
+    from datatunerx_amd.serve.engine import InferenceEngine
+    model = LlamaForCausalLM(LlamaConfig.tiny(), dtype=torch.float32)
+    eng = InferenceEngine(model, tokenizer=tok, template="llama3",
+                          device=torch.device("cpu"))
+    assert {2, 7, 9} <= eng._stop_ids
+    assert eng._is_stop(7) and not eng._is_stop(5)
+
+
+def test_llama3_template_encode():
+    from datatunerx_amd.data.dataset import ByteTokenizer
+    from datatunerx_amd.data.templates import get_template
+    t = get_template("llama3")
+    tok = ByteTokenizer()
+    src, tgt = t.encode_oneturn(tok, "hi", "yo", None, "sys")
+    text = bytes(i - 3 for i in src if i >= 3).decode()
+    assert "<|start_header_id|>system<|end_header_id|>\n\nsys<|eot_id|>" \
+        in text
+    assert text.endswith("assistant<|end_header_id|>\n\n")
+    assert "<|eot_id|>" in t.stop_words
