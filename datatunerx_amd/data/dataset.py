@@ -119,13 +119,43 @@ class SFTDataset:
                              template_name, cutoff_len)
 
     @classmethod
+    def from_rows_pt(cls, rows, tokenizer,
+                     cutoff_len: int = DEFAULT_CUTOFF_LEN):
+        """Pretraining (stage=pt, parser.py:112-221 stage field): pack
+        raw text into fixed cutoff_len blocks with eos separators and
+        train on EVERY position (labels = input_ids, no -100 prompt
+        masking). Rows use the same columns as sft; instruction and
+        response are concatenated as plain text."""
+        eos = ([tokenizer.eos_token_id]
+               if getattr(tokenizer, "eos_token_id", None) is not None
+               else [])
+        stream: List[int] = []
+        for row in rows:
+            text = " ".join(s for s in (row.get("instruction", ""),
+                                        row.get("response", "")) if s)
+            if not text:
+                continue
+            try:
+                ids = tokenizer.encode(text, add_special_tokens=False)
+            except TypeError:
+                ids = tokenizer.encode(text)
+            stream.extend(ids + eos)
+        ex = []
+        for i in range(0, len(stream) - cutoff_len + 1, cutoff_len):
+            blk = stream[i:i + cutoff_len]
+            ex.append({"input_ids": blk, "labels": list(blk)})
+        if not ex and stream:            # corpus shorter than one block
+            ex.append({"input_ids": stream, "labels": list(stream)})
+        return cls(ex)
+
+    @classmethod
     def synthetic(cls, n_examples: int, seq_len: int, vocab_size: int,
                   seed: int = 0, mask_frac: float = 0.25):
         """Fixed-shape synthetic instruction data for benches (BASELINE
         contract: synthetic data, stated in bench output)."""
         g = torch.Generator().manual_seed(seed)
         ex = []
-        n_src = max(1, int(seq_len * mask_frac))
+        n_src = max(1, int(seq_len * mask_frac)) if mask_frac > 0 else 0
         for _ in range(n_examples):
             ids = torch.randint(3, vocab_size, (seq_len,), generator=g)
             labels = ids.clone()

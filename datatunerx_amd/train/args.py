@@ -20,7 +20,11 @@ class ModelArguments:
 
 @dataclass
 class FinetuningArguments:
-    stage: str = "sft"                          # pt/sft supported
+    stage: str = "sft"        # sft (prompt-masked) | pt (packed blocks,
+                              # all positions trained); rm/ppo/dpo are
+                              # rejected with a clear error (the
+                              # reference declares but never runs them
+                              # either — parser.py:112-221)
     finetuning_type: str = "lora"               # lora / full
     lora_rank: int = 8
     lora_alpha: float = 32.0

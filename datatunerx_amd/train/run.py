@@ -112,17 +112,29 @@ def main(argv=None):
 
         tok = load_tokenizer(name)
         vocab = getattr(model.cfg, "vocab_size")
+        if fargs.stage not in ("sft", "pt"):
+            raise SystemExit(
+                f"stage {fargs.stage!r} not supported (sft, pt)")
+        pt = fargs.stage == "pt"
         if dargs.dataset_path and os.path.exists(dargs.dataset_path):
-            ds = SFTDataset.from_csv(
-                dargs.dataset_path, tok,
-                column_map={"instruction": dargs.instruction_column,
-                            "response": dargs.response_column},
-                template_name=dargs.prompt_template,
-                cutoff_len=dargs.block_size)
+            from ..data.dataset import read_csv_rows
+            cmap = {"instruction": dargs.instruction_column,
+                    "response": dargs.response_column}
+            if pt:
+                # pretraining: pack text blocks, train every position
+                ds = SFTDataset.from_rows_pt(
+                    read_csv_rows(dargs.dataset_path, cmap), tok,
+                    cutoff_len=dargs.block_size)
+            else:
+                ds = SFTDataset.from_csv(
+                    dargs.dataset_path, tok, column_map=cmap,
+                    template_name=dargs.prompt_template,
+                    cutoff_len=dargs.block_size)
         else:
             n = dargs.synthetic_examples or 256
             ds = SFTDataset.synthetic(n, dargs.block_size, vocab,
-                                      seed=fargs.seed)
+                                      seed=fargs.seed,
+                                      mask_frac=0.0 if pt else 0.25)
         eval_ds = None
         if dargs.eval_dataset_path and os.path.exists(dargs.eval_dataset_path):
             eval_ds = SFTDataset.from_csv(
