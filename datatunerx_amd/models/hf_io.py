@@ -18,6 +18,11 @@ Name mapping (HF LlamaForCausalLM -> this repo's LlamaForCausalLM):
                                           embed_tokens when tied)
 No permutation is needed: this repo's RoPE is the HF rotate-half
 convention (ops/reference.py:51-61).
+
+Qwen2-family checkpoints load through the same path: `architectures`
+containing "Qwen2" sets attention_bias, and
+  model.layers.N.self_attn.{q,k,v}_proj.bias -> layers.N.self_attn.{q,k,v}_bias
+(Mistral loads as plain Llama geometry.)
 """
 
 from __future__ import annotations
@@ -39,8 +44,10 @@ def load_hf_config(model_dir: str, **lora_kw) -> LlamaConfig:
     with open(os.path.join(model_dir, "config.json")) as f:
         hc = json.load(f)
     archs = hc.get("architectures") or ["LlamaForCausalLM"]
-    if not any("Llama" in a or "Mistral" in a for a in archs):
+    if not any("Llama" in a or "Mistral" in a or "Qwen2" in a
+               for a in archs):
         raise ValueError(f"unsupported architecture {archs} in {model_dir}")
+    qwen2 = any("Qwen2" in a for a in archs)
     kw = dict(
         vocab_size=hc["vocab_size"],
         hidden_size=hc["hidden_size"],
@@ -52,6 +59,9 @@ def load_hf_config(model_dir: str, **lora_kw) -> LlamaConfig:
         max_position_embeddings=hc.get("max_position_embeddings", 4096),
         rms_norm_eps=hc.get("rms_norm_eps", 1e-5),
         rope_theta=hc.get("rope_theta", 10000.0),
+        # Qwen2 hardcodes q/k/v bias (no config field in older
+        # transformers); Llama-family exposes attention_bias explicitly
+        attention_bias=bool(hc.get("attention_bias", qwen2)),
     )
     hd = hc.get("head_dim")
     if hd and hd * kw["num_attention_heads"] != kw["hidden_size"]:
@@ -101,6 +111,10 @@ def _map_name(hf_name: str):
         return n[: -len(".weight")]
     if n.startswith("rotary_emb") or "rotary_emb" in n:
         return None                      # computed, not loaded
+    for p in ("q", "k", "v"):
+        suf = f"self_attn.{p}_proj.bias"
+        if n.endswith(suf):              # Qwen2 qkv bias -> {p}_bias
+            return n[: -len(suf)] + f"self_attn.{p}_bias"
     return n
 
 
@@ -240,6 +254,7 @@ def save_hf_model(model, out_dir: str):
             "max_position_embeddings": cfg.max_position_embeddings,
             "rms_norm_eps": cfg.rms_norm_eps,
             "rope_theta": cfg.rope_theta,
+            "attention_bias": bool(getattr(cfg, "attention_bias", False)),
             "head_dim": head_dim,
             "torch_dtype": "bfloat16",
             "tie_word_embeddings": False,
@@ -250,6 +265,10 @@ def save_hf_model(model, out_dir: str):
             continue                    # adapters export separately
         if name == "norm":
             hf = "model.norm.weight"
+        elif name.endswith(("self_attn.q_bias", "self_attn.k_bias",
+                            "self_attn.v_bias")):
+            # inverse of _map_name's Qwen2 bias rule
+            hf = "model." + name[:-len("_bias")] + "_proj.bias"
         elif name == "lm_head.weight":
             hf = "lm_head.weight"
         elif name.endswith(("input_layernorm", "post_attention_layernorm")):

@@ -34,6 +34,8 @@ class LlamaConfig:
     max_position_embeddings: int = 4096
     rms_norm_eps: float = 1e-5
     rope_theta: float = 10000.0
+    # Qwen2-family: bias on q/k/v projections (o/mlp stay bias-free)
+    attention_bias: bool = False
     # LoRA (reference defaults: parser.py:138-149, finetune_controller.go:482)
     lora_r: int = 8
     lora_alpha: float = 32.0
@@ -112,6 +114,19 @@ class LlamaAttention(nn.Module):
         self.k_proj = _proj(cfg, "k_proj", cfg.hidden_size, Hkv * D, lora, dtype)
         self.v_proj = _proj(cfg, "v_proj", cfg.hidden_size, Hkv * D, lora, dtype)
         self.o_proj = _proj(cfg, "o_proj", H * D, cfg.hidden_size, lora, dtype)
+        if cfg.attention_bias:
+            # Qwen2 convention: bias on q/k/v only. Added as a separate
+            # broadcast add after the (possibly fused) projections —
+            # frozen under LoRA, trainable under train_base like every
+            # other base weight.
+            self.q_bias = nn.Parameter(torch.zeros(H * D, dtype=dtype),
+                                       requires_grad=False)
+            self.k_bias = nn.Parameter(torch.zeros(Hkv * D, dtype=dtype),
+                                       requires_grad=False)
+            self.v_bias = nn.Parameter(torch.zeros(Hkv * D, dtype=dtype),
+                                       requires_grad=False)
+        else:
+            self.q_bias = self.k_bias = self.v_bias = None
 
     def _fused_qkv(self) -> bool:
         """One autograd node for q/k/v (backward dgrads accumulate via
@@ -143,13 +158,17 @@ class LlamaAttention(nn.Module):
             q, k, v = QKVProj.apply(x, qm.weight, qm.lora_A, qm.lora_B,
                                     km.weight, vm.weight, vm.lora_A,
                                     vm.lora_B, qm.scale, sq, sv, keep)
-            q = q.view(B, S, H, D)
-            k = k.view(B, S, Hkv, D)
-            v = v.view(B, S, Hkv, D)
         else:
-            q = self.q_proj(x).view(B, S, H, D)
-            k = self.k_proj(x).view(B, S, Hkv, D)
-            v = self.v_proj(x).view(B, S, Hkv, D)
+            q = self.q_proj(x)
+            k = self.k_proj(x)
+            v = self.v_proj(x)
+        if self.q_bias is not None:
+            q = q + self.q_bias
+            k = k + self.k_bias
+            v = v + self.v_bias
+        q = q.view(B, S, H, D)
+        k = k.view(B, S, Hkv, D)
+        v = v.view(B, S, Hkv, D)
         q = rope(q, cos, sin, pos0, pos_dev)
         k = rope(k, cos, sin, pos0, pos_dev)
         len_dev = None

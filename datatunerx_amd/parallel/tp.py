@@ -290,6 +290,10 @@ def load_hf_weights_tp(model, model_dir: str, cfg: LlamaConfig,
         elif any(k in name for k in ("o_proj", "down_proj")):
             sh = t.shape[1] // ws
             p.copy_(t[:, rank * sh:(rank + 1) * sh].to(p.dtype))
+        elif name.endswith(("q_bias", "k_bias", "v_bias")):
+            # Qwen2 qkv bias: column-parallel like its weight's rows
+            sh = t.shape[0] // ws
+            p.copy_(t[rank * sh:(rank + 1) * sh].to(p.dtype))
         elif name == "lm_head.weight":
             sh = t.shape[0] // ws
             p.copy_(t[rank * sh:(rank + 1) * sh].to(p.dtype))

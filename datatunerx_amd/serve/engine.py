@@ -456,6 +456,11 @@ class InferenceEngine:
     def _generate_stream(self, prompt_ids: List[int],
                          max_new_tokens: int = 64,
                          temperature: float = 0.0, top_p: float = 1.0):
+        limit = getattr(self.model.cfg, "max_position_embeddings", 1 << 30)
+        if len(prompt_ids) >= limit:
+            raise ValueError(
+                f"prompt length {len(prompt_ids)} exceeds the model "
+                f"context ({limit} positions)")
         ids = torch.tensor([prompt_ids], dtype=torch.long,
                            device=self.device)
         n_out = 0

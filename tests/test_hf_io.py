@@ -193,3 +193,101 @@ def test_llama3_template_encode():
         in text
     assert text.endswith("assistant<|end_header_id|>\n\n")
     assert "<|eot_id|>" in t.stop_words
+
+
+def test_qwen2_checkpoint_matches_transformers_logits(tmp_path):
+    """Qwen2-family (q/k/v bias, GQA, tied head) loads through the same
+    HF-dir path and matches transformers' fp32 logits."""
+    d = str(tmp_path / "qwen2")
+    cfg = transformers.Qwen2Config(
+        vocab_size=256, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, max_position_embeddings=128,
+        rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=True, attn_implementation="eager")
+    torch.manual_seed(11)
+    hf = transformers.Qwen2ForCausalLM(cfg)
+    hf.save_pretrained(d, safe_serialization=True)
+
+    mycfg = load_hf_config(d)
+    assert mycfg.attention_bias and mycfg.num_key_value_heads == 2
+    model = LlamaForCausalLM(mycfg, lora=False, dtype=torch.float32)
+    n = load_hf_weights(model, d)
+    assert n >= 2 * 9 + 3          # per-layer tensors + embed/norm/head
+    # bias actually landed (Qwen2 inits bias to zeros=False? randn no —
+    # from_config inits bias zero; force nonzero to prove the add runs)
+    with torch.no_grad():
+        for lyr, mylyr in zip(hf.model.layers, model.layers):
+            for p in ("q", "k", "v"):
+                b = getattr(lyr.self_attn, f"{p}_proj").bias
+                b.copy_(torch.randn_like(b) * 0.5)
+                getattr(mylyr.self_attn, f"{p}_bias").copy_(b)
+    ids = torch.randint(0, 256, (2, 17))
+    with torch.no_grad():
+        ref = hf(ids).logits
+        got = model(ids)
+    assert torch.allclose(ref, got, atol=2e-4, rtol=1e-3), \
+        (ref - got).abs().max()
+
+
+def test_qwen2_export_roundtrip(tmp_path):
+    """save_hf_model writes qkv bias back under HF names; reloading
+    reproduces identical logits."""
+    from datatunerx_amd.models import LlamaConfig
+    from datatunerx_amd.models.hf_io import save_hf_model
+    cfg = LlamaConfig.tiny(attention_bias=True)
+    torch.manual_seed(4)
+    m = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32)
+    m.init_random(seed=9)
+    with torch.no_grad():
+        for lyr in m.layers:
+            for p in ("q", "k", "v"):
+                b = getattr(lyr.self_attn, f"{p}_bias")
+                b.copy_(torch.randn_like(b) * 0.3)
+    d = str(tmp_path / "exp")
+    save_hf_model(m, d)
+    m2 = LlamaForCausalLM(load_hf_config(d), lora=False,
+                          dtype=torch.float32)
+    load_hf_weights(m2, d)
+    ids = torch.randint(0, cfg.vocab_size, (1, 23))
+    with torch.no_grad():
+        assert torch.equal(m(ids), m2(ids))
+
+
+def test_qwen2_lora_train_and_serve(tmp_path):
+    """LoRA fine-tune + engine serve on a Qwen2 checkpoint dir (bias
+    params stay frozen under LoRA; adapter applies on top)."""
+    from datatunerx_amd.serve.engine import InferenceEngine, build_model
+    from datatunerx_amd.train.run import main as train_main
+    d = str(tmp_path / "qwen2")
+    cfg = transformers.Qwen2Config(
+        vocab_size=256, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=2, num_attention_heads=2,
+        num_key_value_heads=2, max_position_embeddings=256,
+        tie_word_embeddings=False, attn_implementation="eager")
+    transformers.Qwen2ForCausalLM(cfg).save_pretrained(
+        d, safe_serialization=True)
+    out = str(tmp_path / "out")
+    train_main([
+        "--model_name_or_path", d, "--output_dir", out,
+        "--max_steps", "2", "--per_device_train_batch_size", "2",
+        "--synthetic_examples", "8", "--block_size", "32",
+        "--lora_rank", "4", "--logging_steps", "0"])
+    ckpt = os.path.join(out, "checkpoint")
+    model = build_model(d, torch.device("cpu"), adapter_dir=ckpt)
+    assert model.layers[0].self_attn.q_bias is not None
+    eng = InferenceEngine(model, template="chatml",
+                          device=torch.device("cpu"))
+    assert isinstance(eng.chat([{"role": "user", "content": "hi"}],
+                               max_tokens=4), str)
+
+
+def test_engine_rejects_overlong_prompt():
+    from datatunerx_amd.models import LlamaConfig
+    from datatunerx_amd.serve.engine import InferenceEngine
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                         dtype=torch.float32)
+    eng = InferenceEngine(m, template="vanilla",
+                          device=torch.device("cpu"))
+    with pytest.raises(ValueError, match="exceeds the model context"):
+        eng.generate(list(range(1, 300)), max_new_tokens=2)
