@@ -22,7 +22,12 @@ convention (ops/reference.py:51-61).
 Qwen2-family checkpoints load through the same path: `architectures`
 containing "Qwen2" sets attention_bias, and
   model.layers.N.self_attn.{q,k,v}_proj.bias -> layers.N.self_attn.{q,k,v}_bias
-(Mistral loads as plain Llama geometry.)
+Qwen3-family sets qk_norm (per-head q/k RMSNorm before RoPE, decoupled
+head_dim), with
+  model.layers.N.self_attn.{q,k}_norm.weight -> layers.N.self_attn.{q,k}_norm
+(Mistral loads as plain Llama geometry.) save_hf_model writes the
+matching architectures/model_type back, so exports reload in
+transformers as the right class.
 """
 
 from __future__ import annotations
@@ -45,9 +50,10 @@ def load_hf_config(model_dir: str, **lora_kw) -> LlamaConfig:
         hc = json.load(f)
     archs = hc.get("architectures") or ["LlamaForCausalLM"]
     if not any("Llama" in a or "Mistral" in a or "Qwen2" in a
-               for a in archs):
+               or "Qwen3" in a for a in archs):
         raise ValueError(f"unsupported architecture {archs} in {model_dir}")
     qwen2 = any("Qwen2" in a for a in archs)
+    qwen3 = any("Qwen3" in a for a in archs)
     kw = dict(
         vocab_size=hc["vocab_size"],
         hidden_size=hc["hidden_size"],
@@ -62,6 +68,8 @@ def load_hf_config(model_dir: str, **lora_kw) -> LlamaConfig:
         # Qwen2 hardcodes q/k/v bias (no config field in older
         # transformers); Llama-family exposes attention_bias explicitly
         attention_bias=bool(hc.get("attention_bias", qwen2)),
+        # Qwen3: per-head q/k RMSNorm before RoPE
+        qk_norm=qwen3,
     )
     hd = hc.get("head_dim")
     if hd and hd * kw["num_attention_heads"] != kw["hidden_size"]:
@@ -115,6 +123,8 @@ def _map_name(hf_name: str):
         suf = f"self_attn.{p}_proj.bias"
         if n.endswith(suf):              # Qwen2 qkv bias -> {p}_bias
             return n[: -len(suf)] + f"self_attn.{p}_bias"
+    if n.endswith(("self_attn.q_norm.weight", "self_attn.k_norm.weight")):
+        return n[: -len(".weight")]      # Qwen3 qk-norm
     return n
 
 
@@ -241,10 +251,16 @@ def save_hf_model(model, out_dir: str):
     cfg = model.cfg
     head_dim = getattr(cfg, "head_dim", None) or (
         cfg.hidden_size // cfg.num_attention_heads)
+    if getattr(cfg, "qk_norm", False):
+        arch, mtype = "Qwen3ForCausalLM", "qwen3"
+    elif getattr(cfg, "attention_bias", False):
+        arch, mtype = "Qwen2ForCausalLM", "qwen2"
+    else:
+        arch, mtype = "LlamaForCausalLM", "llama"
     with open(os.path.join(out_dir, "config.json"), "w") as f:
         json.dump({
-            "architectures": ["LlamaForCausalLM"],
-            "model_type": "llama",
+            "architectures": [arch],
+            "model_type": mtype,
             "vocab_size": cfg.vocab_size,
             "hidden_size": cfg.hidden_size,
             "intermediate_size": cfg.intermediate_size,
@@ -269,6 +285,8 @@ def save_hf_model(model, out_dir: str):
                             "self_attn.v_bias")):
             # inverse of _map_name's Qwen2 bias rule
             hf = "model." + name[:-len("_bias")] + "_proj.bias"
+        elif name.endswith(("self_attn.q_norm", "self_attn.k_norm")):
+            hf = "model." + name + ".weight"
         elif name == "lm_head.weight":
             hf = "lm_head.weight"
         elif name.endswith(("input_layernorm", "post_attention_layernorm")):

@@ -36,6 +36,9 @@ class LlamaConfig:
     rope_theta: float = 10000.0
     # Qwen2-family: bias on q/k/v projections (o/mlp stay bias-free)
     attention_bias: bool = False
+    # Qwen3-family: per-head RMSNorm on q and k (weight [head_dim],
+    # applied BEFORE RoPE)
+    qk_norm: bool = False
     # LoRA (reference defaults: parser.py:138-149, finetune_controller.go:482)
     lora_r: int = 8
     lora_alpha: float = 32.0
@@ -127,6 +130,16 @@ class LlamaAttention(nn.Module):
                                        requires_grad=False)
         else:
             self.q_bias = self.k_bias = self.v_bias = None
+        if cfg.qk_norm:
+            # Qwen3 qk-norm: same fused rmsnorm kernel, rows = every
+            # head vector ([B*S*H, D] through the last-dim-generic
+            # launcher)
+            self.q_norm = nn.Parameter(torch.ones(D, dtype=dtype),
+                                       requires_grad=False)
+            self.k_norm = nn.Parameter(torch.ones(D, dtype=dtype),
+                                       requires_grad=False)
+        else:
+            self.q_norm = self.k_norm = None
 
     def _fused_qkv(self) -> bool:
         """One autograd node for q/k/v (backward dgrads accumulate via
@@ -169,6 +182,9 @@ class LlamaAttention(nn.Module):
         q = q.view(B, S, H, D)
         k = k.view(B, S, Hkv, D)
         v = v.view(B, S, Hkv, D)
+        if self.q_norm is not None:
+            q = rmsnorm(q.contiguous(), self.q_norm, cfg.rms_norm_eps)
+            k = rmsnorm(k.contiguous(), self.k_norm, cfg.rms_norm_eps)
         q = rope(q, cos, sin, pos0, pos_dev)
         k = rope(k, cos, sin, pos0, pos_dev)
         len_dev = None

@@ -21,3 +21,15 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip)
+
+
+def free_port() -> int:
+    """OS-assigned free TCP port for torch.distributed rendezvous.
+    Fixed port numbers collide with TIME_WAIT sockets or straggler
+    workers from a previous suite run and fail the store bind — seen
+    once as a transient test_ddp_gloo failure."""
+    import socket
+    with socket.socket() as s:
+        s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]

@@ -8,6 +8,8 @@ import os
 import subprocess
 import sys
 
+from conftest import free_port
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 BENCH = os.path.join(ROOT, "bench.py")
 
@@ -44,7 +46,7 @@ def test_bench_torchrun_two_ranks(tmp_path):
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29771", BENCH, "--gpus", "2", "--steps", "2",
+         "--master-port", str(free_port()), BENCH, "--gpus", "2", "--steps", "2",
          "--warmup", "1"],
         cwd=ROOT, capture_output=True, text=True, timeout=600, env=env)
     assert out.returncode == 0, out.stderr[-2000:]

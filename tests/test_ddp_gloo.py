@@ -12,6 +12,8 @@ import tempfile
 import pytest
 import torch
 
+from conftest import free_port
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 WORKER = r"""
@@ -71,23 +73,23 @@ def run_workers(nproc, out_dir, port, extra_env=None):
 
 def test_ddp_two_ranks_converge_identically(tmp_path):
     """Both ranks must hold identical trained params after sync steps."""
-    outs = run_workers(2, str(tmp_path), 29712)
+    outs = run_workers(2, str(tmp_path), free_port())
     assert outs[0]["params"] == pytest.approx(outs[1]["params"], abs=1e-7)
     assert outs[0]["param_sum"] == pytest.approx(outs[1]["param_sum"],
                                                  rel=1e-6)
 
 
-@pytest.mark.parametrize("mode,port", [("overlap", 29716), ("zero1", 29718)])
-def test_ddp_modes_match_flat(tmp_path, mode, port):
+@pytest.mark.parametrize("mode", ["overlap", "zero1"])
+def test_ddp_modes_match_flat(tmp_path, mode):
     """Overlapped bucketed all-reduce and ZeRO-1 sharding must produce the
     same trained parameters as the flat fused all-reduce (grad-accum 2
     exercises the final-microbatch hook path)."""
     flat_dir = tmp_path / "flat"
     mode_dir = tmp_path / mode
     flat_dir.mkdir(), mode_dir.mkdir()
-    base = run_workers(2, str(flat_dir), port,
+    base = run_workers(2, str(flat_dir), free_port(),
                        {"DTX_OPT_MODE": "flat", "DTX_ACC": "2"})
-    outs = run_workers(2, str(mode_dir), port + 40,
+    outs = run_workers(2, str(mode_dir), free_port(),
                        {"DTX_OPT_MODE": mode, "DTX_ACC": "2"})
     assert outs[0]["params"] == pytest.approx(outs[1]["params"], abs=1e-7)
     assert outs[0]["params"] == pytest.approx(base[0]["params"], abs=1e-5)
@@ -121,7 +123,8 @@ def test_allreduce_flat_mean():
 
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=worker, args=(r, 2, 29714, q))
+    port = free_port()
+    ps = [ctx.Process(target=worker, args=(r, 2, port, q))
           for r in range(2)]
     for p in ps:
         p.start()
@@ -209,21 +212,21 @@ def _run_resume(nproc, out_dir, port, mode, phase):
         assert p.wait(timeout=300) == 0
 
 
-@pytest.mark.parametrize("mode,port", [("flat", 29722), ("zero1", 29726)])
-def test_save_restart_resume_matches_uninterrupted(tmp_path, mode, port):
+@pytest.mark.parametrize("mode", ["flat", "zero1"])
+def test_save_restart_resume_matches_uninterrupted(tmp_path, mode):
     """4 straight steps == 2 steps -> checkpoint -> restart -> 2 more, for
     both the replicated and the ZeRO-1-sharded optimizer (VERDICT r1
     weak #2: every rank saves its shard; resume all-gathers the restored
     params before the first forward)."""
     out = str(tmp_path)
-    _run_resume(2, out, port, mode, "full")
-    _run_resume(2, out, port + 1, mode, "save")
+    _run_resume(2, out, free_port(), mode, "full")
+    _run_resume(2, out, free_port(), mode, "save")
     assert os.path.exists(os.path.join(out, "shared/checkpoint-2",
                                        "trainer_state.pt"))
     if mode == "zero1":
         assert os.path.exists(os.path.join(
             out, "shared/checkpoint-2", "trainer_state_rank1.pt"))
-    _run_resume(2, out, port + 2, mode, "resume")
+    _run_resume(2, out, free_port(), mode, "resume")
     for rank in range(2):
         full = json.load(open(os.path.join(out, f"full_rank{rank}.json")))
         res = json.load(open(os.path.join(out, f"resume_rank{rank}.json")))
@@ -256,7 +259,7 @@ def test_rccl_check_script_two_ranks(tmp_path):
 def test_ddp_four_ranks_zero1(tmp_path):
     """world 4, zero1 sharded optimizer: all ranks converge to identical
     params (sharding math exercised at a non-trivial world size)."""
-    outs = run_workers(4, str(tmp_path), 29734,
+    outs = run_workers(4, str(tmp_path), free_port(),
                        extra_env={"DTX_OPT_MODE": "zero1"})
     for r in (1, 2, 3):
         assert outs[0]["params"] == pytest.approx(outs[r]["params"],

@@ -291,3 +291,59 @@ def test_engine_rejects_overlong_prompt():
                           device=torch.device("cpu"))
     with pytest.raises(ValueError, match="exceeds the model context"):
         eng.generate(list(range(1, 300)), max_new_tokens=2)
+
+
+def test_qwen3_checkpoint_matches_transformers_logits(tmp_path):
+    """Qwen3-family (per-head q/k RMSNorm before RoPE, decoupled
+    head_dim, no bias) matches transformers' fp32 logits."""
+    d = str(tmp_path / "qwen3")
+    cfg = transformers.Qwen3Config(
+        vocab_size=256, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=32,     # 4*32 != 64: decoupled
+        max_position_embeddings=128, rms_norm_eps=1e-5,
+        rope_theta=10000.0, tie_word_embeddings=False,
+        attn_implementation="eager")
+    torch.manual_seed(13)
+    hf = transformers.Qwen3ForCausalLM(cfg)
+    with torch.no_grad():                       # nontrivial norm weights
+        for lyr in hf.model.layers:
+            lyr.self_attn.q_norm.weight.uniform_(0.5, 1.5)
+            lyr.self_attn.k_norm.weight.uniform_(0.5, 1.5)
+    hf.save_pretrained(d, safe_serialization=True)
+
+    mycfg = load_hf_config(d)
+    assert mycfg.qk_norm and not mycfg.attention_bias
+    assert mycfg.head_dim == 32
+    model = LlamaForCausalLM(mycfg, lora=False, dtype=torch.float32)
+    load_hf_weights(model, d)
+    ids = torch.randint(0, 256, (2, 19))
+    with torch.no_grad():
+        ref = hf(ids).logits
+        got = model(ids)
+    assert torch.allclose(ref, got, atol=2e-4, rtol=1e-3), \
+        (ref - got).abs().max()
+
+
+def test_qwen3_export_reloads_in_transformers(tmp_path):
+    from datatunerx_amd.models import LlamaConfig
+    from datatunerx_amd.models.hf_io import save_hf_model
+    cfg = LlamaConfig.tiny(qk_norm=True)
+    torch.manual_seed(6)
+    m = LlamaForCausalLM(cfg, lora=False, dtype=torch.float32)
+    m.init_random(seed=2)
+    with torch.no_grad():
+        for lyr in m.layers:
+            lyr.self_attn.q_norm.uniform_(0.5, 1.5)
+            lyr.self_attn.k_norm.uniform_(0.5, 1.5)
+    d = str(tmp_path / "exp3")
+    save_hf_model(m, d)
+    with open(os.path.join(d, "config.json")) as f:
+        assert json.load(f)["model_type"] == "qwen3"
+    ref = transformers.AutoModelForCausalLM.from_pretrained(
+        d, torch_dtype=torch.float32, attn_implementation="eager")
+    ref.eval()
+    ids = torch.randint(0, cfg.vocab_size, (1, 21))
+    with torch.no_grad():
+        assert torch.allclose(m(ids), ref(ids).logits, atol=2e-4,
+                              rtol=1e-3)

@@ -11,6 +11,8 @@ import sys
 import pytest
 import torch
 
+from conftest import free_port
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 WORKER = r"""
@@ -47,12 +49,13 @@ dist.destroy_process_group()
 def test_tp_matches_single(tmp_path):
     script = tmp_path / "worker.py"
     script.write_text(WORKER)
+    _port = str(free_port())
     procs = []
     for rank in range(2):
         env = dict(os.environ)
         env.update({"RANK": str(rank), "WORLD_SIZE": "2",
                     "LOCAL_RANK": str(rank),
-                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29762",
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": _port,
                     "DTX_ROOT": ROOT, "DTX_OUT": str(tmp_path)})
         procs.append(subprocess.Popen([sys.executable, str(script)],
                                       env=env))
@@ -69,12 +72,13 @@ def test_tp_server_end_to_end(tmp_path):
     import urllib.request
 
     port = 18973
+    _port = str(free_port())
     procs = []
     for rank in range(2):
         env = dict(os.environ)
         env.update({"RANK": str(rank), "WORLD_SIZE": "2",
                     "LOCAL_RANK": str(rank),
-                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29764",
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": _port,
                     "PYTHONPATH": ROOT})
         procs.append(subprocess.Popen(
             [sys.executable, "-m", "datatunerx_amd.serve.server",
@@ -200,12 +204,13 @@ dist.destroy_process_group()
 """
     script = tmp_path / "w2.py"
     script.write_text(worker)
+    _port = str(free_port())
     procs = []
     for rank in range(2):
         env = dict(os.environ)
         env.update({"RANK": str(rank), "WORLD_SIZE": "2",
                     "LOCAL_RANK": str(rank),
-                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29781",
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": _port,
                     "DTX_ROOT": ROOT, "DTX_OUT": str(tmp_path)})
         procs.append(subprocess.Popen([sys.executable, str(script)],
                                       env=env))
@@ -321,11 +326,12 @@ def test_tp_real_weights_match_single_process(tmp_path):
     script = str(tmp_path / "w.py")
     with open(script, "w") as f:
         f.write(HF_TP_WORKER)
+    _port = str(free_port())
     procs = []
     for rank in range(2):
         env = dict(os.environ)
         env.update({"RANK": str(rank), "WORLD_SIZE": "2",
-                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29741",
+                    "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": _port,
                     "DTX_ROOT": ROOT, "DTX_OUT": str(tmp_path),
                     "DTX_HF_DIR": d})
         procs.append(subprocess.Popen([sys.executable, script], env=env))
