@@ -473,12 +473,17 @@ class SFTTrainer:
     # ------------------------------------------------------------ predict
     @torch.no_grad()
     def predict(self, dataset, max_new_tokens: int = 64,
-                out_file: Optional[str] = None):
+                out_file: Optional[str] = None,
+                eos_token_id: Optional[int] = 2):
         """Greedy generation from each example's prompt (the -100-masked
         prefix), mirroring the reference's prediction_step +
-        save_predictions jsonl (cmd/tuning/trainer.py:405-489)."""
+        save_predictions jsonl (cmd/tuning/trainer.py:405-489).
+        KV-cache decode (prefill once, one token per forward) and stops
+        at eos; eos_token_id=None disables early stop."""
         from ..data.dataset import IGNORE_INDEX
+        from ..serve.engine import KVCache
         self.model.eval()
+        is_llama = hasattr(self.model.cfg, "num_key_value_heads")
         results = []
         for i in range(len(dataset)):
             ex = dataset[i]
@@ -493,13 +498,39 @@ class SFTTrainer:
             cur = torch.tensor([prompt], dtype=torch.long,
                                device=self.device)
             out = []
-            for _ in range(max_new_tokens):
-                logits = self.model(cur)
-                nxt = int(logits[0, -1].argmax())
-                out.append(nxt)
-                cur = torch.cat(
-                    [cur, torch.tensor([[nxt]], dtype=torch.long,
-                                       device=self.device)], dim=1)
+            with torch.no_grad():
+                if is_llama:
+                    cfg = self.model.cfg
+                    max_s = min(cfg.max_position_embeddings,
+                                len(prompt) + max_new_tokens + 1)
+                    caches = [KVCache(
+                        1, cfg.num_key_value_heads, max_s, cfg.head_dim,
+                        self.device, next(self.model.parameters()).dtype)
+                        for _ in range(cfg.num_hidden_layers)]
+                    pos = 0
+                    for _ in range(max_new_tokens):
+                        if pos + cur.shape[1] > max_s:
+                            break
+                        logits = self.model(cur, pos0=pos,
+                                            kv_caches=caches)
+                        pos += cur.shape[1]
+                        nxt = int(logits[0, -1].argmax())
+                        if nxt == eos_token_id:
+                            break
+                        out.append(nxt)
+                        cur = torch.tensor([[nxt]], dtype=torch.long,
+                                           device=self.device)
+                else:
+                    for _ in range(max_new_tokens):
+                        logits = self.model(cur)
+                        nxt = int(logits[0, -1].argmax())
+                        if nxt == eos_token_id:
+                            break
+                        out.append(nxt)
+                        cur = torch.cat(
+                            [cur, torch.tensor(
+                                [[nxt]], dtype=torch.long,
+                                device=self.device)], dim=1)
             results.append({
                 "prompt_ids": prompt,
                 "predict_ids": out,

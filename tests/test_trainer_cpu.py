@@ -203,9 +203,17 @@ def test_predict_writes_jsonl(tmp_path):
     tr = _tiny_trainer(tmp_path, steps=1)
     tr.train()
     ds = SFTDataset.synthetic(3, 24, 512, seed=1)
-    res = tr.predict(ds, max_new_tokens=4)
+    res = tr.predict(ds, max_new_tokens=4, eos_token_id=None)
     assert len(res) == 3
     assert all(len(r["predict_ids"]) == 4 for r in res)
+    # with eos stopping enabled, output never contains eos and is <= 4
+    res2 = tr.predict(ds, max_new_tokens=4)
+    assert all(2 not in r["predict_ids"] and len(r["predict_ids"]) <= 4
+               for r in res2)
+    # eos-stopped output is a prefix of the unstopped greedy output
+    # (same KV-cache decode, so tokens must agree until the stop)
+    for r, q in zip(res, res2):
+        assert r["predict_ids"][:len(q["predict_ids"])] == q["predict_ids"]
     out = os.path.join(str(tmp_path), "generated_predictions.jsonl")
     rows = [json.loads(l) for l in open(out)]
     assert len(rows) == 3 and "prompt_ids" in rows[0]
