@@ -347,3 +347,32 @@ def test_qwen3_export_reloads_in_transformers(tmp_path):
     with torch.no_grad():
         assert torch.allclose(m(ids), ref(ids).logits, atol=2e-4,
                               rtol=1e-3)
+
+
+def test_sentencepiece_tokenizer_backend(tmp_path):
+    """tokenizer.model (SentencePiece) backend: same interface as the
+    tokenizers backend — encode/decode roundtrip, special ids, and the
+    trainer/engine-facing add_special_tokens contract."""
+    spm = pytest.importorskip("sentencepiece")
+    d = str(tmp_path / "spdir")
+    os.makedirs(d)
+    corpus = os.path.join(d, "c.txt")
+    with open(corpus, "w") as f:
+        f.write("the quick brown fox\nllama fine tune\nhello world\n" * 40)
+    spm.SentencePieceTrainer.train(
+        input=corpus, model_prefix=os.path.join(d, "tokenizer"),
+        vocab_size=80, model_type="bpe", minloglevel=2)
+    tok = load_tokenizer(d)
+    assert isinstance(tok, HFTokenizer) and tok.kind == "sp"
+    assert tok.bos_token_id == 1 and tok.eos_token_id == 2
+    ids = tok.encode("quick fox", add_special_tokens=True)
+    assert ids[0] == tok.bos_token_id and len(ids) > 1
+    text = tok.decode(ids)
+    assert "quick" in text and "fox" in text
+    assert tok.vocab_size == 80
+    # dataset pipeline consumes it like any tokenizer
+    from datatunerx_amd.data.dataset import SFTDataset
+    ds = SFTDataset.from_rows(
+        [{"instruction": "quick fox", "response": "hello world"}],
+        tok, template_name="default", cutoff_len=64)
+    assert len(ds) == 1 and any(l != -100 for l in ds[0]["labels"])
