@@ -647,3 +647,28 @@ def test_builtin_model_registry_covers_all_serve_names():
     # GQA geometry reaches the engine path for llama3
     _, c3 = builtin_config("llama3-8b", lora_r=4)
     assert c3.num_key_value_heads == 8 and c3.lora_r == 4
+
+
+def test_sampler_temperature_and_top_p():
+    """Sampler unit behavior: temperature=0 is argmax; top_p keeps only
+    the smallest prefix of the sorted distribution; temperature>0 draws
+    only from kept tokens."""
+    import torch
+
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                         dtype=torch.float32)
+    eng = InferenceEngine(m, template="vanilla",
+                          device=torch.device("cpu"))
+    logits = torch.tensor([0.1, 3.0, 2.0, -1.0, 0.5])
+    assert eng._sample(logits, 0.0, 1.0) == 1
+    torch.manual_seed(0)
+    # top_p=0.5: after softmax(T=0.5) token 1 dominates; only it is kept
+    for _ in range(20):
+        assert eng._sample(logits, 0.5, 0.5) == 1
+    # temperature high, top_p=0.9: draws restricted to the top mass
+    torch.manual_seed(1)
+    seen = {eng._sample(logits, 1.5, 0.9) for _ in range(200)}
+    assert 3 not in seen          # lowest-probability token filtered out
+    assert 1 in seen and len(seen) >= 2
