@@ -168,7 +168,26 @@ register_template("chatglm2", Template(
 register_template("chatglm3", Template(
     prefix=["<|system|>\n{{system}}"],
     prompt=["<|user|>\n{{query}}<|assistant|>\n"]))
-register_template("llama3", Template(
+@dataclass
+class Llama3Template(Template):
+    """Llama-3 header convention; the system block is omitted entirely
+    when no system prompt is given (matching the reference chat
+    format)."""
+
+    def encode_multiturn(self, tokenizer, query, resp, history=None,
+                         system: str = ""):
+        if not system:
+            # non-mutating copy: templates are shared module singletons
+            # and the serving front encodes from multiple threads
+            import dataclasses
+            return Template.encode_multiturn(
+                dataclasses.replace(self, prefix=[]), tokenizer, query,
+                resp, history, system)
+        return super().encode_multiturn(tokenizer, query, resp, history,
+                                        system)
+
+
+register_template("llama3", Llama3Template(
     prefix=["<|start_header_id|>system<|end_header_id|>\n\n"
             "{{system}}<|eot_id|>"],
     prompt=["<|start_header_id|>user<|end_header_id|>\n\n{{query}}<|eot_id|>"

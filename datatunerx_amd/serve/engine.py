@@ -183,9 +183,9 @@ def builtin_config(name: str, **lora_kw):
     if name == "llama-mini":
         return "llama", LlamaConfig.mini(**lora_kw)
     if name in ("gpt2-small", "gpt2"):
-        return "gpt2", GPT2Config.small()
+        return "gpt2", GPT2Config.small(**lora_kw)
     if name == "gpt2-tiny":
-        return "gpt2", GPT2Config.tiny()
+        return "gpt2", GPT2Config.tiny(**lora_kw)
     raise ValueError(f"unknown model {name!r}")
 
 

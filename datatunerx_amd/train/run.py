@@ -137,12 +137,18 @@ def main(argv=None):
                                       mask_frac=0.0 if pt else 0.25)
         eval_ds = None
         if dargs.eval_dataset_path and os.path.exists(dargs.eval_dataset_path):
-            eval_ds = SFTDataset.from_csv(
-                dargs.eval_dataset_path, tok,
-                column_map={"instruction": dargs.instruction_column,
-                            "response": dargs.response_column},
-                template_name=dargs.prompt_template,
-                cutoff_len=dargs.block_size)
+            from ..data.dataset import read_csv_rows
+            cmap = {"instruction": dargs.instruction_column,
+                    "response": dargs.response_column}
+            if pt:               # eval perplexity on packed blocks too
+                eval_ds = SFTDataset.from_rows_pt(
+                    read_csv_rows(dargs.eval_dataset_path, cmap), tok,
+                    cutoff_len=dargs.block_size)
+            else:
+                eval_ds = SFTDataset.from_csv(
+                    dargs.eval_dataset_path, tok, column_map=cmap,
+                    template_name=dargs.prompt_template,
+                    cutoff_len=dargs.block_size)
 
         tcfg = TrainerConfig(
             output_dir=fargs.output_dir,

@@ -155,3 +155,19 @@ def test_run_stage_pt_and_rejects_dpo(tmp_path):
         train_main(["--model_name_or_path", "llama-tiny", "--stage",
                     "dpo", "--output_dir", str(tmp_path / "o2"),
                     "--max_steps", "1"])
+
+
+def test_llama3_template_omits_empty_system():
+    from datatunerx_amd.data.dataset import ByteTokenizer
+    from datatunerx_amd.data.templates import get_template
+    t = get_template("llama3")
+    tok = ByteTokenizer()
+    src, _ = t.encode_oneturn(tok, "hi", "yo", None, "")
+    text = bytes(i - 3 for i in src if i >= 3).decode()
+    assert "system" not in text
+    assert text.startswith("<|start_header_id|>user")
+    # with a system prompt the block is present (and shared state was
+    # not mutated by the empty-system call)
+    src2, _ = t.encode_oneturn(tok, "hi", "yo", None, "sys")
+    text2 = bytes(i - 3 for i in src2 if i >= 3).decode()
+    assert text2.startswith("<|start_header_id|>system")
