@@ -189,11 +189,12 @@ class ShardedLoader:
 
     def __init__(self, dataset, batch_size: int, rank: int = 0,
                  world_size: int = 1, seed: int = 0, pad_token_id: int = 0,
-                 device=None, drop_last: bool = True):
+                 device=None, drop_last: bool = True, collate_fn=None):
         self.ds, self.bs = dataset, batch_size
         self.rank, self.world = rank, world_size
         self.seed, self.pad, self.device = seed, pad_token_id, device
         self.drop_last = drop_last
+        self.collate_fn = collate_fn or collate
 
     def epoch(self, epoch_idx: int) -> Iterator[dict]:
         g = torch.Generator().manual_seed(self.seed + epoch_idx)
@@ -202,8 +203,8 @@ class ShardedLoader:
         n_full = len(shard) // self.bs
         for i in range(n_full):
             idx = shard[i * self.bs:(i + 1) * self.bs]
-            yield collate([self.ds[j] for j in idx], self.pad,
-                          device=self.device)
+            yield self.collate_fn([self.ds[j] for j in idx], self.pad,
+                                  device=self.device)
 
     def __iter__(self):
         e = 0

@@ -334,5 +334,19 @@ class LlamaForCausalLM(nn.Module):
         logits = self.lm_head(h)
         return cross_entropy(logits, targets, ignore_index=-100)
 
+    def sequence_logprobs(self, input_ids, labels):
+        """[B] sum of log p(target token) over each row's non-ignored
+        (shifted) labels — the DPO building block. Gradients flow to
+        trainable params; runs the same masked xent kernels as the
+        loss path, just unreduced."""
+        from ..ops.autograd import per_row_cross_entropy
+        B = input_ids.shape[0]
+        h = self.hidden_states(input_ids)
+        h = h[:, :-1, :].reshape(-1, self.cfg.hidden_size)
+        targets = labels[:, 1:].reshape(-1)
+        logits = self.lm_head(h)
+        nll = per_row_cross_entropy(logits, targets, ignore_index=-100)
+        return -nll.view(B, -1).sum(dim=1)
+
     def trainable_parameters(self):
         return [(n, p) for n, p in self.named_parameters() if p.requires_grad]

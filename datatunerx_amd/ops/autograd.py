@@ -149,6 +149,31 @@ def cross_entropy(logits, targets, ignore_index: int = -100):
     return CrossEntropy.apply(logits, targets, ignore_index)
 
 
+class PerRowCrossEntropy(torch.autograd.Function):
+    """UNREDUCED CE: returns the per-row NLL vector [N] (0 on ignored
+    rows) through the same fused xent kernels — the building block for
+    per-sequence log-probabilities (DPO)."""
+
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        loss, lse = softmax_xent_fwd(logits, targets, ignore_index)
+        ctx.save_for_backward(logits, targets, lse)
+        ctx.ignore_index = ignore_index
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, lse = ctx.saved_tensors
+        dlogits = softmax_xent_bwd(logits, targets, lse,
+                                   dloss.float().contiguous(),
+                                   ctx.ignore_index)
+        return dlogits, None, None
+
+
+def per_row_cross_entropy(logits, targets, ignore_index: int = -100):
+    return PerRowCrossEntropy.apply(logits, targets, ignore_index)
+
+
 def _base_gemm(x2, w, wt):
     """Base-projection GEMM: the hand-written MFMA kernel (gemm.hip) when
     a cached W^T is provided (both fwd and dgrad then run the fast NT

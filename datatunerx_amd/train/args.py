@@ -21,10 +21,12 @@ class ModelArguments:
 @dataclass
 class FinetuningArguments:
     stage: str = "sft"        # sft (prompt-masked) | pt (packed blocks,
-                              # all positions trained); rm/ppo/dpo are
-                              # rejected with a clear error (the
-                              # reference declares but never runs them
-                              # either — parser.py:112-221)
+                              # all positions trained) | dpo (preference
+                              # pairs, adapters-off reference policy);
+                              # rm/ppo are rejected with a clear error
+                              # (the reference declares but never runs
+                              # any of these — parser.py:112-221)
+    dpo_beta: float = 0.1
     finetuning_type: str = "lora"               # lora / full
     lora_rank: int = 8
     lora_alpha: float = 32.0
@@ -65,6 +67,8 @@ class DataArguments:
     eval_dataset_path: Optional[str] = None
     instruction_column: str = "instruction"
     response_column: str = "response"
+    chosen_column: str = "chosen"               # stage=dpo
+    rejected_column: str = "rejected"           # stage=dpo
     prompt_template: str = "llama2"
     block_size: int = 1024                      # cutoff_len (train.py:49-51)
     synthetic_examples: int = 0                 # >0: synthetic data

@@ -112,11 +112,26 @@ def main(argv=None):
 
         tok = load_tokenizer(name)
         vocab = getattr(model.cfg, "vocab_size")
-        if fargs.stage not in ("sft", "pt"):
+        if fargs.stage not in ("sft", "pt", "dpo"):
             raise SystemExit(
-                f"stage {fargs.stage!r} not supported (sft, pt)")
+                f"stage {fargs.stage!r} not supported (sft, pt, dpo)")
         pt = fargs.stage == "pt"
-        if dargs.dataset_path and os.path.exists(dargs.dataset_path):
+        dpo = fargs.stage == "dpo"
+        if dpo:
+            from ..data.preference import PreferenceDataset
+            if dargs.dataset_path and os.path.exists(dargs.dataset_path):
+                ds = PreferenceDataset.from_csv(
+                    dargs.dataset_path, tok,
+                    column_map={"instruction": dargs.instruction_column,
+                                "chosen": dargs.chosen_column,
+                                "rejected": dargs.rejected_column},
+                    template_name=dargs.prompt_template,
+                    cutoff_len=dargs.block_size)
+            else:
+                n = dargs.synthetic_examples or 256
+                ds = PreferenceDataset.synthetic(n, dargs.block_size,
+                                                 vocab, seed=fargs.seed)
+        elif dargs.dataset_path and os.path.exists(dargs.dataset_path):
             from ..data.dataset import read_csv_rows
             cmap = {"instruction": dargs.instruction_column,
                     "response": dargs.response_column}
@@ -136,7 +151,18 @@ def main(argv=None):
                                       seed=fargs.seed,
                                       mask_frac=0.0 if pt else 0.25)
         eval_ds = None
-        if dargs.eval_dataset_path and os.path.exists(dargs.eval_dataset_path):
+        if dpo and dargs.eval_dataset_path and \
+                os.path.exists(dargs.eval_dataset_path):
+            from ..data.preference import PreferenceDataset
+            eval_ds = PreferenceDataset.from_csv(
+                dargs.eval_dataset_path, tok,
+                column_map={"instruction": dargs.instruction_column,
+                            "chosen": dargs.chosen_column,
+                            "rejected": dargs.rejected_column},
+                template_name=dargs.prompt_template,
+                cutoff_len=dargs.block_size)
+        elif dargs.eval_dataset_path and \
+                os.path.exists(dargs.eval_dataset_path):
             from ..data.dataset import read_csv_rows
             cmap = {"instruction": dargs.instruction_column,
                     "response": dargs.response_column}
@@ -168,9 +194,18 @@ def main(argv=None):
             uid=fargs.uid, lora_r=fargs.lora_rank,
             lora_alpha=fargs.lora_alpha, lora_dropout=fargs.lora_dropout,
             lora_targets=fargs.lora_targets, base_model=name)
-        trainer = SFTTrainer(model, ds, tcfg, eval_dataset=eval_ds,
-                             device=device, rank=rank, world_size=world,
-                             pad_token_id=tok.pad_token_id)
+        if dpo:
+            from .trainer import DPOTrainer
+            trainer = DPOTrainer(model, ds, tcfg, eval_dataset=eval_ds,
+                                 device=device, rank=rank,
+                                 world_size=world,
+                                 pad_token_id=tok.pad_token_id,
+                                 beta=fargs.dpo_beta)
+        else:
+            trainer = SFTTrainer(model, ds, tcfg, eval_dataset=eval_ds,
+                                 device=device, rank=rank,
+                                 world_size=world,
+                                 pad_token_id=tok.pad_token_id)
         final_loss = trainer.train()
         ckpt_path = None
         if is_main():

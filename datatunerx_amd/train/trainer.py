@@ -383,6 +383,10 @@ class SFTTrainer:
                 self._exporter.export_eval_metrics(record)
 
     # ------------------------------------------------------------- train
+    def _micro_loss(self, mb):
+        """One micro-batch's loss tensor (overridden by DPOTrainer)."""
+        return self.model(mb["input_ids"], labels=mb["labels"])
+
     def train_step(self, micro_batches) -> float:
         """One optimizer step over `gradient_accumulation_steps` micro
         batches; returns the (local) mean loss."""
@@ -392,7 +396,7 @@ class SFTTrainer:
         for i, mb in enumerate(micro_batches):
             if i == len(micro_batches) - 1:
                 self.opt.mark_final_microbatch()
-            loss = self.model(mb["input_ids"], labels=mb["labels"])
+            loss = self._micro_loss(mb)
             (loss / acc).backward()
             total += float(loss.detach())
         lr = lr_at(self.global_step, self.total_steps, cfg.learning_rate,
@@ -605,3 +609,124 @@ class SFTTrainer:
                            weights_only=False)
         self.opt.load_state_dict(state["optimizer"])
         self.global_step = state["global_step"]
+
+
+# ------------------------------------------------------------------ DPO
+class _AdaptersDisabled:
+    """Zero every LoRA module's scale so the model computes EXACTLY the
+    frozen base — the DPO reference policy without a second weight copy
+    (the adapters start at B=0, so base == pre-DPO policy). The fused
+    QKVProj path reads the same module `scale` attribute, so both
+    dispatch paths are covered. Not thread-safe across concurrent
+    trainers on one model (the trainer owns its model)."""
+
+    def __init__(self, model):
+        self.mods = [m for m in model.modules()
+                     if isinstance(m, LoRALinearModule)]
+
+    def __enter__(self):
+        self.saved = [m.scale for m in self.mods]
+        for m in self.mods:
+            m.scale = 0.0
+        return self
+
+    def __exit__(self, *exc):
+        for m, s in zip(self.mods, self.saved):
+            m.scale = s
+        return False
+
+
+class DPOTrainer(SFTTrainer):
+    """Direct Preference Optimization over (chosen, rejected) pairs —
+    the reference's declared-but-unimplemented stage=dpo
+    (cmd/tuning/parser.py:112-221) made real. LoRA-only: the reference
+    policy is the model with adapters disabled (exact, free — no second
+    model in memory). Loss per pair:
+        -logsigmoid(beta * ((pol_c - ref_c) - (pol_r - ref_r)))
+    where each term is the sum log-probability of that completion's
+    non-masked tokens (model.sequence_logprobs -> the fused masked
+    xent kernels, unreduced)."""
+
+    def __init__(self, model, train_dataset, cfg: TrainerConfig,
+                 device=None, rank: int = 0, world_size: int = 1,
+                 eval_dataset=None, pad_token_id: int = 0,
+                 beta: float = 0.1):
+        from ..data.preference import collate_preference
+        if not any(isinstance(m, LoRALinearModule)
+                   for m in model.modules()):
+            raise ValueError("DPO requires LoRA adapters (the frozen "
+                             "base is the reference policy)")
+        super().__init__(model, train_dataset, cfg, device=device,
+                         rank=rank, world_size=world_size,
+                         eval_dataset=eval_dataset,
+                         pad_token_id=pad_token_id)
+        self.beta = beta
+        self.train_loader.collate_fn = collate_preference
+        self.last_margin = 0.0
+        self.last_acc = 0.0
+
+    def _micro_loss(self, mb):
+        ids = torch.cat([mb["chosen_ids"], mb["rejected_ids"]], dim=0) \
+            if mb["chosen_ids"].shape[1] == mb["rejected_ids"].shape[1] \
+            else None
+        with torch.no_grad(), _AdaptersDisabled(self.model):
+            if ids is not None:
+                labels = torch.cat([mb["chosen_labels"],
+                                    mb["rejected_labels"]], dim=0)
+                ref = self.model.sequence_logprobs(ids, labels)
+                B = mb["chosen_ids"].shape[0]
+                ref_c, ref_r = ref[:B], ref[B:]
+            else:              # chosen/rejected padded to different S
+                ref_c = self.model.sequence_logprobs(
+                    mb["chosen_ids"], mb["chosen_labels"])
+                ref_r = self.model.sequence_logprobs(
+                    mb["rejected_ids"], mb["rejected_labels"])
+        if ids is not None:
+            labels = torch.cat([mb["chosen_labels"],
+                                mb["rejected_labels"]], dim=0)
+            pol = self.model.sequence_logprobs(ids, labels)
+            B = mb["chosen_ids"].shape[0]
+            pol_c, pol_r = pol[:B], pol[B:]
+        else:
+            pol_c = self.model.sequence_logprobs(mb["chosen_ids"],
+                                                 mb["chosen_labels"])
+            pol_r = self.model.sequence_logprobs(mb["rejected_ids"],
+                                                 mb["rejected_labels"])
+        margin = (pol_c - ref_c) - (pol_r - ref_r)
+        self.last_margin = float(margin.detach().mean())
+        self.last_acc = float((margin.detach() > 0).float().mean())
+        return -torch.nn.functional.logsigmoid(
+            self.beta * margin.float()).mean()
+
+    def evaluate(self) -> dict:
+        """Preference accuracy + margin on the eval set (the SFT
+        eval_loss contract does not apply to pairs)."""
+        from ..data.preference import collate_preference
+        self.model.eval()
+        loader = ShardedLoader(self.eval_dataset, self.cfg.micro_batch_size,
+                               self.rank, self.world, seed=0,
+                               pad_token_id=self.pad_token_id,
+                               device=self.device,
+                               collate_fn=collate_preference)
+        margins = []
+        with torch.no_grad():
+            for mb in loader.epoch(0):
+                with _AdaptersDisabled(self.model):
+                    rc = self.model.sequence_logprobs(
+                        mb["chosen_ids"], mb["chosen_labels"])
+                    rr = self.model.sequence_logprobs(
+                        mb["rejected_ids"], mb["rejected_labels"])
+                pc = self.model.sequence_logprobs(
+                    mb["chosen_ids"], mb["chosen_labels"])
+                pr = self.model.sequence_logprobs(
+                    mb["rejected_ids"], mb["rejected_labels"])
+                margins.append(((pc - rc) - (pr - rr)))
+        m = torch.cat(margins) if margins else torch.zeros(1)
+        metrics = {"eval_margin": round(float(m.mean()), 6),
+                   "eval_pref_accuracy": round(
+                       float((m > 0).float().mean()), 6),
+                   "current_steps": self.global_step,
+                   "total_steps": self.total_steps}
+        self._log(metrics, kind="eval")
+        self.model.train()
+        return metrics

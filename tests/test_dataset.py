@@ -127,7 +127,7 @@ def test_synthetic_mask_frac_zero():
 
 
 def test_run_stage_pt_and_rejects_dpo(tmp_path):
-    """train.run CLI: stage=pt trains on a packed CSV corpus; stage=dpo
+    """train.run CLI: stage=pt trains on a packed CSV corpus; stage=ppo
     fails with a clear error (declared-but-unimplemented stages must not
     silently run sft)."""
     import csv
@@ -151,9 +151,9 @@ def test_run_stage_pt_and_rejects_dpo(tmp_path):
     assert last["current_steps"] == 2 and last["loss"] > 0
 
     import pytest as _pytest
-    with _pytest.raises(SystemExit, match="dpo"):
+    with _pytest.raises(SystemExit, match="ppo"):
         train_main(["--model_name_or_path", "llama-tiny", "--stage",
-                    "dpo", "--output_dir", str(tmp_path / "o2"),
+                    "ppo", "--output_dir", str(tmp_path / "o2"),
                     "--max_steps", "1"])
 
 
