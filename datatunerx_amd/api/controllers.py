@@ -117,6 +117,8 @@ class FinetuneController:
         def add(flag, val):
             if val is not None:
                 a.extend([flag, str(val)])
+        add("--stage", p.get("stage"))        # sft | pt | dpo
+        add("--dpo_beta", p.get("dpoBeta"))
         add("--lr_scheduler_type", p.get("scheduler"))
         add("--optim", p.get("optimizer"))
         if p.get("int4"):
@@ -158,6 +160,10 @@ class FinetuneController:
             for feat in info.get("features", []):
                 if feat.get("name") == "instruction":
                     add("--instruction_column", feat.get("mapTo"))
+                elif feat.get("name") == "chosen":
+                    add("--chosen_column", feat.get("mapTo"))
+                elif feat.get("name") == "rejected":
+                    add("--rejected_column", feat.get("mapTo"))
                 if feat.get("name") == "response":
                     add("--response_column", feat.get("mapTo"))
         return a

@@ -97,6 +97,9 @@ def validate_(obj: ApiObject) -> None:
                      "spec.parameters.batchSize must be >= 1", errs)
             _require(not (params.get("int4") and params.get("int8")),
                      "int4 and int8 are mutually exclusive", errs)
+            st = params.get("stage")
+            _require(st is None or st in ("sft", "pt", "dpo"),
+                     "spec.parameters.stage must be sft, pt or dpo", errs)
     elif isinstance(obj, Dataset):
         info = ((obj.spec.get("datasetMetadata") or {})
                 .get("datasetInfo") or {})

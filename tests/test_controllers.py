@@ -672,3 +672,40 @@ def test_sampler_temperature_and_top_p():
     seen = {eng._sample(logits, 1.5, 0.9) for _ in range(200)}
     assert 3 not in seen          # lowest-probability token filtered out
     assert 1 in seen and len(seen) >= 2
+
+
+def test_dpo_job_pipeline(tmp_path):
+    """A FinetuneJob whose Hyperparameter requests stage=dpo runs the
+    DPO trainer through the full control plane (train -> checkpoint ->
+    build -> serve -> score) on synthetic preference data."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store, hp_params={"stage": "dpo",
+                                         "dpoBeta": "0.2",
+                                         "loRA_Dropout": "0.0"})
+    job = FinetuneJob(name="dpojob", spec={
+        "fineTune": {"finetuneSpec": finetune_spec()}})
+    mgr.store.create(job)
+    deadline = time.time() + 300
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        cur = mgr.store.get(FinetuneJob, "default", "dpojob")
+        if cur.status.get("state") in ("Successful", "Failed"):
+            break
+        time.sleep(0.3)
+    cur = mgr.store.get(FinetuneJob, "default", "dpojob")
+    assert cur.status.get("state") == "Successful", cur.status
+    # the trained artifact is a LoRA adapter (DPO trains adapters only)
+    ft = mgr.store.get(Finetune, "default", "dpojob-finetune")
+    ck = (ft.status.get("llmCheckpoint") or {}).get("checkpointPath")
+    assert ck and os.path.exists(os.path.join(ck,
+                                              "adapter_model.safetensors"))
+
+
+def test_validation_rejects_bad_stage(tmp_path):
+    from datatunerx_amd.api.validation import validate_
+    from datatunerx_amd.api.types import Hyperparameter
+    hp = Hyperparameter(name="h", spec={"parameters": {"stage": "ppo"}})
+    with pytest.raises(Exception, match="stage"):
+        validate_(hp)
+    validate_(Hyperparameter(name="h2",
+                             spec={"parameters": {"stage": "dpo"}}))
