@@ -387,6 +387,10 @@ class SFTTrainer:
         """One micro-batch's loss tensor (overridden by DPOTrainer)."""
         return self.model(mb["input_ids"], labels=mb["labels"])
 
+    def _extra_log_metrics(self) -> dict:
+        """Stage-specific additions to the periodic train log."""
+        return {}
+
     def train_step(self, micro_batches) -> float:
         """One optimizer step over `gradient_accumulation_steps` micro
         batches; returns the (local) mean loss."""
@@ -428,6 +432,7 @@ class SFTTrainer:
                 self._log({
                     "current_steps": step, "total_steps": self.total_steps,
                     "loss": round(loss_g, 6), "learning_rate": lr,
+                    **self._extra_log_metrics(),
                     "epoch": round(step / max(1, self.total_steps) *
                                    float(cfg.num_train_epochs), 4),
                     "percentage": round(100 * step / self.total_steps, 2),
@@ -664,6 +669,10 @@ class DPOTrainer(SFTTrainer):
         self.train_loader.collate_fn = collate_preference
         self.last_margin = 0.0
         self.last_acc = 0.0
+
+    def _extra_log_metrics(self) -> dict:
+        return {"reward_margin": round(self.last_margin, 6),
+                "pref_accuracy": round(self.last_acc, 4)}
 
     def _micro_loss(self, mb):
         ids = torch.cat([mb["chosen_ids"], mb["rejected_ids"]], dim=0) \

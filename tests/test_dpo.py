@@ -155,3 +155,37 @@ def test_run_stage_dpo_cli(tmp_path):
     assert rows[-1]["current_steps"] == 3
     assert os.path.exists(os.path.join(out, "checkpoint",
                                        "adapter_model.safetensors"))
+
+
+def test_dpo_two_rank_gloo(tmp_path):
+    """2-rank data-parallel DPO through train.run (gloo): both ranks
+    finish, losses logged, one adapter checkpoint written."""
+    import subprocess
+    import sys
+
+    from conftest import free_port
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = str(tmp_path / "out")
+    port = str(free_port())
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({"RANK": str(rank), "WORLD_SIZE": "2",
+                    "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+                    "MASTER_PORT": port, "PYTHONPATH": root})
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "datatunerx_amd.train.run",
+             "--model_name_or_path", "llama-tiny", "--stage", "dpo",
+             "--output_dir", out, "--max_steps", "3",
+             "--per_device_train_batch_size", "2",
+             "--synthetic_examples", "16", "--block_size", "48",
+             "--logging_steps", "1", "--lora_dropout", "0.0"],
+            env=env))
+    for p in procs:
+        assert p.wait(timeout=300) == 0
+    rows = [json.loads(l) for l in
+            open(os.path.join(out, "watch", "trainer_log.jsonl"))]
+    assert rows[-1]["current_steps"] == 3
+    assert "reward_margin" in rows[-1]
+    assert os.path.exists(os.path.join(out, "checkpoint",
+                                       "adapter_model.safetensors"))
