@@ -136,5 +136,17 @@ class GPT2ForCausalLM(nn.Module):
         targets = labels[:, 1:].reshape(-1)
         return cross_entropy(logits, targets, ignore_index=-100)
 
+    def sequence_logprobs(self, input_ids, labels):
+        """[B] sum log p(target tokens) — DPO building block (same
+        contract as LlamaForCausalLM.sequence_logprobs)."""
+        from ..ops.autograd import per_row_cross_entropy
+        B = input_ids.shape[0]
+        logits = self.forward(input_ids)
+        h = logits[:, :-1, :].reshape(-1, self.cfg.vocab_size)
+        targets = labels[:, 1:].reshape(-1)
+        nll = per_row_cross_entropy(h.contiguous(), targets,
+                                    ignore_index=-100)
+        return -nll.view(B, -1).sum(dim=1)
+
     def trainable_parameters(self):
         return [(n, p) for n, p in self.named_parameters() if p.requires_grad]

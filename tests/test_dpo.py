@@ -189,3 +189,30 @@ def test_dpo_two_rank_gloo(tmp_path):
     assert "reward_margin" in rows[-1]
     assert os.path.exists(os.path.join(out, "checkpoint",
                                        "adapter_model.safetensors"))
+
+
+def test_dpo_gpt2(tmp_path):
+    """DPO on the GPT-2 family (sequence_logprobs parity)."""
+    from datatunerx_amd.models import GPT2Config, GPT2ForCausalLM
+    torch.manual_seed(0)
+    m = GPT2ForCausalLM(GPT2Config.tiny(), dtype=torch.float32)
+    m.init_random(seed=1)
+    ds = PreferenceDataset.synthetic(8, 24, 250, seed=2)
+    tr = DPOTrainer(m, ds, TrainerConfig(
+        output_dir=str(tmp_path), max_steps=2, micro_batch_size=4,
+        logging_steps=0), beta=0.3)
+    it = iter(tr.train_loader)
+    l0 = tr.train_step([next(it)])
+    assert abs(l0 - 0.693) < 0.01          # fresh adapters: margin 0
+    # independent torch check of gpt2 sequence_logprobs
+    ids = torch.randint(3, 250, (2, 17))
+    labels = ids.clone()
+    labels[:, :4] = -100
+    with torch.no_grad():
+        got = m.sequence_logprobs(ids, labels)
+        lp = F.log_softmax(m(ids)[:, :-1].float(), -1)
+        t = labels[:, 1:]
+        msk = t != -100
+        want = (lp.gather(-1, t.clamp(min=0).unsqueeze(-1)).squeeze(-1)
+                * msk).sum(1)
+    assert torch.allclose(got, want, atol=1e-4)
