@@ -161,7 +161,7 @@ class BatchingFront:
                         b["ev"].set()
 
 
-def build_handler(pool, batcher=None):
+def build_handler(pool, batcher=None, model_name: str = "datatunerx"):
     if not isinstance(pool, EnginePool):
         pool = EnginePool([pool])
 
@@ -190,6 +190,11 @@ def build_handler(pool, batcher=None):
         def do_GET(self):
             if self.path == "/health":
                 self._send(200, {"status": "ok"})
+            elif self.path in ("/v1/models", "/models"):
+                # OpenAI-SDK discovery call (clients list models first)
+                self._send(200, {"object": "list", "data": [{
+                    "id": model_name, "object": "model",
+                    "owned_by": "datatunerx-amd"}]})
             else:
                 self._send(404, {"error": "not found"})
 
@@ -201,7 +206,8 @@ def build_handler(pool, batcher=None):
                 self._send(400, {"error": "bad json"})
                 return
             try:
-                if self.path == "/chat/completions":
+                if self.path in ("/chat/completions",
+                                 "/v1/chat/completions"):
                     args = (body.get("messages", []),
                             int(body.get("max_tokens", 64)),
                             float(body.get("temperature", 0.0)),
@@ -273,7 +279,7 @@ def build_handler(pool, batcher=None):
                             "finish_reason": "stop",
                         }],
                     })
-                elif self.path == "/v1/score":
+                elif self.path in ("/v1/score", "/score"):
                     with lock as engine:
                         ppl = engine.perplexity(body.get("texts", []))
                     self._send(200, {"perplexity": ppl})
@@ -290,7 +296,8 @@ def build_handler(pool, batcher=None):
 
 def serve_forever(engine, host: str, port: int, batcher=None):
     httpd = ThreadingHTTPServer((host, port),
-                                build_handler(engine, batcher))
+                                build_handler(engine, batcher,
+                                              model_name=args.model))
     httpd.serve_forever()
 
 
@@ -448,7 +455,8 @@ def main(argv=None):
                             device=device, own_stream=True),
             max_batch=int(os.environ.get("DTX_SERVE_MAX_BATCH", "8")))
     httpd = ThreadingHTTPServer((args.host, args.port),
-                                build_handler(pool, batcher))
+                                build_handler(pool, batcher,
+                                              model_name=args.model))
     httpd.serve_forever()
 
 

@@ -709,3 +709,44 @@ def test_validation_rejects_bad_stage(tmp_path):
         validate_(hp)
     validate_(Hyperparameter(name="h2",
                              spec={"parameters": {"stage": "dpo"}}))
+
+
+def test_openai_compat_endpoints(tmp_path):
+    """OpenAI-SDK-shaped clients work: GET /v1/models lists the served
+    model and POST /v1/chat/completions aliases /chat/completions."""
+    import http.client
+    import json
+    import threading
+    from http.server import ThreadingHTTPServer
+
+    import torch
+
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    from datatunerx_amd.serve.server import build_handler
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                         dtype=torch.float32)
+    m.init_random(seed=1)
+    eng = InferenceEngine(m, template="vanilla",
+                          device=torch.device("cpu"))
+    httpd = ThreadingHTTPServer(("127.0.0.1", 0),
+                                build_handler(eng, model_name="llama-tiny"))
+    port = httpd.server_address[1]
+    t = threading.Thread(target=httpd.serve_forever, daemon=True)
+    t.start()
+    try:
+        c = http.client.HTTPConnection("127.0.0.1", port, timeout=30)
+        c.request("GET", "/v1/models")
+        r = json.loads(c.getresponse().read())
+        assert r["data"][0]["id"] == "llama-tiny"
+        body = json.dumps({"messages": [{"role": "user",
+                                         "content": "hi"}],
+                           "max_tokens": 4})
+        c.request("POST", "/v1/chat/completions", body=body)
+        r1 = json.loads(c.getresponse().read())
+        c.request("POST", "/chat/completions", body=body)
+        r2 = json.loads(c.getresponse().read())
+        assert r1["choices"][0]["message"]["content"] == \
+            r2["choices"][0]["message"]["content"]
+    finally:
+        httpd.shutdown()
