@@ -294,10 +294,11 @@ def build_handler(pool, batcher=None, model_name: str = "datatunerx"):
     return Handler
 
 
-def serve_forever(engine, host: str, port: int, batcher=None):
+def serve_forever(engine, host: str, port: int, batcher=None,
+                  model_name: str = "datatunerx"):
     httpd = ThreadingHTTPServer((host, port),
                                 build_handler(engine, batcher,
-                                              model_name=args.model))
+                                              model_name=model_name))
     httpd.serve_forever()
 
 
@@ -433,7 +434,8 @@ def main(argv=None):
                 # coalesce into batched generations.
                 batcher = BatchingFront(front, max_batch=int(
                     os.environ.get("DTX_SERVE_MAX_BATCH", "8")))
-            serve_forever(front, args.host, args.port, batcher)
+            serve_forever(front, args.host, args.port, batcher,
+                          model_name=args.model)
         else:
             tp_follower_loop(engine)
         return
