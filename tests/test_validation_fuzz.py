@@ -1,0 +1,50 @@
+"""Property test: the validating webhook either accepts or raises
+ValidationError — never an uncontrolled TypeError/AttributeError —
+for ARBITRARY spec shapes (clients send anything over the API)."""
+
+import pytest
+
+hypothesis = pytest.importorskip("hypothesis")
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+from datatunerx_amd.api.types import (Dataset, Finetune,  # noqa: E402
+                                      FinetuneExperiment, FinetuneJob,
+                                      Hyperparameter, LLM, Scoring)
+from datatunerx_amd.api.validation import (ValidationError,  # noqa: E402
+                                           default_, validate_)
+
+json_scalars = st.one_of(st.none(), st.booleans(),
+                         st.integers(-10**6, 10**6),
+                         st.floats(allow_nan=False, allow_infinity=False),
+                         st.text(max_size=20))
+json_values = st.recursive(
+    json_scalars,
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(st.text(max_size=12), children, max_size=4)),
+    max_leaves=12)
+specs = st.dictionaries(
+    st.sampled_from(["fineTune", "finetuneSpec", "llm", "dataset",
+                     "hyperparameter", "parameters", "node", "pending",
+                     "finetuneJobs", "datasetMetadata", "datasetInfo",
+                     "subsets", "features", "serveConfig",
+                     "tensorParallel", "scoringPluginConfig", "plugin",
+                     "stage", "epochs", "batchSize", "int4", "int8"]) |
+    st.text(max_size=12),
+    json_values, max_size=6)
+
+
+@settings(max_examples=150, deadline=None)
+@given(kind=st.sampled_from([FinetuneJob, FinetuneExperiment, Finetune,
+                             Hyperparameter, Dataset, LLM, Scoring]),
+       name=st.text(max_size=20), spec=specs)
+def test_validate_never_crashes(kind, name, spec):
+    obj = kind(name=name, spec=spec)
+    try:
+        default_(obj)
+        validate_(obj)
+    except ValidationError:
+        pass            # the controlled rejection path
+    except (AttributeError, TypeError, KeyError) as e:
+        pytest.fail(f"uncontrolled {type(e).__name__} for "
+                    f"{kind.__name__} spec={spec!r}: {e}")
