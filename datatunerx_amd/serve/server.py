@@ -208,10 +208,27 @@ def build_handler(pool, batcher=None, model_name: str = "datatunerx"):
             try:
                 if self.path in ("/chat/completions",
                                  "/v1/chat/completions"):
-                    args = (body.get("messages", []),
-                            int(body.get("max_tokens", 64)),
-                            float(body.get("temperature", 0.0)),
-                            float(body.get("top_p", 1.0)))
+                    msgs = body.get("messages", []) \
+                        if isinstance(body, dict) else None
+                    if not isinstance(msgs, list) or not all(
+                            isinstance(m, dict) and
+                            isinstance(m.get("role"), str) and
+                            isinstance(m.get("content"), str)
+                            for m in msgs):
+                        self._send(400, {
+                            "error": "messages must be a list of "
+                                     "{role: str, content: str}"})
+                        return
+                    try:
+                        args = (msgs,
+                                int(body.get("max_tokens", 64) or 64),
+                                float(body.get("temperature", 0.0) or 0),
+                                float(body.get("top_p", 1.0) or 1.0))
+                    except (TypeError, ValueError):
+                        self._send(400, {"error": "max_tokens/"
+                                         "temperature/top_p must be "
+                                         "numbers"})
+                        return
                     if body.get("stream"):
                         # SSE token streaming (OpenAI chunk format)
                         self.send_response(200)

@@ -750,3 +750,50 @@ def test_openai_compat_endpoints(tmp_path):
             r2["choices"][0]["message"]["content"]
     finally:
         httpd.shutdown()
+
+
+def test_serve_rejects_malformed_requests(tmp_path):
+    """Malformed /chat/completions bodies get clean 400s, valid ones
+    still work, and the server survives all of them."""
+    import http.client
+    import json as _json
+    import threading
+    from http.server import ThreadingHTTPServer
+
+    import torch
+
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    from datatunerx_amd.serve.server import build_handler
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                         dtype=torch.float32)
+    m.init_random(seed=1)
+    eng = InferenceEngine(m, template="vanilla",
+                          device=torch.device("cpu"))
+    httpd = ThreadingHTTPServer(("127.0.0.1", 0), build_handler(eng))
+    port = httpd.server_address[1]
+    threading.Thread(target=httpd.serve_forever, daemon=True).start()
+    try:
+        def post(obj):
+            c = http.client.HTTPConnection("127.0.0.1", port, timeout=30)
+            c.request("POST", "/chat/completions",
+                      body=_json.dumps(obj) if not isinstance(obj, bytes)
+                      else obj)
+            r = c.getresponse()
+            return r.status, r.read()
+        for bad in [{"messages": [{}]},
+                    {"messages": "hi"},
+                    {"messages": [{"role": 1, "content": "x"}]},
+                    {"messages": [{"role": "user", "content": None}]},
+                    {"messages": [{"role": "user", "content": "x"}],
+                     "max_tokens": "many"},
+                    []]:
+            st, _ = post(bad)
+            assert st == 400, (bad, st)
+        st, _ = post(b"not json at all")
+        assert st == 400
+        st, body = post({"messages": [{"role": "user", "content": "hi"}],
+                         "max_tokens": 4})
+        assert st == 200 and b"choices" in body
+    finally:
+        httpd.shutdown()
