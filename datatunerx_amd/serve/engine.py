@@ -456,6 +456,8 @@ class InferenceEngine:
     def _generate_stream(self, prompt_ids: List[int],
                          max_new_tokens: int = 64,
                          temperature: float = 0.0, top_p: float = 1.0):
+        if not prompt_ids:
+            raise ValueError("empty prompt")
         limit = getattr(self.model.cfg, "max_position_embeddings", 1 << 30)
         if len(prompt_ids) >= limit:
             raise ValueError(
@@ -589,6 +591,8 @@ class InferenceEngine:
         cfg = self.model.cfg
         greedy = not (temperature and temperature > 0)
         lens0 = [len(p) for p in prompts]
+        if not prompts or min(lens0) == 0:
+            raise ValueError("empty prompt in batch")
         gd = None
         if greedy and self.device.type == "cuda" and \
                 len(prompts) <= InferenceEngine.MAX_BATCH:

@@ -797,3 +797,54 @@ def test_serve_rejects_malformed_requests(tmp_path):
         assert st == 200 and b"choices" in body
     finally:
         httpd.shutdown()
+
+
+def test_engine_rejects_empty_prompts():
+    import torch
+
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                         dtype=torch.float32)
+    eng = InferenceEngine(m, template="vanilla",
+                          device=torch.device("cpu"))
+    with pytest.raises(ValueError, match="empty"):
+        eng.generate([], max_new_tokens=2)
+    with pytest.raises(ValueError, match="empty"):
+        eng.generate_batch([[1, 2], []], max_new_tokens=2)
+
+
+def test_multiturn_chat_history_assembly():
+    """chat() folds prior (user, assistant) turns through the template's
+    multiturn encoding — the prompt contains all turns in order."""
+    import torch
+
+    from datatunerx_amd.data.dataset import ByteTokenizer
+    from datatunerx_amd.data.templates import get_template
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                         dtype=torch.float32)
+    m.init_random(seed=0)
+    eng = InferenceEngine(m, template="llama2",
+                          device=torch.device("cpu"))
+    msgs = [{"role": "system", "content": "be brief"},
+            {"role": "user", "content": "first q"},
+            {"role": "assistant", "content": "first a"},
+            {"role": "user", "content": "second q"}]
+    seen = {}
+    orig = eng.generate
+
+    def spy(src, *a, **kw):
+        seen["src"] = list(src)
+        return orig(src, *a, **kw)
+    eng.generate = spy
+    eng.chat(msgs, max_tokens=2)
+    t = get_template("llama2")
+    want, _ = t.encode_oneturn(ByteTokenizer(), "second q", "",
+                               [("first q", "first a")], "be brief")
+    assert seen["src"] == want
+    text = ByteTokenizer().decode(seen["src"])
+    assert text.index("first q") < text.index("first a") < \
+        text.index("second q")
+    assert "be brief" in text
