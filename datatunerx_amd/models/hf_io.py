@@ -54,6 +54,12 @@ def load_hf_config(model_dir: str, **lora_kw) -> LlamaConfig:
         raise ValueError(f"unsupported architecture {archs} in {model_dir}")
     qwen2 = any("Qwen2" in a for a in archs)
     qwen3 = any("Qwen3" in a for a in archs)
+    required = ("vocab_size", "hidden_size", "intermediate_size",
+                "num_hidden_layers", "num_attention_heads")
+    missing = [k for k in required if k not in hc]
+    if missing:
+        raise ValueError(f"{model_dir}/config.json is missing required "
+                         f"fields: {missing}")
     kw = dict(
         vocab_size=hc["vocab_size"],
         hidden_size=hc["hidden_size"],

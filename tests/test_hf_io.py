@@ -376,3 +376,17 @@ def test_sentencepiece_tokenizer_backend(tmp_path):
         [{"instruction": "quick fox", "response": "hello world"}],
         tok, template_name="default", cutoff_len=64)
     assert len(ds) == 1 and any(l != -100 for l in ds[0]["labels"])
+
+
+def test_malformed_config_clear_error(tmp_path):
+    d = str(tmp_path / "bad")
+    os.makedirs(d)
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump({"architectures": ["LlamaForCausalLM"],
+                   "vocab_size": 128}, f)
+    with pytest.raises(ValueError, match="missing required"):
+        load_hf_config(d)
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump({"architectures": ["FalconForCausalLM"]}, f)
+    with pytest.raises(ValueError, match="unsupported architecture"):
+        load_hf_config(d)
