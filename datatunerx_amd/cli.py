@@ -46,6 +46,14 @@ def main(argv=None):
     p_mgr.add_argument("--once", action="store_true",
                        help="reconcile until settled, then exit")
 
+    p_logs = sub.add_parser(
+        "logs", help="print a Finetune's trainer step log (jsonl)")
+    p_logs.add_argument("name", help="Finetune (or FinetuneJob) name")
+    p_logs.add_argument("-n", "--namespace", default="default")
+    p_logs.add_argument("--work-dir", default="./dtx-work")
+    p_logs.add_argument("--tail", type=int, default=0,
+                        help="only the last N lines")
+
     p_run = sub.add_parser(
         "run", help="apply manifest(s), reconcile to completion, print "
                     "final statuses (dtx-ctl-style one-shot)")
@@ -82,6 +90,23 @@ def main(argv=None):
     elif args.cmd == "delete":
         store.delete(_resolve_kind(args.kind), args.namespace, args.name)
         print(f"{args.kind}/{args.name} deleted")
+    elif args.cmd == "logs":
+        import os
+        # accept either the Finetune name or its parent job's name
+        cands = [args.name, f"{args.name}-finetune"]
+        for nm in cands:
+            log = os.path.join(args.work_dir, args.namespace, nm,
+                               "output", "watch", "trainer_log.jsonl")
+            if os.path.exists(log):
+                with open(log) as f:
+                    lines = f.readlines()
+                if args.tail:
+                    lines = lines[-args.tail:]
+                sys.stdout.write("".join(lines))
+                return
+        raise SystemExit(
+            f"no trainer log under {args.work_dir}/{args.namespace}/"
+            f"{{{' | '.join(cands)}}}/output/watch/")
     elif args.cmd == "manager":
         from .api.controllers import ManagerConfig
         from .api.manager import Manager

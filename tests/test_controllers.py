@@ -848,3 +848,27 @@ def test_multiturn_chat_history_assembly():
     assert text.index("first q") < text.index("first a") < \
         text.index("second q")
     assert "be brief" in text
+
+
+def test_cli_logs_command(tmp_path, capsys):
+    """`dtx logs <job>` prints the trainer's jsonl step log from the
+    manager's work dir (finds the job's Finetune child by suffix)."""
+    import json as _json
+
+    from datatunerx_amd.cli import main as cli
+    work = tmp_path / "w" / "default" / "myjob-finetune" / "output" / \
+        "watch"
+    os.makedirs(work)
+    with open(work / "trainer_log.jsonl", "w") as f:
+        for i in range(5):
+            f.write(_json.dumps({"current_steps": i + 1,
+                                 "loss": 1.0 / (i + 1)}) + "\n")
+    st = str(tmp_path / "s")
+    cli(["--state-dir", st, "logs", "myjob",
+         "--work-dir", str(tmp_path / "w"), "--tail", "2"])
+    out = capsys.readouterr().out.strip().splitlines()
+    assert len(out) == 2
+    assert _json.loads(out[-1])["current_steps"] == 5
+    with pytest.raises(SystemExit, match="no trainer log"):
+        cli(["--state-dir", st, "logs", "nosuch",
+             "--work-dir", str(tmp_path / "w")])
