@@ -55,6 +55,13 @@ class FinetuningArguments:
     fp16: bool = False
     gradient_checkpointing: bool = False
     eval_steps: int = 0
+    do_predict: bool = False      # after training, greedy-generate from
+                                  # the eval set's prompts and write
+                                  # generated_predictions.jsonl + token
+                                  # ROUGE-L/BLEU (the reference's
+                                  # GenEvalSeq2SeqTrainer flow,
+                                  # cmd/tuning/trainer.py:29-172)
+    predict_max_new_tokens: int = 64
 
     @property
     def lora_targets(self):

@@ -220,10 +220,22 @@ def main(argv=None):
                 if os.path.abspath(dst) != os.path.abspath(ckpt_path):
                     shutil.copytree(ckpt_path, dst, dirs_exist_ok=True)
                     ckpt_path = dst
+            predict_metrics = None
+            if fargs.do_predict and eval_ds is not None and not dpo:
+                # generation eval: generated_predictions.jsonl + token
+                # ROUGE-L/BLEU (GenEvalSeq2SeqTrainer parity)
+                res = trainer.predict(
+                    eval_ds, max_new_tokens=fargs.predict_max_new_tokens,
+                    eos_token_id=tok.eos_token_id)
+                from .gen_metrics import generation_metrics
+                predict_metrics = generation_metrics(
+                    [r["predict_ids"] for r in res],
+                    [r["label_ids"] for r in res])
             write_status(status_file, "Successful",
                          checkpoint_path=ckpt_path,
                          final_loss=final_loss,
-                         eval=trainer.evaluate() if eval_ds else None)
+                         eval=trainer.evaluate() if eval_ds else None,
+                         predict=predict_metrics)
         import torch.distributed as dist
         if dist.is_initialized():
             dist.barrier()

@@ -340,3 +340,36 @@ def test_fused_qkv_gateup_grads_match_unfused():
             continue                  # gate_proj became "trainable" only
             # to flip the fused gate; it has no fused-side grad to compare
         assert torch.allclose(g_fused[n], p.grad, atol=1e-5), n
+
+
+def test_run_do_predict_cli(tmp_path):
+    """--do_predict true writes generated_predictions.jsonl and reports
+    rouge/bleu in the status file (GenEvalSeq2SeqTrainer flow)."""
+    import csv
+
+    from datatunerx_amd.train.run import main as train_main
+    p = tmp_path / "d.csv"
+    with open(p, "w", newline="") as f:
+        w = csv.writer(f)
+        w.writerow(["instruction", "response"])
+        for i in range(8):
+            w.writerow([f"say number {i}", f"number {i}"])
+    out = str(tmp_path / "out")
+    status = str(tmp_path / "st.json")
+    os.environ["DTX_STATUS_FILE"] = status
+    try:
+        rc = train_main([
+            "--model_name_or_path", "llama-tiny", "--output_dir", out,
+            "--dataset_path", str(p), "--eval_dataset_path", str(p),
+            "--max_steps", "2", "--per_device_train_batch_size", "2",
+            "--block_size", "64", "--logging_steps", "0",
+            "--do_predict", "true", "--predict_max_new_tokens", "4"])
+    finally:
+        del os.environ["DTX_STATUS_FILE"]
+    assert rc == 0
+    st = json.load(open(status))
+    assert st["state"] == "Successful"
+    assert "predict_rouge-l" in st["predict"]
+    gen = os.path.join(out, "generated_predictions.jsonl")
+    rows = [json.loads(l) for l in open(gen)]
+    assert len(rows) == 8 and "predict_ids" in rows[0]
