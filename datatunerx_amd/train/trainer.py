@@ -667,6 +667,12 @@ class DPOTrainer(SFTTrainer):
                          pad_token_id=pad_token_id)
         self.beta = beta
         self.train_loader.collate_fn = collate_preference
+        # standard DPO practice: dropout off — the loss compares policy
+        # and reference LOG-PROBABILITIES of fixed sequences, and
+        # stochastic masks turn that into a noisy estimate
+        for m in model.modules():
+            if isinstance(m, LoRALinearModule):
+                m.dropout = 0.0
         self.last_margin = 0.0
         self.last_acc = 0.0
 

@@ -216,3 +216,17 @@ def test_dpo_gpt2(tmp_path):
         want = (lp.gather(-1, t.clamp(min=0).unsqueeze(-1)).squeeze(-1)
                 * msk).sum(1)
     assert torch.allclose(got, want, atol=1e-4)
+
+
+def test_dpo_disables_dropout():
+    """DPOTrainer forces adapter dropout off (policy/reference logprob
+    comparison must be deterministic — TRL-style disable_dropout)."""
+    from datatunerx_amd.models.lora import LoRALinearModule
+    cfg = LlamaConfig.tiny(lora_dropout=0.3)
+    m = LlamaForCausalLM(cfg, lora=True, dtype=torch.float32)
+    m.init_random(seed=1)
+    DPOTrainer(m, PreferenceDataset.synthetic(4, 16, 500),
+               TrainerConfig(output_dir="/tmp/dpodrop", max_steps=1,
+                             micro_batch_size=2, logging_steps=0))
+    assert all(mod.dropout == 0.0 for mod in m.modules()
+               if isinstance(mod, LoRALinearModule))
