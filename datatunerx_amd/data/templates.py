@@ -170,9 +170,10 @@ register_template("chatglm3", Template(
     prompt=["<|user|>\n{{query}}<|assistant|>\n"]))
 @dataclass
 class Llama3Template(Template):
-    """Llama-3 header convention; the system block is omitted entirely
-    when no system prompt is given (matching the reference chat
-    format)."""
+    """Llama-3 header convention: the system block is omitted entirely
+    when no system prompt is given, and assistant turns terminate with
+    <|eot_id|> (the turn terminator the engine's stop set knows) rather
+    than the tokenizer's plain eos."""
 
     def encode_multiturn(self, tokenizer, query, resp, history=None,
                          system: str = ""):
@@ -180,11 +181,20 @@ class Llama3Template(Template):
             # non-mutating copy: templates are shared module singletons
             # and the serving front encodes from multiple threads
             import dataclasses
-            return Template.encode_multiturn(
+            pairs = Template.encode_multiturn(
                 dataclasses.replace(self, prefix=[]), tokenizer, query,
                 resp, history, system)
-        return super().encode_multiturn(tokenizer, query, resp, history,
-                                        system)
+        else:
+            pairs = super().encode_multiturn(tokenizer, query, resp,
+                                             history, system)
+        eos = getattr(tokenizer, "eos_token_id", None)
+        eot = _tok_encode(tokenizer, "<|eot_id|>")
+        out = []
+        for src, tgt in pairs:
+            if eos is not None and tgt and tgt[-1] == eos:
+                tgt = tgt[:-1] + eot
+            out.append((src, tgt))
+        return out
 
 
 register_template("llama3", Llama3Template(

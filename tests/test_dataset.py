@@ -171,3 +171,19 @@ def test_llama3_template_omits_empty_system():
     src2, _ = t.encode_oneturn(tok, "hi", "yo", None, "sys")
     text2 = bytes(i - 3 for i in src2 if i >= 3).decode()
     assert text2.startswith("<|start_header_id|>system")
+
+
+def test_llama3_targets_end_with_eot():
+    """Assistant turns in llama3 training data terminate with
+    <|eot_id|> (the llama3 turn terminator), not the plain eos id."""
+    from datatunerx_amd.data.dataset import ByteTokenizer
+    from datatunerx_amd.data.templates import get_template
+    t = get_template("llama3")
+    tok = ByteTokenizer()
+    pairs = t.encode_multiturn(tok, "q2", "a2",
+                               history=[("q1", "a1")], system="s")
+    assert len(pairs) == 2
+    eot = tok.encode("<|eot_id|>")
+    for src, tgt in pairs:
+        assert tgt[-len(eot):] == eot
+        assert tok.eos_token_id not in tgt
