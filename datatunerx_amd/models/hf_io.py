@@ -25,7 +25,9 @@ containing "Qwen2" sets attention_bias, and
 Qwen3-family sets qk_norm (per-head q/k RMSNorm before RoPE, decoupled
 head_dim), with
   model.layers.N.self_attn.{q,k}_norm.weight -> layers.N.self_attn.{q,k}_norm
-(Mistral loads as plain Llama geometry.) save_hf_model writes the
+(Mistral loads as plain Llama geometry; its sliding-window attention
+is evaluated as FULL attention — exact for contexts up to the window
+size, an over-attention approximation beyond it.) save_hf_model writes the
 matching architectures/model_type back, so exports reload in
 transformers as the right class.
 """
