@@ -299,9 +299,9 @@ def test_manager_metrics_endpoint(tmp_path):
     mgr = mk_manager(tmp_path)
     seed_resources(mgr.store)
     mgr.reconcile_once()
-    srv = mgr.serve_metrics(port=_PORT[0] + 49)
+    srv = mgr.serve_metrics(port=0)      # OS-assigned (no collisions)
     try:
-        url = f"http://127.0.0.1:{_PORT[0] + 49}"
+        url = f"http://127.0.0.1:{srv.server_address[1]}"
         assert urllib.request.urlopen(url + "/healthz",
                                       timeout=10).read() == b"ok"
         text = urllib.request.urlopen(url + "/metrics",
