@@ -872,3 +872,42 @@ def test_cli_logs_command(tmp_path, capsys):
     with pytest.raises(SystemExit, match="no trainer log"):
         cli(["--state-dir", st, "logs", "nosuch",
              "--work-dir", str(tmp_path / "w")])
+
+
+def test_plugin_scoring_pipeline(tmp_path):
+    """scoringPluginConfig with loadPlugin: the job's score comes from a
+    user 'module:function' plugin called with (endpoint, parameters) —
+    the reference's plugin-scoring pod, in-process."""
+    import sys
+    plug_dir = tmp_path / "plugins"
+    os.makedirs(plug_dir)
+    with open(plug_dir / "dtx_test_scorer.py", "w") as f:
+        f.write(
+            "import json, urllib.request\n"
+            "def my_score(endpoint, params):\n"
+            "    # prove the endpoint is live AND params flow through\n"
+            "    r = urllib.request.urlopen(endpoint + '/health',\n"
+            "                               timeout=30)\n"
+            "    assert json.loads(r.read())['status'] == 'ok'\n"
+            "    return int(params['base']) + 7\n")
+    sys.path.insert(0, str(plug_dir))
+    try:
+        mgr = mk_manager(tmp_path)
+        seed_resources(mgr.store)
+        job = FinetuneJob(name="plugjob", spec={
+            "fineTune": {"finetuneSpec": finetune_spec()},
+            "scoringPluginConfig": {"name": "dtx_test_scorer:my_score",
+                                    "parameters": {"base": 100}}})
+        mgr.store.create(job)
+        deadline = time.time() + 300
+        while time.time() < deadline:
+            mgr.reconcile_once()
+            cur = mgr.store.get(FinetuneJob, "default", "plugjob")
+            if cur.status.get("state") in ("Successful", "Failed"):
+                break
+            time.sleep(0.3)
+        cur = mgr.store.get(FinetuneJob, "default", "plugjob")
+        assert cur.status.get("state") == "Successful", cur.status
+        assert cur.status["result"]["score"] == "107"
+    finally:
+        sys.path.remove(str(plug_dir))
