@@ -911,3 +911,36 @@ def test_plugin_scoring_pipeline(tmp_path):
         assert cur.status["result"]["score"] == "107"
     finally:
         sys.path.remove(str(plug_dir))
+
+
+def test_manager_restart_resumes_inflight_job(tmp_path):
+    """Control-plane crash-restart: a NEW Manager over the same
+    state/work dirs drives a job that was mid-training to Successful
+    (the reference gets this from the apiserver; here the file-backed
+    store + status files are the durable state)."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store, hp_params={"maxSteps": 6})
+    job = FinetuneJob(name="rejob", spec={
+        "fineTune": {"finetuneSpec": finetune_spec()}})
+    mgr.store.create(job)
+    # reconcile until the Finetune exists and training has LAUNCHED
+    deadline = time.time() + 120
+    launched = False
+    while time.time() < deadline and not launched:
+        mgr.reconcile_once()
+        ft = mgr.store.try_get(Finetune, "default", "rejob-finetune")
+        launched = bool(ft and ft.status.get("state") == "Running")
+        time.sleep(0.2)
+    assert launched
+    del mgr                                     # "crash"
+
+    mgr2 = mk_manager(tmp_path)                 # same dirs (mk_manager
+    deadline = time.time() + 300                # is deterministic here)
+    while time.time() < deadline:
+        mgr2.reconcile_once()
+        cur = mgr2.store.get(FinetuneJob, "default", "rejob")
+        if cur.status.get("state") in ("Successful", "Failed"):
+            break
+        time.sleep(0.3)
+    cur = mgr2.store.get(FinetuneJob, "default", "rejob")
+    assert cur.status.get("state") == "Successful", cur.status
