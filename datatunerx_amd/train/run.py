@@ -186,6 +186,7 @@ def main(argv=None):
             num_train_epochs=fargs.num_train_epochs,
             max_steps=fargs.max_steps,
             micro_batch_size=fargs.per_device_train_batch_size,
+            eval_batch_size=fargs.per_device_eval_batch_size,
             gradient_accumulation_steps=fargs.gradient_accumulation_steps,
             logging_steps=fargs.logging_steps,
             eval_steps=fargs.eval_steps, save_steps=fargs.save_steps,

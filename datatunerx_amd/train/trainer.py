@@ -45,6 +45,7 @@ class TrainerConfig:
     num_train_epochs: float = 1.0
     max_steps: int = -1                    # overrides epochs when > 0
     micro_batch_size: int = 4
+    eval_batch_size: int = 0               # 0 = micro_batch_size
     gradient_accumulation_steps: int = 1
     logging_steps: int = 10
     eval_steps: int = 0                    # 0 = eval at end only
@@ -462,7 +463,8 @@ class SFTTrainer:
     @torch.no_grad()
     def evaluate(self) -> dict:
         self.model.eval()
-        loader = ShardedLoader(self.eval_dataset, self.cfg.micro_batch_size,
+        ebs = self.cfg.eval_batch_size or self.cfg.micro_batch_size
+        loader = ShardedLoader(self.eval_dataset, ebs,
                                self.rank, self.world, seed=0,
                                pad_token_id=self.pad_token_id,
                                device=self.device)
@@ -718,7 +720,8 @@ class DPOTrainer(SFTTrainer):
         eval_loss contract does not apply to pairs)."""
         from ..data.preference import collate_preference
         self.model.eval()
-        loader = ShardedLoader(self.eval_dataset, self.cfg.micro_batch_size,
+        ebs = self.cfg.eval_batch_size or self.cfg.micro_batch_size
+        loader = ShardedLoader(self.eval_dataset, ebs,
                                self.rank, self.world, seed=0,
                                pad_token_id=self.pad_token_id,
                                device=self.device,
