@@ -39,6 +39,21 @@ def main(argv=None):
     margs, fargs, dargs = get_train_args(argv)
     status_file = os.environ.get("DTX_STATUS_FILE", "")
     rank, world, local_rank, device = init_distributed()
+    if rank == 0:
+        # declared-for-CR-parity flags this trainer does not act on:
+        # say so loudly rather than silently ignoring a request
+        if fargs.optim not in ("adamw_torch", "adamw"):
+            print(f"warning: --optim {fargs.optim!r} ignored — the "
+                  f"fused AdamW kernel is the only optimizer")
+        if fargs.fp16:
+            print("warning: --fp16 ignored — compute dtype is bf16 "
+                  "(MI355X-native; >= the reference's fp16)")
+        if margs.rope_scaling:
+            print("warning: --rope_scaling ignored (dormant in the "
+                  "reference too, parser.py:60-68)")
+        if margs.shift_attn:
+            print("warning: --shift_attn ignored (dormant in the "
+                  "reference too, parser.py:70-73)")
     try:
         if rank == 0:
             write_status(status_file, "Running")

@@ -373,3 +373,18 @@ def test_run_do_predict_cli(tmp_path):
     gen = os.path.join(out, "generated_predictions.jsonl")
     rows = [json.loads(l) for l in open(gen)]
     assert len(rows) == 8 and "predict_ids" in rows[0]
+
+
+def test_ignored_flags_warn(tmp_path, capsys):
+    from datatunerx_amd.train.run import main as train_main
+    rc = train_main([
+        "--model_name_or_path", "llama-tiny", "--output_dir",
+        str(tmp_path / "o"), "--max_steps", "1",
+        "--per_device_train_batch_size", "2", "--synthetic_examples",
+        "8", "--block_size", "32", "--logging_steps", "0",
+        "--optim", "sgd", "--fp16", "true", "--shift_attn", "true"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "--optim 'sgd' ignored" in out
+    assert "--fp16 ignored" in out
+    assert "--shift_attn ignored" in out
