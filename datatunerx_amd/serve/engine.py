@@ -721,6 +721,13 @@ class InferenceEngine:
         budgets = [int(r.get("max_tokens", 64)) for r in requests]
         temp = float(requests[0].get("temperature", 0.0))
         top_p = float(requests[0].get("top_p", 1.0))
+        for r in requests[1:]:
+            if (float(r.get("temperature", 0.0)),
+                    float(r.get("top_p", 1.0))) != (temp, top_p):
+                # one batched generation samples with ONE setting; the
+                # BatchingFront groups requests by it before batching
+                raise ValueError("batched requests must share "
+                                 "temperature/top_p")
         return prompts, budgets, temp, top_p
 
     def chat_batch(self, requests: List[dict]) -> List[str]:

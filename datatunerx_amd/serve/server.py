@@ -104,61 +104,74 @@ class BatchingFront:
                     batch.append(self._q.get(timeout=left))
                 except queue.Empty:
                     break
-            reqs = [b["req"] for b in batch]
-            streaming = any("tokq" in b for b in batch)
-            try:
-                if len(batch) == 1:
-                    # single request: the padded batched graph would
-                    # decode MAXB rows for one stream — use the
-                    # single-stream (graphed) path instead
-                    b = batch[0]
-                    r = b["req"]
-                    if "tokq" in b:
-                        for d in self.engine.chat_stream(
-                                r["messages"], r["max_tokens"],
-                                r["temperature"], r["top_p"]):
-                            b["tokq"].put(d)
-                        b["tokq"].put(None)
-                    else:
-                        b["out"] = self.engine.chat(
+            # requests only share a batched generation when their
+            # sampling params MATCH — the engine samples the whole
+            # batch with one (temperature, top_p), so a greedy request
+            # must never ride along with a sampled one
+            groups = {}
+            for b in batch:
+                key = (float(b["req"].get("temperature", 0.0)),
+                       float(b["req"].get("top_p", 1.0)))
+                groups.setdefault(key, []).append(b)
+            for batch in groups.values():
+                self._run_batch(batch)
+
+    def _run_batch(self, batch):
+        reqs = [b["req"] for b in batch]
+        streaming = any("tokq" in b for b in batch)
+        try:
+            if len(batch) == 1:
+                # single request: the padded batched graph would
+                # decode MAXB rows for one stream — use the
+                # single-stream (graphed) path instead
+                b = batch[0]
+                r = b["req"]
+                if "tokq" in b:
+                    for d in self.engine.chat_stream(
                             r["messages"], r["max_tokens"],
-                            r["temperature"], r["top_p"])
-                        b["ev"].set()
-                    continue
-                if streaming and hasattr(self.engine,
-                                         "chat_batch_stream"):
-                    texts = [""] * len(batch)
-                    for i, d in self.engine.chat_batch_stream(reqs):
-                        b = batch[i]
-                        if d is None:
-                            if "tokq" in b:
-                                b["tokq"].put(None)
-                            else:
-                                b["out"] = texts[i]
-                                b["ev"].set()
-                        elif "tokq" in b:
-                            b["tokq"].put(d)
-                        else:
-                            texts[i] += d
+                            r["temperature"], r["top_p"]):
+                        b["tokq"].put(d)
+                    b["tokq"].put(None)
                 else:
-                    outs = self.engine.chat_batch(reqs)
-                    for b, o in zip(batch, outs):
+                    b["out"] = self.engine.chat(
+                        r["messages"], r["max_tokens"],
+                        r["temperature"], r["top_p"])
+                    b["ev"].set()
+                return
+            if streaming and hasattr(self.engine,
+                                     "chat_batch_stream"):
+                texts = [""] * len(batch)
+                for i, d in self.engine.chat_batch_stream(reqs):
+                    b = batch[i]
+                    if d is None:
                         if "tokq" in b:
-                            # engine without chat_batch_stream (TP
-                            # front): deliver the whole text as one
-                            # delta so stream slots still resolve
-                            b["tokq"].put(o)
                             b["tokq"].put(None)
                         else:
-                            b["out"] = o
+                            b["out"] = texts[i]
                             b["ev"].set()
-            except Exception as e:      # pragma: no cover
-                for b in batch:
-                    if "tokq" in b:
-                        b["tokq"].put(e)
+                    elif "tokq" in b:
+                        b["tokq"].put(d)
                     else:
-                        b["err"] = e
+                        texts[i] += d
+            else:
+                outs = self.engine.chat_batch(reqs)
+                for b, o in zip(batch, outs):
+                    if "tokq" in b:
+                        # engine without chat_batch_stream (TP
+                        # front): deliver the whole text as one
+                        # delta so stream slots still resolve
+                        b["tokq"].put(o)
+                        b["tokq"].put(None)
+                    else:
+                        b["out"] = o
                         b["ev"].set()
+        except Exception as e:      # pragma: no cover
+            for b in batch:
+                if "tokq" in b:
+                    b["tokq"].put(e)
+                else:
+                    b["err"] = e
+                    b["ev"].set()
 
 
 def build_handler(pool, batcher=None, model_name: str = "datatunerx"):

@@ -944,3 +944,51 @@ def test_manager_restart_resumes_inflight_job(tmp_path):
         time.sleep(0.3)
     cur = mgr2.store.get(FinetuneJob, "default", "rejob")
     assert cur.status.get("state") == "Successful", cur.status
+
+
+def test_batcher_separates_mixed_sampling_params(tmp_path):
+    """A greedy request never rides in the same batched generation as a
+    temperature-sampled one: concurrent mixed requests all succeed and
+    the greedy ones return exactly the sequential greedy output."""
+    import queue as _q
+    import threading
+
+    import torch
+
+    from datatunerx_amd.models import LlamaConfig, LlamaForCausalLM
+    from datatunerx_amd.serve.engine import InferenceEngine
+    from datatunerx_amd.serve.server import BatchingFront
+    m = LlamaForCausalLM(LlamaConfig.tiny(), lora=False,
+                         dtype=torch.float32)
+    m.init_random(seed=3)
+    eng = InferenceEngine(m, template="vanilla",
+                          device=torch.device("cpu"))
+    # engine-level guard: mixed params in one batch are refused
+    with pytest.raises(ValueError, match="share"):
+        eng.chat_batch([
+            {"messages": [{"role": "user", "content": "a"}],
+             "max_tokens": 4, "temperature": 0.0},
+            {"messages": [{"role": "user", "content": "b"}],
+             "max_tokens": 4, "temperature": 0.9}])
+
+    front = BatchingFront(eng, max_batch=8, linger=0.05)
+    msgs = [{"role": "user", "content": "hello there"}]
+    want = eng.chat(msgs, max_tokens=6)          # sequential greedy
+    results = _q.Queue()
+
+    def one(temp):
+        try:
+            results.put((temp, front.chat(msgs, 6, temp, 1.0)))
+        except Exception as e:                   # pragma: no cover
+            results.put((temp, e))
+    threads = [threading.Thread(target=one, args=(t,))
+               for t in (0.0, 0.0, 0.8, 0.0, 0.8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    outs = [results.get(timeout=10) for _ in range(5)]
+    for temp, out in outs:
+        assert not isinstance(out, Exception), out
+        if temp == 0.0:
+            assert out == want                   # greedy stayed greedy
