@@ -69,7 +69,7 @@ def main(argv=None):
     if args.cmd == "apply":
         text = sys.stdin.read() if args.filename == "-" else \
             open(args.filename).read()
-        for obj in store.apply_manifest(text):
+        for obj in _apply(store, text, args.filename):
             print(f"{obj.kind.lower()}/{obj.name} applied")
     elif args.cmd == "get":
         kind = _resolve_kind(args.kind)
@@ -125,7 +125,7 @@ def main(argv=None):
         from .api.types import (FinetuneExperiment, FinetuneJob, Finetune)
         for fn in args.filename:
             text = sys.stdin.read() if fn == "-" else open(fn).read()
-            for obj in store.apply_manifest(text):
+            for obj in _apply(store, text, fn):
                 print(f"{obj.kind.lower()}/{obj.name} applied")
         cfg = ManagerConfig(state_dir=args.state_dir,
                             work_dir=args.work_dir, n_gpus=args.n_gpus,
@@ -142,6 +142,18 @@ def main(argv=None):
                 print(f"{o.kind.lower()}/{o.name}: "
                       f"{o.status.get('state', '')}{extra}")
         sys.exit(0 if ok else 1)
+
+
+def _apply(store, text: str, source: str):
+    """apply_manifest with kubectl-style clean errors (bad YAML, unknown
+    kind, admission rejection) instead of tracebacks."""
+    from .api.validation import ValidationError
+    try:
+        return store.apply_manifest(text)
+    except yaml.YAMLError as e:
+        raise SystemExit(f"error: {source}: invalid YAML: {e}")
+    except (ValidationError, ValueError) as e:
+        raise SystemExit(f"error: {source}: {e}")
 
 
 def _resolve_kind(k: str) -> str:

@@ -992,3 +992,24 @@ def test_batcher_separates_mixed_sampling_params(tmp_path):
         assert not isinstance(out, Exception), out
         if temp == 0.0:
             assert out == want                   # greedy stayed greedy
+
+
+def test_cli_apply_clean_errors(tmp_path, capsys):
+    from datatunerx_amd.cli import main as cli
+    st = str(tmp_path / "s")
+    bad_yaml = tmp_path / "bad.yaml"
+    bad_yaml.write_text("kind: [unclosed")
+    with pytest.raises(SystemExit, match="invalid YAML"):
+        cli(["--state-dir", st, "apply", "-f", str(bad_yaml)])
+    bad_kind = tmp_path / "kind.yaml"
+    bad_kind.write_text("kind: Nope\nmetadata: {name: x}\n")
+    with pytest.raises(SystemExit, match="unknown kind"):
+        cli(["--state-dir", st, "apply", "-f", str(bad_kind)])
+    invalid = tmp_path / "invalid.yaml"
+    invalid.write_text(
+        "apiVersion: finetune.datatunerx.io/v1beta1\n"
+        "kind: Hyperparameter\n"
+        "metadata: {name: h}\n"
+        "spec: {parameters: {stage: ppo}}\n")
+    with pytest.raises(SystemExit, match="stage must be"):
+        cli(["--state-dir", st, "apply", "-f", str(invalid)])
