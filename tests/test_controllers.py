@@ -1013,3 +1013,48 @@ def test_cli_apply_clean_errors(tmp_path, capsys):
         "spec: {parameters: {stage: ppo}}\n")
     with pytest.raises(SystemExit, match="stage must be"):
         cli(["--state-dir", st, "apply", "-f", str(invalid)])
+
+
+def test_experiment_mixed_stage_jobs(tmp_path):
+    """Experiment fan-out where one job OVERRIDES the hyperparameter to
+    stage=dpo: the override merge (updateHyperparameters parity) flows
+    through to the trainer flags, both jobs finish, both score."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store, hp_params={"loRA_Dropout": "0.0"})
+    spec_sft = finetune_spec()
+    spec_dpo = finetune_spec()
+    spec_dpo["hyperparameter"]["overrides"] = {"stage": "dpo",
+                                               "dpoBeta": "0.3"}
+    exp = FinetuneExperiment(name="expmix", spec={
+        "finetuneJobs": [
+            {"name": "mx-sft", "spec": {"fineTune":
+                                        {"finetuneSpec": spec_sft}}},
+            {"name": "mx-dpo", "spec": {"fineTune":
+                                        {"finetuneSpec": spec_dpo}}},
+        ]})
+    mgr.store.create(exp)
+    deadline = time.time() + 300
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        cur = mgr.store.get(FinetuneExperiment, "default", "expmix")
+        if cur.status.get("state") in ("Success", "Failed"):
+            break
+        time.sleep(0.3)
+    cur = mgr.store.get(FinetuneExperiment, "default", "expmix")
+    assert cur.status.get("state") == "Success", cur.status
+    states = {js["name"]: js["finetuneJobStatus"].get("state")
+              for js in cur.status["jobsStatus"]}
+    assert states == {"mx-sft": "Successful", "mx-dpo": "Successful"}
+    # the dpo job's override flowed to the trainer: its Finetune args
+    # carry --stage dpo --dpo_beta 0.3; the sft job's don't
+    ft_dpo = mgr.store.get(Finetune, "default", "mx-dpo-finetune")
+    ft_sft = mgr.store.get(Finetune, "default", "mx-sft-finetune")
+    ov = (ft_dpo.spec.get("hyperparameter") or {}).get("overrides") or {}
+    assert ov.get("stage") == "dpo" and ov.get("dpoBeta") == "0.3", ov
+    assert not ((ft_sft.spec.get("hyperparameter") or {})
+                .get("overrides") or {})
+    # both trained checkpoints exist (adapters in both stages)
+    for ft in (ft_dpo, ft_sft):
+        ck = (ft.status.get("llmCheckpoint") or {}).get("checkpointPath")
+        assert ck and os.path.exists(
+            os.path.join(ck, "adapter_model.safetensors")), ft.status
