@@ -215,7 +215,9 @@ class FinetuneController:
         info = ft.status.get("trainJobInfo") or {}
         for pid in info.get("pids", []):
             if _alive(pid):
-                self.sup.terminate(pid)
+                # SIGTERM + reap (SIGKILL escalation): terminate alone
+                # leaves a zombie the long-lived manager never waits on
+                self.sup.stop_and_reap(pid, 3000)
         self.inv.release_owner(f"ft/{ft.namespace}/{ft.name}")
 
     # ------------------------------------------------------- reconcile
@@ -252,7 +254,7 @@ class FinetuneController:
         if any(c > 0 or c in (-2, -3) for c in codes):
             for pid in info["pids"]:
                 if _alive(pid):
-                    self.sup.terminate(pid)
+                    self.sup.stop_and_reap(pid, 3000)
             self.inv.release_owner(f"ft/{ft.namespace}/{ft.name}")
             # bounded restart (an improvement over the reference, which
             # only propagates failure — SURVEY.md §5 failure detection):

@@ -1058,3 +1058,77 @@ def test_experiment_mixed_stage_jobs(tmp_path):
         ck = (ft.status.get("llmCheckpoint") or {}).get("checkpointPath")
         assert ck and os.path.exists(
             os.path.join(ck, "adapter_model.safetensors")), ft.status
+
+
+def test_delete_job_mid_training_cleans_up(tmp_path):
+    """Deleting a FinetuneJob while training runs: the finalizer path
+    stops the trainer process, releases the gang GPUs, strips
+    back-references and removes the job + GC's its children."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store, hp_params={"maxSteps": 2000})  # long run
+    job = FinetuneJob(name="deljob", spec={
+        "fineTune": {"finetuneSpec": finetune_spec()}})
+    mgr.store.create(job)
+    deadline = time.time() + 120
+    pid = None
+    while time.time() < deadline and pid is None:
+        mgr.reconcile_once()
+        ft = mgr.store.try_get(Finetune, "default", "deljob-finetune")
+        if ft and ft.status.get("state") == "Running":
+            info = ft.status.get("trainJobInfo") or {}
+            pids = info.get("pids") or []
+            pid = pids[0] if pids else None
+        time.sleep(0.2)
+    assert pid is not None, "training never started"
+    assert _alive(pid)
+
+    mgr.store.delete(FinetuneJob, "default", "deljob")
+    deadline = time.time() + 120
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        mgr.store.gc_sweep()
+        if mgr.store.try_get(FinetuneJob, "default", "deljob") is None:
+            break
+        time.sleep(0.2)
+    assert mgr.store.try_get(FinetuneJob, "default", "deljob") is None
+    # children are GC'd over the next reconcile passes (gc_sweep marks
+    # the orphan; the Finetune controller's finalizer removes it)
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        if mgr.store.try_get(Finetune, "default",
+                             "deljob-finetune") is None:
+            break
+        time.sleep(0.2)
+    assert mgr.store.try_get(Finetune, "default", "deljob-finetune") \
+        is None
+    # the trainer process was actually stopped
+    for _ in range(50):
+        if not _alive(pid):
+            break
+        time.sleep(0.2)
+    assert not _alive(pid), f"trainer pid {pid} still alive"
+    # gang GPUs released: a fresh job can allocate immediately
+    hp = mgr.store.get(Hyperparameter, "default", "hp")
+    hp.spec["parameters"]["maxSteps"] = 2
+    mgr.store.update(hp)
+    job2 = FinetuneJob(name="afterjob", spec={
+        "fineTune": {"finetuneSpec": finetune_spec()}})
+    mgr.store.create(job2)
+    deadline = time.time() + 300
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        cur = mgr.store.get(FinetuneJob, "default", "afterjob")
+        if cur.status.get("state") in ("Successful", "Failed"):
+            break
+        time.sleep(0.3)
+    assert mgr.store.get(FinetuneJob, "default",
+                         "afterjob").status.get("state") == "Successful"
+
+
+def _alive(pid) -> bool:
+    try:
+        os.kill(int(pid), 0)
+        return True
+    except OSError:
+        return False
