@@ -1132,3 +1132,48 @@ def _alive(pid) -> bool:
         return True
     except OSError:
         return False
+
+
+def test_delete_experiment_mid_training_cleans_cascade(tmp_path):
+    """Deleting a FinetuneExperiment mid-training cascades: jobs,
+    Finetunes and trainer processes all go; nothing leaks."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store, hp_params={"maxSteps": 2000})
+    exp = FinetuneExperiment(name="delexp", spec={
+        "finetuneJobs": [
+            {"name": "dx-j1", "spec": {"fineTune":
+                                       {"finetuneSpec": finetune_spec()}}},
+        ]})
+    mgr.store.create(exp)
+    deadline = time.time() + 120
+    pid = None
+    while time.time() < deadline and pid is None:
+        mgr.reconcile_once()
+        ft = mgr.store.try_get(Finetune, "default", "dx-j1-finetune")
+        if ft and ft.status.get("state") == "Running":
+            pids = (ft.status.get("trainJobInfo") or {}).get("pids") or []
+            pid = pids[0] if pids else None
+        time.sleep(0.2)
+    assert pid is not None and _alive(pid)
+
+    mgr.store.delete(FinetuneExperiment, "default", "delexp")
+    deadline = time.time() + 120
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        gone = all(mgr.store.try_get(c, "default", n) is None
+                   for c, n in [(FinetuneExperiment, "delexp"),
+                                (FinetuneJob, "dx-j1"),
+                                (Finetune, "dx-j1-finetune")])
+        if gone:
+            break
+        time.sleep(0.2)
+    assert mgr.store.try_get(FinetuneExperiment, "default",
+                             "delexp") is None
+    assert mgr.store.try_get(FinetuneJob, "default", "dx-j1") is None
+    assert mgr.store.try_get(Finetune, "default",
+                             "dx-j1-finetune") is None
+    for _ in range(50):
+        if not _alive(pid):
+            break
+        time.sleep(0.2)
+    assert not _alive(pid), f"trainer pid {pid} leaked"
