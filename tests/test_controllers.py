@@ -1177,3 +1177,40 @@ def test_delete_experiment_mid_training_cleans_cascade(tmp_path):
             break
         time.sleep(0.2)
     assert not _alive(pid), f"trainer pid {pid} leaked"
+
+
+@pytest.mark.slow
+def test_concurrent_jobs_with_mid_flight_deletion(tmp_path):
+    """Churn: three jobs run concurrently through the gang scheduler;
+    one is deleted mid-training; the other two still complete and the
+    deleted one's GPUs return to the pool."""
+    mgr = mk_manager(tmp_path)
+    seed_resources(mgr.store, hp_params={"maxSteps": 40})
+    for i in range(3):
+        mgr.store.create(FinetuneJob(name=f"churn{i}", spec={
+            "fineTune": {"finetuneSpec": finetune_spec()}}))
+    deleted = False
+    deadline = time.time() + 300
+    while time.time() < deadline:
+        mgr.reconcile_once()
+        if not deleted:
+            ft = mgr.store.try_get(Finetune, "default",
+                                   "churn1-finetune")
+            if ft and ft.status.get("state") == "Running":
+                mgr.store.delete(FinetuneJob, "default", "churn1")
+                deleted = True
+        states = [(mgr.store.try_get(FinetuneJob, "default",
+                                     f"churn{i}") or
+                   type("o", (), {"status": {}})).status.get("state")
+                  for i in (0, 2)]
+        gone = mgr.store.try_get(FinetuneJob, "default",
+                                 "churn1") is None
+        if deleted and gone and all(s in ("Successful", "Failed")
+                                    for s in states):
+            break
+        time.sleep(0.3)
+    assert deleted
+    assert mgr.store.try_get(FinetuneJob, "default", "churn1") is None
+    for i in (0, 2):
+        cur = mgr.store.get(FinetuneJob, "default", f"churn{i}")
+        assert cur.status.get("state") == "Successful", (i, cur.status)
