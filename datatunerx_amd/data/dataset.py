@@ -51,7 +51,9 @@ def read_csv_rows(path: str, column_map: Optional[Dict[str, str]] = None
     """column_map: {"instruction": <csv col>, "response": <csv col>}
     (the Dataset CR feature mapping)."""
     rows = []
-    with open(path, newline="", encoding="utf-8") as f:
+    # utf-8-sig: strip the BOM Excel prepends (a plain utf-8 read
+    # leaks \ufeff into the first header and breaks column mapping)
+    with open(path, newline="", encoding="utf-8-sig") as f:
         for raw in csv.DictReader(f):
             if column_map:
                 rows.append({feat: raw.get(col, "")

@@ -225,3 +225,16 @@ def test_preference_rows_dropped_when_fully_masked():
     for ex in ds.examples:
         assert any(l != -100 for l in ex["chosen_labels"])
         assert any(l != -100 for l in ex["rejected_labels"])
+
+
+def test_csv_with_utf8_bom():
+    """Excel-style CSVs (UTF-8 BOM) map columns correctly — a plain
+    utf-8 read leaks the BOM into the first header name."""
+    import tempfile
+    with tempfile.NamedTemporaryFile("wb", suffix=".csv",
+                                     delete=False) as f:
+        f.write("﻿instruction,response\nhola,mundo\n".encode())
+        p = f.name
+    rows = read_csv_rows(p, {"instruction": "instruction",
+                             "response": "response"})
+    assert rows == [{"instruction": "hola", "response": "mundo"}]
