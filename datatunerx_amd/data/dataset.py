@@ -46,21 +46,45 @@ class ByteTokenizer:
                      if 3 <= i < 259).decode("utf-8", errors="replace")
 
 
-def read_csv_rows(path: str, column_map: Optional[Dict[str, str]] = None
-                  ) -> List[Dict[str, str]]:
-    """column_map: {"instruction": <csv col>, "response": <csv col>}
-    (the Dataset CR feature mapping)."""
+def _map_row(raw: dict, column_map: Optional[Dict[str, str]]):
+    if column_map:
+        return {feat: str(raw.get(col, "") or "")
+                for feat, col in column_map.items()}
+    return {k: str(v) if v is not None else "" for k, v in raw.items()}
+
+
+def read_rows(path: str, column_map: Optional[Dict[str, str]] = None
+              ) -> List[Dict[str, str]]:
+    """Rows from a .csv, .json (list of objects) or .jsonl file.
+    column_map: {"instruction": <source col>, ...} (the Dataset CR
+    feature mapping). The reference ingests CSV only (Ray Data,
+    train.py:339-351); json/jsonl cover the common alpaca-style
+    instruction files."""
+    import json as _json
     rows = []
-    # utf-8-sig: strip the BOM Excel prepends (a plain utf-8 read
-    # leaks \ufeff into the first header and breaks column mapping)
+    if path.endswith(".jsonl"):
+        with open(path, encoding="utf-8-sig") as f:
+            for line in f:
+                line = line.strip()
+                if line:
+                    rows.append(_map_row(_json.loads(line), column_map))
+        return rows
+    if path.endswith(".json"):
+        with open(path, encoding="utf-8-sig") as f:
+            data = _json.load(f)
+        if not isinstance(data, list):
+            raise ValueError(f"{path}: expected a JSON list of objects")
+        return [_map_row(r, column_map) for r in data]
+    # default: CSV. utf-8-sig strips the BOM Excel prepends (a plain
+    # utf-8 read leaks \ufeff into the first header and breaks mapping)
     with open(path, newline="", encoding="utf-8-sig") as f:
         for raw in csv.DictReader(f):
-            if column_map:
-                rows.append({feat: raw.get(col, "")
-                             for feat, col in column_map.items()})
-            else:
-                rows.append(dict(raw))
+            rows.append(_map_row(raw, column_map))
     return rows
+
+
+# back-compat name (CSV was the only format through round 2)
+read_csv_rows = read_rows
 
 
 def preprocess_supervised_example(tokenizer, template_name: str,
@@ -117,8 +141,11 @@ class SFTDataset:
     def from_csv(cls, path: str, tokenizer, column_map=None,
                  template_name: str = "llama2",
                  cutoff_len: int = DEFAULT_CUTOFF_LEN):
-        return cls.from_rows(read_csv_rows(path, column_map), tokenizer,
+        """Accepts .csv, .json (list) or .jsonl (see read_rows)."""
+        return cls.from_rows(read_rows(path, column_map), tokenizer,
                              template_name, cutoff_len)
+
+    from_file = from_csv
 
     @classmethod
     def from_rows_pt(cls, rows, tokenizer,

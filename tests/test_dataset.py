@@ -238,3 +238,41 @@ def test_csv_with_utf8_bom():
     rows = read_csv_rows(p, {"instruction": "instruction",
                              "response": "response"})
     assert rows == [{"instruction": "hola", "response": "mundo"}]
+
+
+def test_json_and_jsonl_ingest(tmp_path):
+    """Alpaca-style .json and .jsonl files ingest through the same
+    path as CSV, with the Dataset CR column mapping applied."""
+    import json as _json
+    data = [{"prompt": "q one", "output": "a one"},
+            {"prompt": "q two", "output": "a two"}]
+    pj = tmp_path / "d.json"
+    pj.write_text(_json.dumps(data))
+    pl = tmp_path / "d.jsonl"
+    pl.write_text("\n".join(_json.dumps(r) for r in data) + "\n")
+    cmap = {"instruction": "prompt", "response": "output"}
+    from datatunerx_amd.data.dataset import read_rows
+    want = [{"instruction": "q one", "response": "a one"},
+            {"instruction": "q two", "response": "a two"}]
+    assert read_rows(str(pj), cmap) == want
+    assert read_rows(str(pl), cmap) == want
+    ds = SFTDataset.from_file(str(pj), ByteTokenizer(), column_map=cmap,
+                              template_name="vanilla")
+    assert len(ds) == 2
+    # trainer CLI accepts a jsonl dataset_path directly
+    from datatunerx_amd.train.run import main as train_main
+    out = str(tmp_path / "out")
+    rc = train_main(["--model_name_or_path", "llama-tiny",
+                     "--dataset_path", str(pl),
+                     "--instruction_column", "prompt",
+                     "--response_column", "output",
+                     "--output_dir", out, "--max_steps", "1",
+                     "--per_device_train_batch_size", "2",
+                     "--block_size", "32", "--logging_steps", "0"])
+    assert rc == 0
+    # malformed json -> clear error
+    bad = tmp_path / "bad.json"
+    bad.write_text("{\"not\": \"a list\"}")
+    import pytest as _pytest
+    with _pytest.raises(ValueError, match="JSON list"):
+        read_rows(str(bad))
