@@ -48,3 +48,27 @@ def test_validate_never_crashes(kind, name, spec):
     except (AttributeError, TypeError, KeyError) as e:
         pytest.fail(f"uncontrolled {type(e).__name__} for "
                     f"{kind.__name__} spec={spec!r}: {e}")
+
+
+@settings(max_examples=120, deadline=None)
+@given(tmpl=st.sampled_from(sorted(
+           __import__("datatunerx_amd.data.templates",
+                      fromlist=["TEMPLATES"]).TEMPLATES)),
+       query=st.text(max_size=60), resp=st.text(max_size=40),
+       system=st.text(max_size=30),
+       history=st.lists(st.tuples(st.text(max_size=20),
+                                  st.text(max_size=20)), max_size=3),
+       cutoff=st.integers(4, 200))
+def test_preprocess_never_crashes_and_masks_consistently(
+        tmpl, query, resp, system, history, cutoff):
+    """Property: template encode + masking hold for ARBITRARY text —
+    ids/labels same length, bounded by cutoff, every non-ignored label
+    equals its input id."""
+    from datatunerx_amd.data.dataset import (IGNORE_INDEX, ByteTokenizer,
+                                             preprocess_supervised_example)
+    ids, labels = preprocess_supervised_example(
+        ByteTokenizer(), tmpl, query, resp, history=history,
+        system=system, cutoff_len=cutoff)
+    assert len(ids) == len(labels) <= cutoff
+    for i, l in zip(ids, labels):
+        assert l == IGNORE_INDEX or l == i
