@@ -72,3 +72,33 @@ def test_preprocess_never_crashes_and_masks_consistently(
     assert len(ids) == len(labels) <= cutoff
     for i, l in zip(ids, labels):
         assert l == IGNORE_INDEX or l == i
+
+
+@settings(max_examples=100, deadline=None)
+@given(batch=st.lists(
+           st.lists(st.integers(0, 500), min_size=1, max_size=40),
+           min_size=1, max_size=6),
+       mult=st.integers(1, 8))
+def test_collate_invariants(batch, mult):
+    from datatunerx_amd.data.dataset import IGNORE_INDEX, collate
+    exs = [{"input_ids": ids, "labels": list(ids)} for ids in batch]
+    out = collate(exs, pad_token_id=0, pad_to_multiple_of=mult)
+    B, S = out["input_ids"].shape
+    assert B == len(batch) and S % mult == 0
+    assert S >= max(len(b) for b in batch)
+    for i, ids in enumerate(batch):
+        n = len(ids)
+        assert out["input_ids"][i, :n].tolist() == ids
+        assert (out["input_ids"][i, n:] == 0).all()
+        assert (out["labels"][i, n:] == IGNORE_INDEX).all()
+
+
+@settings(max_examples=100, deadline=None)
+@given(pred=st.lists(st.integers(0, 50), max_size=30),
+       ref=st.lists(st.integers(0, 50), max_size=30))
+def test_gen_metrics_bounded(pred, ref):
+    from datatunerx_amd.train.gen_metrics import bleu, rouge_l
+    for m in (rouge_l(pred, ref), bleu(pred, ref)):
+        assert 0.0 <= m <= 1.0
+    if pred and pred == ref:
+        assert rouge_l(pred, ref) == 1.0
