@@ -3,10 +3,12 @@
 Replaces the reference's Ray Serve `LlamaDeployment` app (baked into the
 checkpoint image — finetunejob_controller.go:378-384):
 native PyTorch-ROCm decode through the same fused HIP kernels as
-training, with a preallocated KV cache. Serving is ephemeral per-job
-evaluation (the job controller tears it down after scoring), so the
-engine favors simplicity + correctness; the TP path for 13B is in
-parallel/tp.py.
+training, with a preallocated KV cache. Single-token decode is
+WEIGHT-bandwidth-bound, so throughput scales by request BATCHING
+(ragged batched decode, hipGraph-captured batched token step) rather
+than by streams; per-stream engine pools cover mixed workloads and the
+TP path for 13B is in parallel/tp.py. The serve process is per-job and
+torn down after scoring.
 """
 
 from __future__ import annotations
