@@ -355,7 +355,6 @@ class SFTTrainer:
         self.total_steps = (cfg.max_steps if cfg.max_steps > 0 else
                             int(steps_per_epoch * cfg.num_train_epochs))
         self.global_step = 0
-        self._log_f = None
         self._exporter = None
         if cfg.metrics_export_address and is_main():
             from ..metrics.remote_write import RemoteWriteExporter
@@ -370,13 +369,11 @@ class SFTTrainer:
             return
         watch = os.path.join(self.cfg.output_dir, "watch")
         os.makedirs(watch, exist_ok=True)
-        if self._log_f is None:
-            self._log_f = {}
-        if kind not in self._log_f:
-            self._log_f[kind] = open(
-                os.path.join(watch, f"{kind}_log.jsonl"), "a")
-        self._log_f[kind].write(json.dumps(record) + "\n")
-        self._log_f[kind].flush()
+        # open-append-close per write: a held handle leaks one fd per
+        # trainer in long-lived processes (and writes are infrequent —
+        # every logging_steps)
+        with open(os.path.join(watch, f"{kind}_log.jsonl"), "a") as f:
+            f.write(json.dumps(record) + "\n")
         if self._exporter is not None:
             if kind == "trainer":
                 self._exporter.export_train_metrics(record)
