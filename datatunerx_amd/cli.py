@@ -67,8 +67,11 @@ def main(argv=None):
     store = Store(args.state_dir)
 
     if args.cmd == "apply":
-        text = sys.stdin.read() if args.filename == "-" else \
-            open(args.filename).read()
+        if args.filename == "-":
+            text = sys.stdin.read()
+        else:
+            with open(args.filename) as fh:
+                text = fh.read()
         for obj in _apply(store, text, args.filename):
             print(f"{obj.kind.lower()}/{obj.name} applied")
     elif args.cmd == "get":
@@ -124,7 +127,11 @@ def main(argv=None):
         from .api.manager import Manager
         from .api.types import (FinetuneExperiment, FinetuneJob, Finetune)
         for fn in args.filename:
-            text = sys.stdin.read() if fn == "-" else open(fn).read()
+            if fn == "-":
+                text = sys.stdin.read()
+            else:
+                with open(fn) as fh:
+                    text = fh.read()
             for obj in _apply(store, text, fn):
                 print(f"{obj.kind.lower()}/{obj.name} applied")
         cfg = ManagerConfig(state_dir=args.state_dir,
