@@ -94,7 +94,7 @@ def test_dpo_micro_loss_matches_pure_torch(tmp_path):
         want = -F.logsigmoid(0.25 * ((pc - rc) - (pr - rr))).mean()
     assert torch.allclose(loss.detach(), want, atol=1e-4), (loss, want)
     # fresh adapters: reference == policy, so loss == -logsigmoid(0)
-    assert abs(float(loss) - 0.6931) < 1e-3
+    assert abs(float(loss.detach()) - 0.6931) < 1e-3
 
 
 def test_dpo_training_learns_preferences(tmp_path):
