@@ -1214,3 +1214,23 @@ def test_concurrent_jobs_with_mid_flight_deletion(tmp_path):
     for i in (0, 2):
         cur = mgr.store.get(FinetuneJob, "default", f"churn{i}")
         assert cur.status.get("state") == "Successful", (i, cur.status)
+
+
+def test_example_manifests_validate(tmp_path):
+    """Every shipped examples/*.yaml passes admission on a fresh store
+    (catches doc drift against the validation rules)."""
+    import glob
+
+    from datatunerx_amd.api.store import Store
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    store = Store(str(tmp_path / "s"))
+    files = sorted(glob.glob(os.path.join(root, "examples", "*.yaml")))
+    assert len(files) >= 4
+    applied = []
+    for fn in files:
+        with open(fn) as f:
+            for obj in store.apply_manifest(f.read()):
+                applied.append(f"{obj.kind}/{obj.name}")
+    assert any(a.startswith("FinetuneJob/") for a in applied)
+    assert any(a.startswith("FinetuneExperiment/") for a in applied)
+    assert any(a.startswith("Hyperparameter/") for a in applied)
