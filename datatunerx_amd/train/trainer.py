@@ -341,6 +341,9 @@ class SFTTrainer:
         sync = GradSynchronizer(world_size) if world_size > 1 else None
         named = model.trainable_parameters()
         mode = cfg.optimizer_mode
+        if mode not in ("auto", "flat", "overlap", "zero1"):
+            raise ValueError(f"optimizer_mode {mode!r} (want auto | "
+                             f"flat | overlap | zero1)")
         if mode == "auto":
             # LoRA-size trainables (≤64M params): one fused latency-bound
             # all-reduce. Full-param models: overlap buckets with backward.

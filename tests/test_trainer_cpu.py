@@ -388,3 +388,9 @@ def test_ignored_flags_warn(tmp_path, capsys):
     assert "--optim 'sgd' ignored" in out
     assert "--fp16 ignored" in out
     assert "--shift_attn ignored" in out
+
+
+def test_invalid_optimizer_mode_rejected(tmp_path):
+    import pytest
+    with pytest.raises(ValueError, match="optimizer_mode"):
+        _tiny_trainer(tmp_path, steps=1, optimizer_mode="sgdish")
