@@ -1234,3 +1234,18 @@ def test_example_manifests_validate(tmp_path):
     assert any(a.startswith("FinetuneJob/") for a in applied)
     assert any(a.startswith("FinetuneExperiment/") for a in applied)
     assert any(a.startswith("Hyperparameter/") for a in applied)
+
+
+def test_entrypoints_help():
+    """All four module entrypoints respond to --help (import-time and
+    argparse wiring stay sound)."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for mod in ("datatunerx_amd.cli", "datatunerx_amd.api.manager",
+                "datatunerx_amd.serve.server", "datatunerx_amd.train.run"):
+        r = subprocess.run([sys.executable, "-m", mod, "--help"],
+                           capture_output=True, text=True, timeout=120,
+                           cwd=root)
+        assert r.returncode == 0, (mod, r.stderr[-300:])
+        assert "usage" in (r.stdout + r.stderr).lower(), mod
