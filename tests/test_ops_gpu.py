@@ -436,7 +436,7 @@ def test_fused_linear_ce_gpu_matches_unfused():
         h.grad = None
         loss2 = cross_entropy(torch.nn.functional.linear(h, w), t)
         loss2.backward()
-        assert abs(float(loss) - float(loss2)) < 2e-3 * max(
+        assert abs(float(loss.detach()) - float(loss2.detach())) < 2e-3 * max(
             1.0, abs(float(loss2)))
         assert_close(g1, h.grad, rtol=3e-2, name="fused-ce dx")
     finally:

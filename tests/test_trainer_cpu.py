@@ -301,9 +301,9 @@ def test_model_loss_fused_ce_env(monkeypatch):
     m = LlamaForCausalLM(LlamaConfig.tiny(), lora=True,
                          dtype=torch.float32).init_random()
     ids = torch.randint(3, 512, (2, 32))
-    base = float(m(ids, labels=ids.clone()))
+    base = float(m(ids, labels=ids.clone()).detach())
     monkeypatch.setenv("DTX_FUSED_CE", "1")
-    fused = float(m(ids, labels=ids.clone()))
+    fused = float(m(ids, labels=ids.clone()).detach())
     assert abs(base - fused) < 1e-4
 
 
