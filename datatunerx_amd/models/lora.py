@@ -5,8 +5,8 @@ Adapter layout contract (what the reference produces via PEFT
 Checkpoint/resume): a directory with `adapter_config.json` and
 `adapter_model.safetensors` whose keys are
 `base_model.model.<module_path>.lora_A.weight` ([r, in]) and
-`...lora_B.weight` ([out, r]). Stock `peft.PeftModel.from_pretrained`
-can load our checkpoints.
+`...lora_B.weight` ([out, r]) — the exact names/shapes PEFT writes
+(layout-verified by tests; peft itself is not in this offline image).
 """
 
 from __future__ import annotations
