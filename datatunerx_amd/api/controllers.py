@@ -481,6 +481,10 @@ class FinetuneJobController:
             args = [self.cfg.python, "-m", "datatunerx_amd.serve.server",
                     "--port", str(port),
                     "--model", ftspec.get("llm") or self.cfg.default_model]
+            if serve_cfg.get("template"):
+                # serveConfig.template: chat template for the served
+                # model (llama2 default; llama3 checkpoints need theirs)
+                args += ["--template", str(serve_cfg["template"])]
             if ck.get("checkpointPath"):
                 args += ["--adapter", ck["checkpointPath"]]
             base_env = {"PYTHONPATH": self.cfg.repo_root,

@@ -92,6 +92,12 @@ def validate_(obj: ApiObject) -> None:
                  1 <= tp <= 8,
                  "spec.serveConfig.tensorParallel must be an int in "
                  "[1, 8] (one 8-GPU MI355X node)", errs)
+        tmpl = sc.get("template")
+        if tmpl is not None:
+            from ..data.templates import TEMPLATES
+            _require(tmpl in TEMPLATES,
+                     f"spec.serveConfig.template must be one of "
+                     f"{sorted(TEMPLATES)}", errs)
     elif isinstance(obj, Finetune):
         _validate_finetune_spec(obj.spec, "spec", errs)
     elif isinstance(obj, FinetuneExperiment):
