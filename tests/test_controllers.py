@@ -1278,3 +1278,26 @@ def test_serve_template_from_cr(tmp_path):
         time.sleep(0.3)
     assert mgr.store.get(FinetuneJob, "default",
                          "tmpljob").status.get("state") == "Successful"
+
+
+def test_cli_get_output_formats(tmp_path, capsys):
+    """`dtx get -o yaml|json` emit loadable documents; the default
+    table lists namespace/name/state."""
+    import json as _json
+
+    import yaml as _yaml
+
+    from datatunerx_amd.cli import main as cli
+    st = str(tmp_path / "s")
+    store = Store(st)
+    seed_resources(store)
+    cli(["--state-dir", st, "get", "hyperparameter", "-o", "json"])
+    docs = _json.loads(capsys.readouterr().out)
+    assert docs[0]["kind"] == "Hyperparameter"
+    assert docs[0]["spec"]["parameters"]["loRA_R"] == 4
+    cli(["--state-dir", st, "get", "dataset", "-o", "yaml"])
+    ydocs = list(_yaml.safe_load_all(capsys.readouterr().out))
+    assert ydocs[0]["kind"] == "Dataset"
+    cli(["--state-dir", st, "get", "llm"])
+    table = capsys.readouterr().out
+    assert "NAMESPACE" in table and "llama-tiny" in table
