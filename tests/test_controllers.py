@@ -1301,3 +1301,13 @@ def test_cli_get_output_formats(tmp_path, capsys):
     cli(["--state-dir", st, "get", "llm"])
     table = capsys.readouterr().out
     assert "NAMESPACE" in table and "llama-tiny" in table
+
+
+def test_cli_delete_missing_and_unknown_kind(tmp_path, capsys):
+    from datatunerx_amd.cli import main as cli
+    st = str(tmp_path / "s")
+    # deleting a nonexistent object is a no-op, not a crash
+    cli(["--state-dir", st, "delete", "finetunejob", "nothere"])
+    assert "deleted" in capsys.readouterr().out
+    with pytest.raises(SystemExit, match="unknown kind"):
+        cli(["--state-dir", st, "get", "gizmo"])
